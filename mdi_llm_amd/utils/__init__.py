@@ -1,0 +1,9 @@
+from .partition import (  # noqa: F401
+    N_LAYERS_NODES,
+    chunk_dir,
+    chunk_file_name,
+    count_transformer_blocks,
+    layer_split,
+    split_and_store,
+    split_parameters,
+)
