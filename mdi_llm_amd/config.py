@@ -173,7 +173,26 @@ configs.extend(
             bias=False,
             norm_class_name="RMSNorm",
             mlp_class_name="LLaMAMLP",
-            intermediate_size=172,
+            intermediate_size=176,
+            norm_eps=1e-5,
+        ),
+        dict(
+            # GPU-testable tiny llama: head_size 64, GQA 2:1 (DecodeEngine
+            # needs head_size in {64,128,256} and hs*q_per_kv >= 64)
+            name="nano-gpu",
+            block_size=256,
+            vocab_size=512,
+            padding_multiple=64,
+            n_layer=4,
+            n_head=4,
+            n_embd=256,
+            n_query_groups=2,
+            rotary_percentage=1.0,
+            parallel_residual=False,
+            bias=False,
+            norm_class_name="RMSNorm",
+            mlp_class_name="LLaMAMLP",
+            intermediate_size=688,
             norm_eps=1e-5,
         ),
         dict(

@@ -77,10 +77,11 @@ def build_rope_cache(
 def apply_rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
     """``x``: (..., T, rope_n_elem); cos/sin: (T, rope_n_elem)."""
     head_size = x.size(-1)
-    x1 = x[..., : head_size // 2]
-    x2 = x[..., head_size // 2 :]
+    xf = x.float()
+    x1 = xf[..., : head_size // 2]
+    x2 = xf[..., head_size // 2 :]
     rotated = torch.cat((-x2, x1), dim=-1)
-    return (x * cos + rotated * sin).to(dtype=x.dtype)
+    return (xf * cos.float() + rotated * sin.float()).to(dtype=x.dtype)
 
 
 # ---------------------------------------------------------------------------
@@ -430,6 +431,21 @@ class GPT(nn.Module):
         # buffers so .to(device) moves them; not persisted
         self.register_buffer("cos", cos, persistent=False)
         self.register_buffer("sin", sin, persistent=False)
+
+    def _apply(self, fn, recurse=True):
+        # keep the RoPE tables fp32 whatever dtype the module is cast to
+        # (the HIP engine and the torch path must read identical tables)
+        ret = super()._apply(fn, recurse)
+        if hasattr(self, "cos") and self.cos.dtype != torch.float32:
+            cos, sin = build_rope_cache(
+                self._max_seq_length,
+                self.config.rope_n_elem,
+                device=self.cos.device,
+                base=self.config.rope_base,
+                condense_ratio=self.config.rope_condense_ratio,
+            )
+            self.cos, self.sin = cos, sin
+        return ret
 
     def set_kv_cache(
         self,

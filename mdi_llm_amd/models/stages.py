@@ -50,6 +50,20 @@ class StageBase(nn.Module):
         self.register_buffer("cos", cos, persistent=False)
         self.register_buffer("sin", sin, persistent=False)
 
+    def _apply(self, fn, recurse=True):
+        # rope tables stay fp32 across .to(dtype=...) — see model.GPT._apply
+        ret = super()._apply(fn, recurse)
+        if hasattr(self, "cos") and self.cos.dtype != torch.float32:
+            cos, sin = build_rope_cache(
+                self._max_seq_length,
+                self.config.rope_n_elem,
+                device=self.cos.device,
+                base=self.config.rope_base,
+                condense_ratio=self.config.rope_condense_ratio,
+            )
+            self.cos, self.sin = cos, sin
+        return ret
+
     @property
     def max_seq_length(self) -> int:
         return self._max_seq_length

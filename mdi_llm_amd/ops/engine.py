@@ -1,0 +1,310 @@
+"""HIP decode engine: sequences the hand-written CDNA4 kernels for one
+pipeline stage's per-token decode step.
+
+This is the MI355X-native replacement for the reference's per-token PyTorch
+forward (/root/reference/src/sub/submodels.py:170-282 driven by
+gptserver.py:788-1110).  One engine serves every in-flight sample: the
+sample slot and sequence position are *device* scalars, so the whole block
+stack is captured once as a hipGraph and replayed per token regardless of
+which sample is flowing through (reference swaps per-sample cache objects
+into the modules instead — gptserver.py:975-978).
+
+Supported natively: RMSNorm/LayerNorm, rope/learned positions, GQA
+attention (head_size multiple of 32, head_size*q_per_kv >= 64), LLaMA
+(SwiGLU) / Gemma / GPT-NeoX MLPs, sequential and parallel residual.
+MoE stages fall back to the torch path.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ..config import ModelConfig
+from ..models.model import KVCachePool
+from . import require_hip_ops
+
+__all__ = ["DecodeEngine", "engine_supported"]
+
+
+def engine_supported(config: ModelConfig) -> bool:
+    if config.mlp_class_name == "LLaMAMoE":
+        return False
+    if config.head_size % 32 != 0 or config.head_size * config.q_per_kv < 64:
+        return False
+    if config.head_size not in (64, 128, 256):
+        return False
+    if config.n_embd % 8 != 0 or (config.intermediate_size or 0) % 8 != 0:
+        return False
+    return True
+
+
+class _BlockWeights:
+    """Contiguous bf16 views of one block's parameters."""
+
+    def __init__(self, block, config: ModelConfig):
+        def p(t):
+            if t is None:
+                return None
+            assert t.dtype == torch.bfloat16, "engine requires bf16 weights"
+            return t.detach().contiguous()
+
+        self.norm1_w = p(block.norm_1.weight)
+        self.norm1_b = p(getattr(block.norm_1, "bias", None))
+        self.attn_w = p(block.attn.attn.weight)
+        self.attn_b = p(block.attn.attn.bias)
+        self.proj_w = p(block.attn.proj.weight)
+        self.proj_b = p(block.attn.proj.bias)
+        if block.norm_2 is not None:
+            self.norm2_w = p(block.norm_2.weight)
+            self.norm2_b = p(getattr(block.norm_2, "bias", None))
+        else:
+            self.norm2_w = self.norm2_b = None
+        mlp = block.mlp
+        if config.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
+            self.fc1_w = p(mlp.fc_1.weight)
+            self.fc2_w = p(mlp.fc_2.weight)
+            self.mlp_proj_w = p(mlp.proj.weight)
+            self.mlp_proj_b = p(mlp.proj.bias)
+        else:  # GptNeoxMLP
+            self.fc_w = p(mlp.fc.weight)
+            self.fc_b = p(mlp.fc.bias)
+            self.mlp_proj_w = p(mlp.proj.weight)
+            self.mlp_proj_b = p(mlp.proj.bias)
+
+
+class DecodeEngine:
+    """Drives the HIP kernels for a stage (Starter or Secondary module)."""
+
+    def __init__(
+        self,
+        stage,
+        kv_pool: KVCachePool,
+        n_chunks: int = 32,
+        use_graphs: bool = True,
+    ) -> None:
+        self.ops = require_hip_ops()
+        self.config: ModelConfig = stage.config
+        cfg = self.config
+        if not engine_supported(cfg):
+            raise ValueError(f"config {cfg.name!r} unsupported by DecodeEngine")
+        self.stage = stage
+        self.kv_pool = kv_pool
+        self.n_chunks = n_chunks
+        self.device = next(stage.parameters()).device
+        self.is_starter = hasattr(stage, "lm_head")
+        self.use_graphs = use_graphs and self.device.type == "cuda"
+
+        dev = self.device
+        self.blocks = [_BlockWeights(b, cfg) for b in stage.transformer.h]
+        if self.is_starter:
+            self.wte = stage.transformer.wte.weight.detach().contiguous()
+            self.lnf_w = stage.transformer.ln_f.weight.detach().contiguous()
+            self.lnf_b = getattr(stage.transformer.ln_f, "bias", None)
+            if self.lnf_b is not None:
+                self.lnf_b = self.lnf_b.detach().contiguous()
+            self.head_w = stage.lm_head.weight.detach().contiguous()
+            self.head_b = stage.lm_head.bias
+            if self.head_b is not None:
+                self.head_b = self.head_b.detach().contiguous()
+            self.wpe = None
+            if cfg.pos_embedding == "learned":
+                self.wpe = stage.transformer.wpe.weight.detach().contiguous()
+
+        # rope tables fp32 (stage buffers, already on device)
+        self.cos = stage.cos.detach().to(torch.float32).contiguous()
+        self.sin = stage.sin.detach().to(torch.float32).contiguous()
+
+        # ---- workspace ----------------------------------------------------
+        bf = dict(device=dev, dtype=torch.bfloat16)
+        E, I = cfg.n_embd, cfg.intermediate_size
+        n_head, hs = cfg.n_head, cfg.head_size
+        self.x = torch.zeros(E, **bf)        # residual stream (graph input)
+        self.xn = torch.zeros(E, **bf)
+        self.qkv = torch.zeros(cfg.qkv_dim, **bf)
+        self.y = torch.zeros(n_head * hs, **bf)   # attention output
+        self.a = torch.zeros(E, **bf)        # proj(attn) (+x)
+        self.hn = torch.zeros(E, **bf)
+        self.act = torch.zeros(I, **bf)
+        self.m_out = torch.zeros(E, **bf)
+        self.part_o = torch.zeros(
+            n_head * n_chunks * hs, device=dev, dtype=torch.float32
+        )
+        self.part_ml = torch.zeros(
+            n_head * n_chunks * 2, device=dev, dtype=torch.float32
+        )
+        if self.is_starter:
+            self.logits = torch.zeros(cfg.padded_vocab_size, **bf)
+            self.token = torch.zeros(1, device=dev, dtype=torch.int32)
+            self.pos_emb = torch.zeros(E, **bf)
+
+        # device-side slot/pos scalars (graph-replayable)
+        self.slot = torch.zeros(1, device=dev, dtype=torch.int32)
+        self.slot_long = torch.zeros(1, device=dev, dtype=torch.int64)
+        self.pos_table = torch.zeros(
+            kv_pool.n_slots, device=dev, dtype=torch.int32
+        )
+        self.pos = torch.zeros(1, device=dev, dtype=torch.int32)
+
+        self._graph_blocks: Optional[torch.cuda.CUDAGraph] = None
+        self._graph_tail: Optional[torch.cuda.CUDAGraph] = None
+
+    # ---------------------------------------------------------------------
+    # slot/pos bookkeeping
+    # ---------------------------------------------------------------------
+    def set_slot_pos(self, slot: int, pos: int) -> None:
+        """Host-side bookkeeping (prefill / sample init)."""
+        self.pos_table[slot] = pos
+
+    def _stage_pos(self) -> None:
+        # pos <- pos_table[slot]   (inside the graph: slot is a device value)
+        self.slot_long.copy_(self.slot)
+        torch.index_select(self.pos_table, 0, self.slot_long, out=self.pos)
+
+    # ---------------------------------------------------------------------
+    # kernel sequence (eager; also what gets captured)
+    # ---------------------------------------------------------------------
+    def _norm(self, out, x, w, b, eps):
+        if self.config.norm_class_name == "RMSNorm":
+            self.ops.rmsnorm(out, x, w, eps)
+        else:
+            self.ops.layernorm(out, x, w, b, eps)
+
+    def _run_blocks(self) -> None:
+        """x -> x through all local blocks (decode, one token)."""
+        cfg = self.config
+        ops = self.ops
+        eps = cfg.norm_eps
+        scale = 1.0 / (cfg.head_size ** 0.5)
+        for li, w in enumerate(self.blocks):
+            self._norm(self.xn, self.x, w.norm1_w, w.norm1_b, eps)
+            ops.gemv(self.qkv, w.attn_w, self.xn, w.attn_b, None, 0)
+            ops.rope_kv_append(
+                self.qkv, self.kv_pool.k, self.kv_pool.v, self.cos, self.sin,
+                self.pos, self.slot, li,
+            )
+            ops.attn_decode(
+                self.y, self.part_o, self.part_ml, self.qkv, self.kv_pool.k,
+                self.kv_pool.v, self.pos, self.slot, li, self.n_chunks, scale,
+            )
+            if cfg.parallel_residual:
+                # x = x + proj(y) + mlp(norm2(x) or xn)
+                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, None, 0)
+                if cfg.shared_attention_norm:
+                    n2 = self.xn
+                else:
+                    self._norm(self.hn, self.x, w.norm2_w, w.norm2_b, eps)
+                    n2 = self.hn
+                self._mlp(n2, w)
+                ops.add(self.a, self.a, self.m_out)
+                ops.add(self.x, self.x, self.a)
+            else:
+                # h = x + proj(y); x = h + mlp(norm2(h))
+                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, self.x, 1)
+                self._norm(self.hn, self.a, w.norm2_w, w.norm2_b, eps)
+                self._mlp(self.hn, w)
+                ops.add(self.x, self.a, self.m_out)
+
+    def _mlp(self, inp, w) -> None:
+        cfg = self.config
+        if cfg.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
+            gelu_gate = cfg.mlp_class_name == "GemmaMLP"
+            self.ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp, gelu_gate)
+        else:
+            self.ops.gemv(self.act, w.fc_w, inp, w.fc_b, None, 2)  # gelu
+        self.ops.gemv(self.m_out, w.mlp_proj_w, self.act, w.mlp_proj_b, None, 0)
+
+    def _embed(self) -> None:
+        cfg = self.config
+        emb_scale = (cfg.n_embd ** 0.5) if cfg.scale_embeddings else 1.0
+        self.ops.embed(self.x, self.wte, self.token, emb_scale)
+        if self.wpe is not None:
+            self.ops.embed(self.pos_emb, self.wpe, self.pos, 1.0)
+            self.ops.add(self.x, self.x, self.pos_emb)
+
+    def _tail_seq(self) -> None:
+        self._norm(self.xn, self.x, self.lnf_w, self.lnf_b,
+                   self.config.norm_eps)
+        self.ops.gemv(self.logits, self.head_w, self.xn, self.head_b, None, 0)
+
+    # ---------------------------------------------------------------------
+    # public decode API
+    # ---------------------------------------------------------------------
+    def capture_graphs(self) -> None:
+        """Capture the per-token block stack (and tail) as hipGraphs."""
+        if not self.use_graphs or self._graph_blocks is not None:
+            return
+        torch.cuda.synchronize()
+        # warm-up in a side stream (allocator requirement for capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._stage_pos()
+                if self.is_starter:
+                    self._embed()
+                self._run_blocks()
+                if self.is_starter:
+                    self._tail_seq()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        # zero any cache rows the warm-up touched
+        self.kv_pool.k.zero_()
+        self.kv_pool.v.zero_()
+
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._stage_pos()
+            if self.is_starter:
+                self._embed()
+            self._run_blocks()
+        self._graph_blocks = g
+        if self.is_starter:
+            gt = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(gt):
+                self._tail_seq()
+            self._graph_tail = gt
+        torch.cuda.synchronize()
+
+    def decode_step_head(self, token: torch.Tensor, slot: int) -> torch.Tensor:
+        """Starter head role: token -> activations (writes self.x).
+
+        ``token``: scalar int tensor on device (or host int).  Returns the
+        activation vector to forward down the ring (bf16, n_embd).
+        """
+        self.token.fill_(int(token)) if not torch.is_tensor(token) else \
+            self.token.copy_(token.view(1).to(torch.int32), non_blocking=True)
+        self.slot.fill_(slot)
+        if self._graph_blocks is not None:
+            self._graph_blocks.replay()
+        else:
+            self._stage_pos()
+            if self.is_starter:
+                self._embed()
+            self._run_blocks()
+        self.pos_table[slot] += 1
+        return self.x
+
+    def decode_step_mid(self, x: torch.Tensor, slot: int) -> torch.Tensor:
+        """Secondary role: activations in -> activations out."""
+        if x.data_ptr() != self.x.data_ptr():
+            self.x.copy_(x.view(-1), non_blocking=True)
+        self.slot.fill_(slot)
+        if self._graph_blocks is not None:
+            self._graph_blocks.replay()
+        else:
+            self._stage_pos()
+            self._run_blocks()
+        self.pos_table[slot] += 1
+        return self.x
+
+    def tail(self, x: Optional[torch.Tensor] = None) -> torch.Tensor:
+        """Starter tail role: activations -> logits (bf16, padded vocab)."""
+        if x is not None and x.data_ptr() != self.x.data_ptr():
+            self.x.copy_(x.view(-1), non_blocking=True)
+        if self._graph_tail is not None:
+            self._graph_tail.replay()
+        else:
+            self._tail_seq()
+        return self.logits

@@ -1,0 +1,198 @@
+// Torch bindings for the hand-written CDNA4 decode kernels.
+// Host-only TU: tensor checks + pointer extraction + launcher calls on the
+// current torch HIP stream.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include "decode_kernels.h"
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+inline void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+inline void check_f32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat32, name, " must be fp32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+inline void check_i32(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kInt32, name, " must be int32");
+}
+
+void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+             double eps) {
+  check_bf16(out, "out");
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int n = (int)x.numel();
+  TORCH_CHECK(n % 8 == 0, "n must be a multiple of 8");
+  TORCH_CHECK(out.numel() == n && w.numel() == n, "size mismatch");
+  launch_rmsnorm(out.data_ptr(), x.data_ptr(), w.data_ptr(), n, (float)eps,
+                 cur_stream());
+}
+
+void layernorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
+               c10::optional<torch::Tensor> b, double eps) {
+  check_bf16(out, "out");
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const int n = (int)x.numel();
+  TORCH_CHECK(n % 8 == 0, "n must be a multiple of 8");
+  const void* bp = nullptr;
+  if (b.has_value()) {
+    check_bf16(*b, "b");
+    bp = b->data_ptr();
+  }
+  launch_layernorm(out.data_ptr(), x.data_ptr(), w.data_ptr(), bp, n,
+                   (float)eps, cur_stream());
+}
+
+void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
+          c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> res,
+          int64_t epilogue) {
+  check_bf16(out, "out");
+  check_bf16(W, "W");
+  check_bf16(x, "x");
+  const int K = (int)x.numel();
+  const int M = (int)out.numel();
+  TORCH_CHECK(W.numel() == (int64_t)M * K, "W shape mismatch");
+  TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+  const void* bp = nullptr;
+  const void* rp = nullptr;
+  if (bias.has_value()) {
+    check_bf16(*bias, "bias");
+    TORCH_CHECK(bias->numel() == M, "bias size");
+    bp = bias->data_ptr();
+  }
+  if (res.has_value()) {
+    check_bf16(*res, "res");
+    TORCH_CHECK(res->numel() == M, "res size");
+    rp = res->data_ptr();
+  }
+  launch_gemv(out.data_ptr(), W.data_ptr(), x.data_ptr(), bp, rp, M, K,
+              (int)epilogue, cur_stream());
+}
+
+void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
+                 torch::Tensor x, bool gelu_gate) {
+  check_bf16(out, "out");
+  check_bf16(Wg, "Wg");
+  check_bf16(Wu, "Wu");
+  check_bf16(x, "x");
+  const int K = (int)x.numel();
+  const int M = (int)out.numel();
+  TORCH_CHECK(Wg.numel() == (int64_t)M * K && Wu.numel() == (int64_t)M * K,
+              "weight shape mismatch");
+  TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+  launch_gemv_swiglu(out.data_ptr(), Wg.data_ptr(), Wu.data_ptr(),
+                     x.data_ptr(), M, K, gelu_gate ? 1 : 0, cur_stream());
+}
+
+void embed(torch::Tensor out, torch::Tensor wte, torch::Tensor token,
+           double scale) {
+  check_bf16(out, "out");
+  check_bf16(wte, "wte");
+  check_i32(token, "token");
+  const int n = (int)out.numel();
+  TORCH_CHECK(n % 8 == 0, "n_embd must be a multiple of 8");
+  launch_embed(out.data_ptr(), wte.data_ptr(), token.data_ptr<int>(), n,
+               (float)scale, cur_stream());
+}
+
+void rope_kv_append(torch::Tensor qkv, torch::Tensor kpool,
+                    torch::Tensor vpool, torch::Tensor cos_t,
+                    torch::Tensor sin_t, torch::Tensor pos,
+                    torch::Tensor slot, int64_t layer) {
+  check_bf16(qkv, "qkv");
+  check_bf16(kpool, "kpool");
+  check_bf16(vpool, "vpool");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
+  check_i32(pos, "pos");
+  check_i32(slot, "slot");
+  // pool: [slots, layers, kv_heads, max_seq, head_size]
+  TORCH_CHECK(kpool.dim() == 5, "kpool must be 5-D");
+  const int n_layers_pool = (int)kpool.size(1);
+  const int n_kv = (int)kpool.size(2);
+  const int max_seq = (int)kpool.size(3);
+  const int hs = (int)kpool.size(4);
+  const int rope_n_elem = (int)cos_t.size(1);
+  const int qkv_dim = (int)qkv.numel();
+  const int qpk = qkv_dim / (n_kv * hs) - 2;
+  TORCH_CHECK(qpk >= 1, "bad qkv length");
+  launch_rope_kv_append(qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+                        cos_t.data_ptr<float>(), sin_t.data_ptr<float>(),
+                        pos.data_ptr<int>(), slot.data_ptr<int>(),
+                        (int)layer, n_layers_pool, n_kv, max_seq, hs,
+                        rope_n_elem, qpk, cur_stream());
+}
+
+void attn_decode(torch::Tensor out, torch::Tensor part_o,
+                 torch::Tensor part_ml, torch::Tensor qkv,
+                 torch::Tensor kpool, torch::Tensor vpool, torch::Tensor pos,
+                 torch::Tensor slot, int64_t layer, int64_t n_chunks,
+                 double scale) {
+  check_bf16(out, "out");
+  check_f32(part_o, "part_o");
+  check_f32(part_ml, "part_ml");
+  check_bf16(qkv, "qkv");
+  check_bf16(kpool, "kpool");
+  check_bf16(vpool, "vpool");
+  check_i32(pos, "pos");
+  check_i32(slot, "slot");
+  const int n_layers_pool = (int)kpool.size(1);
+  const int n_kv = (int)kpool.size(2);
+  const int max_seq = (int)kpool.size(3);
+  const int hs = (int)kpool.size(4);
+  const int qkv_dim = (int)qkv.numel();
+  const int qpk = qkv_dim / (n_kv * hs) - 2;
+  const int n_head = n_kv * qpk;
+  TORCH_CHECK(part_o.numel() >= (int64_t)n_head * n_chunks * hs,
+              "part_o too small");
+  TORCH_CHECK(part_ml.numel() >= (int64_t)n_head * n_chunks * 2,
+              "part_ml too small");
+  TORCH_CHECK(out.numel() == (int64_t)n_head * hs, "out size");
+  int rc = launch_attn_decode(
+      out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
+      qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      pos.data_ptr<int>(), slot.data_ptr<int>(), (int)layer, n_layers_pool,
+      n_kv, max_seq, hs, qpk, (int)n_chunks, (float)scale, cur_stream());
+  TORCH_CHECK(rc == 0, "attn_decode: unsupported geometry qpk=", qpk,
+              " head_size=", hs);
+}
+
+void add(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
+  check_bf16(out, "out");
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  const int n = (int)a.numel();
+  TORCH_CHECK(n % 8 == 0, "n must be a multiple of 8");
+  launch_add(out.data_ptr(), a.data_ptr(), b.data_ptr(), n, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "mdi_llm_amd hand-written CDNA4 (gfx950) decode kernels";
+  m.def("rmsnorm", &rmsnorm, "RMSNorm (decode, bf16)");
+  m.def("layernorm", &layernorm, "LayerNorm (decode, bf16)");
+  m.def("gemv", &gemv, "decode GEMV out = W@x (+bias)(+res)(act)");
+  m.def("gemv_swiglu", &gemv_swiglu, "fused SwiGLU pair GEMV");
+  m.def("embed", &embed, "embedding row gather");
+  m.def("rope_kv_append", &rope_kv_append,
+        "RoPE on interleaved qkv + KV cache append");
+  m.def("attn_decode", &attn_decode, "GQA flash-decode attention (split-S)");
+  m.def("add", &add, "bf16 residual add");
+}

@@ -1,0 +1,41 @@
+// Launcher declarations shared between the HIP kernel TU (hipcc) and the
+// torch bindings TU (host C++). hipStream_t is an opaque pointer on both
+// sides; the bindings include HIP headers for the real typedef.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+void launch_rmsnorm(void* out, const void* x, const void* w, int n, float eps,
+                    hipStream_t stream);
+
+void launch_layernorm(void* out, const void* x, const void* w, const void* b,
+                      int n, float eps, hipStream_t stream);
+
+// epilogue: 0 none, 1 +res, 2 gelu(tanh), 3 silu
+void launch_gemv(void* out, const void* W, const void* x, const void* bias,
+                 const void* res, int M, int K, int epilogue,
+                 hipStream_t stream);
+
+void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
+                        const void* x, int M, int K, int gelu_gate,
+                        hipStream_t stream);
+
+void launch_embed(void* out, const void* wte, const int* token, int n_embd,
+                  float scale, hipStream_t stream);
+
+void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
+                           const float* cos_t, const float* sin_t,
+                           const int* pos, const int* slot, int layer,
+                           int n_layers_pool, int n_kv_heads, int max_seq,
+                           int head_size, int rope_n_elem, int qpk,
+                           hipStream_t stream);
+
+int launch_attn_decode(void* out, float* part_o, float* part_ml,
+                       const void* qkv, const void* kpool, const void* vpool,
+                       const int* pos, const int* slot, int layer,
+                       int n_layers_pool, int n_kv_heads, int max_seq,
+                       int head_size, int qpk, int n_chunks, float scale,
+                       hipStream_t stream);
+
+void launch_add(void* out, const void* a, const void* b, int n,
+                hipStream_t stream);

@@ -1,0 +1,723 @@
+// decode_kernels.hip — hand-written CDNA4 (gfx950 / MI355X) kernels for the
+// per-token decode path of mdi_llm_amd.
+//
+// These implement, natively for MI355X, the ops the reference runs through
+// PyTorch/cuBLAS (survey §2.4; /root/reference/src/sub/model.py): RMSNorm /
+// LayerNorm, the decode GEMVs (fused-QKV, attention proj, SwiGLU MLP, GELU
+// MLP, lm-head), RoPE + KV-cache append, and GQA flash-decode attention
+// (MFMA 16x16x32 bf16 QK^T with online softmax, split-S across workgroups).
+//
+// Design notes (per /opt/skills/guides/cdna_hip_programming.md):
+//  * wave = 64 lanes; all block sizes are multiples of 64.
+//  * bf16 global loads vectorized as int4 (16 B = 8 bf16 per lane).
+//  * decode GEMV streams weights once; x is staged in LDS per block.
+//  * attention uses the "swapped QK^T" MFMA so each lane owns whole
+//    key-scores for one query head (softmax needs only small shfl groups).
+//  * all sequence positions / sample slots arrive via device memory
+//    (int32 tensors) so the whole decode step is hipGraph-replayable.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include "decode_kernels.h"
+
+#define DEVINL __device__ __forceinline__
+
+using bf16 = __hip_bfloat16;
+
+DEVINL float b2f(bf16 v) { return __bfloat162float(v); }
+DEVINL bf16 f2b(float v) { return __float2bfloat16(v); }
+
+// 8 bf16 loaded as one int4 (16 B)
+struct bf16x8 {
+  bf16 v[8];
+};
+
+DEVINL bf16x8 load8(const bf16* p) {
+  bf16x8 r;
+  *reinterpret_cast<int4*>(r.v) = *reinterpret_cast<const int4*>(p);
+  return r;
+}
+
+DEVINL float wave_reduce_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// RMSNorm (decode, one vector): out = x * rsqrt(mean(x^2)+eps) * w
+// fp32 accumulation, matching the torch reference (model.py RMSNorm).
+// One block (256 threads) per vector.
+// ---------------------------------------------------------------------------
+__global__ void rmsnorm_kernel(bf16* __restrict__ out,
+                               const bf16* __restrict__ x,
+                               const bf16* __restrict__ w, int n, float eps) {
+  __shared__ float red[4];
+  const int tid = threadIdx.x;
+  float acc = 0.f;
+  // vectorized: 8 bf16 per step
+  for (int i = tid * 8; i < n; i += blockDim.x * 8) {
+    bf16x8 v = load8(x + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f(v.v[j]);
+      acc += f * f;
+    }
+  }
+  acc = wave_reduce_sum(acc);
+  if ((tid & 63) == 0) red[tid >> 6] = acc;
+  __syncthreads();
+  float total = red[0] + red[1] + red[2] + red[3];
+  float scale = rsqrtf(total / n + eps);
+  for (int i = tid * 8; i < n; i += blockDim.x * 8) {
+    bf16x8 v = load8(x + i);
+    bf16x8 g = load8(w + i);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o.v[j] = f2b(b2f(v.v[j]) * scale * b2f(g.v[j]));
+    *reinterpret_cast<int4*>(out + i) = *reinterpret_cast<const int4*>(o.v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm (decode): out = (x-mean)/sqrt(var+eps) * w + b
+// ---------------------------------------------------------------------------
+__global__ void layernorm_kernel(bf16* __restrict__ out,
+                                 const bf16* __restrict__ x,
+                                 const bf16* __restrict__ w,
+                                 const bf16* __restrict__ b, int n,
+                                 float eps) {
+  __shared__ float red[8];
+  const int tid = threadIdx.x;
+  float s = 0.f, s2 = 0.f;
+  for (int i = tid * 8; i < n; i += blockDim.x * 8) {
+    bf16x8 v = load8(x + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f(v.v[j]);
+      s += f;
+      s2 += f * f;
+    }
+  }
+  s = wave_reduce_sum(s);
+  s2 = wave_reduce_sum(s2);
+  if ((tid & 63) == 0) {
+    red[(tid >> 6) * 2] = s;
+    red[(tid >> 6) * 2 + 1] = s2;
+  }
+  __syncthreads();
+  float mean = (red[0] + red[2] + red[4] + red[6]) / n;
+  float var = (red[1] + red[3] + red[5] + red[7]) / n - mean * mean;
+  float inv = rsqrtf(var + eps);
+  for (int i = tid * 8; i < n; i += blockDim.x * 8) {
+    bf16x8 v = load8(x + i);
+    bf16x8 g = load8(w + i);
+    bf16x8 o;
+    if (b != nullptr) {
+      bf16x8 bb = load8(b + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.v[j] = f2b((b2f(v.v[j]) - mean) * inv * b2f(g.v[j]) + b2f(bb.v[j]));
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.v[j] = f2b((b2f(v.v[j]) - mean) * inv * b2f(g.v[j]));
+    }
+    *reinterpret_cast<int4*>(out + i) = *reinterpret_cast<const int4*>(o.v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Decode GEMV: out[M] = W[M,K] @ x[K] (+ bias) (+ residual) (+ activation)
+//
+// W row-major bf16, streamed once from HBM (the decode bound); x staged in
+// LDS (K*2 bytes, up to ~28 KiB for the 14336-wide MLP). Block = 256
+// threads = 4 waves; each wave owns one output row per grid-stride step;
+// lane reads 16 B of the row per iteration (coalesced 1 KiB per wave-iter).
+// EPI: 0 none, 1 +residual, 2 gelu(tanh), 3 relu-free silu (used standalone)
+// ---------------------------------------------------------------------------
+template <int EPI>
+__global__ void gemv_kernel(bf16* __restrict__ out,
+                            const bf16* __restrict__ W,
+                            const bf16* __restrict__ x,
+                            const bf16* __restrict__ bias,
+                            const bf16* __restrict__ res, int M, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* xs = reinterpret_cast<bf16*>(smem);
+  const int tid = threadIdx.x;
+  for (int i = tid * 8; i < K; i += blockDim.x * 8)
+    *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(x + i);
+  __syncthreads();
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
+
+  for (int row = blockIdx.x * (blockDim.x >> 6) + wave; row < M;
+       row += waves_per_grid) {
+    const bf16* wrow = W + (size_t)row * K;
+    float acc = 0.f;
+    for (int i = lane * 8; i < K; i += 64 * 8) {
+      bf16x8 wv = load8(wrow + i);
+      bf16x8 xv = load8(xs + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc += b2f(wv.v[j]) * b2f(xv.v[j]);
+    }
+    acc = wave_reduce_sum(acc);
+    if (lane == 0) {
+      if (bias != nullptr) acc += b2f(bias[row]);
+      if (EPI == 1 && res != nullptr) acc += b2f(res[row]);
+      if (EPI == 2) {  // gelu tanh approx
+        float c = 0.7978845608028654f * (acc + 0.044715f * acc * acc * acc);
+        acc = 0.5f * acc * (1.f + tanhf(c));
+      }
+      if (EPI == 3) acc = acc / (1.f + expf(-acc));  // silu
+      out[row] = f2b(acc);
+    }
+  }
+}
+
+// SwiGLU pair GEMV: out[i] = silu(Wg_i . x) * (Wu_i . x)
+__global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
+                                   const bf16* __restrict__ Wg,
+                                   const bf16* __restrict__ Wu,
+                                   const bf16* __restrict__ x, int M, int K,
+                                   int gelu_gate) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* xs = reinterpret_cast<bf16*>(smem);
+  const int tid = threadIdx.x;
+  for (int i = tid * 8; i < K; i += blockDim.x * 8)
+    *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(x + i);
+  __syncthreads();
+
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
+
+  for (int row = blockIdx.x * (blockDim.x >> 6) + wave; row < M;
+       row += waves_per_grid) {
+    const bf16* grow = Wg + (size_t)row * K;
+    const bf16* urow = Wu + (size_t)row * K;
+    float ga = 0.f, ua = 0.f;
+    for (int i = lane * 8; i < K; i += 64 * 8) {
+      bf16x8 gv = load8(grow + i);
+      bf16x8 uv = load8(urow + i);
+      bf16x8 xv = load8(xs + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = b2f(xv.v[j]);
+        ga += b2f(gv.v[j]) * xf;
+        ua += b2f(uv.v[j]) * xf;
+      }
+    }
+    ga = wave_reduce_sum(ga);
+    ua = wave_reduce_sum(ua);
+    if (lane == 0) {
+      float act;
+      if (gelu_gate) {
+        float c = 0.7978845608028654f * (ga + 0.044715f * ga * ga * ga);
+        act = 0.5f * ga * (1.f + tanhf(c));
+      } else {
+        act = ga / (1.f + expf(-ga));
+      }
+      out[row] = f2b(act * ua);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Embedding row gather: out[n_embd] = wte[token] (* scale)
+// token id read from device memory (graph-replayable).
+// ---------------------------------------------------------------------------
+__global__ void embed_kernel(bf16* __restrict__ out,
+                             const bf16* __restrict__ wte,
+                             const int* __restrict__ token, int n_embd,
+                             float scale) {
+  const int t = token[0];
+  const bf16* row = wte + (size_t)t * n_embd;
+  for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 8; i < n_embd;
+       i += gridDim.x * blockDim.x * 8) {
+    bf16x8 v = load8(row + i);
+    if (scale != 1.f) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.v[j] = f2b(b2f(v.v[j]) * scale);
+    }
+    *reinterpret_cast<int4*>(out + i) = *reinterpret_cast<const int4*>(v.v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (rotate-half) on the fused interleaved QKV vector + KV-cache append.
+//
+// qkv layout (litGPT interleaved, model.py:686-718): per query group g:
+//   [q_0..q_{qpk-1}, k, v] rows of head_size each.
+// K/V pool layout: [slot, layer, kv_head, max_seq, head_size] bf16.
+// position and slot are device int32 scalars.
+// Grid: n_query_groups blocks, 256 threads; rope_n_elem may be < head_size
+// (partial rotary, e.g. NeoX 25%).
+// ---------------------------------------------------------------------------
+__global__ void rope_kv_append_kernel(
+    bf16* __restrict__ qkv, bf16* __restrict__ kpool,
+    bf16* __restrict__ vpool, const float* __restrict__ cos_t,
+    const float* __restrict__ sin_t, const int* __restrict__ pos_p,
+    const int* __restrict__ slot_p, int layer, int n_layers_pool,
+    int n_kv_heads, int max_seq, int head_size, int rope_n_elem, int qpk) {
+  const int g = blockIdx.x;  // query group == kv head
+  const int pos = pos_p[0];
+  const int slot = slot_p[0];
+  const int tid = threadIdx.x;
+  const int half = rope_n_elem >> 1;
+
+  bf16* base = qkv + (size_t)g * (qpk + 2) * head_size;
+
+  // rope on q rows and the k row (qpk+1 rows)
+  for (int idx = tid; idx < (qpk + 1) * half; idx += blockDim.x) {
+    const int r = idx / half;
+    const int d = idx % half;
+    bf16* row = base + (size_t)r * head_size;
+    float x1 = b2f(row[d]);
+    float x2 = b2f(row[d + half]);
+    float c1 = cos_t[(size_t)pos * rope_n_elem + d];
+    float s1 = sin_t[(size_t)pos * rope_n_elem + d];
+    float c2 = cos_t[(size_t)pos * rope_n_elem + d + half];
+    float s2 = sin_t[(size_t)pos * rope_n_elem + d + half];
+    row[d] = f2b(x1 * c1 - x2 * s1);
+    row[d + half] = f2b(x2 * c2 + x1 * s2);
+  }
+  __syncthreads();
+
+  // append k,v rows into the pool at (slot, layer, g, pos)
+  const size_t cache_off =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+          (size_t)max_seq * head_size +
+      (size_t)pos * head_size;
+  const bf16* krow = base + (size_t)qpk * head_size;
+  const bf16* vrow = base + (size_t)(qpk + 1) * head_size;
+  for (int d = tid; d < head_size; d += blockDim.x) {
+    kpool[cache_off + d] = krow[d];
+    vpool[cache_off + d] = vrow[d];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// GQA flash-decode attention, split-S.
+//
+// One wave handles one (kv_head, chunk): MFMA 16x16x32 bf16 computes
+// scores[key, qhead] (swapped operands so each lane owns 4 keys of ONE
+// query head), online softmax with running (m, l) per qhead in LDS, P*V
+// accumulated in registers. Partials (o, m, l) per chunk go to global fp32;
+// attn_combine_kernel reduces chunks.
+//
+// Launch: grid.x = n_kv_heads * n_chunks / WAVES_PER_BLOCK, block = 256.
+// qpk (query heads per kv head) <= 16.
+// ---------------------------------------------------------------------------
+using f32x4 = __attribute__((__vector_size__(16))) float;
+using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+
+#define ATTN_WAVES 4  // waves per block, each fully independent
+
+// XOR swizzle on the q LDS tile (rows at a power-of-two byte stride would
+// otherwise put a ds_read_b128 lane group on one bank slot — guide §6 G4):
+// byte ^= (row&7)<<4
+template <int HS>
+DEVINL int q_swz(int row, int d) {  // element index into a [16][HS] tile
+  int byte = (row * HS + d) * 2;
+  byte ^= (row & 7) << 4;
+  return byte >> 1;
+}
+
+template <int QPK, int HS>
+__global__ void attn_decode_kernel(
+    float* __restrict__ part_o,   // [n_head, n_chunks, head_size]
+    float* __restrict__ part_ml,  // [n_head, n_chunks, 2]
+    const bf16* __restrict__ qkv, // interleaved, already roped
+    const bf16* __restrict__ kpool, const bf16* __restrict__ vpool,
+    const int* __restrict__ pos_p, const int* __restrict__ slot_p, int layer,
+    int n_layers_pool, int n_kv_heads, int max_seq, int n_chunks,
+    float scale) {
+  constexpr int head_size = HS;
+  // LDS per wave: q tile (16xHS bf16), p tile (16x16 f32 = 1KB),
+  // m/l/alpha (3*16 f32)
+  __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
+  __shared__ float p_lds[ATTN_WAVES][16][16];
+  __shared__ float m_lds[ATTN_WAVES][16];
+  __shared__ float l_lds[ATTN_WAVES][16];
+  __shared__ float a_lds[ATTN_WAVES][16];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wg_id = blockIdx.x * ATTN_WAVES + wave;
+  const int g = wg_id / n_chunks;        // kv head
+  const int chunk = wg_id % n_chunks;
+  if (g >= n_kv_heads) return;  // tail waves exit before any barrier
+
+  const int S = pos_p[0] + 1;  // keys visible this step
+  const int slot = slot_p[0];
+  const int keys_per_chunk = ((S + n_chunks - 1) / n_chunks + 15) & ~15;
+  const int k_begin = chunk * keys_per_chunk;
+
+  const size_t cache_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq * head_size;
+
+  // ---- stage q rows (QPK real, rest zero) into LDS (swizzled) ------------
+  // q row j of group g lives at qkv[(g*(QPK+2)+j)*head_size]
+  for (int i = lane; i < 16 * (head_size / 8); i += 64) {
+    const int r = i / (head_size / 8);
+    const int d8 = (i % (head_size / 8)) * 8;
+    int4 val = {0, 0, 0, 0};
+    if (r < QPK) {
+      const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
+      val = *reinterpret_cast<const int4*>(qrow + d8);
+    }
+    *reinterpret_cast<int4*>(&q_lds[wave][q_swz<HS>(r, d8)]) = val;
+  }
+  if (lane < 16) {
+    m_lds[wave][lane] = -1e30f;
+    l_lds[wave][lane] = 0.f;
+    a_lds[wave][lane] = 1.f;
+  }
+  __syncthreads();
+
+  // ---- per-lane roles ----------------------------------------------------
+  // softmax map (A): qhead qa = lane & 15, key sub-row = lane >> 4
+  // PV map (B): qb = lane % QPK, dim slice d0, ODIM dims per lane
+  static_assert(HS % 32 == 0 && HS * QPK >= 64, "unsupported attn geometry");
+  constexpr int ODIM = HS * QPK / 64;  // output dims per lane
+  const int qa = lane & 15;
+  const int sub = lane >> 4;
+  const int qb = lane % QPK;
+  const int d0 = (lane / QPK) * ODIM;
+
+  float o_acc[ODIM];
+#pragma unroll
+  for (int i = 0; i < ODIM; ++i) o_acc[i] = 0.f;
+
+  const int n_tiles = keys_per_chunk / 16;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int key0 = k_begin + t * 16;
+    // NOTE: no early break — every wave in the block must reach every
+    // __syncthreads(); empty tiles run fully masked.
+
+    // ---- QK^T via MFMA: A = K tile (16 keys x 32 dims), B = q^T ----------
+    f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15;          // key row in tile (and B's qhead col)
+    const int koff = (lane >> 4) * 8;    // dim offset within 32-chunk
+    const bf16* krow = kpool + cache_base + (size_t)(key0 + arow) * head_size;
+    const bool row_valid = (key0 + arow) < S;
+#pragma unroll
+    for (int c = 0; c < HS / 32; ++c) {  // K=32 chunks cover head_size
+      bf16x8_t af = {};
+      if (row_valid)
+        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      const bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
+          &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
+      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc4, 0, 0, 0);
+    }
+    // C[row=key, col=qhead]: lane holds keys (sub*4 + r) for qhead qa
+    float sc[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = key0 + sub * 4 + r;
+      sc[r] = (key < S) ? acc4[r] * scale : -1e30f;
+    }
+
+    // ---- online softmax per qhead (4 lanes per qhead: strides 16,32) -----
+    float tmax = fmaxf(fmaxf(sc[0], sc[1]), fmaxf(sc[2], sc[3]));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    __syncthreads();  // a_lds/p_lds from previous tile fully consumed
+    if (sub == 0) {
+      float m_old = m_lds[wave][qa];
+      float m_new = fmaxf(m_old, tmax);
+      a_lds[wave][qa] = __expf(m_old - m_new);
+      m_lds[wave][qa] = m_new;
+    }
+    __syncthreads();
+    const float m_new = m_lds[wave][qa];
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float p = __expf(sc[r] - m_new);
+      const int key = key0 + sub * 4 + r;
+      if (key >= S) p = 0.f;
+      p_lds[wave][sub * 4 + r][qa] = p;
+      psum += p;
+    }
+    psum += __shfl_xor(psum, 16, 64);
+    psum += __shfl_xor(psum, 32, 64);
+    if (sub == 0)
+      l_lds[wave][qa] = l_lds[wave][qa] * a_lds[wave][qa] + psum;
+    __syncthreads();
+
+    // ---- P*V: o[qb][d0..] += p[s][qb] * V[s][d0..] -----------------------
+    const float alphaB = a_lds[wave][qb];
+#pragma unroll
+    for (int i = 0; i < ODIM; ++i) o_acc[i] *= alphaB;
+    const int smax = min(16, S - key0);
+    for (int s = 0; s < smax; ++s) {
+      const float p = p_lds[wave][s][qb];
+      const bf16* vrow =
+          vpool + cache_base + (size_t)(key0 + s) * head_size + d0;
+      if constexpr (ODIM >= 8) {
+#pragma unroll
+        for (int i = 0; i < ODIM; i += 8) {
+          bf16x8 vv = load8(vrow + i);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) o_acc[i + j] += p * b2f(vv.v[j]);
+        }
+      } else if constexpr (ODIM == 4) {
+        int2 raw = *reinterpret_cast<const int2*>(vrow);
+        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o_acc[j] += p * b2f(vv[j]);
+      } else if constexpr (ODIM == 2) {
+        int raw = *reinterpret_cast<const int*>(vrow);
+        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+        o_acc[0] += p * b2f(vv[0]);
+        o_acc[1] += p * b2f(vv[1]);
+      } else {  // ODIM == 1
+        o_acc[0] += p * b2f(vrow[0]);
+      }
+    }
+  }
+
+  // ---- write partials ----------------------------------------------------
+  // global qhead index for PV map: h = g*QPK + qb
+  if (k_begin < S) {
+    const int hB = g * QPK + qb;
+    float* op = part_o + ((size_t)hB * n_chunks + chunk) * head_size + d0;
+#pragma unroll
+    for (int i = 0; i < ODIM; ++i) op[i] = o_acc[i];
+    if (sub == 0 && qa < QPK) {
+      const int hA = g * QPK + qa;
+      part_ml[((size_t)hA * n_chunks + chunk) * 2 + 0] = m_lds[wave][qa];
+      part_ml[((size_t)hA * n_chunks + chunk) * 2 + 1] = l_lds[wave][qa];
+    }
+  } else if (lane == 0) {
+    // empty chunk: mark invalid via l = 0, m = -inf
+    for (int q = 0; q < QPK; ++q) {
+      const int h = g * QPK + q;
+      part_ml[((size_t)h * n_chunks + chunk) * 2 + 0] = -1e30f;
+      part_ml[((size_t)h * n_chunks + chunk) * 2 + 1] = 0.f;
+    }
+  }
+}
+
+// Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L
+// grid = n_head blocks, 64 threads (heads are independent).
+__global__ void attn_combine_kernel(bf16* __restrict__ out,
+                                    const float* __restrict__ part_o,
+                                    const float* __restrict__ part_ml,
+                                    int n_chunks, int head_size) {
+  const int h = blockIdx.x;
+  const int lane = threadIdx.x;
+  // global max over chunks
+  float M = -1e30f;
+  for (int c = 0; c < n_chunks; ++c)
+    M = fmaxf(M, part_ml[((size_t)h * n_chunks + c) * 2]);
+  float L = 0.f;
+  for (int c = 0; c < n_chunks; ++c) {
+    const float m = part_ml[((size_t)h * n_chunks + c) * 2];
+    const float l = part_ml[((size_t)h * n_chunks + c) * 2 + 1];
+    L += l * __expf(m - M);
+  }
+  const float inv = 1.f / L;
+  const int dpl = (head_size + 63) / 64;  // dims per lane
+  for (int i = 0; i < dpl; ++i) {
+    const int d = lane * dpl + i;
+    if (d >= head_size) break;
+    float acc = 0.f;
+    for (int c = 0; c < n_chunks; ++c) {
+      const float m = part_ml[((size_t)h * n_chunks + c) * 2];
+      const float w = __expf(m - M);
+      acc += w * part_o[((size_t)h * n_chunks + c) * head_size + d];
+    }
+    out[(size_t)h * head_size + d] = f2b(acc * inv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Residual add: out = a + b (bf16, fp32 math)
+// ---------------------------------------------------------------------------
+__global__ void add_kernel(bf16* __restrict__ out, const bf16* __restrict__ a,
+                           const bf16* __restrict__ b, int n) {
+  for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 8; i < n;
+       i += gridDim.x * blockDim.x * 8) {
+    bf16x8 va = load8(a + i);
+    bf16x8 vb = load8(b + i);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o.v[j] = f2b(b2f(va.v[j]) + b2f(vb.v[j]));
+    *reinterpret_cast<int4*>(out + i) = *reinterpret_cast<const int4*>(o.v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host-side launchers (C ABI used by the torch bindings)
+// ---------------------------------------------------------------------------
+
+static inline int gemv_grid(int M) {
+  int blocks = (M + 3) / 4;  // 4 rows per block (1 per wave)
+  return blocks < 4096 ? blocks : 4096;
+}
+
+void launch_rmsnorm(void* out, const void* x, const void* w, int n, float eps,
+                    hipStream_t stream) {
+  hipLaunchKernelGGL(rmsnorm_kernel, dim3(1), dim3(256), 0, stream,
+                     (bf16*)out, (const bf16*)x, (const bf16*)w, n, eps);
+}
+
+void launch_layernorm(void* out, const void* x, const void* w, const void* b,
+                      int n, float eps, hipStream_t stream) {
+  hipLaunchKernelGGL(layernorm_kernel, dim3(1), dim3(256), 0, stream,
+                     (bf16*)out, (const bf16*)x, (const bf16*)w,
+                     (const bf16*)b, n, eps);
+}
+
+void launch_gemv(void* out, const void* W, const void* x, const void* bias,
+                 const void* res, int M, int K, int epilogue,
+                 hipStream_t stream) {
+  const int smem = K * sizeof(bf16);
+  dim3 grid(gemv_grid(M)), block(256);
+  switch (epilogue) {
+    case 0:
+      hipLaunchKernelGGL(gemv_kernel<0>, grid, block, smem, stream,
+                         (bf16*)out, (const bf16*)W, (const bf16*)x,
+                         (const bf16*)bias, (const bf16*)res, M, K);
+      break;
+    case 1:
+      hipLaunchKernelGGL(gemv_kernel<1>, grid, block, smem, stream,
+                         (bf16*)out, (const bf16*)W, (const bf16*)x,
+                         (const bf16*)bias, (const bf16*)res, M, K);
+      break;
+    case 2:
+      hipLaunchKernelGGL(gemv_kernel<2>, grid, block, smem, stream,
+                         (bf16*)out, (const bf16*)W, (const bf16*)x,
+                         (const bf16*)bias, (const bf16*)res, M, K);
+      break;
+    default:
+      hipLaunchKernelGGL(gemv_kernel<3>, grid, block, smem, stream,
+                         (bf16*)out, (const bf16*)W, (const bf16*)x,
+                         (const bf16*)bias, (const bf16*)res, M, K);
+  }
+}
+
+void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
+                        const void* x, int M, int K, int gelu_gate,
+                        hipStream_t stream) {
+  const int smem = K * sizeof(bf16);
+  hipLaunchKernelGGL(gemv_swiglu_kernel, dim3(gemv_grid(M)), dim3(256), smem,
+                     stream, (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,
+                     (const bf16*)x, M, K, gelu_gate);
+}
+
+void launch_embed(void* out, const void* wte, const int* token, int n_embd,
+                  float scale, hipStream_t stream) {
+  int blocks = (n_embd / 8 + 255) / 256;
+  hipLaunchKernelGGL(embed_kernel, dim3(blocks > 0 ? blocks : 1), dim3(256),
+                     0, stream, (bf16*)out, (const bf16*)wte, token, n_embd,
+                     scale);
+}
+
+void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
+                           const float* cos_t, const float* sin_t,
+                           const int* pos, const int* slot, int layer,
+                           int n_layers_pool, int n_kv_heads, int max_seq,
+                           int head_size, int rope_n_elem, int qpk,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(rope_kv_append_kernel, dim3(n_kv_heads), dim3(256), 0,
+                     stream, (bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
+                     sin_t, pos, slot, layer, n_layers_pool, n_kv_heads,
+                     max_seq, head_size, rope_n_elem, qpk);
+}
+
+template <int QPK, int HS>
+static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
+                           const void* kpool, const void* vpool,
+                           const int* pos, const int* slot, int layer,
+                           int n_layers_pool, int n_kv_heads, int max_seq,
+                           int n_chunks, float scale, int blocks,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL((attn_decode_kernel<QPK, HS>), dim3(blocks), dim3(256),
+                     0, stream, part_o, part_ml, (const bf16*)qkv,
+                     (const bf16*)kpool, (const bf16*)vpool, pos, slot, layer,
+                     n_layers_pool, n_kv_heads, max_seq, n_chunks, scale);
+}
+
+template <int QPK>
+static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
+                          const void* qkv, const void* kpool,
+                          const void* vpool, const int* pos, const int* slot,
+                          int layer, int n_layers_pool, int n_kv_heads,
+                          int max_seq, int n_chunks, float scale, int blocks,
+                          hipStream_t stream) {
+#define CASE_HS(H)                                                          \
+  if (head_size == H) {                                                     \
+    if constexpr (QPK * H >= 64) {                                          \
+      attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, pos, slot, \
+                             layer, n_layers_pool, n_kv_heads, max_seq,     \
+                             n_chunks, scale, blocks, stream);              \
+      return 0;                                                             \
+    }                                                                       \
+  }
+  CASE_HS(64)
+  CASE_HS(128)
+  CASE_HS(256)
+#undef CASE_HS
+  return -1;  // unsupported geometry: caller falls back to the torch path
+}
+
+// returns 0 on success, -1 if (qpk, head_size) has no kernel instantiation
+int launch_attn_decode(void* out, float* part_o, float* part_ml,
+                       const void* qkv, const void* kpool, const void* vpool,
+                       const int* pos, const int* slot, int layer,
+                       int n_layers_pool, int n_kv_heads, int max_seq,
+                       int head_size, int qpk, int n_chunks, float scale,
+                       hipStream_t stream) {
+  const int n_wg = n_kv_heads * n_chunks;
+  const int blocks = (n_wg + ATTN_WAVES - 1) / ATTN_WAVES;
+  int rc = -1;
+  switch (qpk) {
+    case 1:
+      rc = attn_dispatch1<1>(head_size, part_o, part_ml, qkv, kpool, vpool,
+                             pos, slot, layer, n_layers_pool, n_kv_heads,
+                             max_seq, n_chunks, scale, blocks, stream);
+      break;
+    case 2:
+      rc = attn_dispatch1<2>(head_size, part_o, part_ml, qkv, kpool, vpool,
+                             pos, slot, layer, n_layers_pool, n_kv_heads,
+                             max_seq, n_chunks, scale, blocks, stream);
+      break;
+    case 4:
+      rc = attn_dispatch1<4>(head_size, part_o, part_ml, qkv, kpool, vpool,
+                             pos, slot, layer, n_layers_pool, n_kv_heads,
+                             max_seq, n_chunks, scale, blocks, stream);
+      break;
+    case 8:
+      rc = attn_dispatch1<8>(head_size, part_o, part_ml, qkv, kpool, vpool,
+                             pos, slot, layer, n_layers_pool, n_kv_heads,
+                             max_seq, n_chunks, scale, blocks, stream);
+      break;
+    case 16:
+      rc = attn_dispatch1<16>(head_size, part_o, part_ml, qkv, kpool, vpool,
+                              pos, slot, layer, n_layers_pool, n_kv_heads,
+                              max_seq, n_chunks, scale, blocks, stream);
+      break;
+    default:
+      return -1;
+  }
+  if (rc != 0) return rc;
+  hipLaunchKernelGGL(attn_combine_kernel, dim3(n_kv_heads * qpk), dim3(64), 0,
+                     stream, (bf16*)out, part_o, part_ml, n_chunks, head_size);
+  return 0;
+}
+
+void launch_add(void* out, const void* a, const void* b, int n,
+                hipStream_t stream) {
+  int blocks = (n / 8 + 255) / 256;
+  hipLaunchKernelGGL(add_kernel, dim3(blocks > 0 ? blocks : 1), dim3(256), 0,
+                     stream, (bf16*)out, (const bf16*)a, (const bf16*)b, n);
+}
