@@ -1,0 +1,123 @@
+"""Activation ring over torch.distributed P2P (RCCL on GPU, gloo on CPU).
+
+MI355X-native replacement for the reference's TCP-socket data plane
+(/root/reference/src/sub/connections.py: 16-byte ASCII header + pickled
+dict per hop).  Here a hop is one (or two) ``isend`` of fixed-layout
+tensors on the torch stream: an int32 header [sample, ntok, stop] and a
+bf16 activation payload.  On ROCm the ``nccl`` backend IS RCCL and adjacent
+ranks map to single xGMI links; on CPU test hosts the same code runs over
+``gloo``.
+
+A headerless "scheduled" mode skips the header entirely when the message
+sequence is deterministic (the bench decode loop): payloads are fixed-size
+(1, n_embd) and arrival order is the seeding order, so no host readback of
+headers is needed on the critical path.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+__all__ = ["RingComm"]
+
+HDR_LEN = 4  # sample_id, ntok, stop, reserved
+
+
+class RingComm:
+    def __init__(
+        self,
+        n_embd: int,
+        max_seq: int,
+        device: torch.device,
+        n_slots: int,
+        dtype: torch.dtype = torch.bfloat16,
+        group: Optional[dist.ProcessGroup] = None,
+    ) -> None:
+        assert dist.is_initialized(), "torch.distributed must be initialized"
+        self.group = group
+        self.rank = dist.get_rank(group)
+        self.world = dist.get_world_size(group)
+        self.next_rank = (self.rank + 1) % self.world
+        self.prev_rank = (self.rank - 1) % self.world
+        self.device = device
+        self.n_embd = n_embd
+        self.max_seq = max_seq
+
+        dev = device
+        # per-slot single-token send buffers (decode) + one shared prefill
+        # buffer, so isend never races buffer reuse
+        self.send_hdr = torch.zeros(n_slots + 1, HDR_LEN, dtype=torch.int32,
+                                    device=dev)
+        self.send_buf = torch.zeros(n_slots + 1, 1, n_embd,
+                                    dtype=dtype, device=dev)
+        self.prefill_buf = torch.zeros(max_seq, n_embd,
+                                       dtype=dtype, device=dev)
+        self.recv_hdr = torch.zeros(HDR_LEN, dtype=torch.int32, device=dev)
+        self.recv_buf = torch.zeros(max_seq, n_embd, dtype=dtype,
+                                    device=dev)
+        self._pending: dict = {}
+        self._PF = "prefill"
+
+    # -- headered path (general generate loop) ----------------------------
+    def send(self, sample: int, x: Optional[torch.Tensor],
+             stop: bool = False) -> None:
+        """Send one activation message (or a stop envelope) to next rank."""
+        slot = sample if sample >= 0 else self.send_hdr.size(0) - 1
+        ntok = 0 if x is None else x.view(-1, self.n_embd).size(0)
+        key = self._PF if ntok > 1 else slot
+        for k in (key, slot):
+            if k in self._pending:
+                for w in self._pending.pop(k):
+                    w.wait()
+        payload = None
+        if ntok > 0:
+            x2 = x.view(-1, self.n_embd)
+            if ntok > 1:
+                payload = self.prefill_buf[:ntok]
+            else:
+                payload = self.send_buf[slot, :1]
+            payload.copy_(x2)
+        hdr = self.send_hdr[slot]
+        hdr[0], hdr[1], hdr[2], hdr[3] = sample, ntok, int(stop), 0
+        works = [dist.isend(hdr, self.next_rank, group=self.group)]
+        if payload is not None:
+            works.append(dist.isend(payload, self.next_rank, group=self.group))
+        self._pending[key] = works
+
+    def recv(self) -> Tuple[int, Optional[torch.Tensor], bool]:
+        """Blocking receive of one message from prev rank.
+
+        Returns (sample, activations (ntok, n_embd) view or None, stop).
+        """
+        dist.recv(self.recv_hdr, self.prev_rank, group=self.group)
+        hdr = self.recv_hdr.cpu()  # host sync: 16 bytes
+        sample, ntok, stop = int(hdr[0]), int(hdr[1]), bool(hdr[2])
+        x = None
+        if ntok > 0:
+            x = self.recv_buf[:ntok]
+            dist.recv(x, self.prev_rank, group=self.group)
+        return sample, x, stop
+
+    # -- scheduled (headerless) path: fixed T=1 payloads -------------------
+    def send_sched(self, slot: int, x: torch.Tensor) -> None:
+        if slot in self._pending:
+            for w in self._pending.pop(slot):
+                w.wait()
+        self.send_buf[slot, 0].copy_(x.view(-1))
+        self._pending[slot] = [
+            dist.isend(self.send_buf[slot, :1], self.next_rank,
+                       group=self.group)
+        ]
+
+    def recv_sched(self) -> torch.Tensor:
+        dist.recv(self.recv_buf[:1], self.prev_rank, group=self.group)
+        return self.recv_buf[0]
+
+    def drain(self) -> None:
+        for works in self._pending.values():
+            for w in works:
+                w.wait()
+        self._pending.clear()

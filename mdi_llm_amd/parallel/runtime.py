@@ -1,0 +1,215 @@
+"""Recurrent pipeline runtime: the scheduler that keeps every stage busy.
+
+Capability parity with the reference's node loops
+(/root/reference/src/sub/gptserver.py: ``_starter_loop`` 788-1019,
+``_secondary_loop`` 1021-1110): >= n_stages samples in flight, each with its
+own KV-cache slot, single-token activation messages between stages, in-band
+per-sample stop envelopes that travel the whole ring back to the starter,
+and the tok/time instrumentation the reference writes to CSV.
+
+Re-designed for one 8xMI355X node: one process per GPU over RCCL/xGMI
+(``RingComm``), the decode step on each stage either replayed as a hipGraph
+(``HipRunner``) or run by the torch stage module (CPU tests use gloo with
+the same code).  Standalone (1 node) short-circuits the ring exactly like
+the reference's queue aliasing (gptserver.py:276-278).
+"""
+
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+
+from ..models.sampling import sample as sample_token
+from .ring import RingComm
+
+__all__ = ["PipelineRuntime", "GenerationResult", "SamplingParams"]
+
+
+@dataclass
+class SamplingParams:
+    temperature: float = 0.8
+    top_k: Optional[int] = 200
+    top_p: float = 1.0
+    seed: Optional[int] = None
+
+
+@dataclass
+class GenerationResult:
+    sequences: List[torch.Tensor] = field(default_factory=list)
+    tok_time: List[Tuple[int, float]] = field(default_factory=list)
+    gen_time: float = 0.0
+    total_new_tokens: int = 0
+
+    @property
+    def tokens_per_second(self) -> float:
+        return self.total_new_tokens / self.gen_time if self.gen_time else 0.0
+
+
+class PipelineRuntime:
+    """One rank of the ring (or the single standalone node)."""
+
+    def __init__(
+        self,
+        runner,
+        rank: int = 0,
+        world: int = 1,
+        comm: Optional[RingComm] = None,
+        device: Optional[torch.device] = None,
+    ) -> None:
+        self.runner = runner
+        self.rank = rank
+        self.world = world
+        self.comm = comm
+        self.is_starter = rank == 0
+        self.device = device or torch.device("cpu")
+        if world > 1:
+            assert comm is not None
+
+    # ------------------------------------------------------------------
+    # starter
+    # ------------------------------------------------------------------
+    def generate(
+        self,
+        prompts: Sequence[torch.Tensor],
+        max_new_tokens: int,
+        sampling: SamplingParams = SamplingParams(),
+        stop_tokens: Sequence[Sequence[int]] = (),
+        token_callback=None,
+    ) -> GenerationResult:
+        """Starter entry: run the recurrent pipeline until every sample
+        produced ``max_new_tokens`` tokens (or hit a stop sequence)."""
+        assert self.is_starter, "generate() runs on the starter"
+        n_samples = len(prompts)
+        runner = self.runner
+        gens = self._generators(sampling, n_samples, self.device)
+
+        res = GenerationResult()
+        seqs: List[List[int]] = [list(map(int, p.tolist())) for p in prompts]
+        new_counts = [0] * n_samples
+        active = [True] * n_samples
+        pending_x = {}
+
+        t_start = time.perf_counter()
+        # ---- prefill: seed every sample into the ring --------------------
+        for s, prompt in enumerate(prompts):
+            ptoks = prompt.to(self.device)
+            x = runner.prefill_head(ptoks, s)
+            if self.world > 1:
+                self.comm.send(s, x, stop=False)
+            else:
+                pending_x[s] = x
+        order = list(range(n_samples))
+
+        # ---- main loop ---------------------------------------------------
+        n_active = n_samples
+        total_new = 0
+        while n_active > 0:
+            if self.world > 1:
+                s, x, stop = self.comm.recv()
+                if stop:
+                    n_active -= 1
+                    continue
+            else:
+                s = order.pop(0)
+                x = pending_x.pop(s)
+            # tail: logits for the last position of this sample
+            x2 = x.view(-1, x.size(-1))
+            logits = runner.tail(x2[-1])
+            tok = sample_token(
+                logits,
+                temperature=sampling.temperature,
+                top_k=sampling.top_k,
+                top_p=sampling.top_p,
+                generator=gens[s],
+            )
+            itok = int(tok)
+            seqs[s].append(itok)
+            new_counts[s] += 1
+            total_new += 1
+            res.tok_time.append((total_new, time.perf_counter() - t_start))
+            if token_callback is not None:
+                token_callback(s, itok)
+
+            done = (
+                new_counts[s] >= max_new_tokens
+                or self._hit_stop(seqs[s], new_counts[s], stop_tokens)
+                or runner.pos[s] + 1 >= self.runner.stage.max_seq_length
+            )
+            if done:
+                active[s] = False
+                if self.world > 1:
+                    self.comm.send(s, None, stop=True)  # travels the ring
+                else:
+                    n_active -= 1
+            else:
+                x = runner.decode_head(tok.view(1).to(self.device), s)
+                if self.world > 1:
+                    self.comm.send(s, x, stop=False)
+                else:
+                    order.append(s)
+                    pending_x[s] = x.clone() if self._needs_clone() else x
+
+        if self.world > 1:
+            self.comm.drain()
+        res.gen_time = time.perf_counter() - t_start
+        res.total_new_tokens = total_new
+        res.sequences = [torch.tensor(s, dtype=torch.int64) for s in seqs]
+        return res
+
+    def _needs_clone(self) -> bool:
+        # HIP runner returns a view of the engine's x buffer; standalone
+        # loopback must snapshot it (the ring path copies into send bufs).
+        return getattr(self.runner, "backend", "") == "hip"
+
+    @staticmethod
+    def _generators(sampling: SamplingParams, n: int, device):
+        if sampling.seed is None:
+            return [None] * n
+        gens = []
+        for s in range(n):
+            g = torch.Generator(device=device)
+            g.manual_seed(sampling.seed + s)
+            gens.append(g)
+        return gens
+
+    @staticmethod
+    def _hit_stop(seq: List[int], n_new: int, stop_tokens) -> bool:
+        for st in stop_tokens:
+            n = len(st)
+            if 0 < n <= n_new and seq[-n:] == list(st):
+                return True
+        return False
+
+    # ------------------------------------------------------------------
+    # secondary
+    # ------------------------------------------------------------------
+    def serve(self) -> int:
+        """Secondary entry: process messages until every sample stopped.
+        Returns the number of activation messages processed."""
+        assert not self.is_starter
+        runner = self.runner
+        seen: set = set()
+        stopped: set = set()
+        processed = 0
+        while True:
+            s, x, stop = self.comm.recv()
+            if stop:
+                stopped.add(s)
+                self.comm.send(s, None, stop=True)  # forward along the ring
+                if seen and stopped >= seen:
+                    break
+                if not seen:
+                    break
+                continue
+            seen.add(s)
+            if x.size(0) > 1:
+                out = runner.prefill_mid(x, s)
+            else:
+                out = runner.decode_mid(x[0], s)
+            self.comm.send(s, out, stop=False)
+            processed += 1
+        self.comm.drain()
+        return processed
