@@ -1,0 +1,185 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Llama-3-8B recurrent-pipeline decode on N MI355X GPUs.
+
+Measures the BASELINE.json metric — generation throughput (tokens/sec,
+whole job) of Llama-3-8B over N pipeline stages with N concurrent samples —
+on synthetic prompts and random-init weights (no network for checkpoints).
+
+Contract (driver):
+  python bench.py --gpus N --steps K --warmup W
+N>1 is launched via torch.distributed.run with one rank per GPU (RCCL).
+One decode "step" = one full pipeline rotation = every sample advances one
+token (N tokens whole-job per step).  Warmup = prefill + W untimed
+rotations.  The timed region is exactly K rotations bracketed by
+barrier + torch.cuda.synchronize() on both sides; the reported time is the
+MAX over ranks.  Rank 0 prints ONE JSON line.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=24)
+    ap.add_argument("--warmup", type=int, default=4)
+    ap.add_argument("--model", default="Meta-Llama-3-8B-Instruct")
+    ap.add_argument("--prompt-len", type=int, default=128)
+    ap.add_argument("--samples", type=int, default=0,
+                    help="in-flight samples (default = n stages)")
+    ap.add_argument("--seq-len", type=int, default=2048,
+                    help="max sequence length (KV budget)")
+    ap.add_argument("--backend", choices=["hip", "torch"], default="hip")
+    ap.add_argument("--no-graphs", action="store_true")
+    return ap.parse_args()
+
+
+def log(msg):
+    print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main():
+    args = parse_args()
+    from mdi_llm_amd.config import ModelConfig
+    from mdi_llm_amd.models.stages import build_stage
+    from mdi_llm_amd.parallel.ring import RingComm
+    from mdi_llm_amd.parallel.runner import make_runner
+    from mdi_llm_amd.parallel.runtime import PipelineRuntime, SamplingParams
+
+    # ---- distributed setup ---------------------------------------------
+    rank = int(os.environ.get("RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    if world == 1 and args.gpus > 1:
+        log("WARNING: --gpus>1 but WORLD_SIZE=1; launch via torchrun")
+    n_stages = world
+    assert torch.cuda.is_available(), "bench requires a GPU"
+    device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+    torch.cuda.set_device(device)
+
+    import torch.distributed as dist
+
+    cpu_group = None
+    if world > 1:
+        dist.init_process_group("nccl")
+        cpu_group = dist.new_group(backend="gloo")
+
+    n_samples = args.samples or max(n_stages, 1)
+    cfg = ModelConfig.from_name(args.model)
+
+    # ---- build this rank's stage (random init, bf16, on device) ---------
+    from mdi_llm_amd.utils import layer_split
+
+    split = layer_split(cfg.n_layer, n_stages)
+    t0 = time.time()
+    torch.manual_seed(1234 + rank)
+    stage = build_stage(cfg, rank, split[rank])
+    stage = stage.to(device=device, dtype=torch.bfloat16)
+    with torch.no_grad():
+        for p in stage.parameters():
+            p.normal_(0.0, 0.02)
+    stage.max_seq_length = min(args.seq_len, cfg.block_size)
+    stage.eval()
+    log(f"rank {rank}: stage built ({split[rank]} layers) "
+        f"in {time.time()-t0:.1f}s")
+
+    runner = make_runner(
+        stage, n_samples, device, use_graphs=not args.no_graphs,
+        force_torch=args.backend == "torch",
+    )
+    log(f"rank {rank}: runner backend={runner.backend}")
+    if args.backend == "hip" and runner.backend != "hip":
+        raise RuntimeError(
+            f"HIP decode engine unavailable for {cfg.name} — refusing to "
+            "silently bench the torch fallback"
+        )
+
+    comm = None
+    if world > 1:
+        comm = RingComm(cfg.n_embd, stage.max_seq_length, device, n_samples)
+    rt = PipelineRuntime(runner, rank=rank, world=world, comm=comm,
+                         device=device)
+
+    def sync_barrier():
+        torch.cuda.synchronize()
+        if world > 1:
+            dist.barrier(group=cpu_group)
+            torch.cuda.synchronize()
+
+    sampling = SamplingParams(temperature=0.8, top_k=200, seed=1234)
+    gens = rt._generators(sampling, n_samples, device)
+
+    # ---- warmup: prefill + W rotations -----------------------------------
+    torch.manual_seed(99)
+    prompts = [
+        torch.randint(0, cfg.vocab_size - 1, (args.prompt_len,))
+        for _ in range(n_samples)
+    ]
+    t0 = time.time()
+    if rank == 0:
+        toks = rt.bench_prefill(prompts)
+        if args.warmup > 0:
+            toks = rt.bench_decode_rounds(toks, args.warmup, sampling, gens)
+    else:
+        rt.bench_serve_prefill(n_samples)
+        if args.warmup > 0:
+            rt.bench_serve_rounds(n_samples, args.warmup)
+    sync_barrier()
+    log(f"rank {rank}: warmup done in {time.time()-t0:.1f}s")
+
+    # ---- timed region: exactly K rotations -------------------------------
+    t_start = time.perf_counter()
+    if rank == 0:
+        toks = rt.bench_decode_rounds(toks, args.steps, sampling, gens)
+    else:
+        rt.bench_serve_rounds(n_samples, args.steps)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t_start
+    sync_barrier()
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX, group=cpu_group)
+        elapsed = float(t[0])
+
+    if rank == 0:
+        total_tokens = n_samples * args.steps
+        tps = total_tokens / elapsed
+        out = {
+            "metric": "llama3_8b_pipeline_decode_tok_per_s",
+            "value": round(tps, 2),
+            "unit": "tokens/s",
+            "n_gpus": n_stages,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": cfg.name,
+                "global_batch": n_samples,
+                "seq_len": args.prompt_len,
+                "max_seq_len": stage.max_seq_length,
+                "parallelism": f"pp{n_stages}",
+                "samples_in_flight": n_samples,
+                "backend": runner.backend,
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if world > 1:
+        dist.barrier(group=cpu_group)
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

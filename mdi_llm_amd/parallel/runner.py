@@ -52,7 +52,8 @@ class TorchRunner:
     @torch.inference_mode()
     def decode_head(self, token: torch.Tensor, slot: int) -> torch.Tensor:
         p = self.pos[slot]
-        x = self.stage.forward_head(token.view(1, 1), slot=slot, input_pos=p)
+        x = self.stage.forward_head(token.long().view(1, 1), slot=slot,
+                                    input_pos=p)
         self.pos[slot] = p + 1
         return x.view(-1)  # (n_embd,)
 

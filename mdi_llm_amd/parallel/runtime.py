@@ -184,6 +184,92 @@ class PipelineRuntime:
         return False
 
     # ------------------------------------------------------------------
+    # bench phases (deterministic schedule, headerless ring, no host syncs
+    # in the decode loop; each phase leaves the ring drained so callers can
+    # barrier/synchronize between phases)
+    # ------------------------------------------------------------------
+    def bench_prefill(self, prompts: Sequence[torch.Tensor]) -> list:
+        """Starter: prefill all samples; returns per-sample device tokens
+        (argmax of the first logits — value irrelevant for timing)."""
+        runner = self.runner
+        n = len(prompts)
+        xs = []
+        for s, p in enumerate(prompts):
+            x = runner.prefill_head(p.to(self.device), s)
+            if self.world > 1:
+                self.comm.send(s, x, stop=False)
+            else:
+                xs.append(x)
+        toks = []
+        for s in range(n):
+            if self.world > 1:
+                _, x, _ = self.comm.recv()
+            else:
+                x = xs[s]
+            logits = runner.tail(x.view(-1, x.size(-1))[-1])
+            toks.append(logits.float().argmax().view(1).to(torch.int32))
+        return toks
+
+    def bench_serve_prefill(self, n_samples: int) -> None:
+        for _ in range(n_samples):
+            s, x, _ = self.comm.recv()
+            out = self.runner.prefill_mid(x, s)
+            self.comm.send(s, out, stop=False)
+        self.comm.drain()
+
+    def bench_decode_rounds(self, toks: list, n_rounds: int,
+                            sampling: SamplingParams, gens=None) -> list:
+        """Starter: n_rounds full rotations; every sample advances one token
+        per rotation.  The final rotation does not re-seed, so the ring is
+        empty on return."""
+        runner = self.runner
+        n = len(toks)
+        if gens is None:
+            gens = self._generators(sampling, n, self.device)
+        if self.world == 1:
+            for _ in range(n_rounds):
+                for s in range(n):
+                    x = runner.decode_head(toks[s], s)
+                    logits = runner.tail(x)
+                    toks[s] = self._draw(logits, sampling, gens[s])
+            return toks
+        # seed one in-flight message per sample
+        for s in range(n):
+            x = runner.decode_head(toks[s], s)
+            self.comm.send_sched(s, x)
+        for r in range(n_rounds):
+            last = r == n_rounds - 1
+            for s in range(n):
+                x = self.comm.recv_sched()
+                logits = runner.tail(x)
+                toks[s] = self._draw(logits, sampling, gens[s])
+                if not last:
+                    x = runner.decode_head(toks[s], s)
+                    self.comm.send_sched(s, x)
+        self.comm.drain()
+        return toks
+
+    def bench_serve_rounds(self, n_samples: int, n_rounds: int) -> None:
+        """Secondary: the matching deterministic message count."""
+        for r in range(n_rounds):
+            for s in range(n_samples):
+                x = self.comm.recv_sched()
+                out = self.runner.decode_mid(x, s)
+                self.comm.send_sched(s, out)
+        self.comm.drain()
+
+    @staticmethod
+    def _draw(logits, sampling: SamplingParams, gen):
+        tok = sample_token(
+            logits,
+            temperature=sampling.temperature,
+            top_k=sampling.top_k,
+            top_p=sampling.top_p,
+            generator=gen,
+        )
+        return tok.view(1).to(torch.int32)
+
+    # ------------------------------------------------------------------
     # secondary
     # ------------------------------------------------------------------
     def serve(self) -> int:
