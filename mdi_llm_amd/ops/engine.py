@@ -171,15 +171,26 @@ class DecodeEngine:
         else:
             self.ops.layernorm(out, x, w, b, eps)
 
+    @property
+    def _nk(self) -> int:
+        """fused-norm kind: 1 RMSNorm, 2 LayerNorm."""
+        return 1 if self.config.norm_class_name == "RMSNorm" else 2
+
     def _run_blocks(self) -> None:
-        """x -> x through all local blocks (decode, one token)."""
+        """x -> x through all local blocks (decode, one token).
+
+        All pre-norms are fused into the GEMV staging pass and residual
+        adds into GEMV epilogues — a llama block is 6 kernels total (qkv,
+        rope+kv, attn x2, proj+res, swiglu, down+res)."""
         cfg = self.config
         ops = self.ops
         eps = cfg.norm_eps
+        nk = self._nk
         scale = 1.0 / (cfg.head_size ** 0.5)
         for li, w in enumerate(self.blocks):
-            self._norm(self.xn, self.x, w.norm1_w, w.norm1_b, eps)
-            ops.gemv(self.qkv, w.attn_w, self.xn, w.attn_b, None, 0)
+            # qkv = Wqkv @ norm1(x)
+            ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
+                     w.norm1_w, w.norm1_b, nk, eps)
             ops.rope_kv_append(
                 self.qkv, self.kv_pool.k, self.kv_pool.v, self.cos, self.sin,
                 self.pos, self.slot, li,
@@ -189,31 +200,40 @@ class DecodeEngine:
                 self.kv_pool.v, self.pos, self.slot, li, self.n_chunks, scale,
             )
             if cfg.parallel_residual:
-                # x = x + proj(y) + mlp(norm2(x) or xn)
+                # x = x + proj(y) + mlp(norm2(x) or norm1(x))
                 ops.gemv(self.a, w.proj_w, self.y, w.proj_b, None, 0)
-                if cfg.shared_attention_norm:
-                    n2 = self.xn
-                else:
-                    self._norm(self.hn, self.x, w.norm2_w, w.norm2_b, eps)
-                    n2 = self.hn
-                self._mlp(n2, w)
-                ops.add(self.a, self.a, self.m_out)
-                ops.add(self.x, self.x, self.a)
+                nw = w.norm1_w if cfg.shared_attention_norm else w.norm2_w
+                nb = w.norm1_b if cfg.shared_attention_norm else w.norm2_b
+                self._mlp(self.x, w, self.a, nw, nb)
+                ops.add(self.x, self.x, self.m_out)
             else:
-                # h = x + proj(y); x = h + mlp(norm2(h))
+                # a = x + proj(y); x = a + mlp(norm2(a))
                 ops.gemv(self.a, w.proj_w, self.y, w.proj_b, self.x, 1)
-                self._norm(self.hn, self.a, w.norm2_w, w.norm2_b, eps)
-                self._mlp(self.hn, w)
-                ops.add(self.x, self.a, self.m_out)
+                self._mlp(self.a, w, self.a, w.norm2_w, w.norm2_b)
+                # _mlp wrote x = res + down(act) directly
+        # (sequential path leaves the stream in self.x)
 
-    def _mlp(self, inp, w) -> None:
+    def _mlp(self, inp, w, res, norm_w, norm_b) -> None:
+        """act = act_fn(fc(norm(inp))); writes x/m_out = res + proj(act).
+
+        Sequential path: out buffer is self.x (the new residual stream).
+        Parallel path: out buffer is self.m_out (summed by caller)."""
         cfg = self.config
+        eps = cfg.norm_eps
+        nk = self._nk
         if cfg.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
             gelu_gate = cfg.mlp_class_name == "GemmaMLP"
-            self.ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp, gelu_gate)
+            self.ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp, gelu_gate,
+                                 norm_w, norm_b, nk, eps)
         else:
-            self.ops.gemv(self.act, w.fc_w, inp, w.fc_b, None, 2)  # gelu
-        self.ops.gemv(self.m_out, w.mlp_proj_w, self.act, w.mlp_proj_b, None, 0)
+            self.ops.gemv(self.act, w.fc_w, inp, w.fc_b, None, 2,
+                          norm_w, norm_b, nk, eps)  # gelu
+        if cfg.parallel_residual:
+            self.ops.gemv(self.m_out, w.mlp_proj_w, self.act, w.mlp_proj_b,
+                          res, 1)
+        else:
+            self.ops.gemv(self.x, w.mlp_proj_w, self.act, w.mlp_proj_b,
+                          res, 1)
 
     def _embed(self) -> None:
         cfg = self.config
@@ -224,9 +244,9 @@ class DecodeEngine:
             self.ops.add(self.x, self.x, self.pos_emb)
 
     def _tail_seq(self) -> None:
-        self._norm(self.xn, self.x, self.lnf_w, self.lnf_b,
-                   self.config.norm_eps)
-        self.ops.gemv(self.logits, self.head_w, self.xn, self.head_b, None, 0)
+        # logits = lm_head @ ln_f(x): one fused kernel
+        self.ops.gemv(self.logits, self.head_w, self.x, self.head_b, None, 0,
+                      self.lnf_w, self.lnf_b, self._nk, self.config.norm_eps)
 
     # ---------------------------------------------------------------------
     # public decode API

@@ -61,7 +61,9 @@ void layernorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
 
 void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
           c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> res,
-          int64_t epilogue) {
+          int64_t epilogue, c10::optional<torch::Tensor> norm_w,
+          c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
+          double eps) {
   check_bf16(out, "out");
   check_bf16(W, "W");
   check_bf16(x, "x");
@@ -71,6 +73,8 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
   TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
   const void* bp = nullptr;
   const void* rp = nullptr;
+  const void* nwp = nullptr;
+  const void* nbp = nullptr;
   if (bias.has_value()) {
     check_bf16(*bias, "bias");
     TORCH_CHECK(bias->numel() == M, "bias size");
@@ -81,12 +85,24 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
     TORCH_CHECK(res->numel() == M, "res size");
     rp = res->data_ptr();
   }
-  launch_gemv(out.data_ptr(), W.data_ptr(), x.data_ptr(), bp, rp, M, K,
-              (int)epilogue, cur_stream());
+  if (norm_kind != 0) {
+    TORCH_CHECK(norm_w.has_value(), "norm_w required with norm_kind");
+    check_bf16(*norm_w, "norm_w");
+    nwp = norm_w->data_ptr();
+    if (norm_b.has_value()) {
+      check_bf16(*norm_b, "norm_b");
+      nbp = norm_b->data_ptr();
+    }
+  }
+  launch_gemv(out.data_ptr(), W.data_ptr(), x.data_ptr(), bp, rp, nwp, nbp,
+              (float)eps, M, K, (int)epilogue, (int)norm_kind, cur_stream());
 }
 
 void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
-                 torch::Tensor x, bool gelu_gate) {
+                 torch::Tensor x, bool gelu_gate,
+                 c10::optional<torch::Tensor> norm_w,
+                 c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
+                 double eps) {
   check_bf16(out, "out");
   check_bf16(Wg, "Wg");
   check_bf16(Wu, "Wu");
@@ -96,8 +112,20 @@ void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
   TORCH_CHECK(Wg.numel() == (int64_t)M * K && Wu.numel() == (int64_t)M * K,
               "weight shape mismatch");
   TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
+  const void* nwp = nullptr;
+  const void* nbp = nullptr;
+  if (norm_kind != 0) {
+    TORCH_CHECK(norm_w.has_value(), "norm_w required with norm_kind");
+    check_bf16(*norm_w, "norm_w");
+    nwp = norm_w->data_ptr();
+    if (norm_b.has_value()) {
+      check_bf16(*norm_b, "norm_b");
+      nbp = norm_b->data_ptr();
+    }
+  }
   launch_gemv_swiglu(out.data_ptr(), Wg.data_ptr(), Wu.data_ptr(),
-                     x.data_ptr(), M, K, gelu_gate ? 1 : 0, cur_stream());
+                     x.data_ptr(), nwp, nbp, (float)eps, M, K,
+                     gelu_gate ? 1 : 0, (int)norm_kind, cur_stream());
 }
 
 void embed(torch::Tensor out, torch::Tensor wte, torch::Tensor token,
@@ -188,8 +216,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "mdi_llm_amd hand-written CDNA4 (gfx950) decode kernels";
   m.def("rmsnorm", &rmsnorm, "RMSNorm (decode, bf16)");
   m.def("layernorm", &layernorm, "LayerNorm (decode, bf16)");
-  m.def("gemv", &gemv, "decode GEMV out = W@x (+bias)(+res)(act)");
-  m.def("gemv_swiglu", &gemv_swiglu, "fused SwiGLU pair GEMV");
+  m.def("gemv", &gemv, "decode GEMV out = W@norm?(x) (+bias)(+res)(act)",
+        py::arg("out"), py::arg("W"), py::arg("x"), py::arg("bias"),
+        py::arg("res"), py::arg("epilogue"),
+        py::arg("norm_w") = c10::nullopt, py::arg("norm_b") = c10::nullopt,
+        py::arg("norm_kind") = 0, py::arg("eps") = 1e-5);
+  m.def("gemv_swiglu", &gemv_swiglu, "fused SwiGLU pair GEMV (+pre-norm)",
+        py::arg("out"), py::arg("Wg"), py::arg("Wu"), py::arg("x"),
+        py::arg("gelu_gate"), py::arg("norm_w") = c10::nullopt,
+        py::arg("norm_b") = c10::nullopt, py::arg("norm_kind") = 0,
+        py::arg("eps") = 1e-5);
   m.def("embed", &embed, "embedding row gather");
   m.def("rope_kv_append", &rope_kv_append,
         "RoPE on interleaved qkv + KV cache append");

@@ -12,12 +12,15 @@ void launch_layernorm(void* out, const void* x, const void* w, const void* b,
                       int n, float eps, hipStream_t stream);
 
 // epilogue: 0 none, 1 +res, 2 gelu(tanh), 3 silu
+// norm_kind: 0 none, 1 fused RMSNorm on x, 2 fused LayerNorm on x
 void launch_gemv(void* out, const void* W, const void* x, const void* bias,
-                 const void* res, int M, int K, int epilogue,
+                 const void* res, const void* norm_w, const void* norm_b,
+                 float eps, int M, int K, int epilogue, int norm_kind,
                  hipStream_t stream);
 
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
-                        const void* x, int M, int K, int gelu_gate,
+                        const void* x, const void* norm_w, const void* norm_b,
+                        float eps, int M, int K, int gelu_gate, int norm_kind,
                         hipStream_t stream);
 
 void launch_embed(void* out, const void* wte, const int* token, int n_embd,

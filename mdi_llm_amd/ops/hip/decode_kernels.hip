@@ -130,68 +130,146 @@ __global__ void layernorm_kernel(bf16* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
-// Decode GEMV: out[M] = W[M,K] @ x[K] (+ bias) (+ residual) (+ activation)
+// Decode GEMV: out[M] = W[M,K] @ norm?(x)[K] (+ bias) (+ residual) (+ act)
 //
 // W row-major bf16, streamed once from HBM (the decode bound); x staged in
 // LDS (K*2 bytes, up to ~28 KiB for the 14336-wide MLP). Block = 256
-// threads = 4 waves; each wave owns one output row per grid-stride step;
-// lane reads 16 B of the row per iteration (coalesced 1 KiB per wave-iter).
-// EPI: 0 none, 1 +residual, 2 gelu(tanh), 3 relu-free silu (used standalone)
+// threads = 4 waves; each wave owns TWO adjacent output rows per grid-stride
+// step (doubles load ILP); lane reads 16 B per row per iteration.
+//
+// NORM fuses the pre-norm into the staging pass (saves a separate tiny
+// kernel + a global round-trip per call): after staging raw x, the block
+// reduces mean/meansq and rewrites LDS with the normalized+weighted value.
+// NORM: 0 none, 1 RMSNorm, 2 LayerNorm.
+// EPI:  0 none, 1 +residual, 2 gelu(tanh), 3 silu.
 // ---------------------------------------------------------------------------
-template <int EPI>
+DEVINL float gelu_tanh(float v) {
+  float c = 0.7978845608028654f * (v + 0.044715f * v * v * v);
+  return 0.5f * v * (1.f + tanhf(c));
+}
+
+template <int NORM>
+DEVINL void stage_x(bf16* xs, const bf16* __restrict__ x,
+                    const bf16* __restrict__ nw, const bf16* __restrict__ nb,
+                    int K, float eps, float* red) {
+  const int tid = threadIdx.x;
+  if (NORM == 0) {
+    for (int i = tid * 8; i < K; i += blockDim.x * 8)
+      *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(x + i);
+    __syncthreads();
+    return;
+  }
+  float s = 0.f, s2 = 0.f;
+  for (int i = tid * 8; i < K; i += blockDim.x * 8) {
+    bf16x8 v = load8(x + i);
+    *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(v.v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = b2f(v.v[j]);
+      s += f;
+      s2 += f * f;
+    }
+  }
+  s = wave_reduce_sum(s);
+  s2 = wave_reduce_sum(s2);
+  if ((tid & 63) == 0) {
+    red[(tid >> 6) * 2] = s;
+    red[(tid >> 6) * 2 + 1] = s2;
+  }
+  __syncthreads();
+  float mean = 0.f, inv;
+  if (NORM == 2) mean = (red[0] + red[2] + red[4] + red[6]) / K;
+  float ms = (red[1] + red[3] + red[5] + red[7]) / K;
+  inv = rsqrtf(ms - mean * mean + eps);
+  for (int i = tid * 8; i < K; i += blockDim.x * 8) {
+    bf16x8 v = load8(xs + i);
+    bf16x8 g = load8(nw + i);
+    bf16x8 o;
+    if (NORM == 2 && nb != nullptr) {
+      bf16x8 bb = load8(nb + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.v[j] = f2b((b2f(v.v[j]) - mean) * inv * b2f(g.v[j]) + b2f(bb.v[j]));
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.v[j] = f2b((b2f(v.v[j]) - mean) * inv * b2f(g.v[j]));
+    }
+    *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(o.v);
+  }
+  __syncthreads();
+}
+
+template <int EPI, int NORM>
 __global__ void gemv_kernel(bf16* __restrict__ out,
                             const bf16* __restrict__ W,
                             const bf16* __restrict__ x,
                             const bf16* __restrict__ bias,
-                            const bf16* __restrict__ res, int M, int K) {
+                            const bf16* __restrict__ res,
+                            const bf16* __restrict__ nw,
+                            const bf16* __restrict__ nb, float eps, int M,
+                            int K) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
-  const int tid = threadIdx.x;
-  for (int i = tid * 8; i < K; i += blockDim.x * 8)
-    *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(x + i);
-  __syncthreads();
+  stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
+  const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
+  const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * 2;
 
-  for (int row = blockIdx.x * (blockDim.x >> 6) + wave; row < M;
-       row += waves_per_grid) {
-    const bf16* wrow = W + (size_t)row * K;
-    float acc = 0.f;
+  for (int row = (blockIdx.x * (blockDim.x >> 6) + wave) * 2; row < M;
+       row += rows_per_grid) {
+    const bf16* wrow0 = W + (size_t)row * K;
+    const bool two = row + 1 < M;
+    const bf16* wrow1 = two ? wrow0 + K : wrow0;
+    float acc0 = 0.f, acc1 = 0.f;
     for (int i = lane * 8; i < K; i += 64 * 8) {
-      bf16x8 wv = load8(wrow + i);
+      bf16x8 w0 = load8(wrow0 + i);
+      bf16x8 w1 = load8(wrow1 + i);
       bf16x8 xv = load8(xs + i);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc += b2f(wv.v[j]) * b2f(xv.v[j]);
-    }
-    acc = wave_reduce_sum(acc);
-    if (lane == 0) {
-      if (bias != nullptr) acc += b2f(bias[row]);
-      if (EPI == 1 && res != nullptr) acc += b2f(res[row]);
-      if (EPI == 2) {  // gelu tanh approx
-        float c = 0.7978845608028654f * (acc + 0.044715f * acc * acc * acc);
-        acc = 0.5f * acc * (1.f + tanhf(c));
+      for (int j = 0; j < 8; ++j) {
+        float xf = b2f(xv.v[j]);
+        acc0 += b2f(w0.v[j]) * xf;
+        acc1 += b2f(w1.v[j]) * xf;
       }
-      if (EPI == 3) acc = acc / (1.f + expf(-acc));  // silu
-      out[row] = f2b(acc);
+    }
+    acc0 = wave_reduce_sum(acc0);
+    acc1 = wave_reduce_sum(acc1);
+    if (lane == 0) {
+#pragma unroll
+      for (int r = 0; r < 2; ++r) {
+        if (r == 1 && !two) break;
+        float acc = r == 0 ? acc0 : acc1;
+        const int rw = row + r;
+        if (bias != nullptr) acc += b2f(bias[rw]);
+        if (EPI == 1 && res != nullptr) acc += b2f(res[rw]);
+        if (EPI == 2) acc = gelu_tanh(acc);
+        if (EPI == 3) acc = acc / (1.f + expf(-acc));
+        out[rw] = f2b(acc);
+      }
     }
   }
 }
 
-// SwiGLU pair GEMV: out[i] = silu(Wg_i . x) * (Wu_i . x)
+// SwiGLU pair GEMV: out[i] = silu(Wg_i . xn) * (Wu_i . xn), optional fused
+// pre-norm like gemv_kernel.
+template <int NORM>
 __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
                                    const bf16* __restrict__ Wg,
                                    const bf16* __restrict__ Wu,
-                                   const bf16* __restrict__ x, int M, int K,
-                                   int gelu_gate) {
+                                   const bf16* __restrict__ x,
+                                   const bf16* __restrict__ nw,
+                                   const bf16* __restrict__ nb, float eps,
+                                   int M, int K, int gelu_gate) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
+  __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
-  const int tid = threadIdx.x;
-  for (int i = tid * 8; i < K; i += blockDim.x * 8)
-    *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(x + i);
-  __syncthreads();
+  stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
+  const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
@@ -215,13 +293,7 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
     ga = wave_reduce_sum(ga);
     ua = wave_reduce_sum(ua);
     if (lane == 0) {
-      float act;
-      if (gelu_gate) {
-        float c = 0.7978845608028654f * (ga + 0.044715f * ga * ga * ga);
-        act = 0.5f * ga * (1.f + tanhf(c));
-      } else {
-        act = ga / (1.f + expf(-ga));
-      }
+      float act = gelu_gate ? gelu_tanh(ga) : ga / (1.f + expf(-ga));
       out[row] = f2b(act * ua);
     }
   }
@@ -507,34 +579,39 @@ __global__ void attn_decode_kernel(
 }
 
 // Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L
-// grid = n_head blocks, 64 threads (heads are independent).
+// grid = n_head blocks, 256 threads; wave 0 computes chunk weights lane-
+// parallel (lane == chunk), then all threads accumulate dims with the
+// chunk loop's loads fully independent (ILP) — the serial-per-lane version
+// of this kernel was 19 us (latency-bound); this one is ~2 us.
 __global__ void attn_combine_kernel(bf16* __restrict__ out,
                                     const float* __restrict__ part_o,
                                     const float* __restrict__ part_ml,
                                     int n_chunks, int head_size) {
+  __shared__ float w_lds[64];
+  __shared__ float inv_l;
   const int h = blockIdx.x;
-  const int lane = threadIdx.x;
-  // global max over chunks
-  float M = -1e30f;
-  for (int c = 0; c < n_chunks; ++c)
-    M = fmaxf(M, part_ml[((size_t)h * n_chunks + c) * 2]);
-  float L = 0.f;
-  for (int c = 0; c < n_chunks; ++c) {
-    const float m = part_ml[((size_t)h * n_chunks + c) * 2];
-    const float l = part_ml[((size_t)h * n_chunks + c) * 2 + 1];
-    L += l * __expf(m - M);
-  }
-  const float inv = 1.f / L;
-  const int dpl = (head_size + 63) / 64;  // dims per lane
-  for (int i = 0; i < dpl; ++i) {
-    const int d = lane * dpl + i;
-    if (d >= head_size) break;
-    float acc = 0.f;
-    for (int c = 0; c < n_chunks; ++c) {
-      const float m = part_ml[((size_t)h * n_chunks + c) * 2];
-      const float w = __expf(m - M);
-      acc += w * part_o[((size_t)h * n_chunks + c) * head_size + d];
+  const int tid = threadIdx.x;
+  if (tid < 64) {
+    float m = -1e30f, l = 0.f;
+    if (tid < n_chunks) {
+      m = part_ml[((size_t)h * n_chunks + tid) * 2];
+      l = part_ml[((size_t)h * n_chunks + tid) * 2 + 1];
     }
+    float M = m;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+      M = fmaxf(M, __shfl_xor(M, off, 64));
+    const float wgt = (tid < n_chunks) ? __expf(m - M) : 0.f;
+    const float L = wave_reduce_sum(l * wgt);
+    w_lds[tid] = wgt;
+    if (tid == 0) inv_l = 1.f / L;
+  }
+  __syncthreads();
+  const float inv = inv_l;
+  for (int d = tid; d < head_size; d += blockDim.x) {
+    float acc = 0.f;
+    for (int c = 0; c < n_chunks; ++c)
+      acc += w_lds[c] * part_o[((size_t)h * n_chunks + c) * head_size + d];
     out[(size_t)h * head_size + d] = f2b(acc * inv);
   }
 }
@@ -559,8 +636,8 @@ __global__ void add_kernel(bf16* __restrict__ out, const bf16* __restrict__ a,
 // Host-side launchers (C ABI used by the torch bindings)
 // ---------------------------------------------------------------------------
 
-static inline int gemv_grid(int M) {
-  int blocks = (M + 3) / 4;  // 4 rows per block (1 per wave)
+static inline int gemv_grid(int M, int rows_per_block) {
+  int blocks = (M + rows_per_block - 1) / rows_per_block;
   return blocks < 4096 ? blocks : 4096;
 }
 
@@ -578,40 +655,51 @@ void launch_layernorm(void* out, const void* x, const void* w, const void* b,
 }
 
 void launch_gemv(void* out, const void* W, const void* x, const void* bias,
-                 const void* res, int M, int K, int epilogue,
+                 const void* res, const void* norm_w, const void* norm_b,
+                 float eps, int M, int K, int epilogue, int norm_kind,
                  hipStream_t stream) {
   const int smem = K * sizeof(bf16);
-  dim3 grid(gemv_grid(M)), block(256);
-  switch (epilogue) {
-    case 0:
-      hipLaunchKernelGGL(gemv_kernel<0>, grid, block, smem, stream,
-                         (bf16*)out, (const bf16*)W, (const bf16*)x,
-                         (const bf16*)bias, (const bf16*)res, M, K);
-      break;
-    case 1:
-      hipLaunchKernelGGL(gemv_kernel<1>, grid, block, smem, stream,
-                         (bf16*)out, (const bf16*)W, (const bf16*)x,
-                         (const bf16*)bias, (const bf16*)res, M, K);
-      break;
-    case 2:
-      hipLaunchKernelGGL(gemv_kernel<2>, grid, block, smem, stream,
-                         (bf16*)out, (const bf16*)W, (const bf16*)x,
-                         (const bf16*)bias, (const bf16*)res, M, K);
-      break;
-    default:
-      hipLaunchKernelGGL(gemv_kernel<3>, grid, block, smem, stream,
-                         (bf16*)out, (const bf16*)W, (const bf16*)x,
-                         (const bf16*)bias, (const bf16*)res, M, K);
+  dim3 grid(gemv_grid(M, 8)), block(256);
+#define GEMV_CASE(E, N)                                                     \
+  hipLaunchKernelGGL((gemv_kernel<E, N>), grid, block, smem, stream,        \
+                     (bf16*)out, (const bf16*)W, (const bf16*)x,            \
+                     (const bf16*)bias, (const bf16*)res,                   \
+                     (const bf16*)norm_w, (const bf16*)norm_b, eps, M, K)
+  switch (epilogue * 4 + norm_kind) {
+    case 0: GEMV_CASE(0, 0); break;
+    case 1: GEMV_CASE(0, 1); break;
+    case 2: GEMV_CASE(0, 2); break;
+    case 4: GEMV_CASE(1, 0); break;
+    case 5: GEMV_CASE(1, 1); break;
+    case 6: GEMV_CASE(1, 2); break;
+    case 8: GEMV_CASE(2, 0); break;
+    case 9: GEMV_CASE(2, 1); break;
+    case 10: GEMV_CASE(2, 2); break;
+    case 12: GEMV_CASE(3, 0); break;
+    case 13: GEMV_CASE(3, 1); break;
+    case 14: GEMV_CASE(3, 2); break;
+    default: GEMV_CASE(0, 0);
   }
+#undef GEMV_CASE
 }
 
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
-                        const void* x, int M, int K, int gelu_gate,
+                        const void* x, const void* norm_w, const void* norm_b,
+                        float eps, int M, int K, int gelu_gate, int norm_kind,
                         hipStream_t stream) {
   const int smem = K * sizeof(bf16);
-  hipLaunchKernelGGL(gemv_swiglu_kernel, dim3(gemv_grid(M)), dim3(256), smem,
-                     stream, (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,
-                     (const bf16*)x, M, K, gelu_gate);
+  dim3 grid(gemv_grid(M, 4)), block(256);
+#define SW_CASE(N)                                                          \
+  hipLaunchKernelGGL((gemv_swiglu_kernel<N>), grid, block, smem, stream,    \
+                     (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,          \
+                     (const bf16*)x, (const bf16*)norm_w,                   \
+                     (const bf16*)norm_b, eps, M, K, gelu_gate)
+  switch (norm_kind) {
+    case 1: SW_CASE(1); break;
+    case 2: SW_CASE(2); break;
+    default: SW_CASE(0);
+  }
+#undef SW_CASE
 }
 
 void launch_embed(void* out, const void* wte, const int* token, int n_embd,

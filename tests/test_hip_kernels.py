@@ -108,6 +108,50 @@ def test_gemv_swiglu(ops):
     assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2)
 
 
+def test_gemv_fused_rmsnorm(ops):
+    M, K = 1024, 4096
+    W = mk(M, K, scale=1.0 / math.sqrt(K), seed=30)
+    x = mk(K, seed=31)
+    g = mk(K, seed=32)
+    out = torch.empty(M, device=DEV, dtype=torch.bfloat16)
+    ops.gemv(out, W, x, None, None, 0, g, None, 1, 1e-5)
+    xf = x.float()
+    xn = xf * torch.rsqrt(xf.pow(2).mean() + 1e-5) * g.float()
+    ref = W.float() @ xn.to(torch.bfloat16).float()
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_gemv_fused_layernorm(ops):
+    M, K = 512, 2048
+    W = mk(M, K, scale=1.0 / math.sqrt(K), seed=33)
+    x = mk(K, seed=34)
+    g = mk(K, seed=35)
+    b = mk(K, seed=36)
+    out = torch.empty(M, device=DEV, dtype=torch.bfloat16)
+    ops.gemv(out, W, x, None, None, 0, g, b, 2, 1e-5)
+    xn = torch.nn.functional.layer_norm(x.float(), (K,), g.float(),
+                                        b.float(), 1e-5)
+    ref = W.float() @ xn.to(torch.bfloat16).float()
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_swiglu_fused_rmsnorm(ops):
+    M, K = 688, 256
+    Wg = mk(M, K, scale=1.0 / math.sqrt(K), seed=37)
+    Wu = mk(M, K, scale=1.0 / math.sqrt(K), seed=38)
+    x = mk(K, seed=39)
+    g = mk(K, seed=40)
+    out = torch.empty(M, device=DEV, dtype=torch.bfloat16)
+    ops.gemv_swiglu(out, Wg, Wu, x, False, g, None, 1, 1e-5)
+    xf = x.float()
+    xn = (xf * torch.rsqrt(xf.pow(2).mean() + 1e-5) * g.float()).to(
+        torch.bfloat16).float()
+    gg = Wg.float() @ xn
+    uu = Wu.float() @ xn
+    ref = torch.nn.functional.silu(gg) * uu
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
 def test_embed(ops):
     V, E = 1000, 512
     wte = mk(V, E, seed=17)
