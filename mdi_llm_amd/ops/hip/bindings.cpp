@@ -63,7 +63,7 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
           c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> res,
           int64_t epilogue, c10::optional<torch::Tensor> norm_w,
           c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
-          double eps) {
+          double eps, int64_t rows) {
   check_bf16(out, "out");
   check_bf16(W, "W");
   check_bf16(x, "x");
@@ -95,7 +95,8 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
     }
   }
   launch_gemv(out.data_ptr(), W.data_ptr(), x.data_ptr(), bp, rp, nwp, nbp,
-              (float)eps, M, K, (int)epilogue, (int)norm_kind, cur_stream());
+              (float)eps, M, K, (int)epilogue, (int)norm_kind, (int)rows,
+              cur_stream());
 }
 
 void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
@@ -220,7 +221,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("W"), py::arg("x"), py::arg("bias"),
         py::arg("res"), py::arg("epilogue"),
         py::arg("norm_w") = c10::nullopt, py::arg("norm_b") = c10::nullopt,
-        py::arg("norm_kind") = 0, py::arg("eps") = 1e-5);
+        py::arg("norm_kind") = 0, py::arg("eps") = 1e-5,
+        py::arg("rows") = 0);
   m.def("gemv_swiglu", &gemv_swiglu, "fused SwiGLU pair GEMV (+pre-norm)",
         py::arg("out"), py::arg("Wg"), py::arg("Wu"), py::arg("x"),
         py::arg("gelu_gate"), py::arg("norm_w") = c10::nullopt,

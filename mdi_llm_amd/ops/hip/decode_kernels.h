@@ -13,10 +13,11 @@ void launch_layernorm(void* out, const void* x, const void* w, const void* b,
 
 // epilogue: 0 none, 1 +res, 2 gelu(tanh), 3 silu
 // norm_kind: 0 none, 1 fused RMSNorm on x, 2 fused LayerNorm on x
+// rows: output rows per wave (1/2/4); 0 = auto by M
 void launch_gemv(void* out, const void* W, const void* x, const void* bias,
                  const void* res, const void* norm_w, const void* norm_b,
                  float eps, int M, int K, int epilogue, int norm_kind,
-                 hipStream_t stream);
+                 int rows, hipStream_t stream);
 
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const void* x, const void* norm_w, const void* norm_b,

@@ -148,17 +148,44 @@ DEVINL float gelu_tanh(float v) {
   return 0.5f * v * (1.f + tanhf(c));
 }
 
+// Stage x into LDS, optionally fusing the pre-norm.  Returns a uniform
+// post-scale to apply to each accumulated dot product:
+//  NORM==1 (RMSNorm): LDS holds x*g; returned scale = rsqrt(mean(x^2)+eps),
+//    exact because the RMS scale is uniform over j (applied post-dot).
+//  NORM==2 (LayerNorm): LDS holds the fully normalized value (two-pass,
+//    mean subtraction is not a uniform post-scale); returns 1.
 template <int NORM>
-DEVINL void stage_x(bf16* xs, const bf16* __restrict__ x,
-                    const bf16* __restrict__ nw, const bf16* __restrict__ nb,
-                    int K, float eps, float* red) {
+DEVINL float stage_x(bf16* xs, const bf16* __restrict__ x,
+                     const bf16* __restrict__ nw, const bf16* __restrict__ nb,
+                     int K, float eps, float* red) {
   const int tid = threadIdx.x;
   if (NORM == 0) {
     for (int i = tid * 8; i < K; i += blockDim.x * 8)
       *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(x + i);
     __syncthreads();
-    return;
+    return 1.f;
   }
+  if (NORM == 1) {
+    float s2 = 0.f;
+    for (int i = tid * 8; i < K; i += blockDim.x * 8) {
+      bf16x8 v = load8(x + i);
+      bf16x8 g = load8(nw + i);
+      bf16x8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = b2f(v.v[j]);
+        s2 += f * f;
+        o.v[j] = f2b(f * b2f(g.v[j]));
+      }
+      *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(o.v);
+    }
+    s2 = wave_reduce_sum(s2);
+    if ((tid & 63) == 0) red[tid >> 6] = s2;
+    __syncthreads();
+    float ms = (red[0] + red[1] + red[2] + red[3]) / K;
+    return rsqrtf(ms + eps);
+  }
+  // NORM == 2: LayerNorm, two-pass
   float s = 0.f, s2 = 0.f;
   for (int i = tid * 8; i < K; i += blockDim.x * 8) {
     bf16x8 v = load8(x + i);
@@ -177,15 +204,14 @@ DEVINL void stage_x(bf16* xs, const bf16* __restrict__ x,
     red[(tid >> 6) * 2 + 1] = s2;
   }
   __syncthreads();
-  float mean = 0.f, inv;
-  if (NORM == 2) mean = (red[0] + red[2] + red[4] + red[6]) / K;
-  float ms = (red[1] + red[3] + red[5] + red[7]) / K;
-  inv = rsqrtf(ms - mean * mean + eps);
+  const float mean = (red[0] + red[2] + red[4] + red[6]) / K;
+  const float ms = (red[1] + red[3] + red[5] + red[7]) / K;
+  const float inv = rsqrtf(ms - mean * mean + eps);
   for (int i = tid * 8; i < K; i += blockDim.x * 8) {
     bf16x8 v = load8(xs + i);
     bf16x8 g = load8(nw + i);
     bf16x8 o;
-    if (NORM == 2 && nb != nullptr) {
+    if (nb != nullptr) {
       bf16x8 bb = load8(nb + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
@@ -198,9 +224,10 @@ DEVINL void stage_x(bf16* xs, const bf16* __restrict__ x,
     *reinterpret_cast<int4*>(xs + i) = *reinterpret_cast<const int4*>(o.v);
   }
   __syncthreads();
+  return 1.f;
 }
 
-template <int EPI, int NORM>
+template <int EPI, int NORM, int ROWS>
 __global__ void gemv_kernel(bf16* __restrict__ out,
                             const bf16* __restrict__ W,
                             const bf16* __restrict__ x,
@@ -212,43 +239,45 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
-  stage_x<NORM>(xs, x, nw, nb, K, eps, red);
+  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * 2;
+  const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * ROWS;
 
-  for (int row = (blockIdx.x * (blockDim.x >> 6) + wave) * 2; row < M;
+  for (int row = (blockIdx.x * (blockDim.x >> 6) + wave) * ROWS; row < M;
        row += rows_per_grid) {
-    const bf16* wrow0 = W + (size_t)row * K;
-    const bool two = row + 1 < M;
-    const bf16* wrow1 = two ? wrow0 + K : wrow0;
-    float acc0 = 0.f, acc1 = 0.f;
+    const bf16* wrow[ROWS];
+    float acc[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      wrow[r] = W + (size_t)min(row + r, M - 1) * K;
+      acc[r] = 0.f;
+    }
     for (int i = lane * 8; i < K; i += 64 * 8) {
-      bf16x8 w0 = load8(wrow0 + i);
-      bf16x8 w1 = load8(wrow1 + i);
       bf16x8 xv = load8(xs + i);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float xf = b2f(xv.v[j]);
-        acc0 += b2f(w0.v[j]) * xf;
-        acc1 += b2f(w1.v[j]) * xf;
+      for (int r = 0; r < ROWS; ++r) {
+        bf16x8 wv = load8(wrow[r] + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[r] += b2f(wv.v[j]) * b2f(xv.v[j]);
       }
     }
-    acc0 = wave_reduce_sum(acc0);
-    acc1 = wave_reduce_sum(acc1);
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) acc[r] = wave_reduce_sum(acc[r]);
     if (lane == 0) {
 #pragma unroll
-      for (int r = 0; r < 2; ++r) {
-        if (r == 1 && !two) break;
-        float acc = r == 0 ? acc0 : acc1;
+      for (int r = 0; r < ROWS; ++r) {
         const int rw = row + r;
-        if (bias != nullptr) acc += b2f(bias[rw]);
-        if (EPI == 1 && res != nullptr) acc += b2f(res[rw]);
-        if (EPI == 2) acc = gelu_tanh(acc);
-        if (EPI == 3) acc = acc / (1.f + expf(-acc));
-        out[rw] = f2b(acc);
+        if (rw >= M) break;
+        float a = acc[r] * nscale;
+        if (bias != nullptr) a += b2f(bias[rw]);
+        if (EPI == 1 && res != nullptr) a += b2f(res[rw]);
+        if (EPI == 2) a = gelu_tanh(a);
+        if (EPI == 3) a = a / (1.f + expf(-a));
+        out[rw] = f2b(a);
       }
     }
   }
@@ -267,7 +296,7 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
-  stage_x<NORM>(xs, x, nw, nb, K, eps, red);
+  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -290,8 +319,8 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
         ua += b2f(uv.v[j]) * xf;
       }
     }
-    ga = wave_reduce_sum(ga);
-    ua = wave_reduce_sum(ua);
+    ga = wave_reduce_sum(ga) * nscale;
+    ua = wave_reduce_sum(ua) * nscale;
     if (lane == 0) {
       float act = gelu_gate ? gelu_tanh(ga) : ga / (1.f + expf(-ga));
       out[row] = f2b(act * ua);
@@ -586,34 +615,39 @@ __global__ void attn_decode_kernel(
 __global__ void attn_combine_kernel(bf16* __restrict__ out,
                                     const float* __restrict__ part_o,
                                     const float* __restrict__ part_ml,
-                                    int n_chunks, int head_size) {
-  __shared__ float w_lds[64];
-  __shared__ float inv_l;
-  const int h = blockIdx.x;
-  const int tid = threadIdx.x;
-  if (tid < 64) {
-    float m = -1e30f, l = 0.f;
-    if (tid < n_chunks) {
-      m = part_ml[((size_t)h * n_chunks + tid) * 2];
-      l = part_ml[((size_t)h * n_chunks + tid) * 2 + 1];
-    }
-    float M = m;
+                                    int n_chunks, int head_size, int n_head) {
+  // one WAVE per (head, 64-dim slice); every lane recomputes the chunk
+  // weights lane-parallel (lane == chunk, shfl broadcast), so blocks are
+  // fully independent and the chunk-accumulate loop is the only serial
+  // part (unrolled for load ILP).
+  const int DS = (head_size + 63) / 64;
+  const int wid = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  const int h = wid / DS;
+  const int d = (wid % DS) * 64 + lane;
+  if (h >= n_head) return;
+
+  float m = -1e30f, l = 0.f;
+  if (lane < n_chunks) {
+    m = part_ml[((size_t)h * n_chunks + lane) * 2];
+    l = part_ml[((size_t)h * n_chunks + lane) * 2 + 1];
+  }
+  float M = m;
 #pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-      M = fmaxf(M, __shfl_xor(M, off, 64));
-    const float wgt = (tid < n_chunks) ? __expf(m - M) : 0.f;
-    const float L = wave_reduce_sum(l * wgt);
-    w_lds[tid] = wgt;
-    if (tid == 0) inv_l = 1.f / L;
+  for (int off = 32; off > 0; off >>= 1)
+    M = fmaxf(M, __shfl_xor(M, off, 64));
+  const float wgt = (lane < n_chunks) ? __expf(m - M) : 0.f;
+  const float inv = 1.f / wave_reduce_sum(l * wgt);
+
+  if (d >= head_size) return;
+  const float* po = part_o + (size_t)h * n_chunks * head_size + d;
+  float acc = 0.f;
+#pragma unroll 4
+  for (int c = 0; c < n_chunks; ++c) {
+    const float w = __shfl(wgt, c, 64);
+    acc += w * po[(size_t)c * head_size];
   }
-  __syncthreads();
-  const float inv = inv_l;
-  for (int d = tid; d < head_size; d += blockDim.x) {
-    float acc = 0.f;
-    for (int c = 0; c < n_chunks; ++c)
-      acc += w_lds[c] * part_o[((size_t)h * n_chunks + c) * head_size + d];
-    out[(size_t)h * head_size + d] = f2b(acc * inv);
-  }
+  out[(size_t)h * head_size + d] = f2b(acc * inv);
 }
 
 // ---------------------------------------------------------------------------
@@ -657,14 +691,21 @@ void launch_layernorm(void* out, const void* x, const void* w, const void* b,
 void launch_gemv(void* out, const void* W, const void* x, const void* bias,
                  const void* res, const void* norm_w, const void* norm_b,
                  float eps, int M, int K, int epilogue, int norm_kind,
-                 hipStream_t stream) {
+                 int rows, hipStream_t stream) {
   const int smem = K * sizeof(bf16);
-  dim3 grid(gemv_grid(M, 8)), block(256);
-#define GEMV_CASE(E, N)                                                     \
-  hipLaunchKernelGGL((gemv_kernel<E, N>), grid, block, smem, stream,        \
+  if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
+  dim3 grid(gemv_grid(M, 4 * rows)), block(256);
+#define GEMV_CASE1(E, N, R)                                                 \
+  hipLaunchKernelGGL((gemv_kernel<E, N, R>), grid, block, smem, stream,     \
                      (bf16*)out, (const bf16*)W, (const bf16*)x,            \
                      (const bf16*)bias, (const bf16*)res,                   \
                      (const bf16*)norm_w, (const bf16*)norm_b, eps, M, K)
+#define GEMV_CASE(E, N)                                                     \
+  do {                                                                      \
+    if (rows == 4) GEMV_CASE1(E, N, 4);                                     \
+    else if (rows == 2) GEMV_CASE1(E, N, 2);                                \
+    else GEMV_CASE1(E, N, 1);                                               \
+  } while (0)
   switch (epilogue * 4 + norm_kind) {
     case 0: GEMV_CASE(0, 0); break;
     case 1: GEMV_CASE(0, 1); break;
@@ -681,6 +722,7 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
     default: GEMV_CASE(0, 0);
   }
 #undef GEMV_CASE
+#undef GEMV_CASE1
 }
 
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
@@ -798,8 +840,12 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
       return -1;
   }
   if (rc != 0) return rc;
-  hipLaunchKernelGGL(attn_combine_kernel, dim3(n_kv_heads * qpk), dim3(64), 0,
-                     stream, (bf16*)out, part_o, part_ml, n_chunks, head_size);
+  const int ds = (head_size + 63) / 64;
+  const int n_waves = n_kv_heads * qpk * ds;
+  const int cblocks = (n_waves * 64 + 255) / 256;
+  hipLaunchKernelGGL(attn_combine_kernel, dim3(cblocks), dim3(256), 0,
+                     stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
+                     n_kv_heads * qpk);
   return 0;
 }
 
