@@ -73,9 +73,11 @@ def main():
     cfg = ModelConfig.from_name(args.model)
 
     # ---- build this rank's stage (random init, bf16, on device) ---------
-    from mdi_llm_amd.utils import layer_split
+    from mdi_llm_amd.utils.partition import balanced_split
 
-    split = layer_split(cfg.n_layer, n_stages)
+    split = balanced_split(cfg, n_stages)
+    if rank == 0:
+        log(f"layer split: {split}")
     t0 = time.time()
     torch.manual_seed(1234 + rank)
     stage = build_stage(cfg, rank, split[rank])

@@ -139,6 +139,13 @@ class DecodeEngine:
             self.token = torch.zeros(1, device=dev, dtype=torch.int32)
             self.pos_emb = torch.zeros(E, **bf)
 
+        if self.is_starter:
+            # fused sampler state
+            self.sample_scratch = torch.zeros(520, device=dev,
+                                              dtype=torch.int32)
+            self.sample_out = torch.zeros(1, device=dev, dtype=torch.int32)
+            self.sample_ctr = torch.zeros(1, device=dev, dtype=torch.int32)
+
         # device-side slot/pos scalars (graph-replayable)
         self.slot = torch.zeros(1, device=dev, dtype=torch.int32)
         self.slot_long = torch.zeros(1, device=dev, dtype=torch.int64)
@@ -146,6 +153,11 @@ class DecodeEngine:
             kv_pool.n_slots, device=dev, dtype=torch.int32
         )
         self.pos = torch.zeros(1, device=dev, dtype=torch.int32)
+
+        self._r_qkv = self._rows(cfg.qkv_dim, E)
+        self._r_proj = self._rows(E, cfg.n_head * cfg.head_size)
+        self._r_down = self._rows(E, I)
+        self._r_head = self._rows(cfg.padded_vocab_size, E)
 
         self._graph_blocks: Optional[torch.cuda.CUDAGraph] = None
         self._graph_tail: Optional[torch.cuda.CUDAGraph] = None
@@ -176,6 +188,30 @@ class DecodeEngine:
         """fused-norm kind: 1 RMSNorm, 2 LayerNorm."""
         return 1 if self.config.norm_class_name == "RMSNorm" else 2
 
+    @staticmethod
+    def _rows(M: int, K: int) -> int:
+        """Output rows per wave for the decode GEMV, chosen from the MI355X
+        micro-bench sweep (tools/bench_kernels.py): long-K shapes want 4-row
+        ILP; small-M and huge-M shapes want max workgroup count."""
+        if K >= 16384:
+            return 4
+        if M <= 4096 or M >= 65536:
+            return 1
+        if M <= 7168:
+            return 4
+        return 2
+
+    def sample_into_token(self, temperature: float, top_k, seed: int) -> torch.Tensor:
+        """Fused on-GPU sampling from self.logits into self.sample_out."""
+        self.sample_scratch.zero_()
+        self.sample_ctr += 1
+        self.ops.sample(
+            self.sample_out, self.logits, self.sample_scratch,
+            float(temperature), int(top_k or 0), temperature > 0.0,
+            int(seed) & 0x7FFFFFFF, self.sample_ctr,
+        )
+        return self.sample_out
+
     def _run_blocks(self) -> None:
         """x -> x through all local blocks (decode, one token).
 
@@ -190,7 +226,7 @@ class DecodeEngine:
         for li, w in enumerate(self.blocks):
             # qkv = Wqkv @ norm1(x)
             ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
-                     w.norm1_w, w.norm1_b, nk, eps)
+                     w.norm1_w, w.norm1_b, nk, eps, self._r_qkv)
             ops.rope_kv_append(
                 self.qkv, self.kv_pool.k, self.kv_pool.v, self.cos, self.sin,
                 self.pos, self.slot, li,
@@ -201,14 +237,16 @@ class DecodeEngine:
             )
             if cfg.parallel_residual:
                 # x = x + proj(y) + mlp(norm2(x) or norm1(x))
-                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, None, 0)
+                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, None, 0,
+                         None, None, 0, eps, self._r_proj)
                 nw = w.norm1_w if cfg.shared_attention_norm else w.norm2_w
                 nb = w.norm1_b if cfg.shared_attention_norm else w.norm2_b
                 self._mlp(self.x, w, self.a, nw, nb)
                 ops.add(self.x, self.x, self.m_out)
             else:
                 # a = x + proj(y); x = a + mlp(norm2(a))
-                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, self.x, 1)
+                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, self.x, 1,
+                         None, None, 0, eps, self._r_proj)
                 self._mlp(self.a, w, self.a, w.norm2_w, w.norm2_b)
                 # _mlp wrote x = res + down(act) directly
         # (sequential path leaves the stream in self.x)
@@ -228,12 +266,9 @@ class DecodeEngine:
         else:
             self.ops.gemv(self.act, w.fc_w, inp, w.fc_b, None, 2,
                           norm_w, norm_b, nk, eps)  # gelu
-        if cfg.parallel_residual:
-            self.ops.gemv(self.m_out, w.mlp_proj_w, self.act, w.mlp_proj_b,
-                          res, 1)
-        else:
-            self.ops.gemv(self.x, w.mlp_proj_w, self.act, w.mlp_proj_b,
-                          res, 1)
+        out = self.m_out if cfg.parallel_residual else self.x
+        self.ops.gemv(out, w.mlp_proj_w, self.act, w.mlp_proj_b, res, 1,
+                      None, None, 0, eps, self._r_down)
 
     def _embed(self) -> None:
         cfg = self.config
@@ -246,7 +281,8 @@ class DecodeEngine:
     def _tail_seq(self) -> None:
         # logits = lm_head @ ln_f(x): one fused kernel
         self.ops.gemv(self.logits, self.head_w, self.x, self.head_b, None, 0,
-                      self.lnf_w, self.lnf_b, self._nk, self.config.norm_eps)
+                      self.lnf_w, self.lnf_b, self._nk, self.config.norm_eps,
+                      self._r_head)
 
     # ---------------------------------------------------------------------
     # public decode API

@@ -202,6 +202,24 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
               " head_size=", hs);
 }
 
+void sample(torch::Tensor out_token, torch::Tensor logits,
+            torch::Tensor scratch, double temperature, int64_t top_k,
+            bool noise, int64_t seed, c10::optional<torch::Tensor> ctr) {
+  check_i32(out_token, "out_token");
+  check_bf16(logits, "logits");
+  TORCH_CHECK(scratch.is_cuda() && scratch.scalar_type() == torch::kInt32 &&
+                  scratch.numel() >= 520,
+              "scratch must be >=520 int32 on GPU");
+  const int* cp = nullptr;
+  if (ctr.has_value()) {
+    check_i32(*ctr, "ctr");
+    cp = ctr->data_ptr<int>();
+  }
+  launch_sample(out_token.data_ptr(), logits.data_ptr(), (int)logits.numel(),
+                scratch.data_ptr(), (float)temperature, (int)top_k,
+                noise ? 1 : 0, (unsigned)(int64_t)seed, cp, cur_stream());
+}
+
 void add(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
   check_bf16(out, "out");
   check_bf16(a, "a");
@@ -233,4 +251,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "RoPE on interleaved qkv + KV cache append");
   m.def("attn_decode", &attn_decode, "GQA flash-decode attention (split-S)");
   m.def("add", &add, "bf16 residual add");
+  m.def("sample", &sample,
+        "fused temperature/top-k/gumbel token sampling (128k vocab ~15us)",
+        py::arg("out_token"), py::arg("logits"), py::arg("scratch"),
+        py::arg("temperature"), py::arg("top_k"), py::arg("noise"),
+        py::arg("seed"), py::arg("ctr") = c10::nullopt);
 }

@@ -43,3 +43,8 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
 
 void launch_add(void* out, const void* a, const void* b, int n,
                 hipStream_t stream);
+
+// fused temperature/top-k/gumbel sampling; scratch: >=520 u32, zeroed
+void launch_sample(void* out_token, const void* logits, int V, void* scratch,
+                   float temperature, int top_k, int noise_on, unsigned seed,
+                   const int* ctr, hipStream_t stream);

@@ -258,8 +258,13 @@ class PipelineRuntime:
                 self.comm.send_sched(s, out)
         self.comm.drain()
 
-    @staticmethod
-    def _draw(logits, sampling: SamplingParams, gen):
+    def _draw(self, logits, sampling: SamplingParams, gen):
+        r = self.runner
+        if getattr(r, "backend", "") == "hip" and sampling.top_p >= 1.0:
+            # fused on-GPU sampler (radix top-k + gumbel), no host sync
+            return r.engine.sample_into_token(
+                sampling.temperature, sampling.top_k, sampling.seed or 0
+            )
         tok = sample_token(
             logits,
             temperature=sampling.temperature,

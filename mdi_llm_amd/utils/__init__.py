@@ -1,5 +1,6 @@
 from .partition import (  # noqa: F401
     N_LAYERS_NODES,
+    balanced_split,
     chunk_dir,
     chunk_file_name,
     count_transformer_blocks,

@@ -104,6 +104,41 @@ def layer_split(n_layer: int, n_nodes: int) -> List[int]:
     return counts
 
 
+def balanced_split(config, n_nodes: int, sampler_bytes: int = 150_000_000,
+                   dtype_bytes: int = 2) -> List[int]:
+    """Byte-balanced per-stage block counts for a given model config.
+
+    Decode is HBM-bandwidth-bound, so per-stage weight BYTES are a direct
+    proxy for per-stage time.  The starter additionally streams the lm_head
+    every token and runs the sampler, so it gets fewer blocks — the general
+    version of the reference's hand-tuned table intuition
+    (/root/reference/src/sub/config.py:56-98, README.md:334-342).
+    """
+    if n_nodes == 1:
+        return [config.n_layer]
+    E, I, V = config.n_embd, config.intermediate_size, config.padded_vocab_size
+    hs, nh, ng = config.head_size, config.n_head, config.n_query_groups
+    layer_b = (E * (nh + 2 * ng) * hs + E * nh * hs + 3 * E * I) * dtype_bytes
+    if config.mlp_class_name == "GptNeoxMLP":
+        layer_b = (E * (nh + 2 * ng) * hs + E * nh * hs + 2 * E * I) * dtype_bytes
+    starter_extra = V * E * dtype_bytes + sampler_bytes
+    extra_layers = starter_extra / max(layer_b, 1)
+    # starter share: solve x + extra = (n_layer - x) / (n_nodes - 1) balance
+    x = (config.n_layer - extra_layers * (n_nodes - 1)) / n_nodes
+    start = max(0, min(config.n_layer - (n_nodes - 1), round(x)))
+    rest = config.n_layer - start
+    base, rem = divmod(rest, n_nodes - 1)
+    counts = [start] + [base + (1 if i < rem else 0)
+                        for i in range(n_nodes - 1)]
+    # every secondary needs at least one block
+    for i in range(1, n_nodes):
+        if counts[i] == 0:
+            counts[i] = 1
+            counts[0] -= 1
+    assert sum(counts) == config.n_layer and all(c >= 0 for c in counts)
+    return counts
+
+
 def count_transformer_blocks(state_dict: dict) -> int:
     """Number of distinct ``transformer.h.<i>.`` indices in a state dict
     (reference utils/utils.py:470-492)."""

@@ -1,0 +1,81 @@
+"""GPU tests for the fused radix-top-k + Gumbel sampling kernel."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from mdi_llm_amd.ops import require_hip_ops
+
+    return require_hip_ops()
+
+
+def _call(ops, logits, temperature, top_k, noise=True, seed=7, ctr_val=0):
+    scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
+    out = torch.zeros(1, device=DEV, dtype=torch.int32)
+    ctr = torch.tensor([ctr_val], device=DEV, dtype=torch.int32)
+    ops.sample(out, logits, scratch, temperature, top_k, noise, seed, ctr)
+    return int(out)
+
+
+def test_greedy_matches_argmax(ops):
+    torch.manual_seed(0)
+    for V in (1000, 128256):
+        logits = torch.randn(V, device=DEV).to(torch.bfloat16)
+        got = _call(ops, logits, 0.0, 0, noise=False)
+        assert got == int(logits.float().argmax())
+
+
+def test_topk_support(ops):
+    """Every sampled token must be inside the true top-k set."""
+    torch.manual_seed(1)
+    V, k = 128256, 200
+    logits = torch.randn(V, device=DEV).to(torch.bfloat16)
+    kth = torch.topk(logits.float(), k).values[-1]
+    topset = set((logits.float() >= kth).nonzero().flatten().tolist())
+    for ctr in range(50):
+        tok = _call(ops, logits, 0.8, k, noise=True, ctr_val=ctr)
+        assert tok in topset, (ctr, tok)
+
+
+def test_deterministic_given_seed_ctr(ops):
+    torch.manual_seed(2)
+    logits = torch.randn(5000, device=DEV).to(torch.bfloat16)
+    a = _call(ops, logits, 1.0, 50, seed=11, ctr_val=3)
+    b = _call(ops, logits, 1.0, 50, seed=11, ctr_val=3)
+    c = _call(ops, logits, 1.0, 50, seed=11, ctr_val=4)
+    assert a == b
+    # different counter should (almost surely) give a different draw
+    # occasionally equal is fine; just check the call runs
+    assert isinstance(c, int)
+
+
+def test_distribution_roughly_matches(ops):
+    """Gumbel-max over temperature-scaled logits == softmax sampling."""
+    probs = torch.tensor([0.6, 0.25, 0.1, 0.03, 0.02])
+    logits = torch.log(probs).to(DEV).to(torch.bfloat16)
+    pad = torch.full((3,), -30.0, device=DEV, dtype=torch.bfloat16)
+    logits = torch.cat([logits, pad])
+    counts = torch.zeros(8)
+    n = 4000
+    for ctr in range(n):
+        tok = _call(ops, logits, 1.0, 0, noise=True, seed=5, ctr_val=ctr)
+        counts[tok] += 1
+    freq = counts / n
+    assert torch.allclose(freq[:5], probs, atol=0.04), freq
+
+
+def test_small_topk(ops):
+    torch.manual_seed(3)
+    logits = torch.randn(1000, device=DEV).to(torch.bfloat16)
+    top2 = set(torch.topk(logits.float(), 2).indices.tolist())
+    for ctr in range(20):
+        tok = _call(ops, logits, 2.0, 2, noise=True, ctr_val=ctr)
+        # allow bf16 ties at the threshold
+        kth = torch.topk(logits.float(), 2).values[-1]
+        assert logits.float()[tok] >= kth

@@ -104,6 +104,16 @@ def bench_sampling():
     print(f"torch sampling (topk200): {us:.1f} us")
     us = timeit(lambda: logits.float().argmax(), iters=20)
     print(f"argmax: {us:.1f} us")
+    scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
+    out = torch.zeros(1, device=DEV, dtype=torch.int32)
+    ctr = torch.zeros(1, device=DEV, dtype=torch.int32)
+
+    def fused():
+        scratch.zero_()
+        ops.sample(out, logits, scratch, 0.8, 200, True, 7, ctr)
+
+    us = timeit(fused, iters=50)
+    print(f"fused HIP sampling (topk200): {us:.1f} us")
 
 
 if __name__ == "__main__":
