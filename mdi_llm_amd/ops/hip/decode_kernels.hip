@@ -677,7 +677,7 @@ __global__ void add_kernel(bf16* __restrict__ out, const bf16* __restrict__ a,
 //   [0..255]   hi-byte histogram
 //   [256..511] lo-byte histogram
 //   [512] bucket_hi, [513] count_above, [514] threshold_u16
-//   [515..516] packed (score,idx) u64 argmax cell
+//   [516..517] packed (score,idx) u64 argmax cell (8-byte aligned)
 // ---------------------------------------------------------------------------
 
 DEVINL unsigned bf16_sortable(unsigned short bits) {
@@ -798,7 +798,7 @@ __global__ void sample_gumbel_argmax_kernel(
     sb = (sb & 0x80000000u) ? ~sb : (sb | 0x80000000u);
     unsigned long long packed =
         ((unsigned long long)sb << 32) | (unsigned)best_i;
-    atomicMax(reinterpret_cast<unsigned long long*>(&scratch[515]), packed);
+    atomicMax(reinterpret_cast<unsigned long long*>(&scratch[516]), packed);
   }
 }
 
@@ -806,7 +806,7 @@ __global__ void sample_unpack_kernel(const unsigned* __restrict__ scratch,
                                      int* __restrict__ out) {
   if (threadIdx.x == 0) {
     unsigned long long packed =
-        *reinterpret_cast<const unsigned long long*>(&scratch[515]);
+        *reinterpret_cast<const unsigned long long*>(&scratch[516]);
     out[0] = (int)(unsigned)(packed & 0xFFFFFFFFull);
   }
 }
