@@ -203,6 +203,22 @@ configs.extend(
             n_layer=2,
             n_head=4,
             n_embd=64,
+            rotary_percentage=0.0,
+            pos_embedding="learned",
+            parallel_residual=False,
+            bias=True,
+            norm_class_name="LayerNorm",
+            mlp_class_name="GptNeoxMLP",
+        ),
+        dict(
+            # NeoX-style tiny config (parallel residual, partial rotary)
+            name="nano-test-neox",
+            block_size=128,
+            vocab_size=256,
+            padding_multiple=64,
+            n_layer=2,
+            n_head=4,
+            n_embd=64,
             rotary_percentage=0.25,
             parallel_residual=True,
             bias=True,

