@@ -124,6 +124,7 @@ def main():
     ]
     t0 = time.time()
     if rank == 0:
+        rt.prepare_bench(sampling)
         toks = rt.bench_prefill(prompts)
         if args.warmup > 0:
             toks = rt.bench_decode_rounds(toks, args.warmup, sampling, gens)

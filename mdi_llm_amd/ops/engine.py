@@ -159,8 +159,18 @@ class DecodeEngine:
         self._r_down = self._rows(E, I)
         self._r_head = self._rows(cfg.padded_vocab_size, E)
 
+        if self.is_starter:
+            # per-slot current-token table for fully-fused step graphs
+            self.token_table = torch.zeros(kv_pool.n_slots, device=dev,
+                                           dtype=torch.int32)
+            self._ones_i32 = torch.ones(1, device=dev, dtype=torch.int32)
+
         self._graph_blocks: Optional[torch.cuda.CUDAGraph] = None
         self._graph_tail: Optional[torch.cuda.CUDAGraph] = None
+        self._graph_standalone: Optional[torch.cuda.CUDAGraph] = None
+        self._graph_starter: Optional[torch.cuda.CUDAGraph] = None
+        self._graph_tail_sample: Optional[torch.cuda.CUDAGraph] = None
+        self._fused_params = None
 
     # ---------------------------------------------------------------------
     # slot/pos bookkeeping
@@ -202,8 +212,8 @@ class DecodeEngine:
         return 2
 
     def sample_into_token(self, temperature: float, top_k, seed: int) -> torch.Tensor:
-        """Fused on-GPU sampling from self.logits into self.sample_out."""
-        self.sample_scratch.zero_()
+        """Fused on-GPU sampling from self.logits into self.sample_out.
+        (scratch is self-cleaning: the unpack kernel zeroes it.)"""
         self.sample_ctr += 1
         self.ops.sample(
             self.sample_out, self.logits, self.sample_scratch,
@@ -224,16 +234,14 @@ class DecodeEngine:
         nk = self._nk
         scale = 1.0 / (cfg.head_size ** 0.5)
         for li, w in enumerate(self.blocks):
-            # qkv = Wqkv @ norm1(x)
+            # qkv = Wqkv @ norm1(x); rope + kv-append are fused inside the
+            # attention kernel (qkv stays raw)
             ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
                      w.norm1_w, w.norm1_b, nk, eps, self._r_qkv)
-            ops.rope_kv_append(
-                self.qkv, self.kv_pool.k, self.kv_pool.v, self.cos, self.sin,
-                self.pos, self.slot, li,
-            )
             ops.attn_decode(
                 self.y, self.part_o, self.part_ml, self.qkv, self.kv_pool.k,
-                self.kv_pool.v, self.pos, self.slot, li, self.n_chunks, scale,
+                self.kv_pool.v, self.cos, self.sin, self.pos, self.slot, li,
+                self.n_chunks, scale,
             )
             if cfg.parallel_residual:
                 # x = x + proj(y) + mlp(norm2(x) or norm1(x))
@@ -322,6 +330,111 @@ class DecodeEngine:
                 self._tail_seq()
             self._graph_tail = gt
         torch.cuda.synchronize()
+
+    # ------------------------------------------------------------------
+    # fully-fused per-token step graphs (bench hot path)
+    # ------------------------------------------------------------------
+    def _sample_seq(self, temperature: float, top_k: int, seed: int) -> None:
+        self.sample_ctr += 1
+        self.ops.sample(self.sample_out, self.logits, self.sample_scratch,
+                        float(temperature), int(top_k or 0),
+                        temperature > 0.0, int(seed) & 0x7FFFFFFF,
+                        self.sample_ctr)
+
+    def _advance(self) -> None:
+        # token_table[slot] = sample_out ; pos_table[slot] += 1 (in-graph)
+        self.token_table.index_copy_(0, self.slot_long, self.sample_out)
+        self.pos_table.index_add_(0, self.slot_long, self._ones_i32)
+
+    def _stage_token(self) -> None:
+        torch.index_select(self.token_table, 0, self.slot_long,
+                           out=self.token)
+
+    def ensure_fused_graphs(self, temperature: float, top_k, seed: int) -> None:
+        """Capture the three starter step graphs for fixed sampling params:
+        standalone (embed->blocks->tail->sample), pipeline-starter
+        (tail->sample->embed->blocks), and tail+sample only (drain rounds).
+        """
+        params = (float(temperature), int(top_k or 0), int(seed))
+        if not self.use_graphs or not self.is_starter:
+            return
+        if self._fused_params == params and self._graph_standalone is not None:
+            return
+        t, k, sd = params
+        torch.cuda.synchronize()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._stage_pos()
+                self._stage_token()
+                self._embed()
+                self._run_blocks()
+                self._tail_seq()
+                self._sample_seq(t, k, sd)
+                self._advance()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        self.kv_pool.k.zero_()
+        self.kv_pool.v.zero_()
+        self.pos_table.zero_()
+        self.token_table.zero_()
+        self.sample_ctr.zero_()
+
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            self._stage_pos()
+            self._stage_token()
+            self._embed()
+            self._run_blocks()
+            self._tail_seq()
+            self._sample_seq(t, k, sd)
+            self._advance()
+        self._graph_standalone = g1
+
+        g2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g2):
+            self._stage_pos()
+            self._tail_seq()
+            self._sample_seq(t, k, sd)
+            self._advance_token_only()
+            self._stage_token()
+            self._embed()
+            self._run_blocks()
+            self.pos_table.index_add_(0, self.slot_long, self._ones_i32)
+        self._graph_starter = g2
+
+        g3 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g3):
+            self._stage_pos()
+            self._tail_seq()
+            self._sample_seq(t, k, sd)
+            self._advance_token_only()
+        self._graph_tail_sample = g3
+        self._fused_params = params
+        torch.cuda.synchronize()
+
+    def _advance_token_only(self) -> None:
+        self.token_table.index_copy_(0, self.slot_long, self.sample_out)
+
+    def standalone_step(self, slot: int) -> None:
+        """One full decode token for `slot` (token_table-chained)."""
+        self.slot.fill_(slot)
+        self._graph_standalone.replay()
+
+    def starter_step(self, x_in: torch.Tensor, slot: int) -> torch.Tensor:
+        """Pipeline starter: tail(x_in)+sample+next head; returns self.x."""
+        if x_in.data_ptr() != self.x.data_ptr():
+            self.x.copy_(x_in.view(-1), non_blocking=True)
+        self.slot.fill_(slot)
+        self._graph_starter.replay()
+        return self.x
+
+    def tail_sample_step(self, x_in: torch.Tensor, slot: int) -> None:
+        if x_in.data_ptr() != self.x.data_ptr():
+            self.x.copy_(x_in.view(-1), non_blocking=True)
+        self.slot.fill_(slot)
+        self._graph_tail_sample.replay()
 
     def decode_step_head(self, token: torch.Tensor, slot: int) -> torch.Tensor:
         """Starter head role: token -> activations (writes self.x).

@@ -170,7 +170,8 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor kpool,
 
 void attn_decode(torch::Tensor out, torch::Tensor part_o,
                  torch::Tensor part_ml, torch::Tensor qkv,
-                 torch::Tensor kpool, torch::Tensor vpool, torch::Tensor pos,
+                 torch::Tensor kpool, torch::Tensor vpool,
+                 torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
                  torch::Tensor slot, int64_t layer, int64_t n_chunks,
                  double scale) {
   check_bf16(out, "out");
@@ -179,12 +180,15 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
   check_bf16(qkv, "qkv");
   check_bf16(kpool, "kpool");
   check_bf16(vpool, "vpool");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
   check_i32(pos, "pos");
   check_i32(slot, "slot");
   const int n_layers_pool = (int)kpool.size(1);
   const int n_kv = (int)kpool.size(2);
   const int max_seq = (int)kpool.size(3);
   const int hs = (int)kpool.size(4);
+  const int rope_ne = cos_t.dim() > 1 ? (int)cos_t.size(1) : 0;
   const int qkv_dim = (int)qkv.numel();
   const int qpk = qkv_dim / (n_kv * hs) - 2;
   const int n_head = n_kv * qpk;
@@ -196,6 +200,8 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
   int rc = launch_attn_decode(
       out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
       qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      rope_ne ? cos_t.data_ptr<float>() : nullptr,
+      rope_ne ? sin_t.data_ptr<float>() : nullptr, rope_ne,
       pos.data_ptr<int>(), slot.data_ptr<int>(), (int)layer, n_layers_pool,
       n_kv, max_seq, hs, qpk, (int)n_chunks, (float)scale, cur_stream());
   TORCH_CHECK(rc == 0, "attn_decode: unsupported geometry qpk=", qpk,

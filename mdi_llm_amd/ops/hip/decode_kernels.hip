@@ -239,15 +239,24 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
-  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * ROWS;
+  const int row0 = (blockIdx.x * (blockDim.x >> 6) + wave) * ROWS;
 
-  for (int row = (blockIdx.x * (blockDim.x >> 6) + wave) * ROWS; row < M;
-       row += rows_per_grid) {
+  // Prefetch the first W chunk of the first row set BEFORE the staging
+  // barrier: plain global loads stay in flight across s_barrier, so HBM
+  // streams W while the block stages/normalizes x in LDS.
+  bf16x8 wpre[ROWS];
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+    wpre[r] = load8(W + (size_t)min(row0 + r, M - 1) * K + lane * 8);
+
+  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
+
+  for (int row = row0; row < M; row += rows_per_grid) {
     const bf16* wrow[ROWS];
     float acc[ROWS];
 #pragma unroll
@@ -259,7 +268,9 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
       bf16x8 xv = load8(xs + i);
 #pragma unroll
       for (int r = 0; r < ROWS; ++r) {
-        bf16x8 wv = load8(wrow[r] + i);
+        bf16x8 wv = (row == row0 && i == lane * 8)
+                        ? wpre[r]
+                        : load8(wrow[r] + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           acc[r] += b2f(wv.v[j]) * b2f(xv.v[j]);
@@ -296,21 +307,28 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
-  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
+  const int row0 = blockIdx.x * (blockDim.x >> 6) + wave;
 
-  for (int row = blockIdx.x * (blockDim.x >> 6) + wave; row < M;
-       row += waves_per_grid) {
+  // W prefetch across the staging barrier (see gemv_kernel)
+  const int rp0 = min(row0, M - 1);
+  bf16x8 gpre = load8(Wg + (size_t)rp0 * K + lane * 8);
+  bf16x8 upre = load8(Wu + (size_t)rp0 * K + lane * 8);
+
+  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
+
+  for (int row = row0; row < M; row += waves_per_grid) {
     const bf16* grow = Wg + (size_t)row * K;
     const bf16* urow = Wu + (size_t)row * K;
     float ga = 0.f, ua = 0.f;
     for (int i = lane * 8; i < K; i += 64 * 8) {
-      bf16x8 gv = load8(grow + i);
-      bf16x8 uv = load8(urow + i);
+      const bool pre = (row == row0 && i == lane * 8);
+      bf16x8 gv = pre ? gpre : load8(grow + i);
+      bf16x8 uv = pre ? upre : load8(urow + i);
       bf16x8 xv = load8(xs + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -429,19 +447,34 @@ DEVINL int q_swz(int row, int d) {  // element index into a [16][HS] tile
   return byte >> 1;
 }
 
+// rotate-half rope of one element d of a head row (raw pointer into qkv)
+DEVINL float rope_elem(const bf16* row, int d, int ne,
+                       const float* __restrict__ cos_t,
+                       const float* __restrict__ sin_t, int pos) {
+  float x = b2f(row[d]);
+  if (d >= ne) return x;
+  const int half = ne >> 1;
+  const float c = cos_t[(size_t)pos * ne + d];
+  const float s = sin_t[(size_t)pos * ne + d];
+  if (d < half) return x * c - b2f(row[d + half]) * s;
+  return x * c + b2f(row[d - half]) * s;
+}
+
 template <int QPK, int HS>
 __global__ void attn_decode_kernel(
     float* __restrict__ part_o,   // [n_head, n_chunks, head_size]
     float* __restrict__ part_ml,  // [n_head, n_chunks, 2]
-    const bf16* __restrict__ qkv, // interleaved, already roped
-    const bf16* __restrict__ kpool, const bf16* __restrict__ vpool,
-    const int* __restrict__ pos_p, const int* __restrict__ slot_p, int layer,
-    int n_layers_pool, int n_kv_heads, int max_seq, int n_chunks,
-    float scale) {
+    const bf16* __restrict__ qkv, // interleaved, RAW (rope fused here)
+    bf16* __restrict__ kpool, bf16* __restrict__ vpool,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    int rope_ne, const int* __restrict__ pos_p,
+    const int* __restrict__ slot_p, int layer, int n_layers_pool,
+    int n_kv_heads, int max_seq, int n_chunks, float scale) {
   constexpr int head_size = HS;
-  // LDS per wave: q tile (16xHS bf16), p tile (16x16 f32 = 1KB),
-  // m/l/alpha (3*16 f32)
+  // LDS per wave: q tile (16xHS bf16), current-token k row, p tile,
+  // m/l/alpha
   __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
+  __shared__ __attribute__((aligned(16))) bf16 k_cur[ATTN_WAVES][HS];
   __shared__ float p_lds[ATTN_WAVES][16][16];
   __shared__ float m_lds[ATTN_WAVES][16];
   __shared__ float l_lds[ATTN_WAVES][16];
@@ -455,6 +488,7 @@ __global__ void attn_decode_kernel(
   if (g >= n_kv_heads) return;  // tail waves exit before any barrier
 
   const int S = pos_p[0] + 1;  // keys visible this step
+  const int pos = S - 1;
   const int slot = slot_p[0];
   const int keys_per_chunk = ((S + n_chunks - 1) / n_chunks + 15) & ~15;
   const int k_begin = chunk * keys_per_chunk;
@@ -462,19 +496,24 @@ __global__ void attn_decode_kernel(
   const size_t cache_base =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
       (size_t)max_seq * head_size;
+  const bf16* krow_cur = qkv + ((size_t)g * (QPK + 2) + QPK) * head_size;
+  const bf16* vrow_cur = krow_cur + head_size;
 
-  // ---- stage q rows (QPK real, rest zero) into LDS (swizzled) ------------
+  // ---- stage q rows (QPK real, rest zero) into LDS, rope fused ----------
   // q row j of group g lives at qkv[(g*(QPK+2)+j)*head_size]
-  for (int i = lane; i < 16 * (head_size / 8); i += 64) {
-    const int r = i / (head_size / 8);
-    const int d8 = (i % (head_size / 8)) * 8;
-    int4 val = {0, 0, 0, 0};
+  for (int i = lane; i < 16 * head_size; i += 64) {
+    const int r = i / head_size;
+    const int d = i % head_size;
+    bf16 val = f2b(0.f);
     if (r < QPK) {
       const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
-      val = *reinterpret_cast<const int4*>(qrow + d8);
+      val = f2b(rope_elem(qrow, d, rope_ne, cos_t, sin_t, pos));
     }
-    *reinterpret_cast<int4*>(&q_lds[wave][q_swz<HS>(r, d8)]) = val;
+    q_lds[wave][q_swz<HS>(r, d)] = val;
   }
+  // current-token k row, roped (the pool does not hold it yet)
+  for (int d = lane; d < head_size; d += 64)
+    k_cur[wave][d] = f2b(rope_elem(krow_cur, d, rope_ne, cos_t, sin_t, pos));
   if (lane < 16) {
     m_lds[wave][lane] = -1e30f;
     l_lds[wave][lane] = 0.f;
@@ -506,12 +545,16 @@ __global__ void attn_decode_kernel(
     f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
     const int arow = lane & 15;          // key row in tile (and B's qhead col)
     const int koff = (lane >> 4) * 8;    // dim offset within 32-chunk
-    const bf16* krow = kpool + cache_base + (size_t)(key0 + arow) * head_size;
-    const bool row_valid = (key0 + arow) < S;
+    const int akey = key0 + arow;
+    const bf16* krow = kpool + cache_base + (size_t)akey * head_size;
+    const bool row_valid = akey < S;
+    const bool row_cur = akey == pos;  // newest key: read from LDS k_cur
 #pragma unroll
     for (int c = 0; c < HS / 32; ++c) {  // K=32 chunks cover head_size
       bf16x8_t af = {};
-      if (row_valid)
+      if (row_cur)
+        af = *reinterpret_cast<const bf16x8_t*>(&k_cur[wave][c * 32 + koff]);
+      else if (row_valid)
         af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
       const bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
           &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
@@ -561,7 +604,9 @@ __global__ void attn_decode_kernel(
     for (int s = 0; s < smax; ++s) {
       const float p = p_lds[wave][s][qb];
       const bf16* vrow =
-          vpool + cache_base + (size_t)(key0 + s) * head_size + d0;
+          (key0 + s == pos)
+              ? vrow_cur + d0
+              : vpool + cache_base + (size_t)(key0 + s) * head_size + d0;
       if constexpr (ODIM >= 8) {
 #pragma unroll
         for (int i = 0; i < ODIM; i += 8) {
@@ -582,6 +627,16 @@ __global__ void attn_decode_kernel(
       } else {  // ODIM == 1
         o_acc[0] += p * b2f(vrow[0]);
       }
+    }
+  }
+
+  // ---- append the current token's k,v to the pool ------------------------
+  // exactly one chunk's range contains pos; its wave owns the append (no
+  // other wave reads pool row pos this step — they use k_cur/vrow_cur)
+  if (k_begin <= pos && pos < k_begin + keys_per_chunk) {
+    for (int d = lane; d < head_size; d += 64) {
+      kpool[cache_base + (size_t)pos * head_size + d] = k_cur[wave][d];
+      vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
     }
   }
 
@@ -802,13 +857,16 @@ __global__ void sample_gumbel_argmax_kernel(
   }
 }
 
-__global__ void sample_unpack_kernel(const unsigned* __restrict__ scratch,
+__global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
                                      int* __restrict__ out) {
   if (threadIdx.x == 0) {
     unsigned long long packed =
         *reinterpret_cast<const unsigned long long*>(&scratch[516]);
     out[0] = (int)(unsigned)(packed & 0xFFFFFFFFull);
   }
+  __syncthreads();  // t0's read precedes the clean
+  // self-clean the scratch for the next call (single block, runs last)
+  for (int i = threadIdx.x; i < 518; i += blockDim.x) scratch[i] = 0;
 }
 
 // ---------------------------------------------------------------------------
@@ -935,30 +993,33 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 
 template <int QPK, int HS>
 static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
-                           const void* kpool, const void* vpool,
-                           const int* pos, const int* slot, int layer,
-                           int n_layers_pool, int n_kv_heads, int max_seq,
-                           int n_chunks, float scale, int blocks,
-                           hipStream_t stream) {
+                           void* kpool, void* vpool, const float* cos_t,
+                           const float* sin_t, int rope_ne, const int* pos,
+                           const int* slot, int layer, int n_layers_pool,
+                           int n_kv_heads, int max_seq, int n_chunks,
+                           float scale, int blocks, hipStream_t stream) {
   hipLaunchKernelGGL((attn_decode_kernel<QPK, HS>), dim3(blocks), dim3(256),
                      0, stream, part_o, part_ml, (const bf16*)qkv,
-                     (const bf16*)kpool, (const bf16*)vpool, pos, slot, layer,
-                     n_layers_pool, n_kv_heads, max_seq, n_chunks, scale);
+                     (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
+                     slot, layer, n_layers_pool, n_kv_heads, max_seq,
+                     n_chunks, scale);
 }
 
 template <int QPK>
 static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
-                          const void* qkv, const void* kpool,
-                          const void* vpool, const int* pos, const int* slot,
+                          const void* qkv, void* kpool, void* vpool,
+                          const float* cos_t, const float* sin_t,
+                          int rope_ne, const int* pos, const int* slot,
                           int layer, int n_layers_pool, int n_kv_heads,
                           int max_seq, int n_chunks, float scale, int blocks,
                           hipStream_t stream) {
 #define CASE_HS(H)                                                          \
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
-      attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, pos, slot, \
-                             layer, n_layers_pool, n_kv_heads, max_seq,     \
-                             n_chunks, scale, blocks, stream);              \
+      attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, cos_t,     \
+                             sin_t, rope_ne, pos, slot, layer,              \
+                             n_layers_pool, n_kv_heads, max_seq, n_chunks,  \
+                             scale, blocks, stream);                        \
       return 0;                                                             \
     }                                                                       \
   }
@@ -971,7 +1032,8 @@ static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
 
 // returns 0 on success, -1 if (qpk, head_size) has no kernel instantiation
 int launch_attn_decode(void* out, float* part_o, float* part_ml,
-                       const void* qkv, const void* kpool, const void* vpool,
+                       const void* qkv, void* kpool, void* vpool,
+                       const float* cos_t, const float* sin_t, int rope_ne,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
                        int head_size, int qpk, int n_chunks, float scale,
@@ -981,29 +1043,34 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
   int rc = -1;
   switch (qpk) {
     case 1:
-      rc = attn_dispatch1<1>(head_size, part_o, part_ml, qkv, kpool, vpool,
-                             pos, slot, layer, n_layers_pool, n_kv_heads,
-                             max_seq, n_chunks, scale, blocks, stream);
+      rc = attn_dispatch1<1>(
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 2:
-      rc = attn_dispatch1<2>(head_size, part_o, part_ml, qkv, kpool, vpool,
-                             pos, slot, layer, n_layers_pool, n_kv_heads,
-                             max_seq, n_chunks, scale, blocks, stream);
+      rc = attn_dispatch1<2>(
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 4:
-      rc = attn_dispatch1<4>(head_size, part_o, part_ml, qkv, kpool, vpool,
-                             pos, slot, layer, n_layers_pool, n_kv_heads,
-                             max_seq, n_chunks, scale, blocks, stream);
+      rc = attn_dispatch1<4>(
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 8:
-      rc = attn_dispatch1<8>(head_size, part_o, part_ml, qkv, kpool, vpool,
-                             pos, slot, layer, n_layers_pool, n_kv_heads,
-                             max_seq, n_chunks, scale, blocks, stream);
+      rc = attn_dispatch1<8>(
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 16:
-      rc = attn_dispatch1<16>(head_size, part_o, part_ml, qkv, kpool, vpool,
-                              pos, slot, layer, n_layers_pool, n_kv_heads,
-                              max_seq, n_chunks, scale, blocks, stream);
+      rc = attn_dispatch1<16>(
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     default:
       return -1;

@@ -188,6 +188,17 @@ class PipelineRuntime:
     # in the decode loop; each phase leaves the ring drained so callers can
     # barrier/synchronize between phases)
     # ------------------------------------------------------------------
+    def prepare_bench(self, sampling: SamplingParams) -> None:
+        """Capture the fused step graphs BEFORE prefill (capture warm-up
+        scribbles on the KV pool, so it must precede cache filling)."""
+        r = self.runner
+        if (self.is_starter and getattr(r, "backend", "") == "hip"
+                and sampling.top_p >= 1.0 and r.engine.use_graphs):
+            r.engine.ensure_fused_graphs(
+                sampling.temperature, sampling.top_k, sampling.seed or 0
+            )
+            r.reset()
+
     def bench_prefill(self, prompts: Sequence[torch.Tensor]) -> list:
         """Starter: prefill all samples; returns per-sample device tokens
         (argmax of the first logits — value irrelevant for timing)."""
@@ -226,27 +237,59 @@ class PipelineRuntime:
         n = len(toks)
         if gens is None:
             gens = self._generators(sampling, n, self.device)
+        fused = (
+            getattr(runner, "backend", "") == "hip"
+            and sampling.top_p >= 1.0
+            and runner.engine.use_graphs
+        )
+        eng = runner.engine if fused else None
+        if fused:
+            eng.ensure_fused_graphs(sampling.temperature, sampling.top_k,
+                                    sampling.seed or 0)
+            for s in range(n):
+                eng.token_table[s] = toks[s].view(())
+
         if self.world == 1:
+            if fused:
+                for _ in range(n_rounds):
+                    for s in range(n):
+                        eng.standalone_step(s)
+                        runner.pos[s] += 1
+                return [eng.token_table[s: s + 1] for s in range(n)]
             for _ in range(n_rounds):
                 for s in range(n):
                     x = runner.decode_head(toks[s], s)
                     logits = runner.tail(x)
                     toks[s] = self._draw(logits, sampling, gens[s])
             return toks
-        # seed one in-flight message per sample
+
+        # ---- pipeline: seed one in-flight message per sample -------------
         for s in range(n):
-            x = runner.decode_head(toks[s], s)
+            if fused:
+                x = runner.decode_head(eng.token_table[s: s + 1], s)
+            else:
+                x = runner.decode_head(toks[s], s)
             self.comm.send_sched(s, x)
         for r in range(n_rounds):
             last = r == n_rounds - 1
             for s in range(n):
                 x = self.comm.recv_sched()
+                if fused:
+                    if last:
+                        eng.tail_sample_step(x, s)
+                    else:
+                        out = eng.starter_step(x, s)
+                        runner.pos[s] += 1
+                        self.comm.send_sched(s, out)
+                    continue
                 logits = runner.tail(x)
                 toks[s] = self._draw(logits, sampling, gens[s])
                 if not last:
                     x = runner.decode_head(toks[s], s)
                     self.comm.send_sched(s, x)
         self.comm.drain()
+        if fused:
+            return [eng.token_table[s: s + 1] for s in range(n)]
         return toks
 
     def bench_serve_rounds(self, n_samples: int, n_rounds: int) -> None:

@@ -221,26 +221,33 @@ def test_rope_kv_append(ops, qpk, n_kv, hs, ne):
 
 
 # ---------------------------------------------------------------------------
-@pytest.mark.parametrize("qpk,n_kv,hs,S", [
-    (4, 8, 128, 1),
-    (4, 8, 128, 500),
-    (4, 8, 128, 2048),
-    (8, 8, 128, 333),
-    (1, 8, 64, 100),
-    (2, 4, 128, 77),
-    (16, 2, 64, 129),
+@pytest.mark.parametrize("qpk,n_kv,hs,S,ne", [
+    (4, 8, 128, 1, 128),
+    (4, 8, 128, 500, 128),
+    (4, 8, 128, 2048, 128),
+    (8, 8, 128, 333, 128),
+    (1, 8, 64, 100, 64),
+    (2, 4, 128, 77, 0),      # no rope (learned-pos models)
+    (16, 2, 64, 129, 16),    # partial rotary
 ])
-def test_attn_decode(ops, qpk, n_kv, hs, S):
+def test_attn_decode(ops, qpk, n_kv, hs, S, ne):
+    """Fused kernel: ropes q and the current k from the RAW qkv buffer,
+    attends over pool[0..S-2] + current, and appends k/v at pos."""
+    from mdi_llm_amd.models.model import build_rope_cache
+
     torch.manual_seed(22)
     n_head = n_kv * qpk
     max_seq = 2048
     n_layers, n_slots, layer, slot_i = 2, 2, 1, 1
     n_chunks = 32
+    pos_i = S - 1
 
     kpool = mk(n_slots, n_layers, n_kv, max_seq, hs, seed=23)
     vpool = mk(n_slots, n_layers, n_kv, max_seq, hs, seed=24)
     qkv = mk(n_kv * (qpk + 2) * hs, seed=25)
-    pos = torch.tensor([S - 1], device=DEV, dtype=torch.int32)
+    cos, sin = build_rope_cache(max_seq, ne, device=DEV)
+    cos, sin = cos.contiguous(), sin.contiguous()
+    pos = torch.tensor([pos_i], device=DEV, dtype=torch.int32)
     slot = torch.tensor([slot_i], device=DEV, dtype=torch.int32)
     out = torch.empty(n_head * hs, device=DEV, dtype=torch.bfloat16)
     part_o = torch.empty(n_head * n_chunks * hs, device=DEV,
@@ -248,20 +255,30 @@ def test_attn_decode(ops, qpk, n_kv, hs, S):
     part_ml = torch.empty(n_head * n_chunks * 2, device=DEV,
                           dtype=torch.float32)
     scale = 1.0 / math.sqrt(hs)
-    ops.attn_decode(out, part_o, part_ml, qkv, kpool, vpool, pos, slot,
-                    layer, n_chunks, scale)
+    ops.attn_decode(out, part_o, part_ml, qkv, kpool, vpool, cos, sin, pos,
+                    slot, layer, n_chunks, scale)
 
     # fp32 reference
-    grp = qkv.view(n_kv, qpk + 2, hs).float()
-    K = kpool[slot_i, layer, :, :S].float()  # (n_kv, S, hs)
-    V = vpool[slot_i, layer, :, :S].float()
+    c, s = cos[pos_i:pos_i + 1], sin[pos_i:pos_i + 1]
+    grp = qkv.view(n_kv, qpk + 2, hs)
     ref = torch.empty(n_head, hs, device=DEV)
     for g in range(n_kv):
+        k_cur = _rope_ref(grp[g, qpk].view(1, -1), c, s)[0] if ne else \
+            grp[g, qpk].float()
+        K = torch.cat([kpool[slot_i, layer, g, :pos_i].float(),
+                       k_cur.view(1, -1)])
+        V = torch.cat([vpool[slot_i, layer, g, :pos_i].float(),
+                       grp[g, qpk + 1].float().view(1, -1)])
         for j in range(qpk):
-            q = grp[g, j]
-            att = (K[g] @ q) * scale
+            q = _rope_ref(grp[g, j].view(1, -1), c, s)[0] if ne else \
+                grp[g, j].float()
+            att = (K @ q) * scale
             p = torch.softmax(att, dim=0)
-            ref[g * qpk + j] = p @ V[g]
+            ref[g * qpk + j] = p @ V
+        # append happened
+        assert torch.allclose(kpool[slot_i, layer, g, pos_i].float(), k_cur,
+                              atol=2e-2, rtol=2e-2)
+        assert torch.equal(vpool[slot_i, layer, g, pos_i], grp[g, qpk + 1])
     got = out.view(n_head, hs).float()
     assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2), \
         (got - ref).abs().max()
