@@ -250,9 +250,11 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
   // barrier: plain global loads stay in flight across s_barrier, so HBM
   // streams W while the block stages/normalizes x in LDS.
   bf16x8 wpre[ROWS];
+  const bool pre_ok = lane * 8 < K;  // K is a multiple of 8
 #pragma unroll
   for (int r = 0; r < ROWS; ++r)
-    wpre[r] = load8(W + (size_t)min(row0 + r, M - 1) * K + lane * 8);
+    if (pre_ok)
+      wpre[r] = load8(W + (size_t)min(row0 + r, M - 1) * K + lane * 8);
 
   const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
@@ -264,13 +266,26 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
       wrow[r] = W + (size_t)min(row + r, M - 1) * K;
       acc[r] = 0.f;
     }
-    for (int i = lane * 8; i < K; i += 64 * 8) {
+    int istart = lane * 8;
+    if (row == row0 && pre_ok) {
+      // peeled first chunk: consume the pre-barrier prefetch
+      bf16x8 xv = load8(xs + istart);
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+        bf16x8 wv;
+        *reinterpret_cast<int4*>(wv.v) =
+            *reinterpret_cast<const int4*>(wpre[r].v);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[r] += b2f(wv.v[j]) * b2f(xv.v[j]);
+      }
+      istart += 64 * 8;
+    }
+    for (int i = istart; i < K; i += 64 * 8) {
       bf16x8 xv = load8(xs + i);
 #pragma unroll
       for (int r = 0; r < ROWS; ++r) {
-        bf16x8 wv = (row == row0 && i == lane * 8)
-                        ? wpre[r]
-                        : load8(wrow[r] + i);
+        bf16x8 wv = load8(wrow[r] + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           acc[r] += b2f(wv.v[j]) * b2f(xv.v[j]);
@@ -316,8 +331,12 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
 
   // W prefetch across the staging barrier (see gemv_kernel)
   const int rp0 = min(row0, M - 1);
-  bf16x8 gpre = load8(Wg + (size_t)rp0 * K + lane * 8);
-  bf16x8 upre = load8(Wu + (size_t)rp0 * K + lane * 8);
+  const bool pre_ok = lane * 8 < K;
+  bf16x8 gpre, upre;
+  if (pre_ok) {
+    gpre = load8(Wg + (size_t)rp0 * K + lane * 8);
+    upre = load8(Wu + (size_t)rp0 * K + lane * 8);
+  }
 
   const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
@@ -325,10 +344,20 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
     const bf16* grow = Wg + (size_t)row * K;
     const bf16* urow = Wu + (size_t)row * K;
     float ga = 0.f, ua = 0.f;
-    for (int i = lane * 8; i < K; i += 64 * 8) {
-      const bool pre = (row == row0 && i == lane * 8);
-      bf16x8 gv = pre ? gpre : load8(grow + i);
-      bf16x8 uv = pre ? upre : load8(urow + i);
+    int istart = lane * 8;
+    if (row == row0 && pre_ok) {
+      bf16x8 xv = load8(xs + istart);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xf = b2f(xv.v[j]);
+        ga += b2f(gpre.v[j]) * xf;
+        ua += b2f(upre.v[j]) * xf;
+      }
+      istart += 64 * 8;
+    }
+    for (int i = istart; i < K; i += 64 * 8) {
+      bf16x8 gv = load8(grow + i);
+      bf16x8 uv = load8(urow + i);
       bf16x8 xv = load8(xs + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
