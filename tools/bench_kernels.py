@@ -78,6 +78,8 @@ def bench_attn():
     n_head = n_kv * qpk
     max_seq = 8192
     n_chunks = 32
+    from mdi_llm_amd.models.model import build_rope_cache
+
     kpool = torch.randn(1, 1, n_kv, max_seq, hs, device=DEV).to(torch.bfloat16)
     vpool = torch.randn_like(kpool)
     qkv = torch.randn(n_kv * (qpk + 2) * hs, device=DEV).to(torch.bfloat16)
@@ -85,11 +87,13 @@ def bench_attn():
     part_o = torch.empty(n_head * n_chunks * hs, device=DEV)
     part_ml = torch.empty(n_head * n_chunks * 2, device=DEV)
     slot = torch.zeros(1, device=DEV, dtype=torch.int32)
+    cos, sin = build_rope_cache(max_seq, hs, device=DEV)
+    cos, sin = cos.contiguous(), sin.contiguous()
     for S in (128, 512, 2048, 8192):
         pos = torch.tensor([S - 1], device=DEV, dtype=torch.int32)
         us = timeit(lambda: ops.attn_decode(out, part_o, part_ml, qkv, kpool,
-                                            vpool, pos, slot, 0, n_chunks,
-                                            1 / math.sqrt(hs)))
+                                            vpool, cos, sin, pos, slot, 0,
+                                            n_chunks, 1 / math.sqrt(hs)))
         gb = 2 * n_kv * S * hs * 2 / 1e12
         print(f"attn S={S:5d}: {us:7.1f} us  KV-read {gb/(us*1e-6):5.2f} TB/s"
               f"  (incl. combine)")
