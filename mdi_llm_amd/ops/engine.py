@@ -134,9 +134,6 @@ class DecodeEngine:
         self.part_ml = torch.zeros(
             n_head * n_chunks * 2, device=dev, dtype=torch.float32
         )
-        # monotonic split-S arrival counters (in-launch combine; never reset)
-        self.attn_ticket = torch.zeros(cfg.n_query_groups, device=dev,
-                                       dtype=torch.int32)
         if self.is_starter:
             self.logits = torch.zeros(cfg.padded_vocab_size, **bf)
             self.token = torch.zeros(1, device=dev, dtype=torch.int32)
@@ -242,9 +239,9 @@ class DecodeEngine:
             ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
                      w.norm1_w, w.norm1_b, nk, eps, self._r_qkv)
             ops.attn_decode(
-                self.y, self.attn_ticket, self.part_o, self.part_ml,
-                self.qkv, self.kv_pool.k, self.kv_pool.v, self.cos, self.sin,
-                self.pos, self.slot, li, self.n_chunks, scale,
+                self.y, self.part_o, self.part_ml, self.qkv, self.kv_pool.k,
+                self.kv_pool.v, self.cos, self.sin, self.pos, self.slot, li,
+                self.n_chunks, scale,
             )
             if cfg.parallel_residual:
                 # x = x + proj(y) + mlp(norm2(x) or norm1(x))

@@ -168,13 +168,12 @@ void rope_kv_append(torch::Tensor qkv, torch::Tensor kpool,
                         rope_n_elem, qpk, cur_stream());
 }
 
-void attn_decode(torch::Tensor out, torch::Tensor ticket,
-                 torch::Tensor part_o, torch::Tensor part_ml,
-                 torch::Tensor qkv, torch::Tensor kpool, torch::Tensor vpool,
+void attn_decode(torch::Tensor out, torch::Tensor part_o,
+                 torch::Tensor part_ml, torch::Tensor qkv,
+                 torch::Tensor kpool, torch::Tensor vpool,
                  torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
                  torch::Tensor slot, int64_t layer, int64_t n_chunks,
                  double scale) {
-  check_i32(ticket, "ticket");
   check_bf16(out, "out");
   check_f32(part_o, "part_o");
   check_f32(part_ml, "part_ml");
@@ -198,10 +197,8 @@ void attn_decode(torch::Tensor out, torch::Tensor ticket,
   TORCH_CHECK(part_ml.numel() >= (int64_t)n_head * n_chunks * 2,
               "part_ml too small");
   TORCH_CHECK(out.numel() == (int64_t)n_head * hs, "out size");
-  TORCH_CHECK(ticket.numel() >= n_kv, "ticket too small");
   int rc = launch_attn_decode(
-      out.data_ptr(), (unsigned*)ticket.data_ptr<int>(),
-      part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
+      out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
       qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
       rope_ne ? cos_t.data_ptr<float>() : nullptr,
       rope_ne ? sin_t.data_ptr<float>() : nullptr, rope_ne,

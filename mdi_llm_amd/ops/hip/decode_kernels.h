@@ -35,16 +35,14 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
                            hipStream_t stream);
 
 // qkv is RAW (rope applied inside); the kernel also appends the current
-// token's roped k and raw v into the pools at pos, and combines the
-// split-S partials in-launch (ticket: [n_kv_heads] u32, monotonic, never
-// reset).
-int launch_attn_decode(void* out, unsigned* ticket, float* part_o,
-                       float* part_ml, const void* qkv, void* kpool,
-                       void* vpool, const float* cos_t, const float* sin_t,
-                       int rope_ne, const int* pos, const int* slot,
-                       int layer, int n_layers_pool, int n_kv_heads,
-                       int max_seq, int head_size, int qpk, int n_chunks,
-                       float scale, hipStream_t stream);
+// token's roped k and raw v into the pools at pos.
+int launch_attn_decode(void* out, float* part_o, float* part_ml,
+                       const void* qkv, void* kpool, void* vpool,
+                       const float* cos_t, const float* sin_t, int rope_ne,
+                       const int* pos, const int* slot, int layer,
+                       int n_layers_pool, int n_kv_heads, int max_seq,
+                       int head_size, int qpk, int n_chunks, float scale,
+                       hipStream_t stream);
 
 void launch_add(void* out, const void* a, const void* b, int n,
                 hipStream_t stream);

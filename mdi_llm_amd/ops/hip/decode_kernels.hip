@@ -491,8 +491,6 @@ DEVINL float rope_elem(const bf16* row, int d, int ne,
 
 template <int QPK, int HS>
 __global__ void attn_decode_kernel(
-    bf16* __restrict__ out,       // [n_head, head_size] (combined)
-    unsigned* __restrict__ ticket,  // [n_kv_heads] monotonic arrival counter
     float* __restrict__ part_o,   // [n_head, n_chunks, head_size]
     float* __restrict__ part_ml,  // [n_head, n_chunks, 2]
     const bf16* __restrict__ qkv, // interleaved, RAW (rope fused here)
@@ -691,48 +689,6 @@ __global__ void attn_decode_kernel(
       part_ml[((size_t)h * n_chunks + chunk) * 2 + 1] = 0.f;
     }
   }
-
-  // ---- in-launch split-S combine (guide §6 Guideline 16 recipe) ----------
-  // Publish this wave's partials (plain stores + agent release), take a
-  // ticket on the per-kv-head counter; the LAST arriving chunk wave
-  // acquires and reduces all chunks for its head group.  The counter is
-  // monotonic — "last" is (old % n_chunks == n_chunks-1) — so no per-call
-  // reset is needed (graph-replay safe).
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // drain partial stores
-  __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // restate post-fence wait
-  unsigned old = 0;
-  if (lane == 0)
-    old = __hip_atomic_fetch_add(&ticket[g], 1u, __ATOMIC_RELAXED,
-                                 __HIP_MEMORY_SCOPE_AGENT);
-  old = __shfl(old, 0, 64);
-  if (old % (unsigned)n_chunks != (unsigned)n_chunks - 1) return;
-
-  // last arriver for kv-head g: reduce the chunks
-  __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-  // lane -> (head qb, dims d0..d0+ODIM) exactly like the PV map
-  const int h = g * QPK + qb;
-  const float* mlb = part_ml + (size_t)h * n_chunks * 2;
-  float M = -1e30f;
-  for (int c = 0; c < n_chunks; ++c) M = fmaxf(M, mlb[c * 2]);
-  float L = 0.f;
-  for (int c = 0; c < n_chunks; ++c)
-    L += mlb[c * 2 + 1] * __expf(mlb[c * 2] - M);
-  const float inv = 1.f / L;
-  const float* pob = part_o + (size_t)h * n_chunks * head_size + d0;
-  float acc[ODIM];
-#pragma unroll
-  for (int i = 0; i < ODIM; ++i) acc[i] = 0.f;
-#pragma unroll 4
-  for (int c = 0; c < n_chunks; ++c) {
-    const float wgt = __expf(mlb[c * 2] - M);
-    const float* po = pob + (size_t)c * head_size;
-#pragma unroll
-    for (int i = 0; i < ODIM; ++i) acc[i] += wgt * po[i];
-  }
-#pragma unroll
-  for (int i = 0; i < ODIM; ++i)
-    out[(size_t)h * head_size + d0 + i] = f2b(acc[i] * inv);
 }
 
 // Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L
@@ -1065,35 +1021,34 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 }
 
 template <int QPK, int HS>
-static void attn_dispatch2(void* out, unsigned* ticket, float* part_o,
-                           float* part_ml, const void* qkv, void* kpool,
-                           void* vpool, const float* cos_t,
+static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
+                           void* kpool, void* vpool, const float* cos_t,
                            const float* sin_t, int rope_ne, const int* pos,
                            const int* slot, int layer, int n_layers_pool,
                            int n_kv_heads, int max_seq, int n_chunks,
                            float scale, int blocks, hipStream_t stream) {
   hipLaunchKernelGGL((attn_decode_kernel<QPK, HS>), dim3(blocks), dim3(256),
-                     0, stream, (bf16*)out, ticket, part_o, part_ml,
-                     (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
-                     sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-                     n_kv_heads, max_seq, n_chunks, scale);
+                     0, stream, part_o, part_ml, (const bf16*)qkv,
+                     (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
+                     slot, layer, n_layers_pool, n_kv_heads, max_seq,
+                     n_chunks, scale);
 }
 
 template <int QPK>
-static int attn_dispatch1(int head_size, void* out, unsigned* ticket,
-                          float* part_o, float* part_ml, const void* qkv,
-                          void* kpool, void* vpool, const float* cos_t,
-                          const float* sin_t, int rope_ne, const int* pos,
-                          const int* slot, int layer, int n_layers_pool,
-                          int n_kv_heads, int max_seq, int n_chunks,
-                          float scale, int blocks, hipStream_t stream) {
+static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
+                          const void* qkv, void* kpool, void* vpool,
+                          const float* cos_t, const float* sin_t,
+                          int rope_ne, const int* pos, const int* slot,
+                          int layer, int n_layers_pool, int n_kv_heads,
+                          int max_seq, int n_chunks, float scale, int blocks,
+                          hipStream_t stream) {
 #define CASE_HS(H)                                                          \
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
-      attn_dispatch2<QPK, H>(out, ticket, part_o, part_ml, qkv, kpool,    \
-                             vpool, cos_t, sin_t, rope_ne, pos, slot,       \
-                             layer, n_layers_pool, n_kv_heads, max_seq,     \
-                             n_chunks, scale, blocks, stream);              \
+      attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, cos_t,     \
+                             sin_t, rope_ne, pos, slot, layer,              \
+                             n_layers_pool, n_kv_heads, max_seq, n_chunks,  \
+                             scale, blocks, stream);                        \
       return 0;                                                             \
     }                                                                       \
   }
@@ -1105,51 +1060,58 @@ static int attn_dispatch1(int head_size, void* out, unsigned* ticket,
 }
 
 // returns 0 on success, -1 if (qpk, head_size) has no kernel instantiation
-int launch_attn_decode(void* out, unsigned* ticket, float* part_o,
-                       float* part_ml, const void* qkv, void* kpool,
-                       void* vpool, const float* cos_t, const float* sin_t,
-                       int rope_ne, const int* pos, const int* slot,
-                       int layer, int n_layers_pool, int n_kv_heads,
-                       int max_seq, int head_size, int qpk, int n_chunks,
-                       float scale, hipStream_t stream) {
+int launch_attn_decode(void* out, float* part_o, float* part_ml,
+                       const void* qkv, void* kpool, void* vpool,
+                       const float* cos_t, const float* sin_t, int rope_ne,
+                       const int* pos, const int* slot, int layer,
+                       int n_layers_pool, int n_kv_heads, int max_seq,
+                       int head_size, int qpk, int n_chunks, float scale,
+                       hipStream_t stream) {
   const int n_wg = n_kv_heads * n_chunks;
   const int blocks = (n_wg + ATTN_WAVES - 1) / ATTN_WAVES;
   int rc = -1;
   switch (qpk) {
     case 1:
       rc = attn_dispatch1<1>(
-          head_size, out, ticket, part_o, part_ml, qkv, kpool, vpool,
-          cos_t, sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-          n_kv_heads, max_seq, n_chunks, scale, blocks, stream);
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 2:
       rc = attn_dispatch1<2>(
-          head_size, out, ticket, part_o, part_ml, qkv, kpool, vpool,
-          cos_t, sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-          n_kv_heads, max_seq, n_chunks, scale, blocks, stream);
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 4:
       rc = attn_dispatch1<4>(
-          head_size, out, ticket, part_o, part_ml, qkv, kpool, vpool,
-          cos_t, sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-          n_kv_heads, max_seq, n_chunks, scale, blocks, stream);
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 8:
       rc = attn_dispatch1<8>(
-          head_size, out, ticket, part_o, part_ml, qkv, kpool, vpool,
-          cos_t, sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-          n_kv_heads, max_seq, n_chunks, scale, blocks, stream);
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     case 16:
       rc = attn_dispatch1<16>(
-          head_size, out, ticket, part_o, part_ml, qkv, kpool, vpool,
-          cos_t, sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-          n_kv_heads, max_seq, n_chunks, scale, blocks, stream);
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
+          n_chunks, scale, blocks, stream);
       break;
     default:
       return -1;
   }
-  return rc;
+  if (rc != 0) return rc;
+  const int ds = (head_size + 63) / 64;
+  const int n_waves = n_kv_heads * qpk * ds;
+  const int cblocks = (n_waves * 64 + 255) / 256;
+  hipLaunchKernelGGL(attn_combine_kernel, dim3(cblocks), dim3(256), 0,
+                     stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
+                     n_kv_heads * qpk);
+  return 0;
 }
 
 void launch_add(void* out, const void* a, const void* b, int n,

@@ -254,10 +254,9 @@ def test_attn_decode(ops, qpk, n_kv, hs, S, ne):
                          dtype=torch.float32)
     part_ml = torch.empty(n_head * n_chunks * 2, device=DEV,
                           dtype=torch.float32)
-    ticket = torch.zeros(n_kv, device=DEV, dtype=torch.int32)
     scale = 1.0 / math.sqrt(hs)
-    ops.attn_decode(out, ticket, part_o, part_ml, qkv, kpool, vpool, cos,
-                    sin, pos, slot, layer, n_chunks, scale)
+    ops.attn_decode(out, part_o, part_ml, qkv, kpool, vpool, cos, sin, pos,
+                    slot, layer, n_chunks, scale)
 
     # fp32 reference
     c, s = cos[pos_i:pos_i + 1], sin[pos_i:pos_i + 1]

@@ -87,15 +87,13 @@ def bench_attn():
     part_o = torch.empty(n_head * n_chunks * hs, device=DEV)
     part_ml = torch.empty(n_head * n_chunks * 2, device=DEV)
     slot = torch.zeros(1, device=DEV, dtype=torch.int32)
-    ticket = torch.zeros(n_kv, device=DEV, dtype=torch.int32)
     cos, sin = build_rope_cache(max_seq, hs, device=DEV)
     cos, sin = cos.contiguous(), sin.contiguous()
     for S in (128, 512, 2048, 8192):
         pos = torch.tensor([S - 1], device=DEV, dtype=torch.int32)
-        us = timeit(lambda: ops.attn_decode(out, ticket, part_o, part_ml,
-                                            qkv, kpool, vpool, cos, sin, pos,
-                                            slot, 0, n_chunks,
-                                            1 / math.sqrt(hs)))
+        us = timeit(lambda: ops.attn_decode(out, part_o, part_ml, qkv, kpool,
+                                            vpool, cos, sin, pos, slot, 0,
+                                            n_chunks, 1 / math.sqrt(hs)))
         gb = 2 * n_kv * S * hs * 2 / 1e12
         print(f"attn S={S:5d}: {us:7.1f} us  KV-read {gb/(us*1e-6):5.2f} TB/s"
               f"  (incl. combine)")
