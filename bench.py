@@ -58,15 +58,22 @@ def main():
     if world == 1 and args.gpus > 1:
         log("WARNING: --gpus>1 but WORLD_SIZE=1; launch via torchrun")
     n_stages = world
-    assert torch.cuda.is_available(), "bench requires a GPU"
-    device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
-    torch.cuda.set_device(device)
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        torch.cuda.set_device(device)
+    else:
+        # CPU fallback so the exact driver invocation is testable without a
+        # GPU (gloo ring, torch backend, fp32)
+        device = torch.device("cpu")
+        log("no GPU: falling back to CPU/gloo/torch-backend (testing mode)")
+        args.backend = "torch"
 
     import torch.distributed as dist
 
     cpu_group = None
     if world > 1:
-        dist.init_process_group("nccl")
+        dist.init_process_group("nccl" if on_gpu else "gloo")
         cpu_group = dist.new_group(backend="gloo")
 
     n_samples = args.samples or max(n_stages, 1)
@@ -81,7 +88,8 @@ def main():
     t0 = time.time()
     torch.manual_seed(1234 + rank)
     stage = build_stage(cfg, rank, split[rank])
-    stage = stage.to(device=device, dtype=torch.bfloat16)
+    model_dtype = torch.bfloat16 if on_gpu else torch.float32
+    stage = stage.to(device=device, dtype=model_dtype)
     with torch.no_grad():
         for p in stage.parameters():
             p.normal_(0.0, 0.02)
@@ -103,15 +111,18 @@ def main():
 
     comm = None
     if world > 1:
-        comm = RingComm(cfg.n_embd, stage.max_seq_length, device, n_samples)
+        comm = RingComm(cfg.n_embd, stage.max_seq_length, device, n_samples,
+                        dtype=model_dtype)
     rt = PipelineRuntime(runner, rank=rank, world=world, comm=comm,
                          device=device)
 
     def sync_barrier():
-        torch.cuda.synchronize()
+        if on_gpu:
+            torch.cuda.synchronize()
         if world > 1:
             dist.barrier(group=cpu_group)
-            torch.cuda.synchronize()
+            if on_gpu:
+                torch.cuda.synchronize()
 
     sampling = SamplingParams(temperature=0.8, top_k=200, seed=1234)
     gens = rt._generators(sampling, n_samples, device)
@@ -141,7 +152,8 @@ def main():
         toks = rt.bench_decode_rounds(toks, args.steps, sampling, gens)
     else:
         rt.bench_serve_rounds(n_samples, args.steps)
-    torch.cuda.synchronize()
+    if on_gpu:
+        torch.cuda.synchronize()
     elapsed = time.perf_counter() - t_start
     sync_barrier()
 
@@ -165,7 +177,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": "bf16" if on_gpu else "float32",
             "data": "synthetic",
             "config": {
                 "model": cfg.name,
