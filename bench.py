@@ -32,7 +32,10 @@ def parse_args():
     ap.add_argument("--model", default="Meta-Llama-3-8B-Instruct")
     ap.add_argument("--prompt-len", type=int, default=128)
     ap.add_argument("--samples", type=int, default=0,
-                    help="in-flight samples (default = n stages)")
+                    help="in-flight samples (default = n stages * group)")
+    ap.add_argument("--group-size", type=int, default=1,
+                    help="batch B samples per stage pass (grouped decode; "
+                         "weights stream once per B tokens)")
     ap.add_argument("--seq-len", type=int, default=2048,
                     help="max sequence length (KV budget)")
     ap.add_argument("--backend", choices=["hip", "torch"], default="hip")
@@ -76,7 +79,8 @@ def main():
         dist.init_process_group("nccl" if on_gpu else "gloo")
         cpu_group = dist.new_group(backend="gloo")
 
-    n_samples = args.samples or max(n_stages, 1)
+    B = max(args.group_size, 1)
+    n_samples = args.samples or max(n_stages, 1) * B
     cfg = ModelConfig.from_name(args.model)
 
     # ---- build this rank's stage (random init, bf16, on device) ---------
@@ -116,6 +120,29 @@ def main():
     rt = PipelineRuntime(runner, rank=rank, world=world, comm=comm,
                          device=device)
 
+    # grouped decode engine (B samples per stage pass)
+    geng = None
+    group_slots = None
+    if B > 1:
+        from mdi_llm_amd.ops.group_engine import (
+            GroupDecodeEngine,
+            group_engine_supported,
+        )
+
+        if not (on_gpu and runner.backend == "hip"
+                and group_engine_supported(cfg)):
+            raise RuntimeError("--group-size>1 requires the HIP engine and "
+                               "a llama-family config on GPU")
+        G = n_samples // B
+        geng = GroupDecodeEngine(stage, stage.kv_pool, B)
+        group_slots = [
+            torch.arange(g * B, (g + 1) * B, device=device,
+                         dtype=torch.int32)
+            for g in range(G)
+        ]
+        if world > 1:
+            comm.alloc_groups(G, B)
+
     def sync_barrier():
         if on_gpu:
             torch.cuda.synchronize()
@@ -134,14 +161,30 @@ def main():
         for _ in range(n_samples)
     ]
     t0 = time.time()
+    if geng is not None:
+        geng.ensure_graphs(sampling.temperature, sampling.top_k,
+                           sampling.seed or 0)
+        runner.reset()
     if rank == 0:
-        rt.prepare_bench(sampling)
+        if geng is None:
+            rt.prepare_bench(sampling)
         toks = rt.bench_prefill(prompts)
-        if args.warmup > 0:
+        if geng is not None:
+            for si in range(n_samples):
+                geng.set_slot_pos(si, runner.pos[si])
+                geng.token_table[si] = toks[si].view(())
+            if args.warmup > 0:
+                rt.bench_group_rounds(geng, group_slots, args.warmup)
+        elif args.warmup > 0:
             toks = rt.bench_decode_rounds(toks, args.warmup, sampling, gens)
     else:
         rt.bench_serve_prefill(n_samples)
-        if args.warmup > 0:
+        if geng is not None:
+            for si in range(n_samples):
+                geng.set_slot_pos(si, runner.pos[si])
+            if args.warmup > 0:
+                rt.bench_group_serve(geng, group_slots, args.warmup)
+        elif args.warmup > 0:
             rt.bench_serve_rounds(n_samples, args.warmup)
     sync_barrier()
     log(f"rank {rank}: warmup done in {time.time()-t0:.1f}s")
@@ -149,9 +192,15 @@ def main():
     # ---- timed region: exactly K rotations -------------------------------
     t_start = time.perf_counter()
     if rank == 0:
-        toks = rt.bench_decode_rounds(toks, args.steps, sampling, gens)
+        if geng is not None:
+            rt.bench_group_rounds(geng, group_slots, args.steps)
+        else:
+            toks = rt.bench_decode_rounds(toks, args.steps, sampling, gens)
     else:
-        rt.bench_serve_rounds(n_samples, args.steps)
+        if geng is not None:
+            rt.bench_group_serve(geng, group_slots, args.steps)
+        else:
+            rt.bench_serve_rounds(n_samples, args.steps)
     if on_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t_start
@@ -186,6 +235,7 @@ def main():
                 "max_seq_len": stage.max_seq_length,
                 "parallelism": f"pp{n_stages}",
                 "samples_in_flight": n_samples,
+                "group_size": B,
                 "backend": runner.backend,
             },
         }

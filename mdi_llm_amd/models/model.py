@@ -422,9 +422,12 @@ class GPT(nn.Module):
                 f"seq length {value} > block_size {self.config.block_size}"
             )
         self._max_seq_length = value
+        # rebuild on whatever device the model currently lives on
+        dev = self.cos.device if hasattr(self, "cos") else None
         cos, sin = build_rope_cache(
             value,
             self.config.rope_n_elem,
+            device=dev,
             base=self.config.rope_base,
             condense_ratio=self.config.rope_condense_ratio,
         )

@@ -173,7 +173,7 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
                  torch::Tensor kpool, torch::Tensor vpool,
                  torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
                  torch::Tensor slot, int64_t layer, int64_t n_chunks,
-                 double scale) {
+                 double scale, int64_t n_batch) {
   check_bf16(out, "out");
   check_f32(part_o, "part_o");
   check_f32(part_ml, "part_ml");
@@ -189,21 +189,28 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
   const int max_seq = (int)kpool.size(3);
   const int hs = (int)kpool.size(4);
   const int rope_ne = cos_t.dim() > 1 ? (int)cos_t.size(1) : 0;
-  const int qkv_dim = (int)qkv.numel();
+  const int nb = (int)n_batch;
+  const int beff = nb > 0 ? nb : 1;
+  const int qkv_dim = (int)(qkv.numel() / beff);
   const int qpk = qkv_dim / (n_kv * hs) - 2;
   const int n_head = n_kv * qpk;
-  TORCH_CHECK(part_o.numel() >= (int64_t)n_head * n_chunks * hs,
+  if (nb > 0) {
+    TORCH_CHECK(pos.numel() >= nb && slot.numel() >= nb,
+                "pos/slot arrays too small for batch");
+  }
+  TORCH_CHECK(part_o.numel() >= (int64_t)beff * n_head * n_chunks * hs,
               "part_o too small");
-  TORCH_CHECK(part_ml.numel() >= (int64_t)n_head * n_chunks * 2,
+  TORCH_CHECK(part_ml.numel() >= (int64_t)beff * n_head * n_chunks * 2,
               "part_ml too small");
-  TORCH_CHECK(out.numel() == (int64_t)n_head * hs, "out size");
+  TORCH_CHECK(out.numel() == (int64_t)beff * n_head * hs, "out size");
   int rc = launch_attn_decode(
       out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
       qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
       rope_ne ? cos_t.data_ptr<float>() : nullptr,
       rope_ne ? sin_t.data_ptr<float>() : nullptr, rope_ne,
       pos.data_ptr<int>(), slot.data_ptr<int>(), (int)layer, n_layers_pool,
-      n_kv, max_seq, hs, qpk, (int)n_chunks, (float)scale, cur_stream());
+      n_kv, max_seq, hs, qpk, (int)n_chunks, (float)scale, nb,
+      cur_stream());
   TORCH_CHECK(rc == 0, "attn_decode: unsupported geometry qpk=", qpk,
               " head_size=", hs);
 }
@@ -255,7 +262,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embed", &embed, "embedding row gather");
   m.def("rope_kv_append", &rope_kv_append,
         "RoPE on interleaved qkv + KV cache append");
-  m.def("attn_decode", &attn_decode, "GQA flash-decode attention (split-S)");
+  m.def("attn_decode", &attn_decode,
+        "GQA flash-decode attention (split-S, fused rope+append)",
+        py::arg("out"), py::arg("part_o"), py::arg("part_ml"),
+        py::arg("qkv"), py::arg("kpool"), py::arg("vpool"), py::arg("cos"),
+        py::arg("sin"), py::arg("pos"), py::arg("slot"), py::arg("layer"),
+        py::arg("n_chunks"), py::arg("scale"), py::arg("n_batch") = 0);
   m.def("add", &add, "bf16 residual add");
   m.def("sample", &sample,
         "fused temperature/top-k/gumbel token sampling (128k vocab ~15us)",

@@ -36,13 +36,16 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 
 // qkv is RAW (rope applied inside); the kernel also appends the current
 // token's roped k and raw v into the pools at pos.
+// n_batch == 0: single-token mode (pos/slot device scalars);
+// n_batch > 0: batched mode (pos/slot arrays [n_batch], leading batch dim
+// on qkv/part_o/part_ml/out).
 int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        const void* qkv, void* kpool, void* vpool,
                        const float* cos_t, const float* sin_t, int rope_ne,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
                        int head_size, int qpk, int n_chunks, float scale,
-                       hipStream_t stream);
+                       int n_batch, hipStream_t stream);
 
 void launch_add(void* out, const void* a, const void* b, int n,
                 hipStream_t stream);

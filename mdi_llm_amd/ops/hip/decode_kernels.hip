@@ -489,16 +489,19 @@ DEVINL float rope_elem(const bf16* row, int d, int ne,
   return x * c + b2f(row[d - half]) * s;
 }
 
-template <int QPK, int HS>
+// BATCH: 0 -> single-token mode (pos_p/slot_p are device scalars);
+//        1 -> batched mode: pos_p/slot_p are arrays [n_batch], and
+//             qkv/part_o/part_ml/out get a leading batch dimension.
+template <int QPK, int HS, int BATCH>
 __global__ void attn_decode_kernel(
-    float* __restrict__ part_o,   // [n_head, n_chunks, head_size]
-    float* __restrict__ part_ml,  // [n_head, n_chunks, 2]
-    const bf16* __restrict__ qkv, // interleaved, RAW (rope fused here)
+    float* __restrict__ part_o,   // [B?, n_head, n_chunks, head_size]
+    float* __restrict__ part_ml,  // [B?, n_head, n_chunks, 2]
+    const bf16* __restrict__ qkv, // [B?, qkv_dim] interleaved, RAW
     bf16* __restrict__ kpool, bf16* __restrict__ vpool,
     const float* __restrict__ cos_t, const float* __restrict__ sin_t,
     int rope_ne, const int* __restrict__ pos_p,
     const int* __restrict__ slot_p, int layer, int n_layers_pool,
-    int n_kv_heads, int max_seq, int n_chunks, float scale) {
+    int n_kv_heads, int max_seq, int n_chunks, float scale, int n_batch) {
   constexpr int head_size = HS;
   // LDS per wave: q tile (16xHS bf16), current-token k row, p tile,
   // m/l/alpha
@@ -512,15 +515,26 @@ __global__ void attn_decode_kernel(
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wg_id = blockIdx.x * ATTN_WAVES + wave;
-  const int g = wg_id / n_chunks;        // kv head
-  const int chunk = wg_id % n_chunks;
-  if (g >= n_kv_heads) return;  // tail waves exit before any barrier
+  const int per_b = n_kv_heads * n_chunks;
+  const int b = BATCH ? wg_id / per_b : 0;
+  const int rem = BATCH ? wg_id % per_b : wg_id;
+  const int g = rem / n_chunks;          // kv head
+  const int chunk = rem % n_chunks;
+  if (g >= n_kv_heads || b >= n_batch)
+    return;  // tail waves exit before any barrier
 
-  const int S = pos_p[0] + 1;  // keys visible this step
+  const int S = pos_p[b] + 1;  // keys visible this step
   const int pos = S - 1;
-  const int slot = slot_p[0];
+  const int slot = slot_p[b];
   const int keys_per_chunk = ((S + n_chunks - 1) / n_chunks + 15) & ~15;
   const int k_begin = chunk * keys_per_chunk;
+  const int n_head_all = n_kv_heads * QPK;
+  if (BATCH) {
+    const int qkv_dim = n_kv_heads * (QPK + 2) * head_size;
+    qkv += (size_t)b * qkv_dim;
+    part_o += (size_t)b * n_head_all * n_chunks * head_size;
+    part_ml += (size_t)b * n_head_all * n_chunks * 2;
+  }
 
   const size_t cache_base =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
@@ -1026,12 +1040,21 @@ static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
                            const float* sin_t, int rope_ne, const int* pos,
                            const int* slot, int layer, int n_layers_pool,
                            int n_kv_heads, int max_seq, int n_chunks,
-                           float scale, int blocks, hipStream_t stream) {
-  hipLaunchKernelGGL((attn_decode_kernel<QPK, HS>), dim3(blocks), dim3(256),
-                     0, stream, part_o, part_ml, (const bf16*)qkv,
-                     (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
-                     slot, layer, n_layers_pool, n_kv_heads, max_seq,
-                     n_chunks, scale);
+                           float scale, int n_batch, int blocks,
+                           hipStream_t stream) {
+  if (n_batch > 0) {
+    hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 1>), dim3(blocks),
+                       dim3(256), 0, stream, part_o, part_ml,
+                       (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
+                       sin_t, rope_ne, pos, slot, layer, n_layers_pool,
+                       n_kv_heads, max_seq, n_chunks, scale, n_batch);
+  } else {
+    hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 0>), dim3(blocks),
+                       dim3(256), 0, stream, part_o, part_ml,
+                       (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
+                       sin_t, rope_ne, pos, slot, layer, n_layers_pool,
+                       n_kv_heads, max_seq, n_chunks, scale, 1);
+  }
 }
 
 template <int QPK>
@@ -1040,15 +1063,15 @@ static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
                           const float* cos_t, const float* sin_t,
                           int rope_ne, const int* pos, const int* slot,
                           int layer, int n_layers_pool, int n_kv_heads,
-                          int max_seq, int n_chunks, float scale, int blocks,
-                          hipStream_t stream) {
+                          int max_seq, int n_chunks, float scale,
+                          int n_batch, int blocks, hipStream_t stream) {
 #define CASE_HS(H)                                                          \
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
       attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, cos_t,     \
                              sin_t, rope_ne, pos, slot, layer,              \
                              n_layers_pool, n_kv_heads, max_seq, n_chunks,  \
-                             scale, blocks, stream);                        \
+                             scale, n_batch, blocks, stream);               \
       return 0;                                                             \
     }                                                                       \
   }
@@ -1060,14 +1083,15 @@ static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
 }
 
 // returns 0 on success, -1 if (qpk, head_size) has no kernel instantiation
+// n_batch == 0 -> single-token mode; > 0 -> batched (pos/slot arrays)
 int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        const void* qkv, void* kpool, void* vpool,
                        const float* cos_t, const float* sin_t, int rope_ne,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
                        int head_size, int qpk, int n_chunks, float scale,
-                       hipStream_t stream) {
-  const int n_wg = n_kv_heads * n_chunks;
+                       int n_batch, hipStream_t stream) {
+  const int n_wg = n_kv_heads * n_chunks * (n_batch > 0 ? n_batch : 1);
   const int blocks = (n_wg + ATTN_WAVES - 1) / ATTN_WAVES;
   int rc = -1;
   switch (qpk) {
@@ -1075,42 +1099,45 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
       rc = attn_dispatch1<1>(
           head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, blocks, stream);
+          n_chunks, scale, n_batch, blocks, stream);
       break;
     case 2:
       rc = attn_dispatch1<2>(
           head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, blocks, stream);
+          n_chunks, scale, n_batch, blocks, stream);
       break;
     case 4:
       rc = attn_dispatch1<4>(
           head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, blocks, stream);
+          n_chunks, scale, n_batch, blocks, stream);
       break;
     case 8:
       rc = attn_dispatch1<8>(
           head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, blocks, stream);
+          n_chunks, scale, n_batch, blocks, stream);
       break;
     case 16:
       rc = attn_dispatch1<16>(
           head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, blocks, stream);
+          n_chunks, scale, n_batch, blocks, stream);
       break;
     default:
       return -1;
   }
   if (rc != 0) return rc;
+  // the combine kernel is batch-agnostic: [B, n_head, chunks, hs] is just
+  // B*n_head heads
+  const int n_head_eff = n_kv_heads * qpk * (n_batch > 0 ? n_batch : 1);
   const int ds = (head_size + 63) / 64;
-  const int n_waves = n_kv_heads * qpk * ds;
+  const int n_waves = n_head_eff * ds;
   const int cblocks = (n_waves * 64 + 255) / 256;
   hipLaunchKernelGGL(attn_combine_kernel, dim3(cblocks), dim3(256), 0,
                      stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
-                     n_kv_heads * qpk);
+                     n_head_eff);
   return 0;
 }
 

@@ -116,6 +116,29 @@ class RingComm:
         dist.recv(self.recv_buf[:1], self.prev_rank, group=self.group)
         return self.recv_buf[0]
 
+    # -- grouped (batched) scheduled path: fixed (B, n_embd) payloads ------
+    def alloc_groups(self, n_groups: int, group_size: int,
+                     dtype: torch.dtype = torch.bfloat16) -> None:
+        self.gsend = torch.zeros(n_groups, group_size, self.n_embd,
+                                 dtype=dtype, device=self.device)
+        self.grecv = torch.zeros(group_size, self.n_embd, dtype=dtype,
+                                 device=self.device)
+
+    def send_group(self, gi: int, X: torch.Tensor) -> None:
+        key = ("g", gi)
+        if key in self._pending:
+            for w in self._pending.pop(key):
+                w.wait()
+        buf = self.gsend[gi]
+        buf.copy_(X.view(buf.shape))
+        self._pending[key] = [
+            dist.isend(buf, self.next_rank, group=self.group)
+        ]
+
+    def recv_group(self) -> torch.Tensor:
+        dist.recv(self.grecv, self.prev_rank, group=self.group)
+        return self.grecv
+
     def drain(self) -> None:
         for works in self._pending.values():
             for w in works:

@@ -292,6 +292,46 @@ class PipelineRuntime:
             return [eng.token_table[s: s + 1] for s in range(n)]
         return toks
 
+    # ------------------------------------------------------------------
+    # grouped (batched) bench phases — GroupDecodeEngine drives B samples
+    # per pass; a rotation advances all n_groups*B samples by one token
+    # ------------------------------------------------------------------
+    def bench_group_rounds(self, geng, group_slots, n_rounds: int) -> None:
+        """Starter: n_rounds rotations over all groups (graphs captured by
+        geng.ensure_graphs beforehand)."""
+        G = len(group_slots)
+        if self.world == 1:
+            for _ in range(n_rounds):
+                for g in range(G):
+                    geng.set_group(group_slots[g])
+                    geng.standalone_step()
+            return
+        for g in range(G):
+            geng.set_group(group_slots[g])
+            X = geng.head_step()
+            self.comm.send_group(g, X)
+        for r in range(n_rounds):
+            last = r == n_rounds - 1
+            for g in range(G):
+                X = self.comm.recv_group()
+                geng.set_group(group_slots[g])
+                if last:
+                    geng.tail_step(X)
+                else:
+                    out = geng.starter_step(X)
+                    self.comm.send_group(g, out)
+        self.comm.drain()
+
+    def bench_group_serve(self, geng, group_slots, n_rounds: int) -> None:
+        G = len(group_slots)
+        for r in range(n_rounds):
+            for g in range(G):
+                X = self.comm.recv_group()
+                geng.set_group(group_slots[g])
+                out = geng.mid_step(X)
+                self.comm.send_group(g, out)
+        self.comm.drain()
+
     def bench_serve_rounds(self, n_samples: int, n_rounds: int) -> None:
         """Secondary: the matching deterministic message count."""
         for r in range(n_rounds):
