@@ -74,6 +74,8 @@ class GroupDecodeEngine:
         E, I = cfg.n_embd, cfg.intermediate_size
         n_head, hs = cfg.n_head, cfg.head_size
         self.X = torch.zeros(B, E, **bf)
+        self.XN = torch.zeros(B, E, **bf)
+        self.HN = torch.zeros(B, E, **bf)
         self.QKV = torch.zeros(B, cfg.qkv_dim, **bf)
         self.Y = torch.zeros(B, n_head * hs, **bf)
         self.part_o = torch.zeros(B * n_head * n_chunks * hs, device=dev,
@@ -82,7 +84,7 @@ class GroupDecodeEngine:
                                    dtype=torch.float32)
         if self.is_starter:
             self.LOGITS = torch.zeros(B, cfg.padded_vocab_size, **bf)
-            self.sample_scratch = torch.zeros(520, device=dev,
+            self.sample_scratch = torch.zeros(520 * B, device=dev,
                                               dtype=torch.int32)
             self.sample_ctr = torch.zeros(1, device=dev, dtype=torch.int32)
             self.tokens = torch.zeros(B, device=dev, dtype=torch.int32)
@@ -110,11 +112,11 @@ class GroupDecodeEngine:
         self.slots_long.copy_(self.slots)
         torch.index_select(self.pos_table, 0, self.slots_long, out=self.pos)
 
-    def _rms(self, X: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-        xf = X.float()
-        xn = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True)
-                              + self.config.norm_eps)
-        return (xn.to(torch.bfloat16) * w)
+    def _rms(self, X: torch.Tensor, w: torch.Tensor,
+             out: torch.Tensor) -> torch.Tensor:
+        # one batched HIP kernel (replaces a ~6-kernel torch composition)
+        self.ops.rmsnorm(out, X, w, self.config.norm_eps)
+        return out
 
     def _run_blocks(self) -> None:
         cfg = self.config
@@ -122,16 +124,15 @@ class GroupDecodeEngine:
         gelu_gate = cfg.mlp_class_name == "GemmaMLP"
         X = self.X
         for li, w in enumerate(self.blocks):
-            xn = self._rms(X, w.norm1_w)
-            qkv = F.linear(xn, w.attn_w, w.attn_b)
-            self.QKV.copy_(qkv)
+            xn = self._rms(X, w.norm1_w, self.XN)
+            self.QKV.copy_(F.linear(xn, w.attn_w, w.attn_b))
             self.ops.attn_decode(
                 self.Y, self.part_o, self.part_ml, self.QKV, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slots, li,
                 self.n_chunks, scale, self.B,
             )
             a = X + F.linear(self.Y, w.proj_w, w.proj_b)
-            hn = self._rms(a, w.norm2_w)
+            hn = self._rms(a, w.norm2_w, self.HN)
             gate = F.linear(hn, w.fc1_w)
             up = F.linear(hn, w.fc2_w)
             act = (F.gelu(gate, approximate="tanh") if gelu_gate
@@ -140,17 +141,16 @@ class GroupDecodeEngine:
         self.X.copy_(X)
 
     def _tail_seq(self) -> None:
-        xn = self._rms(self.X, self.lnf_w)
+        xn = self._rms(self.X, self.lnf_w, self.XN)
         torch.matmul(xn, self.head_w.t(), out=self.LOGITS)
 
     def _sample_seq(self, temperature, top_k, seed) -> None:
-        for b in range(self.B):
-            self.sample_ctr += 1
-            self.ops.sample(
-                self.tokens[b: b + 1], self.LOGITS[b], self.sample_scratch,
-                float(temperature), int(top_k or 0), temperature > 0.0,
-                int(seed) & 0x7FFFFFFF, self.sample_ctr,
-            )
+        self.sample_ctr += 1
+        self.ops.sample(
+            self.tokens, self.LOGITS, self.sample_scratch,
+            float(temperature), int(top_k or 0), temperature > 0.0,
+            int(seed) & 0x7FFFFFFF, self.sample_ctr, self.B,
+        )
 
     def _embed_seq(self) -> None:
         torch.index_select(self.token_table, 0, self.slots_long,

@@ -36,11 +36,13 @@ void rmsnorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
   check_bf16(out, "out");
   check_bf16(x, "x");
   check_bf16(w, "w");
-  const int n = (int)x.numel();
+  const int n = (int)w.numel();
+  const int nb = (int)(x.numel() / n);
   TORCH_CHECK(n % 8 == 0, "n must be a multiple of 8");
-  TORCH_CHECK(out.numel() == n && w.numel() == n, "size mismatch");
+  TORCH_CHECK(x.numel() == (int64_t)nb * n && out.numel() == x.numel(),
+              "size mismatch");
   launch_rmsnorm(out.data_ptr(), x.data_ptr(), w.data_ptr(), n, (float)eps,
-                 cur_stream());
+                 nb, cur_stream());
 }
 
 void layernorm(torch::Tensor out, torch::Tensor x, torch::Tensor w,
@@ -217,20 +219,25 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
 
 void sample(torch::Tensor out_token, torch::Tensor logits,
             torch::Tensor scratch, double temperature, int64_t top_k,
-            bool noise, int64_t seed, c10::optional<torch::Tensor> ctr) {
+            bool noise, int64_t seed, c10::optional<torch::Tensor> ctr,
+            int64_t n_batch) {
   check_i32(out_token, "out_token");
   check_bf16(logits, "logits");
+  const int nb = n_batch > 0 ? (int)n_batch : 1;
   TORCH_CHECK(scratch.is_cuda() && scratch.scalar_type() == torch::kInt32 &&
-                  scratch.numel() >= 520,
-              "scratch must be >=520 int32 on GPU");
+                  scratch.numel() >= 520 * nb,
+              "scratch must be >=520 int32 per sample on GPU");
+  TORCH_CHECK(out_token.numel() >= nb, "out_token too small");
   const int* cp = nullptr;
   if (ctr.has_value()) {
     check_i32(*ctr, "ctr");
     cp = ctr->data_ptr<int>();
   }
-  launch_sample(out_token.data_ptr(), logits.data_ptr(), (int)logits.numel(),
+  const int V = (int)(logits.numel() / nb);
+  launch_sample(out_token.data_ptr(), logits.data_ptr(), V,
                 scratch.data_ptr(), (float)temperature, (int)top_k,
-                noise ? 1 : 0, (unsigned)(int64_t)seed, cp, cur_stream());
+                noise ? 1 : 0, (unsigned)(int64_t)seed, cp, (int)n_batch,
+                cur_stream());
 }
 
 void add(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
@@ -273,5 +280,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused temperature/top-k/gumbel token sampling (128k vocab ~15us)",
         py::arg("out_token"), py::arg("logits"), py::arg("scratch"),
         py::arg("temperature"), py::arg("top_k"), py::arg("noise"),
-        py::arg("seed"), py::arg("ctr") = c10::nullopt);
+        py::arg("seed"), py::arg("ctr") = c10::nullopt,
+        py::arg("n_batch") = 0);
 }

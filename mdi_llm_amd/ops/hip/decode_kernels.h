@@ -5,8 +5,9 @@
 
 #include <hip/hip_runtime.h>
 
+// n_batch rows of [n] normalized independently (n_batch<=0 -> 1)
 void launch_rmsnorm(void* out, const void* x, const void* w, int n, float eps,
-                    hipStream_t stream);
+                    int n_batch, hipStream_t stream);
 
 void launch_layernorm(void* out, const void* x, const void* w, const void* b,
                       int n, float eps, hipStream_t stream);
@@ -50,7 +51,8 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
 void launch_add(void* out, const void* a, const void* b, int n,
                 hipStream_t stream);
 
-// fused temperature/top-k/gumbel sampling; scratch: >=520 u32, zeroed
+// fused temperature/top-k/gumbel sampling; scratch: >=520 u32 PER SAMPLE,
+// zeroed initially (self-cleaning); n_batch draws from [n_batch, V] logits
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, int noise_on, unsigned seed,
-                   const int* ctr, hipStream_t stream);
+                   const int* ctr, int n_batch, hipStream_t stream);

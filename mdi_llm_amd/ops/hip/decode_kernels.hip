@@ -54,6 +54,9 @@ __global__ void rmsnorm_kernel(bf16* __restrict__ out,
                                const bf16* __restrict__ x,
                                const bf16* __restrict__ w, int n, float eps) {
   __shared__ float red[4];
+  // batched: row blockIdx.y of [B, n]
+  out += (size_t)blockIdx.y * n;
+  x += (size_t)blockIdx.y * n;
   const int tid = threadIdx.x;
   float acc = 0.f;
   // vectorized: 8 bf16 per step
@@ -795,6 +798,8 @@ DEVINL unsigned hash_u32(unsigned x) {
 __global__ void sample_hist_hi_kernel(const bf16* __restrict__ logits, int V,
                                       unsigned* __restrict__ scratch) {
   __shared__ unsigned h[256];
+  logits += (size_t)blockIdx.y * V;
+  scratch += (size_t)blockIdx.y * 520;
   const int tid = threadIdx.x;
   if (tid < 256) h[tid] = 0;
   __syncthreads();
@@ -812,6 +817,7 @@ __global__ void sample_select_hi_kernel(unsigned* __restrict__ scratch,
                                         int top_k) {
   // single wave: serial scan from the top bucket down (256 iterations of
   // LDS-free register work — trivial)
+  scratch += (size_t)blockIdx.y * 520;
   if (threadIdx.x != 0) return;
   unsigned cum = 0;
   int b = 255;
@@ -828,6 +834,8 @@ __global__ void sample_select_hi_kernel(unsigned* __restrict__ scratch,
 __global__ void sample_hist_lo_kernel(const bf16* __restrict__ logits, int V,
                                       unsigned* __restrict__ scratch) {
   __shared__ unsigned h[256];
+  logits += (size_t)blockIdx.y * V;
+  scratch += (size_t)blockIdx.y * 520;
   const int tid = threadIdx.x;
   if (tid < 256) h[tid] = 0;
   __syncthreads();
@@ -844,6 +852,7 @@ __global__ void sample_hist_lo_kernel(const bf16* __restrict__ logits, int V,
 
 __global__ void sample_select_lo_kernel(unsigned* __restrict__ scratch,
                                         int top_k) {
+  scratch += (size_t)blockIdx.y * 520;
   if (threadIdx.x != 0) return;
   unsigned cum = scratch[513];
   const unsigned bucket = scratch[512];
@@ -861,8 +870,11 @@ __global__ void sample_gumbel_argmax_kernel(
     const bf16* __restrict__ logits, int V, unsigned* __restrict__ scratch,
     float inv_temp, int use_threshold, int noise, unsigned seed,
     const int* __restrict__ ctr) {
+  logits += (size_t)blockIdx.y * V;
+  scratch += (size_t)blockIdx.y * 520;
   const unsigned t = use_threshold ? scratch[514] : 0u;
-  const unsigned salt = seed ^ (unsigned)(ctr ? ctr[0] : 0) * 0x9E3779B9u;
+  const unsigned salt = (seed + blockIdx.y * 0x85EBCA6Bu) ^
+                        (unsigned)(ctr ? ctr[0] : 0) * 0x9E3779B9u;
   float best = -1e38f;
   int best_i = 0;
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < V;
@@ -902,6 +914,8 @@ __global__ void sample_gumbel_argmax_kernel(
 
 __global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
                                      int* __restrict__ out) {
+  scratch += (size_t)blockIdx.y * 520;
+  out += blockIdx.y;
   if (threadIdx.x == 0) {
     unsigned long long packed =
         *reinterpret_cast<const unsigned long long*>(&scratch[516]);
@@ -918,26 +932,27 @@ __global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
 
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, int noise_on, unsigned seed,
-                   const int* ctr, hipStream_t stream) {
+                   const int* ctr, int n_batch, hipStream_t stream) {
   unsigned* sc = (unsigned*)scratch;
-  const int blocks = 128;
+  const int B = n_batch > 0 ? n_batch : 1;
+  const int blocks = B > 1 ? 32 : 128;
   const int use_thresh = (top_k > 0 && top_k < V) ? 1 : 0;
   if (use_thresh) {
-    hipLaunchKernelGGL(sample_hist_hi_kernel, dim3(blocks), dim3(256), 0,
+    hipLaunchKernelGGL(sample_hist_hi_kernel, dim3(blocks, B), dim3(256), 0,
                        stream, (const bf16*)logits, V, sc);
-    hipLaunchKernelGGL(sample_select_hi_kernel, dim3(1), dim3(64), 0, stream,
-                       sc, top_k);
-    hipLaunchKernelGGL(sample_hist_lo_kernel, dim3(blocks), dim3(256), 0,
+    hipLaunchKernelGGL(sample_select_hi_kernel, dim3(1, B), dim3(64), 0,
+                       stream, sc, top_k);
+    hipLaunchKernelGGL(sample_hist_lo_kernel, dim3(blocks, B), dim3(256), 0,
                        stream, (const bf16*)logits, V, sc);
-    hipLaunchKernelGGL(sample_select_lo_kernel, dim3(1), dim3(64), 0, stream,
-                       sc, top_k);
+    hipLaunchKernelGGL(sample_select_lo_kernel, dim3(1, B), dim3(64), 0,
+                       stream, sc, top_k);
   }
   float inv_t = temperature > 0.f ? 1.f / temperature : 1.f;
-  hipLaunchKernelGGL(sample_gumbel_argmax_kernel, dim3(blocks), dim3(256), 0,
-                     stream, (const bf16*)logits, V, sc, inv_t, use_thresh,
+  hipLaunchKernelGGL(sample_gumbel_argmax_kernel, dim3(blocks, B), dim3(256),
+                     0, stream, (const bf16*)logits, V, sc, inv_t, use_thresh,
                      noise_on, seed, ctr);
-  hipLaunchKernelGGL(sample_unpack_kernel, dim3(1), dim3(64), 0, stream, sc,
-                     (int*)out_token);
+  hipLaunchKernelGGL(sample_unpack_kernel, dim3(1, B), dim3(64), 0, stream,
+                     sc, (int*)out_token);
 }
 
 static inline int gemv_grid(int M, int rows_per_block) {
@@ -946,9 +961,10 @@ static inline int gemv_grid(int M, int rows_per_block) {
 }
 
 void launch_rmsnorm(void* out, const void* x, const void* w, int n, float eps,
-                    hipStream_t stream) {
-  hipLaunchKernelGGL(rmsnorm_kernel, dim3(1), dim3(256), 0, stream,
-                     (bf16*)out, (const bf16*)x, (const bf16*)w, n, eps);
+                    int n_batch, hipStream_t stream) {
+  hipLaunchKernelGGL(rmsnorm_kernel, dim3(1, n_batch > 0 ? n_batch : 1),
+                     dim3(256), 0, stream, (bf16*)out, (const bf16*)x,
+                     (const bf16*)w, n, eps);
 }
 
 void launch_layernorm(void* out, const void* x, const void* w, const void* b,
