@@ -91,9 +91,16 @@ def main():
         log(f"layer split: {split}")
     t0 = time.time()
     torch.manual_seed(1234 + rank)
-    stage = build_stage(cfg, rank, split[rank])
     model_dtype = torch.bfloat16 if on_gpu else torch.float32
-    stage = stage.to(device=device, dtype=model_dtype)
+    if on_gpu:
+        # build directly on-device in bf16 (a 70B stage would not survive
+        # an fp32 detour: 2x the bytes)
+        torch.set_default_dtype(model_dtype)
+        with torch.device(device):
+            stage = build_stage(cfg, rank, split[rank])
+        torch.set_default_dtype(torch.float32)
+    else:
+        stage = build_stage(cfg, rank, split[rank]).to(dtype=model_dtype)
     with torch.no_grad():
         for p in stage.parameters():
             p.normal_(0.0, 0.02)
@@ -215,8 +222,11 @@ def main():
     if rank == 0:
         total_tokens = n_samples * args.steps
         tps = total_tokens / elapsed
+        metric = ("llama3_8b_pipeline_decode_tok_per_s"
+                  if "Llama-3-8B" in cfg.name
+                  else f"{cfg.name}_pipeline_decode_tok_per_s")
         out = {
-            "metric": "llama3_8b_pipeline_decode_tok_per_s",
+            "metric": metric,
             "value": round(tps, 2),
             "unit": "tokens/s",
             "n_gpus": n_stages,
