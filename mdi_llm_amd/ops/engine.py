@@ -332,6 +332,60 @@ class DecodeEngine:
         torch.cuda.synchronize()
 
     # ------------------------------------------------------------------
+    # HIP prefill: torch GEMMs + hand-written rope-append and causal flash
+    # attention (no S^2 score materialization)
+    # ------------------------------------------------------------------
+    @property
+    def supports_hip_prefill(self) -> bool:
+        cfg = self.config
+        return (cfg.norm_class_name == "RMSNorm"
+                and cfg.mlp_class_name in ("LLaMAMLP", "GemmaMLP")
+                and not cfg.parallel_residual)
+
+    @torch.inference_mode()
+    def prefill_hidden(self, x: torch.Tensor, slot: int,
+                       pos0: int = 0) -> torch.Tensor:
+        """Run the local blocks over a [T, n_embd] prompt chunk, filling the
+        KV pool at positions pos0..pos0+T-1."""
+        import torch.nn.functional as F
+
+        cfg = self.config
+        ops = self.ops
+        T = x.size(0)
+        scale = 1.0 / (cfg.head_size ** 0.5)
+        Y = torch.empty(T, cfg.n_head * cfg.head_size, device=self.device,
+                        dtype=torch.bfloat16)
+        xn = torch.empty_like(x)
+        for li, w in enumerate(self.blocks):
+            ops.rmsnorm(xn, x, w.norm1_w, cfg.norm_eps)
+            qkv = F.linear(xn, w.attn_w, w.attn_b)
+            ops.rope_prefill_append(qkv, self.kv_pool.k, self.kv_pool.v,
+                                    self.cos, self.sin, pos0, slot, li)
+            ops.prefill_attn(Y, qkv, self.kv_pool.k, self.kv_pool.v, pos0,
+                             slot, li, scale)
+            a = x + F.linear(Y, w.proj_w, w.proj_b)
+            hn = xn
+            ops.rmsnorm(hn, a, w.norm2_w, cfg.norm_eps)
+            gelu_gate = cfg.mlp_class_name == "GemmaMLP"
+            gate = F.linear(hn, w.fc1_w)
+            up = F.linear(hn, w.fc2_w)
+            act = (F.gelu(gate, approximate="tanh") if gelu_gate
+                   else F.silu(gate)) * up
+            x = a + F.linear(act, w.mlp_proj_w, w.mlp_proj_b)
+        return x
+
+    @torch.inference_mode()
+    def prefill_prompt(self, tokens: torch.Tensor, slot: int,
+                       pos0: int = 0) -> torch.Tensor:
+        """Starter: embed prompt tokens then run the local blocks."""
+        import torch.nn.functional as F
+
+        x = F.embedding(tokens.long().view(-1), self.wte)
+        if self.config.scale_embeddings:
+            x = x * (self.config.n_embd ** 0.5)
+        return self.prefill_hidden(x, slot, pos0)
+
+    # ------------------------------------------------------------------
     # fully-fused per-token step graphs (bench hot path)
     # ------------------------------------------------------------------
     def _sample_seq(self, temperature: float, top_k: int, seed: int) -> None:

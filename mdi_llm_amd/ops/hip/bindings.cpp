@@ -240,6 +240,53 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
                 cur_stream());
 }
 
+void rope_prefill_append(torch::Tensor qkv, torch::Tensor kpool,
+                         torch::Tensor vpool, torch::Tensor cos_t,
+                         torch::Tensor sin_t, int64_t pos0, int64_t slot,
+                         int64_t layer) {
+  check_bf16(qkv, "qkv");
+  check_bf16(kpool, "kpool");
+  check_bf16(vpool, "vpool");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
+  const int n_layers_pool = (int)kpool.size(1);
+  const int n_kv = (int)kpool.size(2);
+  const int max_seq = (int)kpool.size(3);
+  const int hs = (int)kpool.size(4);
+  const int rope_ne = cos_t.dim() > 1 ? (int)cos_t.size(1) : 0;
+  TORCH_CHECK(qkv.dim() == 2, "qkv must be [T, qkv_dim]");
+  const int T = (int)qkv.size(0);
+  const int qpk = (int)qkv.size(1) / (n_kv * hs) - 2;
+  TORCH_CHECK((int64_t)pos0 + T <= max_seq, "prefill overflows the pool");
+  launch_rope_prefill_append(
+      qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      rope_ne ? cos_t.data_ptr<float>() : nullptr,
+      rope_ne ? sin_t.data_ptr<float>() : nullptr, (int)pos0, (int)slot,
+      (int)layer, n_layers_pool, n_kv, max_seq, hs, rope_ne, qpk, T,
+      cur_stream());
+}
+
+void prefill_attn(torch::Tensor out, torch::Tensor qkv, torch::Tensor kpool,
+                  torch::Tensor vpool, int64_t pos0, int64_t slot,
+                  int64_t layer, double scale) {
+  check_bf16(out, "out");
+  check_bf16(qkv, "qkv");
+  check_bf16(kpool, "kpool");
+  check_bf16(vpool, "vpool");
+  const int n_layers_pool = (int)kpool.size(1);
+  const int n_kv = (int)kpool.size(2);
+  const int max_seq = (int)kpool.size(3);
+  const int hs = (int)kpool.size(4);
+  const int T = (int)qkv.size(0);
+  const int qpk = (int)qkv.size(1) / (n_kv * hs) - 2;
+  TORCH_CHECK(out.numel() == (int64_t)T * n_kv * qpk * hs, "out size");
+  int rc = launch_prefill_attn(
+      out.data_ptr(), qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      (int)pos0, (int)slot, (int)layer, n_layers_pool, n_kv, max_seq, hs,
+      qpk, T, (float)scale, cur_stream());
+  TORCH_CHECK(rc == 0, "prefill_attn: unsupported head_size ", hs);
+}
+
 void add(torch::Tensor out, torch::Tensor a, torch::Tensor b) {
   check_bf16(out, "out");
   check_bf16(a, "a");
@@ -276,6 +323,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("sin"), py::arg("pos"), py::arg("slot"), py::arg("layer"),
         py::arg("n_chunks"), py::arg("scale"), py::arg("n_batch") = 0);
   m.def("add", &add, "bf16 residual add");
+  m.def("rope_prefill_append", &rope_prefill_append,
+        "prefill: rope q/k for T positions + append k/v to the pool");
+  m.def("prefill_attn", &prefill_attn,
+        "causal GQA prefill flash attention (MFMA, online softmax)");
   m.def("sample", &sample,
         "fused temperature/top-k/gumbel token sampling (128k vocab ~15us)",
         py::arg("out_token"), py::arg("logits"), py::arg("scratch"),

@@ -56,3 +56,18 @@ void launch_add(void* out, const void* a, const void* b, int n,
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, int noise_on, unsigned seed,
                    const int* ctr, int n_batch, hipStream_t stream);
+
+// prefill: rope+append all T positions (grid covers T)
+void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,
+                                const float* cos_t, const float* sin_t,
+                                int pos0, int slot, int layer,
+                                int n_layers_pool, int n_kv_heads,
+                                int max_seq, int head_size, int rope_ne,
+                                int qpk, int T, hipStream_t stream);
+
+// prefill flash attention (causal, GQA); out [T, n_head*head_size]
+int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
+                        const void* vpool, int pos0, int slot, int layer,
+                        int n_layers_pool, int n_kv_heads, int max_seq,
+                        int head_size, int qpk, int T, float scale,
+                        hipStream_t stream);

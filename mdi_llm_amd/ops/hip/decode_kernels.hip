@@ -752,6 +752,202 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
+// Prefill: rope q/k for ALL T prompt positions + append k/v to the pool.
+// qkv: [T, qkv_dim] (roped in place); grid (n_kv_heads, T).
+// ---------------------------------------------------------------------------
+__global__ void rope_prefill_append_kernel(
+    bf16* __restrict__ qkv, bf16* __restrict__ kpool,
+    bf16* __restrict__ vpool, const float* __restrict__ cos_t,
+    const float* __restrict__ sin_t, int pos0, int slot, int layer,
+    int n_layers_pool, int n_kv_heads, int max_seq, int head_size,
+    int rope_ne, int qpk) {
+  const int g = blockIdx.x;
+  const int t = blockIdx.y;
+  const int pos = pos0 + t;
+  const int tid = threadIdx.x;
+  const int half = rope_ne >> 1;
+  const int qkv_dim = n_kv_heads * (qpk + 2) * head_size;
+
+  bf16* base = qkv + (size_t)t * qkv_dim + (size_t)g * (qpk + 2) * head_size;
+  // rope q rows + k row: process pairs (d, d+half)
+  for (int idx = tid; idx < (qpk + 1) * half; idx += blockDim.x) {
+    const int r = idx / half;
+    const int d = idx % half;
+    bf16* row = base + (size_t)r * head_size;
+    const float x1 = b2f(row[d]);
+    const float x2 = b2f(row[d + half]);
+    const float c1 = cos_t[(size_t)pos * rope_ne + d];
+    const float s1 = sin_t[(size_t)pos * rope_ne + d];
+    const float c2 = cos_t[(size_t)pos * rope_ne + d + half];
+    const float s2 = sin_t[(size_t)pos * rope_ne + d + half];
+    row[d] = f2b(x1 * c1 - x2 * s1);
+    row[d + half] = f2b(x2 * c2 + x1 * s2);
+  }
+  __syncthreads();
+  const size_t cache_off =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+          (size_t)max_seq * head_size +
+      (size_t)pos * head_size;
+  const bf16* krow = base + (size_t)qpk * head_size;
+  const bf16* vrow = base + (size_t)(qpk + 1) * head_size;
+  for (int d = tid; d < head_size; d += blockDim.x) {
+    kpool[cache_off + d] = krow[d];
+    vpool[cache_off + d] = vrow[d];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Prefill flash attention (causal, GQA): MFMA QK^T with online softmax.
+//
+// One WAVE handles one (query head h, 16-query-row tile): Q tile staged in
+// LDS (roped already), K/V streamed from the pool in 16-key tiles, causal
+// per-row masking, online (m,l) rescaling, O normalized and written
+// directly — no S^2 score materialization, no split-S partials.
+// grid: ceil(n_head * n_qtiles / 4) blocks of 256 threads.
+// ---------------------------------------------------------------------------
+template <int HS>
+__global__ void prefill_attn_kernel(
+    bf16* __restrict__ out,        // [T, n_head*HS]
+    const bf16* __restrict__ qkv,  // [T, qkv_dim], q already roped
+    const bf16* __restrict__ kpool, const bf16* __restrict__ vpool,
+    int pos0, int slot, int layer, int n_layers_pool, int n_kv_heads,
+    int max_seq, int qpk, int T, float scale) {
+  __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
+  __shared__ float p_lds[ATTN_WAVES][16][16];
+  __shared__ float m_lds[ATTN_WAVES][16];
+  __shared__ float l_lds[ATTN_WAVES][16];
+  __shared__ float a_lds[ATTN_WAVES][16];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int n_head = n_kv_heads * qpk;
+  const int n_qtiles = (T + 15) / 16;
+  const int wg_id = blockIdx.x * ATTN_WAVES + wave;
+  const int h = wg_id / n_qtiles;        // query head
+  const int qtile = wg_id % n_qtiles;
+  if (h >= n_head) return;
+  const int g = h / qpk;                 // kv head
+  const int hj = h % qpk;                // q row within group
+  const int q_base = qtile * 16;         // first q row (relative to prompt)
+  const int qkv_dim = n_kv_heads * (qpk + 2) * HS;
+
+  const size_t cache_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq * HS;
+
+  // stage 16 q rows (t = q_base..q_base+15) of head h into LDS
+  for (int i = lane; i < 16 * (HS / 8); i += 64) {
+    const int r = i / (HS / 8);
+    const int d8 = (i % (HS / 8)) * 8;
+    int4 val = {0, 0, 0, 0};
+    const int t = q_base + r;
+    if (t < T) {
+      const bf16* qrow =
+          qkv + (size_t)t * qkv_dim + ((size_t)g * (qpk + 2) + hj) * HS + d8;
+      val = *reinterpret_cast<const int4*>(qrow);
+    }
+    *reinterpret_cast<int4*>(&q_lds[wave][q_swz<HS>(r, d8)]) = val;
+  }
+  if (lane < 16) {
+    m_lds[wave][lane] = -1e30f;
+    l_lds[wave][lane] = 0.f;
+    a_lds[wave][lane] = 1.f;
+  }
+  __syncthreads();
+
+  constexpr int ODIM = HS * 16 / 64;  // dims per lane in the PV map
+  const int qa = lane & 15;   // q row (softmax map)
+  const int sub = lane >> 4;
+  const int qb = lane % 16;   // q row (PV map)
+  const int d0 = (lane / 16) * ODIM;
+
+  float o_acc[ODIM];
+#pragma unroll
+  for (int i = 0; i < ODIM; ++i) o_acc[i] = 0.f;
+
+  // keys visible to the LAST row of this tile (causal upper bound)
+  const int k_last = pos0 + min(q_base + 15, T - 1) + 1;
+  const int n_tiles = (k_last + 15) / 16;
+  for (int kt = 0; kt < n_tiles; ++kt) {
+    const int key0 = kt * 16;
+    // ---- scores via MFMA: A = K tile, B = Q^T --------------------------
+    f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15;
+    const int koff = (lane >> 4) * 8;
+    const bf16* krow = kpool + cache_base + (size_t)(key0 + arow) * HS;
+    const bool row_valid = (key0 + arow) < k_last;
+#pragma unroll
+    for (int c = 0; c < HS / 32; ++c) {
+      bf16x8_t af = {};
+      if (row_valid)
+        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      const bf16x8_t bfr = *reinterpret_cast<const bf16x8_t*>(
+          &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
+      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc4, 0, 0, 0);
+    }
+    // causal mask: key (abs) must be <= q position (abs)
+    float sc[4];
+    const int q_abs = pos0 + q_base + qa;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = key0 + sub * 4 + r;
+      sc[r] = (key <= q_abs && key < k_last) ? acc4[r] * scale : -1e30f;
+    }
+    float tmax = fmaxf(fmaxf(sc[0], sc[1]), fmaxf(sc[2], sc[3]));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    __syncthreads();
+    if (sub == 0) {
+      const float m_old = m_lds[wave][qa];
+      const float m_new = fmaxf(m_old, tmax);
+      a_lds[wave][qa] = __expf(m_old - m_new);
+      m_lds[wave][qa] = m_new;
+    }
+    __syncthreads();
+    const float m_new = m_lds[wave][qa];
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float p = __expf(sc[r] - m_new);
+      const int key = key0 + sub * 4 + r;
+      if (key > q_abs || key >= k_last) p = 0.f;
+      p_lds[wave][sub * 4 + r][qa] = p;
+      psum += p;
+    }
+    psum += __shfl_xor(psum, 16, 64);
+    psum += __shfl_xor(psum, 32, 64);
+    if (sub == 0)
+      l_lds[wave][qa] = l_lds[wave][qa] * a_lds[wave][qa] + psum;
+    __syncthreads();
+
+    const float alphaB = a_lds[wave][qb];
+#pragma unroll
+    for (int i = 0; i < ODIM; ++i) o_acc[i] *= alphaB;
+    const int smax = min(16, k_last - key0);
+    for (int s = 0; s < smax; ++s) {
+      const float p = p_lds[wave][s][qb];
+      const bf16* vrow = vpool + cache_base + (size_t)(key0 + s) * HS + d0;
+#pragma unroll
+      for (int i = 0; i < ODIM; i += 8) {
+        bf16x8 vv = load8(vrow + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (i + j < ODIM) o_acc[i + j] += p * b2f(vv.v[j]);
+      }
+    }
+  }
+
+  // normalize + write
+  const int t_out = q_base + qb;
+  if (t_out < T) {
+    const float inv = 1.f / l_lds[wave][qb];
+    bf16* op = out + (size_t)t_out * (n_head * HS) + (size_t)h * HS + d0;
+#pragma unroll
+    for (int i = 0; i < ODIM; ++i) op[i] = f2b(o_acc[i] * inv);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Residual add: out = a + b (bf16, fp32 math)
 // ---------------------------------------------------------------------------
 __global__ void add_kernel(bf16* __restrict__ out, const bf16* __restrict__ a,
@@ -1155,6 +1351,43 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
                      stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
                      n_head_eff);
   return 0;
+}
+
+void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,
+                                const float* cos_t, const float* sin_t,
+                                int pos0, int slot, int layer,
+                                int n_layers_pool, int n_kv_heads,
+                                int max_seq, int head_size, int rope_ne,
+                                int qpk, int T, hipStream_t stream) {
+  hipLaunchKernelGGL(rope_prefill_append_kernel, dim3(n_kv_heads, T),
+                     dim3(256), 0, stream, (bf16*)qkv, (bf16*)kpool,
+                     (bf16*)vpool, cos_t, sin_t, pos0, slot, layer,
+                     n_layers_pool, n_kv_heads, max_seq, head_size, rope_ne,
+                     qpk);
+}
+
+int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
+                        const void* vpool, int pos0, int slot, int layer,
+                        int n_layers_pool, int n_kv_heads, int max_seq,
+                        int head_size, int qpk, int T, float scale,
+                        hipStream_t stream) {
+  const int n_head = n_kv_heads * qpk;
+  const int n_qtiles = (T + 15) / 16;
+  const int blocks = (n_head * n_qtiles + ATTN_WAVES - 1) / ATTN_WAVES;
+#define PF_CASE(H)                                                          \
+  if (head_size == H) {                                                     \
+    hipLaunchKernelGGL((prefill_attn_kernel<H>), dim3(blocks), dim3(256),   \
+                       0, stream, (bf16*)out, (const bf16*)qkv,             \
+                       (const bf16*)kpool, (const bf16*)vpool, pos0, slot,  \
+                       layer, n_layers_pool, n_kv_heads, max_seq, qpk, T,   \
+                       scale);                                              \
+    return 0;                                                               \
+  }
+  PF_CASE(64)
+  PF_CASE(128)
+  PF_CASE(256)
+#undef PF_CASE
+  return -1;
 }
 
 void launch_add(void* out, const void* a, const void* b, int n,

@@ -93,13 +93,23 @@ class HipRunner(TorchRunner):
 
     @torch.inference_mode()
     def prefill_head(self, tokens: torch.Tensor, slot: int) -> torch.Tensor:
-        out = super().prefill_head(tokens, slot)
+        if self.engine.supports_hip_prefill:
+            out = self.engine.prefill_prompt(tokens, slot, 0)
+            self.pos[slot] = tokens.numel()
+        else:
+            out = super().prefill_head(tokens, slot)
         self.engine.set_slot_pos(slot, self.pos[slot])
         return out
 
     @torch.inference_mode()
     def prefill_mid(self, x: torch.Tensor, slot: int) -> torch.Tensor:
-        out = super().prefill_mid(x, slot)
+        if self.engine.supports_hip_prefill:
+            out = self.engine.prefill_hidden(
+                x.view(-1, self.config.n_embd).to(torch.bfloat16), slot, 0
+            )
+            self.pos[slot] = x.view(-1, self.config.n_embd).size(0)
+        else:
+            out = super().prefill_mid(x, slot)
         self.engine.set_slot_pos(slot, self.pos[slot])
         return out
 
