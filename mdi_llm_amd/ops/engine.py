@@ -121,11 +121,9 @@ class DecodeEngine:
         E, I = cfg.n_embd, cfg.intermediate_size
         n_head, hs = cfg.n_head, cfg.head_size
         self.x = torch.zeros(E, **bf)        # residual stream (graph input)
-        self.xn = torch.zeros(E, **bf)
         self.qkv = torch.zeros(cfg.qkv_dim, **bf)
         self.y = torch.zeros(n_head * hs, **bf)   # attention output
         self.a = torch.zeros(E, **bf)        # proj(attn) (+x)
-        self.hn = torch.zeros(E, **bf)
         self.act = torch.zeros(I, **bf)
         self.m_out = torch.zeros(E, **bf)
         self.part_o = torch.zeros(
@@ -187,12 +185,6 @@ class DecodeEngine:
     # ---------------------------------------------------------------------
     # kernel sequence (eager; also what gets captured)
     # ---------------------------------------------------------------------
-    def _norm(self, out, x, w, b, eps):
-        if self.config.norm_class_name == "RMSNorm":
-            self.ops.rmsnorm(out, x, w, eps)
-        else:
-            self.ops.layernorm(out, x, w, b, eps)
-
     @property
     def _nk(self) -> int:
         """fused-norm kind: 1 RMSNorm, 2 LayerNorm."""

@@ -362,12 +362,21 @@ class PipelineRuntime:
     # ------------------------------------------------------------------
     def serve(self) -> int:
         """Secondary entry: process messages until every sample stopped.
-        Returns the number of activation messages processed."""
+        Returns the number of activation messages processed.  Exceptions /
+        Ctrl-C shut the loop down cleanly (reference
+        utils/context_managers.py:16-56 semantics)."""
+        from ..utils.context_managers import catch_loop_errors
+
         assert not self.is_starter
         runner = self.runner
         seen: set = set()
         stopped: set = set()
         processed = 0
+        with catch_loop_errors(label=f"secondary-{self.rank}"):
+            return self._serve_loop(runner, seen, stopped, processed)
+        return processed
+
+    def _serve_loop(self, runner, seen, stopped, processed) -> int:
         while True:
             s, x, stop = self.comm.recv()
             if stop:
