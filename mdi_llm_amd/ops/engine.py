@@ -203,14 +203,15 @@ class DecodeEngine:
             return 4
         return 2
 
-    def sample_into_token(self, temperature: float, top_k, seed: int) -> torch.Tensor:
+    def sample_into_token(self, temperature: float, top_k, seed: int,
+                          top_p: float = 1.0) -> torch.Tensor:
         """Fused on-GPU sampling from self.logits into self.sample_out.
         (scratch is self-cleaning: the unpack kernel zeroes it.)"""
         self.sample_ctr += 1
         self.ops.sample(
             self.sample_out, self.logits, self.sample_scratch,
             float(temperature), int(top_k or 0), temperature > 0.0,
-            int(seed) & 0x7FFFFFFF, self.sample_ctr,
+            int(seed) & 0x7FFFFFFF, self.sample_ctr, 0, float(top_p),
         )
         return self.sample_out
 
@@ -380,12 +381,13 @@ class DecodeEngine:
     # ------------------------------------------------------------------
     # fully-fused per-token step graphs (bench hot path)
     # ------------------------------------------------------------------
-    def _sample_seq(self, temperature: float, top_k: int, seed: int) -> None:
+    def _sample_seq(self, temperature: float, top_k: int, seed: int,
+                    top_p: float = 1.0) -> None:
         self.sample_ctr += 1
         self.ops.sample(self.sample_out, self.logits, self.sample_scratch,
                         float(temperature), int(top_k or 0),
                         temperature > 0.0, int(seed) & 0x7FFFFFFF,
-                        self.sample_ctr)
+                        self.sample_ctr, 0, float(top_p))
 
     def _advance(self) -> None:
         # token_table[slot] = sample_out ; pos_table[slot] += 1 (in-graph)

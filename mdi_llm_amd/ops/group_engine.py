@@ -144,12 +144,12 @@ class GroupDecodeEngine:
         xn = self._rms(self.X, self.lnf_w, self.XN)
         torch.matmul(xn, self.head_w.t(), out=self.LOGITS)
 
-    def _sample_seq(self, temperature, top_k, seed) -> None:
+    def _sample_seq(self, temperature, top_k, seed, top_p=1.0) -> None:
         self.sample_ctr += 1
         self.ops.sample(
             self.tokens, self.LOGITS, self.sample_scratch,
             float(temperature), int(top_k or 0), temperature > 0.0,
-            int(seed) & 0x7FFFFFFF, self.sample_ctr, self.B,
+            int(seed) & 0x7FFFFFFF, self.sample_ctr, self.B, float(top_p),
         )
 
     def _embed_seq(self) -> None:

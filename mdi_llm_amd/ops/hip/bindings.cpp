@@ -220,7 +220,7 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
 void sample(torch::Tensor out_token, torch::Tensor logits,
             torch::Tensor scratch, double temperature, int64_t top_k,
             bool noise, int64_t seed, c10::optional<torch::Tensor> ctr,
-            int64_t n_batch) {
+            int64_t n_batch, double top_p) {
   check_i32(out_token, "out_token");
   check_bf16(logits, "logits");
   const int nb = n_batch > 0 ? (int)n_batch : 1;
@@ -236,8 +236,8 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
   const int V = (int)(logits.numel() / nb);
   launch_sample(out_token.data_ptr(), logits.data_ptr(), V,
                 scratch.data_ptr(), (float)temperature, (int)top_k,
-                noise ? 1 : 0, (unsigned)(int64_t)seed, cp, (int)n_batch,
-                cur_stream());
+                (float)top_p, noise ? 1 : 0, (unsigned)(int64_t)seed, cp,
+                (int)n_batch, cur_stream());
 }
 
 void rope_prefill_append(torch::Tensor qkv, torch::Tensor kpool,
@@ -332,5 +332,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out_token"), py::arg("logits"), py::arg("scratch"),
         py::arg("temperature"), py::arg("top_k"), py::arg("noise"),
         py::arg("seed"), py::arg("ctr") = c10::nullopt,
-        py::arg("n_batch") = 0);
+        py::arg("n_batch") = 0, py::arg("top_p") = 1.0);
 }

@@ -54,8 +54,9 @@ void launch_add(void* out, const void* a, const void* b, int n,
 // fused temperature/top-k/gumbel sampling; scratch: >=520 u32 PER SAMPLE,
 // zeroed initially (self-cleaning); n_batch draws from [n_batch, V] logits
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
-                   float temperature, int top_k, int noise_on, unsigned seed,
-                   const int* ctr, int n_batch, hipStream_t stream);
+                   float temperature, int top_k, float top_p, int noise_on,
+                   unsigned seed, const int* ctr, int n_batch,
+                   hipStream_t stream);
 
 // prefill: rope+append all T positions (grid covers T)
 void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,

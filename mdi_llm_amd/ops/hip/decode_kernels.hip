@@ -999,14 +999,118 @@ __global__ void sample_hist_hi_kernel(const bf16* __restrict__ logits, int V,
   const int tid = threadIdx.x;
   if (tid < 256) h[tid] = 0;
   __syncthreads();
+  unsigned umax = 0;
   for (int i = blockIdx.x * blockDim.x + tid; i < V;
        i += gridDim.x * blockDim.x) {
     unsigned u = bf16_sortable(
         reinterpret_cast<const unsigned short*>(logits)[i]);
     atomicAdd(&h[u >> 8], 1u);
+    umax = max(umax, u);
   }
   __syncthreads();
   if (tid < 256 && h[tid]) atomicAdd(&scratch[tid], h[tid]);
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    umax = max(umax, (unsigned)__shfl_xor((int)umax, off, 64));
+  if ((tid & 63) == 0) atomicMax(&scratch[515], umax);
+}
+
+// decode a sortable u16 code back to the bf16 float value
+DEVINL float sortable_to_float(unsigned u) {
+  // inverse of bf16_sortable: positive had 0x8000 OR'd, negative was ~x
+  unsigned short bits = (u & 0x8000u) ? (unsigned short)(u & 0x7FFFu)
+                                      : (unsigned short)(~u & 0xFFFFu);
+  bf16 v = *reinterpret_cast<bf16*>(&bits);
+  return b2f(v);
+}
+
+// ---- top-p (nucleus) mass-radix: exp((l-m)/T) histograms over the same
+// sortable buckets; threshold = code where top-down cumulative mass
+// crosses top_p * total.  Uses float views of the hist slots (zeroed by
+// the preceding count-select kernels).
+__global__ void sample_mass_hist_hi_kernel(const bf16* __restrict__ logits,
+                                           int V,
+                                           unsigned* __restrict__ scratch,
+                                           float inv_temp) {
+  __shared__ float h[256];
+  logits += (size_t)blockIdx.y * V;
+  scratch += (size_t)blockIdx.y * 520;
+  float* fh = reinterpret_cast<float*>(scratch);
+  const int tid = threadIdx.x;
+  if (tid < 256) h[tid] = 0.f;
+  __syncthreads();
+  const float m = sortable_to_float(scratch[515]) * inv_temp;
+  for (int i = blockIdx.x * blockDim.x + tid; i < V;
+       i += gridDim.x * blockDim.x) {
+    unsigned short bits = reinterpret_cast<const unsigned short*>(logits)[i];
+    unsigned u = bf16_sortable(bits);
+    float l = b2f(*reinterpret_cast<bf16*>(&bits)) * inv_temp;
+    atomicAdd(&h[u >> 8], __expf(l - m));
+  }
+  __syncthreads();
+  if (tid < 256 && h[tid] != 0.f) atomicAdd(&fh[tid], h[tid]);
+}
+
+__global__ void sample_select_hi_mass_kernel(unsigned* __restrict__ scratch,
+                                             float top_p) {
+  scratch += (size_t)blockIdx.y * 520;
+  float* fh = reinterpret_cast<float*>(scratch);
+  if (threadIdx.x != 0) return;
+  float total = 0.f;
+  for (int b = 0; b < 256; ++b) total += fh[b];
+  const float target = top_p * total;
+  float cum = 0.f;
+  int b = 255;
+  for (; b > 0; --b) {
+    if (cum + fh[b] >= target) break;
+    cum += fh[b];
+  }
+  scratch[512] = (unsigned)b;  // bucket for the lo pass
+  // stash remaining target as float bits in [513]
+  float rem = target - cum;
+  scratch[513] = __float_as_uint(rem);
+}
+
+__global__ void sample_mass_hist_lo_kernel(const bf16* __restrict__ logits,
+                                           int V,
+                                           unsigned* __restrict__ scratch,
+                                           float inv_temp) {
+  __shared__ float h[256];
+  logits += (size_t)blockIdx.y * V;
+  scratch += (size_t)blockIdx.y * 520;
+  float* fh = reinterpret_cast<float*>(scratch);
+  const int tid = threadIdx.x;
+  if (tid < 256) h[tid] = 0.f;
+  __syncthreads();
+  const unsigned bucket = scratch[512];
+  const float m = sortable_to_float(scratch[515]) * inv_temp;
+  for (int i = blockIdx.x * blockDim.x + tid; i < V;
+       i += gridDim.x * blockDim.x) {
+    unsigned short bits = reinterpret_cast<const unsigned short*>(logits)[i];
+    unsigned u = bf16_sortable(bits);
+    if ((u >> 8) != bucket) continue;
+    float l = b2f(*reinterpret_cast<bf16*>(&bits)) * inv_temp;
+    atomicAdd(&h[u & 255], __expf(l - m));
+  }
+  __syncthreads();
+  if (tid < 256 && h[tid] != 0.f) atomicAdd(&fh[256 + tid], h[tid]);
+}
+
+__global__ void sample_select_lo_mass_kernel(unsigned* __restrict__ scratch) {
+  scratch += (size_t)blockIdx.y * 520;
+  float* fh = reinterpret_cast<float*>(scratch);
+  if (threadIdx.x != 0) return;
+  const float rem = __uint_as_float(scratch[513]);
+  const unsigned bucket = scratch[512];
+  float cum = 0.f;
+  int b = 255;
+  for (; b > 0; --b) {
+    if (cum + fh[256 + b] >= rem) break;
+    cum += fh[256 + b];
+  }
+  const unsigned tp = (bucket << 8) | (unsigned)b;
+  // final threshold = max(count-radix top-k threshold, nucleus threshold)
+  if (tp > scratch[514]) scratch[514] = tp;
 }
 
 __global__ void sample_select_hi_kernel(unsigned* __restrict__ scratch,
@@ -1014,17 +1118,18 @@ __global__ void sample_select_hi_kernel(unsigned* __restrict__ scratch,
   // single wave: serial scan from the top bucket down (256 iterations of
   // LDS-free register work — trivial)
   scratch += (size_t)blockIdx.y * 520;
-  if (threadIdx.x != 0) return;
-  unsigned cum = 0;
-  int b = 255;
-  for (; b >= 0; --b) {
-    unsigned c = scratch[b];
-    if (cum + c >= (unsigned)top_k) break;
-    cum += c;
+  if (threadIdx.x == 0) {
+    unsigned cum = 0;
+    int b = 255;
+    for (; b >= 0; --b) {
+      unsigned c = scratch[b];
+      if (cum + c >= (unsigned)top_k) break;
+      cum += c;
+    }
+    if (b < 0) b = 0;
+    scratch[512] = (unsigned)b;
+    scratch[513] = cum;
   }
-  if (b < 0) b = 0;
-  scratch[512] = (unsigned)b;
-  scratch[513] = cum;
 }
 
 __global__ void sample_hist_lo_kernel(const bf16* __restrict__ logits, int V,
@@ -1049,17 +1154,21 @@ __global__ void sample_hist_lo_kernel(const bf16* __restrict__ logits, int V,
 __global__ void sample_select_lo_kernel(unsigned* __restrict__ scratch,
                                         int top_k) {
   scratch += (size_t)blockIdx.y * 520;
-  if (threadIdx.x != 0) return;
-  unsigned cum = scratch[513];
-  const unsigned bucket = scratch[512];
-  int b = 255;
-  for (; b >= 0; --b) {
-    unsigned c = scratch[256 + b];
-    if (cum + c >= (unsigned)top_k) break;
-    cum += c;
+  if (threadIdx.x == 0) {
+    unsigned cum = scratch[513];
+    const unsigned bucket = scratch[512];
+    int b = 255;
+    for (; b >= 0; --b) {
+      unsigned c = scratch[256 + b];
+      if (cum + c >= (unsigned)top_k) break;
+      cum += c;
+    }
+    if (b < 0) b = 0;
+    scratch[514] = (bucket << 8) | (unsigned)b;  // threshold: keep u >= t
   }
-  if (b < 0) b = 0;
-  scratch[514] = (bucket << 8) | (unsigned)b;  // threshold: keep u >= t
+  __syncthreads();
+  // zero the count hists so the (optional) mass phase can reuse the slots
+  for (int i = threadIdx.x; i < 512; i += blockDim.x) scratch[i] = 0;
 }
 
 __global__ void sample_gumbel_argmax_kernel(
@@ -1127,26 +1236,43 @@ __global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
 // ---------------------------------------------------------------------------
 
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
-                   float temperature, int top_k, int noise_on, unsigned seed,
-                   const int* ctr, int n_batch, hipStream_t stream) {
+                   float temperature, int top_k, float top_p, int noise_on,
+                   unsigned seed, const int* ctr, int n_batch,
+                   hipStream_t stream) {
   unsigned* sc = (unsigned*)scratch;
   const int B = n_batch > 0 ? n_batch : 1;
   const int blocks = B > 1 ? 32 : 128;
-  const int use_thresh = (top_k > 0 && top_k < V) ? 1 : 0;
-  if (use_thresh) {
+  float inv_t = temperature > 0.f ? 1.f / temperature : 1.f;
+  const int use_k = (top_k > 0 && top_k < V) ? 1 : 0;
+  const int use_p = (top_p > 0.f && top_p < 1.f) ? 1 : 0;
+  if (use_k || use_p) {
+    // count radix (also records the global max for the mass phase);
+    // with only top-p the count pass still runs (top_k=V keeps everything)
+    const int k_eff = use_k ? top_k : V;
     hipLaunchKernelGGL(sample_hist_hi_kernel, dim3(blocks, B), dim3(256), 0,
                        stream, (const bf16*)logits, V, sc);
     hipLaunchKernelGGL(sample_select_hi_kernel, dim3(1, B), dim3(64), 0,
-                       stream, sc, top_k);
+                       stream, sc, k_eff);
     hipLaunchKernelGGL(sample_hist_lo_kernel, dim3(blocks, B), dim3(256), 0,
                        stream, (const bf16*)logits, V, sc);
-    hipLaunchKernelGGL(sample_select_lo_kernel, dim3(1, B), dim3(64), 0,
-                       stream, sc, top_k);
+    hipLaunchKernelGGL(sample_select_lo_kernel, dim3(1, B), dim3(256), 0,
+                       stream, sc, k_eff);
   }
-  float inv_t = temperature > 0.f ? 1.f / temperature : 1.f;
+  if (use_p) {
+    hipLaunchKernelGGL(sample_mass_hist_hi_kernel, dim3(blocks, B),
+                       dim3(256), 0, stream, (const bf16*)logits, V, sc,
+                       inv_t);
+    hipLaunchKernelGGL(sample_select_hi_mass_kernel, dim3(1, B), dim3(64), 0,
+                       stream, sc, top_p);
+    hipLaunchKernelGGL(sample_mass_hist_lo_kernel, dim3(blocks, B),
+                       dim3(256), 0, stream, (const bf16*)logits, V, sc,
+                       inv_t);
+    hipLaunchKernelGGL(sample_select_lo_mass_kernel, dim3(1, B), dim3(64), 0,
+                       stream, sc);
+  }
   hipLaunchKernelGGL(sample_gumbel_argmax_kernel, dim3(blocks, B), dim3(256),
-                     0, stream, (const bf16*)logits, V, sc, inv_t, use_thresh,
-                     noise_on, seed, ctr);
+                     0, stream, (const bf16*)logits, V, sc, inv_t,
+                     use_k || use_p, noise_on, seed, ctr);
   hipLaunchKernelGGL(sample_unpack_kernel, dim3(1, B), dim3(64), 0, stream,
                      sc, (int*)out_token);
 }

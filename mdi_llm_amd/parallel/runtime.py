@@ -343,10 +343,12 @@ class PipelineRuntime:
 
     def _draw(self, logits, sampling: SamplingParams, gen):
         r = self.runner
-        if getattr(r, "backend", "") == "hip" and sampling.top_p >= 1.0:
-            # fused on-GPU sampler (radix top-k + gumbel), no host sync
+        if getattr(r, "backend", "") == "hip":
+            # fused on-GPU sampler (radix top-k + mass-radix top-p +
+            # gumbel), no host sync
             return r.engine.sample_into_token(
-                sampling.temperature, sampling.top_k, sampling.seed or 0
+                sampling.temperature, sampling.top_k, sampling.seed or 0,
+                sampling.top_p,
             )
         tok = sample_token(
             logits,

@@ -70,6 +70,46 @@ def test_distribution_roughly_matches(ops):
     assert torch.allclose(freq[:5], probs, atol=0.04), freq
 
 
+def _call_p(ops, logits, temperature, top_k, top_p, seed=7, ctr_val=0):
+    scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
+    out = torch.zeros(1, device=DEV, dtype=torch.int32)
+    ctr = torch.tensor([ctr_val], device=DEV, dtype=torch.int32)
+    ops.sample(out, logits, scratch, temperature, top_k, True, seed, ctr,
+               0, top_p)
+    return int(out)
+
+
+def test_top_p_nucleus_support(ops):
+    """Every draw must come from the nucleus (smallest top set with
+    cumulative probability >= p, boundary token included)."""
+    torch.manual_seed(4)
+    V = 4096
+    logits = (torch.randn(V, device=DEV) * 3).to(torch.bfloat16)
+    T, P = 1.0, 0.6
+    probs = torch.softmax(logits.float() / T, dim=0)
+    sp, si = probs.sort(descending=True)
+    cum = sp.cumsum(0)
+    n_keep = int((cum < P).sum()) + 1  # boundary token included
+    nucleus = set(si[:n_keep].tolist())
+    # allow bf16-level ties at the boundary value
+    bval = sp[n_keep - 1]
+    nucleus |= set((probs >= bval * 0.999).nonzero().flatten().tolist())
+    for ctr in range(60):
+        tok = _call_p(ops, logits, T, 0, P, ctr_val=ctr)
+        assert tok in nucleus, (ctr, tok, n_keep)
+
+
+def test_top_p_and_top_k_combined(ops):
+    torch.manual_seed(5)
+    V = 8192
+    logits = (torch.randn(V, device=DEV) * 2).to(torch.bfloat16)
+    kth = torch.topk(logits.float(), 50).values[-1]
+    topset = set((logits.float() >= kth).nonzero().flatten().tolist())
+    for ctr in range(30):
+        tok = _call_p(ops, logits, 0.8, 50, 0.9, ctr_val=ctr)
+        assert tok in topset
+
+
 def test_small_topk(ops):
     torch.manual_seed(3)
     logits = torch.randn(1000, device=DEV).to(torch.bfloat16)
