@@ -196,6 +196,22 @@ configs.extend(
             norm_eps=1e-5,
         ),
         dict(
+            # GPU-testable NeoX-style tiny config: parallel residual,
+            # LayerNorm, partial rotary — exercises those engine paths
+            name="nano-neox-gpu",
+            block_size=256,
+            vocab_size=512,
+            padding_multiple=64,
+            n_layer=3,
+            n_head=4,
+            n_embd=256,
+            rotary_percentage=0.25,
+            parallel_residual=True,
+            bias=True,
+            norm_class_name="LayerNorm",
+            mlp_class_name="GptNeoxMLP",
+        ),
+        dict(
             name="nano-test-gpt2",
             block_size=128,
             vocab_size=256,

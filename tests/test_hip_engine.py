@@ -75,6 +75,43 @@ def test_engine_matches_torch_decode(use_graphs):
 
 
 @torch.inference_mode()
+@pytest.mark.parametrize("name", ["nano-neox-gpu"])
+def test_engine_parallel_residual_matches_torch(name):
+    """Parallel-residual + LayerNorm + partial-rotary decode on the HIP
+    engine vs the torch model."""
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine, engine_supported
+
+    cfg, m = _build(name, seed=21)
+    assert engine_supported(cfg)
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV, dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.eval()
+    stage.set_kv_cache(1)
+    m.set_kv_cache(1)
+
+    torch.manual_seed(22)
+    prompt = torch.randint(0, 511, (10,), device=DEV)
+    ref_logits = m(prompt.view(1, -1), input_pos=0, slot=0)
+    stage.forward_head(prompt.view(1, -1), slot=0, input_pos=0)
+    eng = DecodeEngine(stage, stage.kv_pool, n_chunks=8, use_graphs=False)
+    eng.set_slot_pos(0, 10)
+
+    tok = ref_logits[0, -1].float().argmax()
+    pos = 10
+    for i in range(6):
+        ref_logits = m(tok.view(1, 1), input_pos=pos, slot=0)
+        x = eng.decode_step_head(tok.to(torch.int32), slot=0)
+        logits = eng.tail(x)
+        diff = (logits.float() - ref_logits[0, -1].float()).abs().max()
+        assert diff < 0.5, (i, float(diff))
+        assert int(logits.float().argmax()) == int(
+            ref_logits[0, -1].float().argmax()), i
+        tok = ref_logits[0, -1].float().argmax()
+        pos += 1
+
+
+@torch.inference_mode()
 def test_hip_prefill_matches_torch():
     """prefill_prompt (torch GEMMs + rope-append + MFMA causal flash
     attention) must agree with the torch stage forward: hidden states, KV
