@@ -125,19 +125,27 @@ class GroupDecodeEngine:
         X = self.X
         for li, w in enumerate(self.blocks):
             xn = self._rms(X, w.norm1_w, self.XN)
-            self.QKV.copy_(F.linear(xn, w.attn_w, w.attn_b))
+            qkv = F.linear(xn, w.attn_w, w.attn_b)
+            # contiguous capture-pool temp: stable address under graph replay
             self.ops.attn_decode(
-                self.Y, self.part_o, self.part_ml, self.QKV, self.kv_pool.k,
+                self.Y, self.part_o, self.part_ml, qkv, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slots, li,
                 self.n_chunks, scale, self.B,
             )
-            a = X + F.linear(self.Y, w.proj_w, w.proj_b)
+            # residual adds fused into the GEMMs (addmm beta=1)
+            if w.proj_b is None:
+                a = torch.addmm(X, self.Y, w.proj_w.t())
+            else:
+                a = X + F.linear(self.Y, w.proj_w, w.proj_b)
             hn = self._rms(a, w.norm2_w, self.HN)
             gate = F.linear(hn, w.fc1_w)
             up = F.linear(hn, w.fc2_w)
             act = (F.gelu(gate, approximate="tanh") if gelu_gate
                    else F.silu(gate)) * up
-            X = a + F.linear(act, w.mlp_proj_w, w.mlp_proj_b)
+            if w.mlp_proj_b is None:
+                X = torch.addmm(a, act, w.mlp_proj_w.t())
+            else:
+                X = a + F.linear(act, w.mlp_proj_w, w.mlp_proj_b)
         self.X.copy_(X)
 
     def _tail_seq(self) -> None:
