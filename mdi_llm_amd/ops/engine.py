@@ -84,6 +84,9 @@ class DecodeEngine:
         n_chunks: int = 32,
         use_graphs: bool = True,
     ) -> None:
+        import os
+
+        n_chunks = int(os.environ.get("MDI_ATTN_CHUNKS", n_chunks))
         self.ops = require_hip_ops()
         self.config: ModelConfig = stage.config
         cfg = self.config
