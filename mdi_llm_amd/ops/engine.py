@@ -81,7 +81,7 @@ class DecodeEngine:
         self,
         stage,
         kv_pool: KVCachePool,
-        n_chunks: int = 32,
+        n_chunks: int = 16,
         use_graphs: bool = True,
     ) -> None:
         import os

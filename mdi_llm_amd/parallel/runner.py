@@ -75,7 +75,7 @@ class HipRunner(TorchRunner):
 
     backend = "hip"
 
-    def __init__(self, stage, n_slots: int, n_chunks: int = 32,
+    def __init__(self, stage, n_slots: int, n_chunks: int = 16,
                  use_graphs: bool = True) -> None:
         super().__init__(stage, n_slots)
         from ..ops.engine import DecodeEngine
@@ -129,7 +129,7 @@ class HipRunner(TorchRunner):
 
 
 def make_runner(stage, n_slots: int, device: torch.device,
-                n_chunks: int = 32, use_graphs: bool = True,
+                n_chunks: int = 16, use_graphs: bool = True,
                 force_torch: bool = False):
     """Pick the HIP runner on GPU when the config is supported; fail loudly
     if a GPU is present but the extension is missing (no silent eager
