@@ -13,6 +13,7 @@ from __future__ import annotations
 import os
 import sys
 import time
+from datetime import timedelta
 from pathlib import Path
 from typing import List, Optional
 
@@ -179,7 +180,11 @@ class MDIRuntime:
             os.environ.setdefault("MASTER_ADDR", self.topology.master_addr)
             os.environ.setdefault("MASTER_PORT", str(self.topology.master_port))
             backend = "nccl" if self.device.type == "cuda" else "gloo"
-            dist.init_process_group(backend, rank=0, world_size=self.world)
+            # bounded timeout: a dead peer surfaces as an error instead of a
+            # wedged ring (reference detects peer death via zero-byte recv,
+            # connections.py:174-182)
+            dist.init_process_group(backend, rank=0, world_size=self.world,
+                                    timeout=timedelta(seconds=600))
 
         # 3. build the local stage
         self.stage = build_stage(config, 0, split[0])
@@ -258,7 +263,8 @@ class MDIRuntime:
         os.environ.setdefault("MASTER_PORT", str(msg["master_port"]))
         backend = "nccl" if self.device.type == "cuda" else "gloo"
         dist.init_process_group(backend, rank=msg["rank"],
-                                world_size=msg["world"])
+                                world_size=msg["world"],
+                                timeout=timedelta(seconds=600))
 
         self.stage = build_stage(config, self.rank, n_local)
         hint = Path(msg["chunk_path"]) if msg.get("chunk_path") else None
