@@ -39,6 +39,9 @@ def parse_args():
     ap.add_argument("--seq-len", type=int, default=2048,
                     help="max sequence length (KV budget)")
     ap.add_argument("--backend", choices=["hip", "torch"], default="hip")
+    ap.add_argument("--weights", choices=["bf16", "fp8"], default="bf16",
+                    help="decode weight dtype (fp8 = e4m3 per-row-scaled; "
+                         "NOT the headline config — reported in dtype)")
     ap.add_argument("--no-graphs", action="store_true")
     return ap.parse_args()
 
@@ -49,6 +52,8 @@ def log(msg):
 
 def main():
     args = parse_args()
+    if args.weights == "fp8":
+        os.environ["MDI_WEIGHT_DTYPE"] = "fp8"
     from mdi_llm_amd.config import ModelConfig
     from mdi_llm_amd.models.stages import build_stage
     from mdi_llm_amd.parallel.ring import RingComm
@@ -236,7 +241,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if on_gpu else "float32",
+            "dtype": ("bf16_act_fp8_weights" if args.weights == "fp8" else ("bf16" if on_gpu else "float32")),
             "data": "synthetic",
             "config": {
                 "model": cfg.name,

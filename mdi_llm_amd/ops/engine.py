@@ -40,6 +40,15 @@ def engine_supported(config: ModelConfig) -> bool:
     return True
 
 
+def quantize_fp8_rowwise(w: torch.Tensor):
+    """bf16 [M,K] -> (fp8 e4m3 tensor, fp32 per-row scales).  OCP e4m3
+    (torch.float8_e4m3fn == gfx950's fp8) with absmax/448 scaling."""
+    wf = w.float()
+    s = (wf.abs().amax(dim=1).clamp(min=1e-12) / 448.0).float()
+    q = (wf / s[:, None]).to(torch.float8_e4m3fn)
+    return q.contiguous(), s.contiguous()
+
+
 class _BlockWeights:
     """Contiguous bf16 views of one block's parameters."""
 
@@ -73,6 +82,18 @@ class _BlockWeights:
             self.mlp_proj_w = p(mlp.proj.weight)
             self.mlp_proj_b = p(mlp.proj.bias)
 
+    def quantize_fp8(self, config) -> None:
+        """Attach fp8 copies + per-row scales for the decode GEMVs."""
+        self.attn_w8, self.attn_s = quantize_fp8_rowwise(self.attn_w)
+        self.proj_w8, self.proj_s = quantize_fp8_rowwise(self.proj_w)
+        self.mlp_proj_w8, self.mlp_proj_s = quantize_fp8_rowwise(
+            self.mlp_proj_w)
+        if config.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
+            self.fc1_w8, self.fc1_s = quantize_fp8_rowwise(self.fc1_w)
+            self.fc2_w8, self.fc2_s = quantize_fp8_rowwise(self.fc2_w)
+        else:
+            self.fc_w8, self.fc_s = quantize_fp8_rowwise(self.fc_w)
+
 
 class DecodeEngine:
     """Drives the HIP kernels for a stage (Starter or Secondary module)."""
@@ -87,6 +108,7 @@ class DecodeEngine:
         import os
 
         n_chunks = int(os.environ.get("MDI_ATTN_CHUNKS", n_chunks))
+        self.weight_dtype = os.environ.get("MDI_WEIGHT_DTYPE", "bf16")
         self.ops = require_hip_ops()
         self.config: ModelConfig = stage.config
         cfg = self.config
@@ -101,6 +123,11 @@ class DecodeEngine:
 
         dev = self.device
         self.blocks = [_BlockWeights(b, cfg) for b in stage.transformer.h]
+        self.fp8 = (self.weight_dtype == "fp8"
+                    and self.device.type == "cuda")
+        if self.fp8:
+            for w in self.blocks:
+                w.quantize_fp8(cfg)
         if self.is_starter:
             self.wte = stage.transformer.wte.weight.detach().contiguous()
             self.lnf_w = stage.transformer.ln_f.weight.detach().contiguous()
@@ -108,6 +135,8 @@ class DecodeEngine:
             if self.lnf_b is not None:
                 self.lnf_b = self.lnf_b.detach().contiguous()
             self.head_w = stage.lm_head.weight.detach().contiguous()
+            if self.fp8:
+                self.head_w8, self.head_s = quantize_fp8_rowwise(self.head_w)
             self.head_b = stage.lm_head.bias
             if self.head_b is not None:
                 self.head_b = self.head_b.detach().contiguous()
@@ -232,8 +261,13 @@ class DecodeEngine:
         for li, w in enumerate(self.blocks):
             # qkv = Wqkv @ norm1(x); rope + kv-append are fused inside the
             # attention kernel (qkv stays raw)
-            ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
-                     w.norm1_w, w.norm1_b, nk, eps, self._r_qkv)
+            if self.fp8:
+                ops.gemv_fp8(self.qkv, w.attn_w8, w.attn_s, self.x,
+                             w.attn_b, None, 0, w.norm1_w, w.norm1_b, nk,
+                             eps, self._r_qkv)
+            else:
+                ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
+                         w.norm1_w, w.norm1_b, nk, eps, self._r_qkv)
             ops.attn_decode(
                 self.y, self.part_o, self.part_ml, self.qkv, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slot, li,
@@ -241,16 +275,26 @@ class DecodeEngine:
             )
             if cfg.parallel_residual:
                 # x = x + proj(y) + mlp(norm2(x) or norm1(x))
-                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, None, 0,
-                         None, None, 0, eps, self._r_proj)
+                if self.fp8:
+                    ops.gemv_fp8(self.a, w.proj_w8, w.proj_s, self.y,
+                                 w.proj_b, None, 0, None, None, 0, eps,
+                                 self._r_proj)
+                else:
+                    ops.gemv(self.a, w.proj_w, self.y, w.proj_b, None, 0,
+                             None, None, 0, eps, self._r_proj)
                 nw = w.norm1_w if cfg.shared_attention_norm else w.norm2_w
                 nb = w.norm1_b if cfg.shared_attention_norm else w.norm2_b
                 self._mlp(self.x, w, self.a, nw, nb)
                 ops.add(self.x, self.x, self.m_out)
             else:
                 # a = x + proj(y); x = a + mlp(norm2(a))
-                ops.gemv(self.a, w.proj_w, self.y, w.proj_b, self.x, 1,
-                         None, None, 0, eps, self._r_proj)
+                if self.fp8:
+                    ops.gemv_fp8(self.a, w.proj_w8, w.proj_s, self.y,
+                                 w.proj_b, self.x, 1, None, None, 0, eps,
+                                 self._r_proj)
+                else:
+                    ops.gemv(self.a, w.proj_w, self.y, w.proj_b, self.x, 1,
+                             None, None, 0, eps, self._r_proj)
                 self._mlp(self.a, w, self.a, w.norm2_w, w.norm2_b)
                 # _mlp wrote x = res + down(act) directly
         # (sequential path leaves the stream in self.x)
@@ -265,14 +309,27 @@ class DecodeEngine:
         nk = self._nk
         if cfg.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
             gelu_gate = cfg.mlp_class_name == "GemmaMLP"
-            self.ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp, gelu_gate,
-                                 norm_w, norm_b, nk, eps)
+            if self.fp8:
+                self.ops.gemv_swiglu_fp8(self.act, w.fc1_w8, w.fc1_s,
+                                         w.fc2_w8, w.fc2_s, inp, gelu_gate,
+                                         norm_w, norm_b, nk, eps)
+            else:
+                self.ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp,
+                                     gelu_gate, norm_w, norm_b, nk, eps)
+        elif self.fp8:
+            self.ops.gemv_fp8(self.act, w.fc_w8, w.fc_s, inp, w.fc_b, None,
+                              2, norm_w, norm_b, nk, eps, 0)
         else:
             self.ops.gemv(self.act, w.fc_w, inp, w.fc_b, None, 2,
                           norm_w, norm_b, nk, eps)  # gelu
         out = self.m_out if cfg.parallel_residual else self.x
-        self.ops.gemv(out, w.mlp_proj_w, self.act, w.mlp_proj_b, res, 1,
-                      None, None, 0, eps, self._r_down)
+        if self.fp8:
+            self.ops.gemv_fp8(out, w.mlp_proj_w8, w.mlp_proj_s, self.act,
+                              w.mlp_proj_b, res, 1, None, None, 0, eps,
+                              self._r_down)
+        else:
+            self.ops.gemv(out, w.mlp_proj_w, self.act, w.mlp_proj_b, res, 1,
+                          None, None, 0, eps, self._r_down)
 
     def _embed(self) -> None:
         cfg = self.config
@@ -284,9 +341,15 @@ class DecodeEngine:
 
     def _tail_seq(self) -> None:
         # logits = lm_head @ ln_f(x): one fused kernel
-        self.ops.gemv(self.logits, self.head_w, self.x, self.head_b, None, 0,
-                      self.lnf_w, self.lnf_b, self._nk, self.config.norm_eps,
-                      self._r_head)
+        if self.fp8:
+            self.ops.gemv_fp8(self.logits, self.head_w8, self.head_s,
+                              self.x, self.head_b, None, 0, self.lnf_w,
+                              self.lnf_b, self._nk, self.config.norm_eps,
+                              self._r_head)
+        else:
+            self.ops.gemv(self.logits, self.head_w, self.x, self.head_b,
+                          None, 0, self.lnf_w, self.lnf_b, self._nk,
+                          self.config.norm_eps, self._r_head)
 
     # ---------------------------------------------------------------------
     # public decode API

@@ -101,6 +101,67 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
               cur_stream());
 }
 
+void gemv_fp8(torch::Tensor out, torch::Tensor W, torch::Tensor wscale,
+              torch::Tensor x, c10::optional<torch::Tensor> bias,
+              c10::optional<torch::Tensor> res, int64_t epilogue,
+              c10::optional<torch::Tensor> norm_w,
+              c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
+              double eps, int64_t rows) {
+  check_bf16(out, "out");
+  check_bf16(x, "x");
+  check_f32(wscale, "wscale");
+  TORCH_CHECK(W.is_cuda() && W.element_size() == 1 && W.is_contiguous(),
+              "W must be contiguous fp8/uint8 on GPU");
+  const int K = (int)x.numel();
+  const int M = (int)out.numel();
+  TORCH_CHECK(W.numel() == (int64_t)M * K, "W shape mismatch");
+  TORCH_CHECK(K % 16 == 0, "K must be a multiple of 16 for fp8 weights");
+  TORCH_CHECK(wscale.numel() == M, "wscale size");
+  const void* bp = nullptr;
+  const void* rp = nullptr;
+  const void* nwp = nullptr;
+  const void* nbp = nullptr;
+  if (bias.has_value()) { check_bf16(*bias, "bias"); bp = bias->data_ptr(); }
+  if (res.has_value()) { check_bf16(*res, "res"); rp = res->data_ptr(); }
+  if (norm_kind != 0) {
+    check_bf16(*norm_w, "norm_w");
+    nwp = norm_w->data_ptr();
+    if (norm_b.has_value()) { check_bf16(*norm_b, "norm_b"); nbp = norm_b->data_ptr(); }
+  }
+  launch_gemv_fp8(out.data_ptr(), W.data_ptr(), wscale.data_ptr<float>(),
+                  x.data_ptr(), bp, rp, nwp, nbp, (float)eps, M, K,
+                  (int)epilogue, (int)norm_kind, (int)rows, cur_stream());
+}
+
+void gemv_swiglu_fp8(torch::Tensor out, torch::Tensor Wg,
+                     torch::Tensor gscale, torch::Tensor Wu,
+                     torch::Tensor uscale, torch::Tensor x, bool gelu_gate,
+                     c10::optional<torch::Tensor> norm_w,
+                     c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
+                     double eps) {
+  check_bf16(out, "out");
+  check_bf16(x, "x");
+  check_f32(gscale, "gscale");
+  check_f32(uscale, "uscale");
+  TORCH_CHECK(Wg.element_size() == 1 && Wu.element_size() == 1,
+              "fp8 weights must be 1-byte");
+  const int K = (int)x.numel();
+  const int M = (int)out.numel();
+  TORCH_CHECK(K % 16 == 0, "K must be a multiple of 16 for fp8 weights");
+  const void* nwp = nullptr;
+  const void* nbp = nullptr;
+  if (norm_kind != 0) {
+    check_bf16(*norm_w, "norm_w");
+    nwp = norm_w->data_ptr();
+    if (norm_b.has_value()) { check_bf16(*norm_b, "norm_b"); nbp = norm_b->data_ptr(); }
+  }
+  launch_gemv_swiglu_fp8(out.data_ptr(), Wg.data_ptr(),
+                         gscale.data_ptr<float>(), Wu.data_ptr(),
+                         uscale.data_ptr<float>(), x.data_ptr(), nwp, nbp,
+                         (float)eps, M, K, gelu_gate ? 1 : 0,
+                         (int)norm_kind, cur_stream());
+}
+
 void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
                  torch::Tensor x, bool gelu_gate,
                  c10::optional<torch::Tensor> norm_w,
@@ -308,6 +369,19 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("norm_w") = c10::nullopt, py::arg("norm_b") = c10::nullopt,
         py::arg("norm_kind") = 0, py::arg("eps") = 1e-5,
         py::arg("rows") = 0);
+  m.def("gemv_fp8", &gemv_fp8,
+        "decode GEMV with fp8(e4m3) weights + per-row scales",
+        py::arg("out"), py::arg("W"), py::arg("wscale"), py::arg("x"),
+        py::arg("bias"), py::arg("res"), py::arg("epilogue"),
+        py::arg("norm_w") = c10::nullopt, py::arg("norm_b") = c10::nullopt,
+        py::arg("norm_kind") = 0, py::arg("eps") = 1e-5,
+        py::arg("rows") = 0);
+  m.def("gemv_swiglu_fp8", &gemv_swiglu_fp8,
+        "fused SwiGLU pair GEMV with fp8 weights",
+        py::arg("out"), py::arg("Wg"), py::arg("gscale"), py::arg("Wu"),
+        py::arg("uscale"), py::arg("x"), py::arg("gelu_gate"),
+        py::arg("norm_w") = c10::nullopt, py::arg("norm_b") = c10::nullopt,
+        py::arg("norm_kind") = 0, py::arg("eps") = 1e-5);
   m.def("gemv_swiglu", &gemv_swiglu, "fused SwiGLU pair GEMV (+pre-norm)",
         py::arg("out"), py::arg("Wg"), py::arg("Wu"), py::arg("x"),
         py::arg("gelu_gate"), py::arg("norm_w") = c10::nullopt,

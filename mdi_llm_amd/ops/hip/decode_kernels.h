@@ -20,6 +20,20 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
                  float eps, int M, int K, int epilogue, int norm_kind,
                  int rows, hipStream_t stream);
 
+// fp8(e4m3)-weight variants: W is bytes, wscale fp32 per output row
+void launch_gemv_fp8(void* out, const void* W, const float* wscale,
+                     const void* x, const void* bias, const void* res,
+                     const void* norm_w, const void* norm_b, float eps,
+                     int M, int K, int epilogue, int norm_kind, int rows,
+                     hipStream_t stream);
+
+void launch_gemv_swiglu_fp8(void* out, const void* Wg, const float* gscale,
+                            const void* Wu, const float* uscale,
+                            const void* x, const void* norm_w,
+                            const void* norm_b, float eps, int M, int K,
+                            int gelu_gate, int norm_kind,
+                            hipStream_t stream);
+
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const void* x, const void* norm_w, const void* norm_b,
                         float eps, int M, int K, int gelu_gate, int norm_kind,

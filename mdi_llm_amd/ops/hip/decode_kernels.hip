@@ -39,6 +39,21 @@ DEVINL bf16x8 load8(const bf16* p) {
   return r;
 }
 
+// nontemporal variant for streamed-once weight reads (no L2 retention);
+// the builtin needs a clang ext_vector type, not HIP's int4 struct
+using i32x4_t = __attribute__((ext_vector_type(4))) int;
+
+DEVINL bf16x8 load8_nt(const bf16* p) {
+  bf16x8 r;
+  *reinterpret_cast<i32x4_t*>(r.v) =
+      __builtin_nontemporal_load(reinterpret_cast<const i32x4_t*>(p));
+  return r;
+}
+
+DEVINL i32x4_t load16_nt_u8(const unsigned char* p) {
+  return __builtin_nontemporal_load(reinterpret_cast<const i32x4_t*>(p));
+}
+
 DEVINL float wave_reduce_sum(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
@@ -257,7 +272,7 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
 #pragma unroll
   for (int r = 0; r < ROWS; ++r)
     if (pre_ok)
-      wpre[r] = load8(W + (size_t)min(row0 + r, M - 1) * K + lane * 8);
+      wpre[r] = load8_nt(W + (size_t)min(row0 + r, M - 1) * K + lane * 8);
 
   const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
 
@@ -288,7 +303,7 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
       bf16x8 xv = load8(xs + i);
 #pragma unroll
       for (int r = 0; r < ROWS; ++r) {
-        bf16x8 wv = load8(wrow[r] + i);
+        bf16x8 wv = load8_nt(wrow[r] + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           acc[r] += b2f(wv.v[j]) * b2f(xv.v[j]);
@@ -308,6 +323,138 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
         if (EPI == 3) a = a / (1.f + expf(-a));
         out[rw] = f2b(a);
       }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// FP8 (OCP e4m3) weight variants: W stored as fp8 with one fp32 scale per
+// output row (absmax/448 quantization).  Halves the decode weight stream;
+// dequant via the gfx950 packed converts (v_cvt_pk_f32_fp8).
+// ---------------------------------------------------------------------------
+using f32x2 = __attribute__((__vector_size__(8))) float;
+
+DEVINL float dot16_fp8(const unsigned* wq, const bf16* xs) {
+  float acc = 0.f;
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(wq[q], false);
+    const f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(wq[q], true);
+    const int b = q * 4;
+    acc += lo[0] * b2f(xs[b + 0]) + lo[1] * b2f(xs[b + 1]) +
+           hi[0] * b2f(xs[b + 2]) + hi[1] * b2f(xs[b + 3]);
+  }
+  return acc;
+}
+
+template <int EPI, int NORM, int ROWS>
+__global__ void gemv_fp8_kernel(bf16* __restrict__ out,
+                                const unsigned char* __restrict__ W,
+                                const float* __restrict__ wscale,  // [M]
+                                const bf16* __restrict__ x,
+                                const bf16* __restrict__ bias,
+                                const bf16* __restrict__ res,
+                                const bf16* __restrict__ nw,
+                                const bf16* __restrict__ nb, float eps,
+                                int M, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __shared__ float red[8];
+  bf16* xs = reinterpret_cast<bf16*>(smem);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * ROWS;
+  const int row0 = (blockIdx.x * (blockDim.x >> 6) + wave) * ROWS;
+
+  // prefetch across the staging barrier (16 fp8 = one dwordx4 per lane)
+  i32x4_t wpre[ROWS];
+  const bool pre_ok = lane * 16 < K;
+#pragma unroll
+  for (int r = 0; r < ROWS; ++r)
+    if (pre_ok)
+      wpre[r] = load16_nt_u8(W + (size_t)min(row0 + r, M - 1) * K +
+                             lane * 16);
+
+  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
+
+  for (int row = row0; row < M; row += rows_per_grid) {
+    const unsigned char* wrow[ROWS];
+    float acc[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      wrow[r] = W + (size_t)min(row + r, M - 1) * K;
+      acc[r] = 0.f;
+    }
+    int istart = lane * 16;
+    if (row == row0 && pre_ok) {
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r)
+        acc[r] += dot16_fp8(reinterpret_cast<const unsigned*>(&wpre[r]),
+                            xs + istart);
+      istart += 64 * 16;
+    }
+    for (int i = istart; i < K; i += 64 * 16) {
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+        const i32x4_t wq = load16_nt_u8(wrow[r] + i);
+        acc[r] += dot16_fp8(reinterpret_cast<const unsigned*>(&wq), xs + i);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) acc[r] = wave_reduce_sum(acc[r]);
+    if (lane == 0) {
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+        const int rw = row + r;
+        if (rw >= M) break;
+        float a = acc[r] * nscale * wscale[rw];
+        if (bias != nullptr) a += b2f(bias[rw]);
+        if (EPI == 1 && res != nullptr) a += b2f(res[rw]);
+        if (EPI == 2) a = gelu_tanh(a);
+        if (EPI == 3) a = a / (1.f + expf(-a));
+        out[rw] = f2b(a);
+      }
+    }
+  }
+}
+
+template <int NORM>
+__global__ void gemv_swiglu_fp8_kernel(
+    bf16* __restrict__ out, const unsigned char* __restrict__ Wg,
+    const float* __restrict__ gscale, const unsigned char* __restrict__ Wu,
+    const float* __restrict__ uscale, const bf16* __restrict__ x,
+    const bf16* __restrict__ nw, const bf16* __restrict__ nb, float eps,
+    int M, int K, int gelu_gate) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __shared__ float red[8];
+  bf16* xs = reinterpret_cast<bf16*>(smem);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
+  const int row0 = blockIdx.x * (blockDim.x >> 6) + wave;
+
+  const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
+
+  for (int row = row0; row < M; row += waves_per_grid) {
+    const unsigned char* grow = Wg + (size_t)row * K;
+    const unsigned char* urow = Wu + (size_t)row * K;
+    float ga = 0.f, ua = 0.f;
+    for (int i = lane * 16; i < K; i += 64 * 16) {
+      const i32x4_t gq = load16_nt_u8(grow + i);
+      const i32x4_t uq = load16_nt_u8(urow + i);
+      ga += dot16_fp8(reinterpret_cast<const unsigned*>(&gq), xs + i);
+      ua += dot16_fp8(reinterpret_cast<const unsigned*>(&uq), xs + i);
+    }
+    ga = wave_reduce_sum(ga) * nscale;
+    ua = wave_reduce_sum(ua) * nscale;
+    if (lane == 0) {
+      ga *= gscale[row];
+      ua *= uscale[row];
+      float act = gelu_gate ? gelu_tanh(ga) : ga / (1.f + expf(-ga));
+      out[row] = f2b(act * ua);
     }
   }
 }
@@ -337,8 +484,8 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
   const bool pre_ok = lane * 8 < K;
   bf16x8 gpre, upre;
   if (pre_ok) {
-    gpre = load8(Wg + (size_t)rp0 * K + lane * 8);
-    upre = load8(Wu + (size_t)rp0 * K + lane * 8);
+    gpre = load8_nt(Wg + (size_t)rp0 * K + lane * 8);
+    upre = load8_nt(Wu + (size_t)rp0 * K + lane * 8);
   }
 
   const float nscale = stage_x<NORM>(xs, x, nw, nb, K, eps, red);
@@ -359,8 +506,8 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
       istart += 64 * 8;
     }
     for (int i = istart; i < K; i += 64 * 8) {
-      bf16x8 gv = load8(grow + i);
-      bf16x8 uv = load8(urow + i);
+      bf16x8 gv = load8_nt(grow + i);
+      bf16x8 uv = load8_nt(urow + i);
       bf16x8 xv = load8(xs + i);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -1331,6 +1478,66 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
   }
 #undef GEMV_CASE
 #undef GEMV_CASE1
+}
+
+void launch_gemv_fp8(void* out, const void* W, const float* wscale,
+                     const void* x, const void* bias, const void* res,
+                     const void* norm_w, const void* norm_b, float eps,
+                     int M, int K, int epilogue, int norm_kind, int rows,
+                     hipStream_t stream) {
+  const int smem = K * sizeof(bf16);
+  if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
+  dim3 grid(gemv_grid(M, 4 * rows)), block(256);
+#define GEMV8_CASE1(E, N, R)                                                \
+  hipLaunchKernelGGL((gemv_fp8_kernel<E, N, R>), grid, block, smem, stream, \
+                     (bf16*)out, (const unsigned char*)W, wscale,           \
+                     (const bf16*)x, (const bf16*)bias, (const bf16*)res,   \
+                     (const bf16*)norm_w, (const bf16*)norm_b, eps, M, K)
+#define GEMV8_CASE(E, N)                                                    \
+  do {                                                                      \
+    if (rows == 4) GEMV8_CASE1(E, N, 4);                                    \
+    else if (rows == 2) GEMV8_CASE1(E, N, 2);                               \
+    else GEMV8_CASE1(E, N, 1);                                              \
+  } while (0)
+  switch (epilogue * 4 + norm_kind) {
+    case 0: GEMV8_CASE(0, 0); break;
+    case 1: GEMV8_CASE(0, 1); break;
+    case 2: GEMV8_CASE(0, 2); break;
+    case 4: GEMV8_CASE(1, 0); break;
+    case 5: GEMV8_CASE(1, 1); break;
+    case 6: GEMV8_CASE(1, 2); break;
+    case 8: GEMV8_CASE(2, 0); break;
+    case 9: GEMV8_CASE(2, 1); break;
+    case 10: GEMV8_CASE(2, 2); break;
+    case 12: GEMV8_CASE(3, 0); break;
+    case 13: GEMV8_CASE(3, 1); break;
+    case 14: GEMV8_CASE(3, 2); break;
+    default: GEMV8_CASE(0, 0);
+  }
+#undef GEMV8_CASE
+#undef GEMV8_CASE1
+}
+
+void launch_gemv_swiglu_fp8(void* out, const void* Wg, const float* gscale,
+                            const void* Wu, const float* uscale,
+                            const void* x, const void* norm_w,
+                            const void* norm_b, float eps, int M, int K,
+                            int gelu_gate, int norm_kind,
+                            hipStream_t stream) {
+  const int smem = K * sizeof(bf16);
+  dim3 grid(gemv_grid(M, 4)), block(256);
+#define SW8_CASE(N)                                                         \
+  hipLaunchKernelGGL((gemv_swiglu_fp8_kernel<N>), grid, block, smem,        \
+                     stream, (bf16*)out, (const unsigned char*)Wg, gscale,  \
+                     (const unsigned char*)Wu, uscale, (const bf16*)x,      \
+                     (const bf16*)norm_w, (const bf16*)norm_b, eps, M, K,   \
+                     gelu_gate)
+  switch (norm_kind) {
+    case 1: SW8_CASE(1); break;
+    case 2: SW8_CASE(2); break;
+    default: SW8_CASE(0);
+  }
+#undef SW8_CASE
 }
 
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,

@@ -152,6 +152,83 @@ def test_swiglu_fused_rmsnorm(ops):
     assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2)
 
 
+def test_gemv_fp8(ops):
+    M, K = 4096, 4096
+    torch.manual_seed(50)
+    W = mk(M, K, scale=1.0 / math.sqrt(K), seed=50)
+    x = mk(K, seed=51)
+    from mdi_llm_amd.ops.engine import quantize_fp8_rowwise
+
+    Wq, s = quantize_fp8_rowwise(W)
+    out = torch.empty(M, device=DEV, dtype=torch.bfloat16)
+    ops.gemv_fp8(out, Wq, s, x, None, None, 0)
+    # exact reference of the dequantized weights
+    ref = (Wq.float() * s[:, None]) @ x.float()
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2), \
+        (out.float() - ref).abs().max()
+    # and close to the bf16 weights' result (quantization error bounded)
+    full = W.float() @ x.float()
+    rel = (out.float() - full).abs().mean() / full.abs().mean()
+    assert rel < 0.05, float(rel)
+
+
+def test_gemv_swiglu_fp8(ops):
+    M, K = 2048, 1024
+    Wg = mk(M, K, scale=1.0 / math.sqrt(K), seed=52)
+    Wu = mk(M, K, scale=1.0 / math.sqrt(K), seed=53)
+    x = mk(K, seed=54)
+    from mdi_llm_amd.ops.engine import quantize_fp8_rowwise
+
+    Wgq, gs = quantize_fp8_rowwise(Wg)
+    Wuq, us = quantize_fp8_rowwise(Wu)
+    out = torch.empty(M, device=DEV, dtype=torch.bfloat16)
+    ops.gemv_swiglu_fp8(out, Wgq, gs, Wuq, us, x, False)
+    g = (Wgq.float() * gs[:, None]) @ x.float()
+    u = (Wuq.float() * us[:, None]) @ x.float()
+    ref = torch.nn.functional.silu(g) * u
+    assert torch.allclose(out.float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_engine_fp8_mode():
+    """fp8-weight engine decode stays close to the bf16 engine."""
+    import os
+
+    from mdi_llm_amd import GPT, ModelConfig
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine
+
+    torch.manual_seed(55)
+    cfg = ModelConfig.from_name("nano-gpu")
+    m = GPT(cfg)
+    m.apply_init()
+    m = m.to(device=DEV, dtype=torch.bfloat16)
+    m.eval()
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV, dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.eval()
+    stage.set_kv_cache(1)
+
+    prompt = torch.randint(0, 511, (8,), device=DEV)
+    stage.forward_head(prompt.view(1, -1), slot=0, input_pos=0)
+    os.environ["MDI_WEIGHT_DTYPE"] = "fp8"
+    try:
+        eng = DecodeEngine(stage, stage.kv_pool, n_chunks=8, use_graphs=False)
+    finally:
+        del os.environ["MDI_WEIGHT_DTYPE"]
+    assert eng.fp8
+    eng.set_slot_pos(0, 8)
+    tok = torch.tensor([5], device=DEV, dtype=torch.int32)
+    x = eng.decode_step_head(tok, 0)
+    logits8 = eng.tail(x).float().clone()
+
+    m.set_kv_cache(1)
+    m(prompt.view(1, -1), input_pos=0, slot=0)
+    ref = m(torch.tensor([[5]], device=DEV), input_pos=8, slot=0)
+    ref = ref[0, -1].float()
+    rel = (logits8 - ref).abs().mean() / ref.abs().std()
+    assert rel < 0.2, float(rel)  # fp8 quantization-level agreement
+
+
 def test_embed(ops):
     V, E = 1000, 512
     wte = mk(V, E, seed=17)
