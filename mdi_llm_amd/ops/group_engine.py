@@ -47,10 +47,13 @@ class GroupDecodeEngine:
         n_chunks: int = 16,
         use_graphs: bool = True,
     ) -> None:
+        import os
+
         self.ops = require_hip_ops()
         cfg: ModelConfig = stage.config
         if not group_engine_supported(cfg):
             raise ValueError(f"{cfg.name!r} unsupported by GroupDecodeEngine")
+        self.fp8 = os.environ.get("MDI_WEIGHT_DTYPE", "bf16") == "fp8"
         self.config = cfg
         self.stage = stage
         self.kv_pool = kv_pool
@@ -63,10 +66,19 @@ class GroupDecodeEngine:
         dev = self.device
         B = self.B
         self.blocks = [_BlockWeights(b, cfg) for b in stage.transformer.h]
+        if self.fp8:
+            from .engine import quantize_fp8_rowwise
+
+            for w in self.blocks:
+                w.quantize_fp8(cfg)
         if self.is_starter:
             self.wte = stage.transformer.wte.weight.detach().contiguous()
             self.lnf_w = stage.transformer.ln_f.weight.detach().contiguous()
             self.head_w = stage.lm_head.weight.detach().contiguous()
+            if self.fp8:
+                from .engine import quantize_fp8_rowwise
+
+                self.head_w8, self.head_s = quantize_fp8_rowwise(self.head_w)
         self.cos = stage.cos.detach().to(torch.float32).contiguous()
         self.sin = stage.sin.detach().to(torch.float32).contiguous()
 
@@ -118,31 +130,60 @@ class GroupDecodeEngine:
         self.ops.rmsnorm(out, X, w, self.config.norm_eps)
         return out
 
+    def _mm(self, x: torch.Tensor, w8: torch.Tensor, ws: torch.Tensor,
+            wbf: torch.Tensor):
+        """fp8 GEMM via hipBLASLt _scaled_mm with dynamic per-token
+        activation scales; falls back to bf16 F.linear."""
+        if not self.fp8:
+            return F.linear(x, wbf)
+        ax = (x.float().abs().amax(dim=1, keepdim=True)
+              .clamp(min=1e-12) / 448.0)
+        xq = (x.float() / ax).to(torch.float8_e4m3fn)
+        return torch._scaled_mm(xq, w8.t(), scale_a=ax,
+                                scale_b=ws[None, :],
+                                out_dtype=torch.bfloat16)
+
     def _run_blocks(self) -> None:
         cfg = self.config
         scale = 1.0 / (cfg.head_size ** 0.5)
         gelu_gate = cfg.mlp_class_name == "GemmaMLP"
         X = self.X
+        fp8 = self.fp8
         for li, w in enumerate(self.blocks):
             xn = self._rms(X, w.norm1_w, self.XN)
-            qkv = F.linear(xn, w.attn_w, w.attn_b)
+            if fp8:
+                qkv = self._mm(xn, w.attn_w8, w.attn_s, w.attn_w)
+                if w.attn_b is not None:
+                    qkv = qkv + w.attn_b
+                qkv = qkv.contiguous()
+            else:
+                qkv = F.linear(xn, w.attn_w, w.attn_b)
             # contiguous capture-pool temp: stable address under graph replay
             self.ops.attn_decode(
                 self.Y, self.part_o, self.part_ml, qkv, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slots, li,
                 self.n_chunks, scale, self.B,
             )
-            # residual adds fused into the GEMMs (addmm beta=1)
-            if w.proj_b is None:
+            # residual adds fused into the GEMMs (addmm beta=1) on bf16
+            if fp8:
+                a = X + self._mm(self.Y, w.proj_w8, w.proj_s, w.proj_w)
+            elif w.proj_b is None:
                 a = torch.addmm(X, self.Y, w.proj_w.t())
             else:
                 a = X + F.linear(self.Y, w.proj_w, w.proj_b)
             hn = self._rms(a, w.norm2_w, self.HN)
-            gate = F.linear(hn, w.fc1_w)
-            up = F.linear(hn, w.fc2_w)
+            if fp8:
+                gate = self._mm(hn, w.fc1_w8, w.fc1_s, w.fc1_w)
+                up = self._mm(hn, w.fc2_w8, w.fc2_s, w.fc2_w)
+            else:
+                gate = F.linear(hn, w.fc1_w)
+                up = F.linear(hn, w.fc2_w)
             act = (F.gelu(gate, approximate="tanh") if gelu_gate
                    else F.silu(gate)) * up
-            if w.mlp_proj_b is None:
+            if fp8:
+                X = a + self._mm(act, w.mlp_proj_w8, w.mlp_proj_s,
+                                 w.mlp_proj_w)
+            elif w.mlp_proj_b is None:
                 X = torch.addmm(a, act, w.mlp_proj_w.t())
             else:
                 X = a + F.linear(act, w.mlp_proj_w, w.mlp_proj_b)
@@ -150,7 +191,11 @@ class GroupDecodeEngine:
 
     def _tail_seq(self) -> None:
         xn = self._rms(self.X, self.lnf_w, self.XN)
-        torch.matmul(xn, self.head_w.t(), out=self.LOGITS)
+        if self.fp8:
+            self.LOGITS.copy_(
+                self._mm(xn, self.head_w8, self.head_s, self.head_w))
+        else:
+            torch.matmul(xn, self.head_w.t(), out=self.LOGITS)
 
     def _sample_seq(self, temperature, top_k, seed, top_p=1.0) -> None:
         self.sample_ctr += 1
