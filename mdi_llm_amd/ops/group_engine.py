@@ -53,7 +53,15 @@ class GroupDecodeEngine:
         cfg: ModelConfig = stage.config
         if not group_engine_supported(cfg):
             raise ValueError(f"{cfg.name!r} unsupported by GroupDecodeEngine")
-        self.fp8 = os.environ.get("MDI_WEIGHT_DTYPE", "bf16") == "fp8"
+        # fp8 grouped GEMMs measured SLOWER than bf16 hipBLASLt on ROCm 7
+        # (skinny-M _scaled_mm + dynamic-quant overhead): keep groups bf16.
+        # fp8 weights remain available for B=1 decode (DecodeEngine).
+        if os.environ.get("MDI_WEIGHT_DTYPE", "bf16") == "fp8":
+            import warnings
+
+            warnings.warn("grouped decode ignores MDI_WEIGHT_DTYPE=fp8 "
+                          "(bf16 is faster for batched GEMMs; see ROADMAP)")
+        self.fp8 = False
         self.config = cfg
         self.stage = stage
         self.kv_pool = kv_pool
