@@ -103,6 +103,22 @@ GPT2_MAP = {  # Conv1D weights need transposition (handled below)
     "ln_f.bias": "transformer.ln_f.bias",
 }
 
+PHI_MAP = {
+    "model.embed_tokens.weight": "transformer.wte.weight",
+    "model.layers.{}.input_layernorm.weight": "transformer.h.{}.norm_1.weight",
+    "model.layers.{}.input_layernorm.bias": "transformer.h.{}.norm_1.bias",
+    "model.layers.{}.self_attn.dense.weight": "transformer.h.{}.attn.proj.weight",
+    "model.layers.{}.self_attn.dense.bias": "transformer.h.{}.attn.proj.bias",
+    "model.layers.{}.mlp.fc1.weight": "transformer.h.{}.mlp.fc.weight",
+    "model.layers.{}.mlp.fc1.bias": "transformer.h.{}.mlp.fc.bias",
+    "model.layers.{}.mlp.fc2.weight": "transformer.h.{}.mlp.proj.weight",
+    "model.layers.{}.mlp.fc2.bias": "transformer.h.{}.mlp.proj.bias",
+    "model.final_layernorm.weight": "transformer.ln_f.weight",
+    "model.final_layernorm.bias": "transformer.ln_f.bias",
+    "lm_head.weight": "lm_head.weight",
+    "lm_head.bias": "lm_head.bias",
+}
+
 GPT2_TRANSPOSE = (".attn.c_proj.weight", ".mlp.c_fc.weight",
                   ".mlp.c_proj.weight", ".attn.c_attn.weight")
 
@@ -113,6 +129,8 @@ def _family(config: ModelConfig) -> str:
         return "llama"
     if n.startswith("gpt2") or config.pos_embedding == "learned":
         return "gpt2"
+    if "phi" in n:
+        return "phi"
     return "neox"
 
 
@@ -170,7 +188,8 @@ def convert_hf_checkpoint(
     out_dir = Path(out_dir) if out_dir else hf_dir
     config = ModelConfig.from_name(model_name or hf_dir.name)
     fam = _family(config)
-    tmap = {"llama": LLAMA_MAP, "neox": NEOX_MAP, "gpt2": GPT2_MAP}[fam]
+    tmap = {"llama": LLAMA_MAP, "neox": NEOX_MAP, "gpt2": GPT2_MAP,
+            "phi": PHI_MAP}[fam]
 
     sd: dict = {}
     pending_qkv: Dict[int, dict] = {}
@@ -180,12 +199,13 @@ def convert_hf_checkpoint(
             key = key.removeprefix("transformer.") if fam == "gpt2" else key
             if dtype is not None and t.is_floating_point():
                 t = t.to(dtype)
-            if fam == "llama" and ".self_attn." in key and (
+            if fam in ("llama", "phi") and ".self_attn." in key and (
                 "q_proj" in key or "k_proj" in key or "v_proj" in key
             ):
                 layer = int(key.split(".")[2])
                 which = key.split(".")[4][0]  # q/k/v
-                pending_qkv.setdefault(layer, {})[which] = t
+                suffix = "b" if key.endswith("bias") else ""
+                pending_qkv.setdefault(layer, {})[which + suffix] = t
                 continue
             if fam == "gpt2" and ".attn.c_attn." in key:
                 layer = int(key.split(".")[1])
@@ -212,7 +232,7 @@ def convert_hf_checkpoint(
             sd[f"transformer.h.{layer}.attn.attn.weight"] = weave_qkv(
                 d["q"], d["k"], d["v"], config
             )
-        if "qb" in d:
+        if "qb" in d and "kb" in d and "vb" in d:
             sd[f"transformer.h.{layer}.attn.attn.bias"] = weave_qkv(
                 d["qb"].unsqueeze(1), d["kb"].unsqueeze(1),
                 d["vb"].unsqueeze(1), config
