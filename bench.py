@@ -87,6 +87,13 @@ def main():
     B = max(args.group_size, 1)
     n_samples = args.samples or max(n_stages, 1) * B
     cfg = ModelConfig.from_name(args.model)
+    need = args.prompt_len + args.warmup + args.steps + 2
+    if need > min(args.seq_len, cfg.block_size):
+        raise SystemExit(
+            f"[bench] prompt+warmup+steps = {need} exceeds the sequence "
+            f"budget {min(args.seq_len, cfg.block_size)}; raise --seq-len "
+            "or lower --steps"
+        )
 
     # ---- build this rank's stage (random init, bf16, on device) ---------
     from mdi_llm_amd.utils.partition import balanced_split
