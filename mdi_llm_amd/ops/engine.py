@@ -107,6 +107,13 @@ class DecodeEngine:
     ) -> None:
         import os
 
+        # adaptive split-S: 16 chunks up to 4k contexts (measured best at
+        # short S), scale up for longer KV budgets so waves cover the chip
+        default_chunks = max(16, min(128, stage.max_seq_length // 64))
+        if stage.max_seq_length <= 4096:
+            default_chunks = 16
+        if n_chunks == 16:
+            n_chunks = default_chunks
         n_chunks = int(os.environ.get("MDI_ATTN_CHUNKS", n_chunks))
         self.weight_dtype = os.environ.get("MDI_WEIGHT_DTYPE", "bf16")
         self.ops = require_hip_ops()

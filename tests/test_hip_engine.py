@@ -243,3 +243,33 @@ def test_engine_multi_slot_graph_replay():
                 ref_logits[0, -1].float().argmax()), (step, s)
             toks[s] = ref_logits[0, -1].float().argmax()
             pos[s] += 1
+
+
+@torch.inference_mode()
+def test_engine_edge_cases():
+    """Length-1 prompt, highest slot index, decode from pos 1."""
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine
+
+    cfg, m = _build(seed=31)
+    n_slots = 5
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV, dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.eval()
+    stage.set_kv_cache(n_slots)
+    m.set_kv_cache(n_slots)
+
+    eng = DecodeEngine(stage, stage.kv_pool, n_chunks=8, use_graphs=False)
+    slot = n_slots - 1
+    prompt = torch.randint(0, 511, (1,), device=DEV)  # single-token prompt
+    ref = m(prompt.view(1, 1), input_pos=0, slot=slot)
+    stage.forward_head(prompt.view(1, 1), slot=slot, input_pos=0)
+    eng.set_slot_pos(slot, 1)
+    tok = ref[0, -1].float().argmax()
+    for i in range(3):
+        ref = m(tok.view(1, 1), input_pos=1 + i, slot=slot)
+        x = eng.decode_step_head(tok.to(torch.int32), slot=slot)
+        logits = eng.tail(x)
+        assert int(logits.float().argmax()) == int(
+            ref[0, -1].float().argmax()), i
+        tok = ref[0, -1].float().argmax()
