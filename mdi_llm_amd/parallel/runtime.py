@@ -84,6 +84,12 @@ class PipelineRuntime:
         assert self.is_starter, "generate() runs on the starter"
         n_samples = len(prompts)
         runner = self.runner
+        if (self.world == 1 and getattr(runner, "backend", "") == "hip"
+                and runner.engine.use_graphs and sampling.top_p >= 1.0):
+            return self._generate_standalone_fused(
+                prompts, max_new_tokens, sampling, stop_tokens,
+                token_callback,
+            )
         gens = self._generators(sampling, n_samples, self.device)
 
         res = GenerationResult()
@@ -157,6 +163,55 @@ class PipelineRuntime:
         res.gen_time = time.perf_counter() - t_start
         res.total_new_tokens = total_new
         res.sequences = [torch.tensor(s, dtype=torch.int64) for s in seqs]
+        return res
+
+    def _generate_standalone_fused(
+        self, prompts, max_new_tokens, sampling: SamplingParams,
+        stop_tokens, token_callback,
+    ) -> GenerationResult:
+        """Standalone generation on the fused hipGraph step (one graph
+        replay + one 4-byte readback per token).  Sampling runs on-GPU
+        (radix top-k + gumbel) with the run's seed."""
+        runner = self.runner
+        eng = runner.engine
+        n_samples = len(prompts)
+        # graphs must be captured before prefill (capture scribbles caches)
+        eng.ensure_fused_graphs(sampling.temperature, sampling.top_k,
+                                sampling.seed or 0)
+        runner.reset()
+
+        res = GenerationResult()
+        seqs = [list(map(int, p.tolist())) for p in prompts]
+        t_start = time.perf_counter()
+        for s, prompt in enumerate(prompts):
+            x = runner.prefill_head(prompt.to(self.device), s)
+            eng.slot.fill_(s)
+            eng.tail_sample_step(x.view(-1, x.size(-1))[-1], s)
+
+        new_counts = [0] * n_samples
+        active = set(range(n_samples))
+        total_new = 0
+        while active:
+            for s in list(active):
+                tok = int(eng.token_table[s])  # 4-byte D2H sync
+                seqs[s].append(tok)
+                new_counts[s] += 1
+                total_new += 1
+                res.tok_time.append(
+                    (total_new, time.perf_counter() - t_start))
+                if token_callback is not None:
+                    token_callback(s, tok)
+                if (new_counts[s] >= max_new_tokens
+                        or self._hit_stop(seqs[s], new_counts[s],
+                                          stop_tokens)
+                        or runner.pos[s] + 1 >= runner.stage.max_seq_length):
+                    active.discard(s)
+                    continue
+                eng.standalone_step(s)
+                runner.pos[s] += 1
+        res.gen_time = time.perf_counter() - t_start
+        res.total_new_tokens = total_new
+        res.sequences = [torch.tensor(q, dtype=torch.int64) for q in seqs]
         return res
 
     def _needs_clone(self) -> bool:
