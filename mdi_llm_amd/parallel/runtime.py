@@ -91,6 +91,19 @@ class PipelineRuntime:
                 token_callback,
             )
         gens = self._generators(sampling, n_samples, self.device)
+        # pipelined fused path: tail+sample+next-head in one graph replay;
+        # graphs must be captured BEFORE prefill (capture wipes caches)
+        fused = (
+            self.world > 1
+            and getattr(runner, "backend", "") == "hip"
+            and runner.engine.use_graphs
+            and sampling.top_p >= 1.0
+        )
+        if fused:
+            runner.engine.ensure_fused_graphs(
+                sampling.temperature, sampling.top_k, sampling.seed or 0
+            )
+            runner.reset()
 
         res = GenerationResult()
         seqs: List[List[int]] = [list(map(int, p.tolist())) for p in prompts]
@@ -123,15 +136,22 @@ class PipelineRuntime:
                 x = pending_x.pop(s)
             # tail: logits for the last position of this sample
             x2 = x.view(-1, x.size(-1))
-            logits = runner.tail(x2[-1])
-            tok = sample_token(
-                logits,
-                temperature=sampling.temperature,
-                top_k=sampling.top_k,
-                top_p=sampling.top_p,
-                generator=gens[s],
-            )
-            itok = int(tok)
+            if fused:
+                # one graph: tail + on-GPU sample + embed + blocks
+                runner.engine.starter_step(x2[-1], s)
+                runner.pos[s] += 1
+                itok = int(runner.engine.token_table[s])
+                self._fused_x = runner.engine.x
+            else:
+                logits = runner.tail(x2[-1])
+                tok = sample_token(
+                    logits,
+                    temperature=sampling.temperature,
+                    top_k=sampling.top_k,
+                    top_p=sampling.top_p,
+                    generator=gens[s],
+                )
+                itok = int(tok)
             seqs[s].append(itok)
             new_counts[s] += 1
             total_new += 1
@@ -150,6 +170,10 @@ class PipelineRuntime:
                     self.comm.send(s, None, stop=True)  # travels the ring
                 else:
                     n_active -= 1
+            elif fused:
+                # the graph already embedded the sampled token and ran the
+                # local blocks; just forward the activations
+                self.comm.send(s, self._fused_x, stop=False)
             else:
                 x = runner.decode_head(tok.view(1).to(self.device), s)
                 if self.world > 1:
