@@ -69,3 +69,37 @@ def test_balanced_split_llama3():
     cfg70 = ModelConfig.from_name("Meta-Llama-3-70B-Instruct")
     s = balanced_split(cfg70, 8)
     assert sum(s) == 80 and s[0] <= min(s[1:])
+
+
+def test_generation_utils():
+    import torch
+
+    from mdi_llm_amd.utils.generation import (
+        detect_stop_tokens,
+        find_eot,
+        get_obj_size,
+    )
+
+    assert detect_stop_tokens([1, 2, 3, 4], [(3, 4)])
+    assert not detect_stop_tokens([1, 2, 3, 4], [(2, 3)])
+    toks = torch.tensor([5, 6, 7, 99, 100, 8])
+    assert find_eot(toks, [(99, 100)], prompt_len=1) == 3
+    assert find_eot(toks, [(42,)]) == 6
+    sd = {"a": torch.zeros(10, 10), "b": torch.zeros(4, dtype=torch.float64)}
+    assert get_obj_size(sd) == 400 + 32
+
+
+def test_console_utils(capsys):
+    import threading
+    import time
+
+    from mdi_llm_amd.utils.console import loading_bar, waiting_animation
+
+    bar = loading_bar(5, 10, width=10)
+    assert "50.0%" in bar
+    ev = threading.Event()
+    t = waiting_animation(ev, "busy", interval=0.01)
+    time.sleep(0.05)
+    ev.set()
+    t.join(timeout=1)
+    assert not t.is_alive()
