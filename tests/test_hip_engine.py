@@ -273,3 +273,42 @@ def test_engine_edge_cases():
         assert int(logits.float().argmax()) == int(
             ref[0, -1].float().argmax()), i
         tok = ref[0, -1].float().argmax()
+
+
+@torch.inference_mode()
+def test_fused_generate_reproducible_and_stops():
+    """The fused standalone generate(): same seed -> identical sequences;
+    stop tokens truncate; engine path actually used."""
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.parallel.runner import make_runner
+    from mdi_llm_amd.parallel.runtime import PipelineRuntime, SamplingParams
+
+    cfg, m = _build(seed=41)
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV, dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.eval()
+    runner = make_runner(stage, 2, torch.device(DEV))
+    assert runner.backend == "hip"
+    rt = PipelineRuntime(runner, device=torch.device(DEV))
+
+    torch.manual_seed(42)
+    prompts = [torch.randint(0, 511, (6,), device=DEV),
+               torch.randint(0, 511, (9,), device=DEV)]
+    sp = SamplingParams(temperature=0.8, top_k=50, seed=77)
+    r1 = rt.generate(prompts, 12, sp)
+    r2 = rt.generate(prompts, 12, sp)
+    # deterministic RNG is (seed, ctr)-based; ctr continues across calls,
+    # so reseed determinism is checked via a fresh runner
+    runner2 = make_runner(stage, 2, torch.device(DEV))
+    rt2 = PipelineRuntime(runner2, device=torch.device(DEV))
+    r3 = rt2.generate(prompts, 12, sp)
+    assert [s.tolist() for s in r1.sequences] == \
+        [s.tolist() for s in r3.sequences]
+    assert all(s.numel() == p.numel() + 12
+               for s, p in zip(r1.sequences, prompts))
+
+    # stop token: use the first generated token of sample 0
+    stop = (int(r1.sequences[0][6]),)
+    r4 = rt2.generate(prompts, 12, sp, stop_tokens=[stop])
+    assert r4.sequences[0].numel() == 7  # truncated after 1 token
+    assert r4.sequences[1].tolist() == r1.sequences[1].tolist()
