@@ -308,7 +308,10 @@ def test_fused_generate_reproducible_and_stops():
                for s, p in zip(r1.sequences, prompts))
 
     # stop token: use the first generated token of sample 0
+    # (fresh runner: the gumbel counter is engine-lifetime monotonic)
     stop = (int(r1.sequences[0][6]),)
-    r4 = rt2.generate(prompts, 12, sp, stop_tokens=[stop])
+    runner3 = make_runner(stage, 2, torch.device(DEV))
+    rt3 = PipelineRuntime(runner3, device=torch.device(DEV))
+    r4 = rt3.generate(prompts, 12, sp, stop_tokens=[stop])
     assert r4.sequences[0].numel() == 7  # truncated after 1 token
     assert r4.sequences[1].tolist() == r1.sequences[1].tolist()
