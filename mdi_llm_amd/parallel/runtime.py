@@ -195,7 +195,13 @@ class PipelineRuntime:
     ) -> GenerationResult:
         """Standalone generation on the fused hipGraph step (one graph
         replay + one 4-byte readback per token).  Sampling runs on-GPU
-        (radix top-k + gumbel) with the run's seed."""
+        (radix top-k + gumbel) with the run's seed.
+
+        Determinism note: draws come from a (seed, engine-lifetime draw
+        counter) hash, so token streams are reproducible for a fresh
+        engine + identical schedule, but one sample stopping early shifts
+        the other samples' later draws (the torch path keeps fully
+        independent per-sample generator streams instead)."""
         runner = self.runner
         eng = runner.engine
         n_samples = len(prompts)

