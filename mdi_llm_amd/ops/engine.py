@@ -181,7 +181,6 @@ class DecodeEngine:
             self.sample_scratch = torch.zeros(520, device=dev,
                                               dtype=torch.int32)
             self.sample_out = torch.zeros(1, device=dev, dtype=torch.int32)
-            self.sample_ctr = torch.zeros(1, device=dev, dtype=torch.int32)
 
         # device-side slot/pos scalars (graph-replayable)
         self.slot = torch.zeros(1, device=dev, dtype=torch.int32)
@@ -245,12 +244,12 @@ class DecodeEngine:
     def sample_into_token(self, temperature: float, top_k, seed: int,
                           top_p: float = 1.0) -> torch.Tensor:
         """Fused on-GPU sampling from self.logits into self.sample_out.
-        (scratch is self-cleaning: the unpack kernel zeroes it.)"""
-        self.sample_ctr += 1
+        (scratch is self-cleaning; the gumbel stream is keyed by
+        (seed, slot, pos) — reproducible and schedule-independent.)"""
         self.ops.sample(
             self.sample_out, self.logits, self.sample_scratch,
             float(temperature), int(top_k or 0), temperature > 0.0,
-            int(seed) & 0x7FFFFFFF, self.sample_ctr, 0, float(top_p),
+            int(seed) & 0x7FFFFFFF, self.pos, self.slot, 0, float(top_p),
         )
         return self.sample_out
 
@@ -456,11 +455,10 @@ class DecodeEngine:
     # ------------------------------------------------------------------
     def _sample_seq(self, temperature: float, top_k: int, seed: int,
                     top_p: float = 1.0) -> None:
-        self.sample_ctr += 1
         self.ops.sample(self.sample_out, self.logits, self.sample_scratch,
                         float(temperature), int(top_k or 0),
                         temperature > 0.0, int(seed) & 0x7FFFFFFF,
-                        self.sample_ctr, 0, float(top_p))
+                        self.pos, self.slot, 0, float(top_p))
 
     def _advance(self) -> None:
         # token_table[slot] = sample_out ; pos_table[slot] += 1 (in-graph)
@@ -500,7 +498,6 @@ class DecodeEngine:
         self.kv_pool.v.zero_()
         self.pos_table.zero_()
         self.token_table.zero_()
-        self.sample_ctr.zero_()
 
         g1 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g1):

@@ -106,7 +106,6 @@ class GroupDecodeEngine:
             self.LOGITS = torch.zeros(B, cfg.padded_vocab_size, **bf)
             self.sample_scratch = torch.zeros(520 * B, device=dev,
                                               dtype=torch.int32)
-            self.sample_ctr = torch.zeros(1, device=dev, dtype=torch.int32)
             self.tokens = torch.zeros(B, device=dev, dtype=torch.int32)
             self.token_table = torch.zeros(kv_pool.n_slots, device=dev,
                                            dtype=torch.int32)
@@ -206,11 +205,11 @@ class GroupDecodeEngine:
             torch.matmul(xn, self.head_w.t(), out=self.LOGITS)
 
     def _sample_seq(self, temperature, top_k, seed, top_p=1.0) -> None:
-        self.sample_ctr += 1
         self.ops.sample(
             self.tokens, self.LOGITS, self.sample_scratch,
             float(temperature), int(top_k or 0), temperature > 0.0,
-            int(seed) & 0x7FFFFFFF, self.sample_ctr, self.B, float(top_p),
+            int(seed) & 0x7FFFFFFF, self.pos, self.slots, self.B,
+            float(top_p),
         )
 
     def _embed_seq(self) -> None:
@@ -254,7 +253,6 @@ class GroupDecodeEngine:
         self.pos_table.zero_()
         if self.is_starter:
             self.token_table.zero_()
-            self.sample_ctr.zero_()
 
         if self.is_starter:
             g1 = torch.cuda.CUDAGraph()

@@ -280,8 +280,9 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
 
 void sample(torch::Tensor out_token, torch::Tensor logits,
             torch::Tensor scratch, double temperature, int64_t top_k,
-            bool noise, int64_t seed, c10::optional<torch::Tensor> ctr,
-            int64_t n_batch, double top_p) {
+            bool noise, int64_t seed, c10::optional<torch::Tensor> pos,
+            c10::optional<torch::Tensor> slot, int64_t n_batch,
+            double top_p) {
   check_i32(out_token, "out_token");
   check_bf16(logits, "logits");
   const int nb = n_batch > 0 ? (int)n_batch : 1;
@@ -289,15 +290,22 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
                   scratch.numel() >= 520 * nb,
               "scratch must be >=520 int32 per sample on GPU");
   TORCH_CHECK(out_token.numel() >= nb, "out_token too small");
-  const int* cp = nullptr;
-  if (ctr.has_value()) {
-    check_i32(*ctr, "ctr");
-    cp = ctr->data_ptr<int>();
+  const int* pp = nullptr;
+  const int* sp = nullptr;
+  if (pos.has_value()) {
+    check_i32(*pos, "pos");
+    TORCH_CHECK(pos->numel() >= nb, "pos too small");
+    pp = pos->data_ptr<int>();
+  }
+  if (slot.has_value()) {
+    check_i32(*slot, "slot");
+    TORCH_CHECK(slot->numel() >= nb, "slot too small");
+    sp = slot->data_ptr<int>();
   }
   const int V = (int)(logits.numel() / nb);
   launch_sample(out_token.data_ptr(), logits.data_ptr(), V,
                 scratch.data_ptr(), (float)temperature, (int)top_k,
-                (float)top_p, noise ? 1 : 0, (unsigned)(int64_t)seed, cp,
+                (float)top_p, noise ? 1 : 0, (unsigned)(int64_t)seed, pp, sp,
                 (int)n_batch, cur_stream());
 }
 
@@ -405,6 +413,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused temperature/top-k/gumbel token sampling (128k vocab ~15us)",
         py::arg("out_token"), py::arg("logits"), py::arg("scratch"),
         py::arg("temperature"), py::arg("top_k"), py::arg("noise"),
-        py::arg("seed"), py::arg("ctr") = c10::nullopt,
-        py::arg("n_batch") = 0, py::arg("top_p") = 1.0);
+        py::arg("seed"), py::arg("pos") = c10::nullopt,
+        py::arg("slot") = c10::nullopt, py::arg("n_batch") = 0,
+        py::arg("top_p") = 1.0);
 }

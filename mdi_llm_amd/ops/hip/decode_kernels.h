@@ -65,12 +65,14 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
 void launch_add(void* out, const void* a, const void* b, int n,
                 hipStream_t stream);
 
-// fused temperature/top-k/gumbel sampling; scratch: >=520 u32 PER SAMPLE,
-// zeroed initially (self-cleaning); n_batch draws from [n_batch, V] logits
+// fused temperature/top-k/top-p/gumbel sampling; scratch: >=520 u32 PER
+// SAMPLE, zeroed initially (self-cleaning); n_batch draws from
+// [n_batch, V] logits.  The gumbel stream is keyed by (seed, slot, pos) —
+// reproducible and schedule-independent.
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, float top_p, int noise_on,
-                   unsigned seed, const int* ctr, int n_batch,
-                   hipStream_t stream);
+                   unsigned seed, const int* pos, const int* slot,
+                   int n_batch, hipStream_t stream);
 
 // prefill: rope+append all T positions (grid covers T)
 void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,

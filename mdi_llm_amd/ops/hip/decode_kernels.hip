@@ -1321,12 +1321,16 @@ __global__ void sample_select_lo_kernel(unsigned* __restrict__ scratch,
 __global__ void sample_gumbel_argmax_kernel(
     const bf16* __restrict__ logits, int V, unsigned* __restrict__ scratch,
     float inv_temp, int use_threshold, int noise, unsigned seed,
-    const int* __restrict__ ctr) {
+    const int* __restrict__ pos_p, const int* __restrict__ slot_p) {
   logits += (size_t)blockIdx.y * V;
   scratch += (size_t)blockIdx.y * 520;
   const unsigned t = use_threshold ? scratch[514] : 0u;
-  const unsigned salt = (seed + blockIdx.y * 0x85EBCA6Bu) ^
-                        (unsigned)(ctr ? ctr[0] : 0) * 0x9E3779B9u;
+  // counter-based RNG keyed by (seed, slot, position): reproducible and
+  // independent of the scheduling order across samples
+  const unsigned pos_v = pos_p ? (unsigned)pos_p[blockIdx.y] : 0u;
+  const unsigned slot_v = slot_p ? (unsigned)slot_p[blockIdx.y] : blockIdx.y;
+  const unsigned salt = seed ^ (pos_v * 0x9E3779B9u) ^
+                        (slot_v * 0x85EBCA6Bu);
   float best = -1e38f;
   int best_i = 0;
   for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < V;
@@ -1384,8 +1388,8 @@ __global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
 
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, float top_p, int noise_on,
-                   unsigned seed, const int* ctr, int n_batch,
-                   hipStream_t stream) {
+                   unsigned seed, const int* pos, const int* slot,
+                   int n_batch, hipStream_t stream) {
   unsigned* sc = (unsigned*)scratch;
   const int B = n_batch > 0 ? n_batch : 1;
   const int blocks = B > 1 ? 32 : 128;
@@ -1419,7 +1423,7 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
   }
   hipLaunchKernelGGL(sample_gumbel_argmax_kernel, dim3(blocks, B), dim3(256),
                      0, stream, (const bf16*)logits, V, sc, inv_t,
-                     use_k || use_p, noise_on, seed, ctr);
+                     use_k || use_p, noise_on, seed, pos, slot);
   hipLaunchKernelGGL(sample_unpack_kernel, dim3(1, B), dim3(64), 0, stream,
                      sc, (int*)out_token);
 }

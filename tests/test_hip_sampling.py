@@ -16,10 +16,13 @@ def ops():
 
 
 def _call(ops, logits, temperature, top_k, noise=True, seed=7, ctr_val=0):
+    # ctr_val is realised as the (pos, slot) RNG key of the draw
     scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
     out = torch.zeros(1, device=DEV, dtype=torch.int32)
-    ctr = torch.tensor([ctr_val], device=DEV, dtype=torch.int32)
-    ops.sample(out, logits, scratch, temperature, top_k, noise, seed, ctr)
+    pos = torch.tensor([ctr_val], device=DEV, dtype=torch.int32)
+    slot = torch.zeros(1, device=DEV, dtype=torch.int32)
+    ops.sample(out, logits, scratch, temperature, top_k, noise, seed,
+               pos=pos, slot=slot)
     return int(out)
 
 
@@ -43,14 +46,14 @@ def test_topk_support(ops):
         assert tok in topset, (ctr, tok)
 
 
-def test_deterministic_given_seed_ctr(ops):
+def test_deterministic_given_seed_pos(ops):
     torch.manual_seed(2)
     logits = torch.randn(5000, device=DEV).to(torch.bfloat16)
     a = _call(ops, logits, 1.0, 50, seed=11, ctr_val=3)
     b = _call(ops, logits, 1.0, 50, seed=11, ctr_val=3)
     c = _call(ops, logits, 1.0, 50, seed=11, ctr_val=4)
     assert a == b
-    # different counter should (almost surely) give a different draw
+    # different position should (almost surely) give a different draw
     # occasionally equal is fine; just check the call runs
     assert isinstance(c, int)
 
@@ -73,9 +76,9 @@ def test_distribution_roughly_matches(ops):
 def _call_p(ops, logits, temperature, top_k, top_p, seed=7, ctr_val=0):
     scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
     out = torch.zeros(1, device=DEV, dtype=torch.int32)
-    ctr = torch.tensor([ctr_val], device=DEV, dtype=torch.int32)
-    ops.sample(out, logits, scratch, temperature, top_k, True, seed, ctr,
-               0, top_p)
+    pos = torch.tensor([ctr_val], device=DEV, dtype=torch.int32)
+    ops.sample(out, logits, scratch, temperature, top_k, True, seed,
+               pos=pos, top_p=top_p)
     return int(out)
 
 

@@ -110,11 +110,11 @@ def bench_sampling():
     print(f"argmax: {us:.1f} us")
     scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
     out = torch.zeros(1, device=DEV, dtype=torch.int32)
-    ctr = torch.zeros(1, device=DEV, dtype=torch.int32)
+    pos = torch.zeros(1, device=DEV, dtype=torch.int32)
 
     def fused():
         scratch.zero_()
-        ops.sample(out, logits, scratch, 0.8, 200, True, 7, ctr)
+        ops.sample(out, logits, scratch, 0.8, 200, True, 7, pos=pos)
 
     us = timeit(fused, iters=50)
     print(f"fused HIP sampling (topk200): {us:.1f} us")
