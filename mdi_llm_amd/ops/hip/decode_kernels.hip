@@ -1115,7 +1115,7 @@ __global__ void add_kernel(bf16* __restrict__ out, const bf16* __restrict__ a,
 // Gumbel-max draw.  Replaces the reference's torch sample()
 // (/root/reference/src/sub/model.py:34-90) on the decode hot path — the
 // torch composition costs ~166 us/token on a 128k vocab; this pipeline of
-// 5 small kernels costs ~15 us.  Deterministic given (seed, ctr).
+// 5 small kernels costs ~15 us.  Deterministic given (seed, slot, pos).
 //
 // scratch layout (int32/u32, >= 520 entries, zeroed before each call):
 //   [0..255]   hi-byte histogram

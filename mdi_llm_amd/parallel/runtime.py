@@ -197,11 +197,11 @@ class PipelineRuntime:
         replay + one 4-byte readback per token).  Sampling runs on-GPU
         (radix top-k + gumbel) with the run's seed.
 
-        Determinism note: draws come from a (seed, engine-lifetime draw
-        counter) hash, so token streams are reproducible for a fresh
-        engine + identical schedule, but one sample stopping early shifts
-        the other samples' later draws (the torch path keeps fully
-        independent per-sample generator streams instead)."""
+        Determinism note: draws come from a (seed, slot, position) hash,
+        so each sample's token stream is reproducible and independent of
+        the scheduling order across samples — one sample stopping early
+        does not shift any other sample's draws (matching the torch
+        path's per-sample generator streams in spirit)."""
         runner = self.runner
         eng = runner.engine
         n_samples = len(prompts)
