@@ -242,6 +242,23 @@ configs.extend(
             mlp_class_name="GptNeoxMLP",
         ),
         dict(
+            # falcon-7b-style tiny config: MQA + shared attention norm
+            name="nano-test-falcon",
+            block_size=128,
+            vocab_size=256,
+            padding_multiple=64,
+            n_layer=2,
+            n_head=4,
+            n_embd=64,
+            n_query_groups=1,
+            rotary_percentage=1.0,
+            parallel_residual=True,
+            shared_attention_norm=True,
+            bias=False,
+            norm_class_name="LayerNorm",
+            mlp_class_name="GptNeoxMLP",
+        ),
+        dict(
             name="nano-test-moe",
             block_size=128,
             vocab_size=256,
@@ -537,6 +554,61 @@ for size, n_layer, n_head, n_embd, interm, groups, hs in (
             intermediate_size=interm,
             scale_embeddings=True,
             norm_eps=1e-6,
+        )
+    )
+
+# -- Falcon (reference: config.py falcon block, lines 476-534) -----------
+# 7b is MQA (one kv group) with a single shared attention/MLP norm;
+# 40b/180B use 8 kv groups and separate ln_attn/ln_mlp norms.
+for base in (
+    dict(
+        name="falcon-7b{}",
+        hf_config=dict(org="tiiuae", name="falcon-7b{}"),
+        block_size=2048,
+        vocab_size=65024,
+        padded_vocab_size=65024,
+        n_layer=32,
+        n_head=71,
+        n_embd=4544,
+        rotary_percentage=1.0,
+        n_query_groups=1,
+        bias=False,
+        shared_attention_norm=True,
+    ),
+    dict(
+        name="falcon-40b{}",
+        hf_config=dict(org="tiiuae", name="falcon-40b{}"),
+        block_size=2048,
+        vocab_size=65024,
+        padded_vocab_size=65024,
+        n_layer=60,
+        n_head=128,
+        n_embd=8192,
+        rotary_percentage=1.0,
+        n_query_groups=8,
+        bias=False,
+    ),
+):
+    for kind in ("", "-instruct"):
+        c = dict(base)
+        c["name"] = base["name"].format(kind)
+        c["hf_config"] = dict(org="tiiuae",
+                              name=base["hf_config"]["name"].format(kind))
+        configs.append(c)
+for kind in ("", "-chat"):
+    configs.append(
+        dict(
+            name=f"falcon-180B{kind}",
+            hf_config=dict(org="tiiuae", name=f"falcon-180B{kind}"),
+            block_size=2048,
+            vocab_size=65024,
+            padded_vocab_size=65024,
+            n_layer=80,
+            n_head=232,
+            n_embd=14848,
+            rotary_percentage=1.0,
+            n_query_groups=8,
+            bias=False,
         )
     )
 

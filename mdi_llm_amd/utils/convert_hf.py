@@ -119,12 +119,41 @@ PHI_MAP = {
     "lm_head.bias": "lm_head.bias",
 }
 
+# Falcon's HF query_key_value layout is already grouped-interleaved
+# ([q*qpk | k | v] per kv group), identical to the lit attn.attn layout,
+# so it maps directly with no weave (reference
+# convert_hf_checkpoint.py:61-119).  The map carries both norm namings:
+# 7b has a single shared input_layernorm, 40b/180B split ln_attn/ln_mlp.
+FALCON_MAP = {
+    "transformer.word_embeddings.weight": "transformer.wte.weight",
+    "transformer.h.{}.self_attention.query_key_value.weight":
+        "transformer.h.{}.attn.attn.weight",
+    "transformer.h.{}.self_attention.dense.weight":
+        "transformer.h.{}.attn.proj.weight",
+    "transformer.h.{}.mlp.dense_h_to_4h.weight":
+        "transformer.h.{}.mlp.fc.weight",
+    "transformer.h.{}.mlp.dense_4h_to_h.weight":
+        "transformer.h.{}.mlp.proj.weight",
+    "transformer.h.{}.input_layernorm.weight":
+        "transformer.h.{}.norm_1.weight",
+    "transformer.h.{}.input_layernorm.bias": "transformer.h.{}.norm_1.bias",
+    "transformer.h.{}.ln_attn.weight": "transformer.h.{}.norm_1.weight",
+    "transformer.h.{}.ln_attn.bias": "transformer.h.{}.norm_1.bias",
+    "transformer.h.{}.ln_mlp.weight": "transformer.h.{}.norm_2.weight",
+    "transformer.h.{}.ln_mlp.bias": "transformer.h.{}.norm_2.bias",
+    "transformer.ln_f.weight": "transformer.ln_f.weight",
+    "transformer.ln_f.bias": "transformer.ln_f.bias",
+    "lm_head.weight": "lm_head.weight",
+}
+
 GPT2_TRANSPOSE = (".attn.c_proj.weight", ".mlp.c_fc.weight",
                   ".mlp.c_proj.weight", ".attn.c_attn.weight")
 
 
 def _family(config: ModelConfig) -> str:
     n = config.name.lower()
+    if "falcon" in n:
+        return "falcon"
     if config.mlp_class_name in ("LLaMAMLP", "GemmaMLP", "LLaMAMoE"):
         return "llama"
     if n.startswith("gpt2") or config.pos_embedding == "learned":
@@ -189,7 +218,7 @@ def convert_hf_checkpoint(
     config = ModelConfig.from_name(model_name or hf_dir.name)
     fam = _family(config)
     tmap = {"llama": LLAMA_MAP, "neox": NEOX_MAP, "gpt2": GPT2_MAP,
-            "phi": PHI_MAP}[fam]
+            "phi": PHI_MAP, "falcon": FALCON_MAP}[fam]
 
     sd: dict = {}
     pending_qkv: Dict[int, dict] = {}
@@ -255,12 +284,35 @@ def convert_lit_checkpoint(ckpt_dir: PathLike, out_path: PathLike,
         (ModelConfig.from_name(model_name),
          load_from_pt(ckpt_dir, ModelConfig.from_name(model_name))[1])
     fam = _family(config)
-    if fam != "llama":
-        raise NotImplementedError("lit->HF export currently supports llama")
+    if fam == "falcon":
+        # qkv copies straight through (layouts identical); norm naming
+        # depends on the variant: shared norm -> input_layernorm (7b),
+        # split norms -> ln_attn/ln_mlp (40b/180B)
+        rev = {v: k for k, v in FALCON_MAP.items()
+               if "layernorm" not in k and "ln_attn" not in k
+               and "ln_mlp" not in k}
+        if config.shared_attention_norm:
+            rev["transformer.h.{}.norm_1.weight"] = \
+                "transformer.h.{}.input_layernorm.weight"
+            rev["transformer.h.{}.norm_1.bias"] = \
+                "transformer.h.{}.input_layernorm.bias"
+        else:
+            rev["transformer.h.{}.norm_1.weight"] = \
+                "transformer.h.{}.ln_attn.weight"
+            rev["transformer.h.{}.norm_1.bias"] = \
+                "transformer.h.{}.ln_attn.bias"
+            rev["transformer.h.{}.norm_2.weight"] = \
+                "transformer.h.{}.ln_mlp.weight"
+            rev["transformer.h.{}.norm_2.bias"] = \
+                "transformer.h.{}.ln_mlp.bias"
+    elif fam == "llama":
+        rev = {v: k for k, v in LLAMA_MAP.items()}
+    else:
+        raise NotImplementedError(
+            "lit->HF export currently supports llama and falcon")
     inv = {}
-    rev = {v: k for k, v in LLAMA_MAP.items()}
     for key, t in sd.items():
-        if key.endswith(".attn.attn.weight"):
+        if fam == "llama" and key.endswith(".attn.attn.weight"):
             layer = key.split(".")[2]
             q, k, v = unweave_qkv(t, config)
             inv[f"model.layers.{layer}.self_attn.q_proj.weight"] = q
