@@ -242,6 +242,24 @@ configs.extend(
             mlp_class_name="GptNeoxMLP",
         ),
         dict(
+            # 8-layer variant: splits across up to 8 pipeline stages in
+            # CPU tests of the multi-GPU bench path
+            name="nano-test-deep",
+            block_size=128,
+            vocab_size=256,
+            padding_multiple=64,
+            n_layer=8,
+            n_head=4,
+            n_embd=64,
+            n_query_groups=4,
+            rotary_percentage=1.0,
+            parallel_residual=False,
+            bias=False,
+            norm_class_name="RMSNorm",
+            mlp_class_name="LLaMAMLP",
+            intermediate_size=128,
+        ),
+        dict(
             # falcon-7b-style tiny config: MQA + shared attention norm
             name="nano-test-falcon",
             block_size=128,

@@ -77,3 +77,18 @@ def test_bench_cpu_mode(tmp_path):
              "--prompt-len", "8", "--seq-len", "64"])
     assert r.returncode == 0, r.stderr[-2000:]
     assert '"n_gpus": 2' in r.stdout
+
+
+@pytest.mark.parametrize("world", [4, 8])
+def test_bench_cpu_mode_wide(world):
+    """The exact flow the driver runs on the 8-GPU node (torchrun, one
+    rank per device, balanced split) on an 8-layer model, CPU/gloo."""
+    r = run(["-m", "torch.distributed.run", "--nnodes=1",
+             "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+             "--master-port", str(29680 + world), "bench.py",
+             "--gpus", str(world), "--steps", "3", "--warmup", "1",
+             "--model", "nano-test-deep", "--prompt-len", "8",
+             "--seq-len", "64"], timeout=420)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert f'"n_gpus": {world}' in r.stdout
+    assert '"value"' in r.stdout
