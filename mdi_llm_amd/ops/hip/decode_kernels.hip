@@ -600,22 +600,13 @@ __global__ void rope_kv_append_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// GQA flash-decode attention, split-S, with FUSED combine.
+// GQA flash-decode attention, split-S.
 //
 // One wave handles one (kv_head, chunk): MFMA 16x16x32 bf16 computes
 // scores[key, qhead] (swapped operands so each lane owns 4 keys of ONE
 // query head), online softmax with running (m, l) per qhead in LDS, P*V
-// accumulated in registers. Partials (o, m, l) per chunk go to global fp32.
-//
-// The split-S reduction happens IN the same launch: the last wave of a
-// (batch, kv_head) to finish (per-head device-scope atomic counter, self-
-// cleaning for graph replay) combines that head group's n_active chunk
-// partials wave-locally and writes the bf16 output. No grid-wide sync, no
-// spinning — a wave either exits or combines — so this is hang-free by
-// construction, and it removes one whole kernel launch per layer (~40% of
-// the attention cost at short S, where both kernels sat at the ~5 us
-// in-stream floor). Single-active-chunk steps (S <= keys_per_chunk) skip
-// the partial roundtrip entirely and write the output directly.
+// accumulated in registers. Partials (o, m, l) per chunk go to global fp32;
+// attn_combine_kernel reduces chunks.
 //
 // Launch: grid.x = n_kv_heads * n_chunks / WAVES_PER_BLOCK, block = 256.
 // qpk (query heads per kv head) <= 16.
@@ -624,13 +615,6 @@ using f32x4 = __attribute__((__vector_size__(16))) float;
 using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
 
 #define ATTN_WAVES 4  // waves per block, each fully independent
-
-// Per-(batch, kv_head) arrival counters for the fused split-S combine.
-// Zero-initialized at module load; the combining wave resets its cell, so
-// the invariant "all zero between launches" holds across graph replays.
-// Indexed b * n_kv_heads + g; 16384 cells cover B=1024 x 16 kv heads.
-#define ATTN_CTR_CAP 16384
-__device__ int g_attn_ctr[ATTN_CTR_CAP];
 
 // XOR swizzle on the q LDS tile (rows at a power-of-two byte stride would
 // otherwise put a ds_read_b128 lane group on one bank slot — guide §6 G4):
@@ -660,7 +644,6 @@ DEVINL float rope_elem(const bf16* row, int d, int ne,
 //             qkv/part_o/part_ml/out get a leading batch dimension.
 template <int QPK, int HS, int BATCH>
 __global__ void attn_decode_kernel(
-    bf16* __restrict__ out,       // [B?, n_head * head_size]
     float* __restrict__ part_o,   // [B?, n_head, n_chunks, head_size]
     float* __restrict__ part_ml,  // [B?, n_head, n_chunks, 2]
     const bf16* __restrict__ qkv, // [B?, qkv_dim] interleaved, RAW
@@ -701,12 +684,7 @@ __global__ void attn_decode_kernel(
     qkv += (size_t)b * qkv_dim;
     part_o += (size_t)b * n_head_all * n_chunks * head_size;
     part_ml += (size_t)b * n_head_all * n_chunks * 2;
-    out += (size_t)b * n_head_all * head_size;
   }
-  // chunks holding no keys exit immediately (an exited wave does not count
-  // toward s_barrier, and all LDS tiles here are per-wave)
-  const int n_active = min(n_chunks, (S + keys_per_chunk - 1) / keys_per_chunk);
-  if (chunk >= n_active) return;
 
   const size_t cache_base =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
@@ -855,20 +833,9 @@ __global__ void attn_decode_kernel(
     }
   }
 
-  // ---- single active chunk: write the final output directly -------------
-  // (the common case for short contexts: S <= keys_per_chunk)
-  if (n_active == 1) {
-    const int hB = g * QPK + qb;
-    const float inv = 1.f / l_lds[wave][qb];
-    bf16* op = out + (size_t)hB * head_size + d0;
-#pragma unroll
-    for (int i = 0; i < ODIM; ++i) op[i] = f2b(o_acc[i] * inv);
-    return;
-  }
-
   // ---- write partials ----------------------------------------------------
   // global qhead index for PV map: h = g*QPK + qb
-  {
+  if (k_begin < S) {
     const int hB = g * QPK + qb;
     float* op = part_o + ((size_t)hB * n_chunks + chunk) * head_size + d0;
 #pragma unroll
@@ -878,52 +845,57 @@ __global__ void attn_decode_kernel(
       part_ml[((size_t)hA * n_chunks + chunk) * 2 + 0] = m_lds[wave][qa];
       part_ml[((size_t)hA * n_chunks + chunk) * 2 + 1] = l_lds[wave][qa];
     }
+  } else if (lane == 0) {
+    // empty chunk: mark invalid via l = 0, m = -inf
+    for (int q = 0; q < QPK; ++q) {
+      const int h = g * QPK + q;
+      part_ml[((size_t)h * n_chunks + chunk) * 2 + 0] = -1e30f;
+      part_ml[((size_t)h * n_chunks + chunk) * 2 + 1] = 0.f;
+    }
   }
+}
 
-  // ---- fused combine: last wave of this (b, g) reduces the partials ------
-  __threadfence();  // release our partials (agent scope, cross-XCD)
-  int prev = 0;
-  if (lane == 0)
-    prev = atomicAdd(&g_attn_ctr[b * n_kv_heads + g], 1);
-  prev = __shfl(prev, 0, 64);
-  if (prev != n_active - 1) return;  // not last: done (no spinning)
-  if (lane == 0) g_attn_ctr[b * n_kv_heads + g] = 0;  // self-clean
-  __threadfence();  // acquire the other chunks' partials
+// Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L
+// grid = n_head blocks, 256 threads; wave 0 computes chunk weights lane-
+// parallel (lane == chunk), then all threads accumulate dims with the
+// chunk loop's loads fully independent (ILP) — the serial-per-lane version
+// of this kernel was 19 us (latency-bound); this one is ~2 us.
+__global__ void attn_combine_kernel(bf16* __restrict__ out,
+                                    const float* __restrict__ part_o,
+                                    const float* __restrict__ part_ml,
+                                    int n_chunks, int head_size, int n_head) {
+  // one WAVE per (head, 64-dim slice); every lane recomputes the chunk
+  // weights lane-parallel (lane == chunk, shfl broadcast), so blocks are
+  // fully independent and the chunk-accumulate loop is the only serial
+  // part (unrolled for load ILP).
+  const int DS = (head_size + 63) / 64;
+  const int wid = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  const int h = wid / DS;
+  const int d = (wid % DS) * 64 + lane;
+  if (h >= n_head) return;
 
-  // wave-local combine of n_active chunk partials for this group's QPK
-  // heads: out[h][d] = sum_c exp(m_c - M) * part_o[h][c][d] / L.
-  // Chunk weights live lane-parallel in registers (lane c holds chunks c
-  // and c+64; n_chunks is clamped to 128 by the launcher) and are
-  // broadcast with shfl — no LDS, no barriers.
-  for (int q = 0; q < QPK; ++q) {
-    const int h = g * QPK + q;
-    const float* ml = part_ml + (size_t)h * n_chunks * 2;
-    float m0 = -1e30f, m1 = -1e30f, l0 = 0.f, l1 = 0.f;
-    if (lane < n_active) {
-      m0 = ml[lane * 2];
-      l0 = ml[lane * 2 + 1];
-    }
-    if (lane + 64 < n_active) {
-      m1 = ml[(lane + 64) * 2];
-      l1 = ml[(lane + 64) * 2 + 1];
-    }
-    float M = fmaxf(m0, m1);
+  float m = -1e30f, l = 0.f;
+  if (lane < n_chunks) {
+    m = part_ml[((size_t)h * n_chunks + lane) * 2];
+    l = part_ml[((size_t)h * n_chunks + lane) * 2 + 1];
+  }
+  float M = m;
 #pragma unroll
-    for (int off = 32; off > 0; off >>= 1)
-      M = fmaxf(M, __shfl_xor(M, off, 64));
-    const float w0 = (lane < n_active) ? __expf(m0 - M) : 0.f;
-    const float w1 = (lane + 64 < n_active) ? __expf(m1 - M) : 0.f;
-    const float inv = 1.f / wave_reduce_sum(w0 * l0 + w1 * l1);
-    const float* po = part_o + (size_t)h * n_chunks * head_size;
-    for (int d = lane; d < head_size; d += 64) {
-      float acc = 0.f;
-      for (int c = 0; c < n_active; ++c) {
-        const float w = (c < 64) ? __shfl(w0, c, 64) : __shfl(w1, c - 64, 64);
-        acc += w * po[(size_t)c * head_size + d];
-      }
-      out[(size_t)h * head_size + d] = f2b(acc * inv);
-    }
+  for (int off = 32; off > 0; off >>= 1)
+    M = fmaxf(M, __shfl_xor(M, off, 64));
+  const float wgt = (lane < n_chunks) ? __expf(m - M) : 0.f;
+  const float inv = 1.f / wave_reduce_sum(l * wgt);
+
+  if (d >= head_size) return;
+  const float* po = part_o + (size_t)h * n_chunks * head_size + d;
+  float acc = 0.f;
+#pragma unroll 4
+  for (int c = 0; c < n_chunks; ++c) {
+    const float w = __shfl(wgt, c, 64);
+    acc += w * po[(size_t)c * head_size];
   }
+  out[(size_t)h * head_size + d] = f2b(acc * inv);
 }
 
 // ---------------------------------------------------------------------------
@@ -1612,22 +1584,22 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 }
 
 template <int QPK, int HS>
-static void attn_dispatch2(void* out, float* part_o, float* part_ml,
-                           const void* qkv, void* kpool, void* vpool,
-                           const float* cos_t, const float* sin_t,
-                           int rope_ne, const int* pos, const int* slot,
-                           int layer, int n_layers_pool, int n_kv_heads,
-                           int max_seq, int n_chunks, float scale,
-                           int n_batch, int blocks, hipStream_t stream) {
+static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
+                           void* kpool, void* vpool, const float* cos_t,
+                           const float* sin_t, int rope_ne, const int* pos,
+                           const int* slot, int layer, int n_layers_pool,
+                           int n_kv_heads, int max_seq, int n_chunks,
+                           float scale, int n_batch, int blocks,
+                           hipStream_t stream) {
   if (n_batch > 0) {
     hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 1>), dim3(blocks),
-                       dim3(256), 0, stream, (bf16*)out, part_o, part_ml,
+                       dim3(256), 0, stream, part_o, part_ml,
                        (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
                        sin_t, rope_ne, pos, slot, layer, n_layers_pool,
                        n_kv_heads, max_seq, n_chunks, scale, n_batch);
   } else {
     hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 0>), dim3(blocks),
-                       dim3(256), 0, stream, (bf16*)out, part_o, part_ml,
+                       dim3(256), 0, stream, part_o, part_ml,
                        (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
                        sin_t, rope_ne, pos, slot, layer, n_layers_pool,
                        n_kv_heads, max_seq, n_chunks, scale, 1);
@@ -1635,19 +1607,18 @@ static void attn_dispatch2(void* out, float* part_o, float* part_ml,
 }
 
 template <int QPK>
-static int attn_dispatch1(int head_size, void* out, float* part_o,
-                          float* part_ml, const void* qkv, void* kpool,
-                          void* vpool, const float* cos_t,
-                          const float* sin_t, int rope_ne, const int* pos,
-                          const int* slot, int layer, int n_layers_pool,
-                          int n_kv_heads, int max_seq, int n_chunks,
-                          float scale, int n_batch, int blocks,
-                          hipStream_t stream) {
+static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
+                          const void* qkv, void* kpool, void* vpool,
+                          const float* cos_t, const float* sin_t,
+                          int rope_ne, const int* pos, const int* slot,
+                          int layer, int n_layers_pool, int n_kv_heads,
+                          int max_seq, int n_chunks, float scale,
+                          int n_batch, int blocks, hipStream_t stream) {
 #define CASE_HS(H)                                                          \
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
-      attn_dispatch2<QPK, H>(out, part_o, part_ml, qkv, kpool, vpool,       \
-                             cos_t, sin_t, rope_ne, pos, slot, layer,       \
+      attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, cos_t,     \
+                             sin_t, rope_ne, pos, slot, layer,              \
                              n_layers_pool, n_kv_heads, max_seq, n_chunks,  \
                              scale, n_batch, blocks, stream);               \
       return 0;                                                             \
@@ -1669,47 +1640,54 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        int n_layers_pool, int n_kv_heads, int max_seq,
                        int head_size, int qpk, int n_chunks, float scale,
                        int n_batch, hipStream_t stream) {
-  // the fused combine broadcasts chunk weights from two registers per lane
-  if (n_chunks > 128) n_chunks = 128;
-  if ((n_batch > 0 ? n_batch : 1) * n_kv_heads > ATTN_CTR_CAP) return -1;
   const int n_wg = n_kv_heads * n_chunks * (n_batch > 0 ? n_batch : 1);
   const int blocks = (n_wg + ATTN_WAVES - 1) / ATTN_WAVES;
   int rc = -1;
   switch (qpk) {
     case 1:
       rc = attn_dispatch1<1>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 2:
       rc = attn_dispatch1<2>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 4:
       rc = attn_dispatch1<4>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 8:
       rc = attn_dispatch1<8>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 16:
       rc = attn_dispatch1<16>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     default:
       return -1;
   }
-  return rc;
+  if (rc != 0) return rc;
+  // the combine kernel is batch-agnostic: [B, n_head, chunks, hs] is just
+  // B*n_head heads
+  const int n_head_eff = n_kv_heads * qpk * (n_batch > 0 ? n_batch : 1);
+  const int ds = (head_size + 63) / 64;
+  const int n_waves = n_head_eff * ds;
+  const int cblocks = (n_waves * 64 + 255) / 256;
+  hipLaunchKernelGGL(attn_combine_kernel, dim3(cblocks), dim3(256), 0,
+                     stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
+                     n_head_eff);
+  return 0;
 }
 
 void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,

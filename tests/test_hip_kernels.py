@@ -298,26 +298,25 @@ def test_rope_kv_append(ops, qpk, n_kv, hs, ne):
 
 
 # ---------------------------------------------------------------------------
-@pytest.mark.parametrize("qpk,n_kv,hs,S,ne,n_chunks", [
-    (4, 8, 128, 1, 128, 32),     # single active chunk: direct-write path
-    (4, 8, 128, 500, 128, 32),
-    (4, 8, 128, 2048, 128, 32),
-    (8, 8, 128, 333, 128, 32),
-    (1, 8, 64, 100, 64, 32),
-    (2, 4, 128, 77, 0, 32),      # no rope (learned-pos models)
-    (16, 2, 64, 129, 16, 32),    # partial rotary
-    (4, 8, 128, 2048, 128, 128), # >64 active chunks (second shfl register)
+@pytest.mark.parametrize("qpk,n_kv,hs,S,ne", [
+    (4, 8, 128, 1, 128),
+    (4, 8, 128, 500, 128),
+    (4, 8, 128, 2048, 128),
+    (8, 8, 128, 333, 128),
+    (1, 8, 64, 100, 64),
+    (2, 4, 128, 77, 0),      # no rope (learned-pos models)
+    (16, 2, 64, 129, 16),    # partial rotary
 ])
-def test_attn_decode(ops, qpk, n_kv, hs, S, ne, n_chunks):
+def test_attn_decode(ops, qpk, n_kv, hs, S, ne):
     """Fused kernel: ropes q and the current k from the RAW qkv buffer,
-    attends over pool[0..S-2] + current, appends k/v at pos, and combines
-    the split-S partials in the same launch (last-wave reduction)."""
+    attends over pool[0..S-2] + current, and appends k/v at pos."""
     from mdi_llm_amd.models.model import build_rope_cache
 
     torch.manual_seed(22)
     n_head = n_kv * qpk
     max_seq = 2048
     n_layers, n_slots, layer, slot_i = 2, 2, 1, 1
+    n_chunks = 32
     pos_i = S - 1
 
     kpool = mk(n_slots, n_layers, n_kv, max_seq, hs, seed=23)

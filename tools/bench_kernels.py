@@ -1,6 +1,6 @@
 #!/usr/bin/env python3
 """Micro-benchmark of the decode kernels (per-shape GEMV variants, attn,
-fused split-S combine) on one MI355X.  Prints achieved TB/s per variant so per-shape
+combine) on one MI355X.  Prints achieved TB/s per variant so per-shape
 `rows` choices in the engine can be set from measurement, not guesses.
 
 Usage (on a GPU box):  python tools/bench_kernels.py
@@ -96,7 +96,7 @@ def bench_attn():
                                             n_chunks, 1 / math.sqrt(hs)))
         gb = 2 * n_kv * S * hs * 2 / 1e12
         print(f"attn S={S:5d}: {us:7.1f} us  KV-read {gb/(us*1e-6):5.2f} TB/s"
-              f"  (combine fused)")
+              f"  (incl. combine)")
 
 
 def bench_sampling():
