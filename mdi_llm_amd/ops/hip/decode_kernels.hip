@@ -600,21 +600,36 @@ __global__ void rope_kv_append_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// GQA flash-decode attention, split-S.
+// GQA flash-decode attention. Two variants share the MFMA tile core:
 //
-// One wave handles one (kv_head, chunk): MFMA 16x16x32 bf16 computes
-// scores[key, qhead] (swapped operands so each lane owns 4 keys of ONE
-// query head), online softmax with running (m, l) per qhead in LDS, P*V
-// accumulated in registers. Partials (o, m, l) per chunk go to global fp32;
-// attn_combine_kernel reduces chunks.
+//  * attn_decode_block_kernel — used when max_seq <= 4096 (chosen
+//    statically at launch, so hipGraphs stay shape-stable): ONE block per
+//    (batch, kv_head); its 4 waves split the keys tile-interleaved, each
+//    wave's online-softmax state (m, l) lives in registers keyed to the
+//    lane's query head and every cross-lane exchange is a shfl, so the
+//    key loop has NO LDS traffic and NO barriers; the 4 per-wave partials
+//    are combined through LDS by the whole block at the end. The entire
+//    attention step is one launch with zero global scratch — at decode
+//    context lengths both the old split-S kernel and its combine kernel
+//    sat at the ~5 us in-stream launch floor, and a fence-based in-kernel
+//    combine was measured SLOWER (agent-scope release = cross-XCD L2
+//    writeback per wave); block-local combine avoids both.
 //
-// Launch: grid.x = n_kv_heads * n_chunks / WAVES_PER_BLOCK, block = 256.
-// qpk (query heads per kv head) <= 16.
+//  * attn_decode_kernel (split-S) — for long contexts (max_seq > 4096):
+//    one wave per (kv_head, chunk) over n_chunks chunks for full-chip
+//    parallelism, fp32 partials to global, attn_combine_kernel reduces.
+//
+// MFMA 16x16x32 bf16 computes scores[key, qhead] (swapped operands so
+// each lane owns 4 keys of ONE query head). qpk (query heads per kv
+// head) <= 16.
 // ---------------------------------------------------------------------------
 using f32x4 = __attribute__((__vector_size__(16))) float;
 using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
 
 #define ATTN_WAVES 4  // waves per block, each fully independent
+// max_seq at/below which the one-launch block-local variant is used; the
+// choice is static per engine (max_seq), so hipGraph shapes never change
+#define ATTN_BLOCK_MAX_SEQ 4096
 
 // XOR swizzle on the q LDS tile (rows at a power-of-two byte stride would
 // otherwise put a ds_read_b128 lane group on one bank slot — guide §6 G4):
@@ -637,6 +652,202 @@ DEVINL float rope_elem(const bf16* row, int d, int ne,
   const float s = sin_t[(size_t)pos * ne + d];
   if (d < half) return x * c - b2f(row[d + half]) * s;
   return x * c + b2f(row[d - half]) * s;
+}
+
+// Block-local variant: grid.x = (n_batch) * n_kv_heads, block = 256.
+// BATCH semantics as in attn_decode_kernel below.
+template <int QPK, int HS, int BATCH>
+__global__ void attn_decode_block_kernel(
+    bf16* __restrict__ out,       // [B?, n_head * head_size]
+    const bf16* __restrict__ qkv, // [B?, qkv_dim] interleaved, RAW
+    bf16* __restrict__ kpool, bf16* __restrict__ vpool,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    int rope_ne, const int* __restrict__ pos_p,
+    const int* __restrict__ slot_p, int layer, int n_layers_pool,
+    int n_kv_heads, int max_seq, float scale, int n_batch) {
+  constexpr int head_size = HS;
+  static_assert(HS % 32 == 0 && HS * QPK >= 64, "unsupported attn geometry");
+  constexpr int ODIM = HS * QPK / 64;  // output dims per lane (PV map)
+
+  // shared q tile + current k row (staged once per block) and the 4
+  // per-wave partials for the final in-block combine
+  __shared__ __attribute__((aligned(16))) bf16 q_lds[16 * HS];
+  __shared__ __attribute__((aligned(16))) bf16 k_cur[HS];
+  __shared__ float o_part[ATTN_WAVES][QPK][HS];
+  __shared__ float ml_part[ATTN_WAVES][QPK][2];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int b = BATCH ? (int)blockIdx.x / n_kv_heads : 0;
+  const int g = BATCH ? (int)blockIdx.x % n_kv_heads : (int)blockIdx.x;
+  if (b >= n_batch) return;
+
+  const int S = pos_p[b] + 1;
+  const int pos = S - 1;
+  const int slot = slot_p[b];
+  const int n_head_all = n_kv_heads * QPK;
+  if (BATCH) {
+    const int qkv_dim = n_kv_heads * (QPK + 2) * head_size;
+    qkv += (size_t)b * qkv_dim;
+    out += (size_t)b * n_head_all * head_size;
+  }
+
+  const size_t cache_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq * head_size;
+  const bf16* krow_cur = qkv + ((size_t)g * (QPK + 2) + QPK) * head_size;
+  const bf16* vrow_cur = krow_cur + head_size;
+
+  // ---- block-cooperative staging: q rows (roped), current k row ----------
+  for (int i = threadIdx.x; i < 16 * head_size; i += 256) {
+    const int r = i / head_size;
+    const int d = i % head_size;
+    bf16 val = f2b(0.f);
+    if (r < QPK) {
+      const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
+      val = f2b(rope_elem(qrow, d, rope_ne, cos_t, sin_t, pos));
+    }
+    q_lds[q_swz<HS>(r, d)] = val;
+  }
+  for (int d = threadIdx.x; d < head_size; d += 256)
+    k_cur[d] = f2b(rope_elem(krow_cur, d, rope_ne, cos_t, sin_t, pos));
+  __syncthreads();  // the only barrier before the combine
+
+  // append the current token's k,v (no other block touches (slot, g))
+  for (int d = threadIdx.x; d < head_size; d += 256) {
+    kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
+    vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
+  }
+
+  // ---- per-lane roles ----------------------------------------------------
+  // softmax map (A): qhead qa = lane & 15, key sub-row = lane >> 4
+  // PV map (B): qb = lane % QPK, dim slice d0, ODIM dims per lane
+  const int qa = lane & 15;
+  const int sub = lane >> 4;
+  const int qb = lane % QPK;
+  const int d0 = (lane / QPK) * ODIM;
+
+  // every lane redundantly tracks (m, l) of ITS query head qa — the 4
+  // lanes sharing a qa stay in sync because they see identical reduced
+  // values, so no LDS round-trips are needed in the loop
+  float m_st = -1e30f, l_st = 0.f;
+  float o_acc[ODIM];
+#pragma unroll
+  for (int i = 0; i < ODIM; ++i) o_acc[i] = 0.f;
+
+  const int n_tiles = (S + 15) / 16;
+  for (int t = wave; t < n_tiles; t += ATTN_WAVES) {
+    const int key0 = t * 16;
+
+    // ---- QK^T via MFMA: A = K tile (16 keys x 32 dims), B = q^T ----------
+    f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15;
+    const int koff = (lane >> 4) * 8;
+    const int akey = key0 + arow;
+    const bf16* krow = kpool + cache_base + (size_t)akey * head_size;
+    const bool row_valid = akey < S;
+    const bool row_cur = akey == pos;
+#pragma unroll
+    for (int c = 0; c < HS / 32; ++c) {
+      bf16x8_t af = {};
+      if (row_cur)
+        af = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
+      else if (row_valid)
+        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      const bf16x8_t bq = *reinterpret_cast<const bf16x8_t*>(
+          &q_lds[q_swz<HS>(arow, c * 32 + koff)]);
+      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bq, acc4, 0, 0, 0);
+    }
+    float sc[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = key0 + sub * 4 + r;
+      sc[r] = (key < S) ? acc4[r] * scale : -1e30f;
+    }
+
+    // ---- online softmax, all state in registers, exchanges via shfl ------
+    float tmax = fmaxf(fmaxf(sc[0], sc[1]), fmaxf(sc[2], sc[3]));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float m_new = fmaxf(m_st, tmax);
+    const float alpha = __expf(m_st - m_new);
+    m_st = m_new;
+    float e[4];
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = key0 + sub * 4 + r;
+      e[r] = (key < S) ? __expf(sc[r] - m_new) : 0.f;
+      psum += e[r];
+    }
+    psum += __shfl_xor(psum, 16, 64);
+    psum += __shfl_xor(psum, 32, 64);
+    l_st = l_st * alpha + psum;
+
+    // ---- P*V: o[qb][d0..] += p[s][qb] * V[s][d0..] -----------------------
+    // p for (key s, head qb) lives in lane ((s>>2)<<4 | qb), register s&3
+    const float alphaB = __shfl(alpha, qb, 64);
+#pragma unroll
+    for (int i = 0; i < ODIM; ++i) o_acc[i] *= alphaB;
+    // fixed 16-iteration loop so it fully unrolls (e[s&3] must stay in
+    // registers); keys past S have e == 0, their V row is clamped to pos
+#pragma unroll
+    for (int s = 0; s < 16; ++s) {
+      const float p = __shfl(e[s & 3], ((s >> 2) << 4) | qb, 64);
+      const int skey = min(key0 + s, pos);
+      const bf16* vrow =
+          (skey == pos)
+              ? vrow_cur + d0
+              : vpool + cache_base + (size_t)skey * head_size + d0;
+      if constexpr (ODIM >= 8) {
+#pragma unroll
+        for (int i = 0; i < ODIM; i += 8) {
+          bf16x8 vv = load8(vrow + i);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) o_acc[i + j] += p * b2f(vv.v[j]);
+        }
+      } else if constexpr (ODIM == 4) {
+        int2 raw = *reinterpret_cast<const int2*>(vrow);
+        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o_acc[j] += p * b2f(vv[j]);
+      } else if constexpr (ODIM == 2) {
+        int raw = *reinterpret_cast<const int*>(vrow);
+        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+        o_acc[0] += p * b2f(vv[0]);
+        o_acc[1] += p * b2f(vv[1]);
+      } else {  // ODIM == 1
+        o_acc[0] += p * b2f(vrow[0]);
+      }
+    }
+  }
+
+  // ---- per-wave partials to LDS, then block combine ----------------------
+  // (a wave that got no tiles contributes m=-inf, l=0, o=0 — weight 0)
+#pragma unroll
+  for (int i = 0; i < ODIM; ++i) o_part[wave][qb][d0 + i] = o_acc[i];
+  if (lane < QPK) {
+    ml_part[wave][lane][0] = m_st;  // lane < 16 => qa == lane
+    ml_part[wave][lane][1] = l_st;
+  }
+  __syncthreads();
+
+  for (int i = threadIdx.x; i < QPK * head_size; i += 256) {
+    const int h = i / head_size;
+    const int d = i % head_size;
+    float M = -1e30f;
+#pragma unroll
+    for (int w = 0; w < ATTN_WAVES; ++w)
+      M = fmaxf(M, ml_part[w][h][0]);
+    float den = 0.f, num = 0.f;
+#pragma unroll
+    for (int w = 0; w < ATTN_WAVES; ++w) {
+      const float wgt = __expf(ml_part[w][h][0] - M);
+      den += wgt * ml_part[w][h][1];
+      num += wgt * o_part[w][h][d];
+    }
+    out[(size_t)(g * QPK + h) * head_size + d] = f2b(num / den);
+  }
 }
 
 // BATCH: 0 -> single-token mode (pos_p/slot_p are device scalars);
@@ -875,24 +1086,31 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
   const int d = (wid % DS) * 64 + lane;
   if (h >= n_head) return;
 
-  float m = -1e30f, l = 0.f;
+  // chunk weights lane-parallel in TWO registers (lane c holds chunks c
+  // and c+64), so up to 128 chunks combine correctly
+  float m0 = -1e30f, m1 = -1e30f, l0 = 0.f, l1 = 0.f;
   if (lane < n_chunks) {
-    m = part_ml[((size_t)h * n_chunks + lane) * 2];
-    l = part_ml[((size_t)h * n_chunks + lane) * 2 + 1];
+    m0 = part_ml[((size_t)h * n_chunks + lane) * 2];
+    l0 = part_ml[((size_t)h * n_chunks + lane) * 2 + 1];
   }
-  float M = m;
+  if (lane + 64 < n_chunks) {
+    m1 = part_ml[((size_t)h * n_chunks + lane + 64) * 2];
+    l1 = part_ml[((size_t)h * n_chunks + lane + 64) * 2 + 1];
+  }
+  float M = fmaxf(m0, m1);
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1)
     M = fmaxf(M, __shfl_xor(M, off, 64));
-  const float wgt = (lane < n_chunks) ? __expf(m - M) : 0.f;
-  const float inv = 1.f / wave_reduce_sum(l * wgt);
+  const float w0 = (lane < n_chunks) ? __expf(m0 - M) : 0.f;
+  const float w1 = (lane + 64 < n_chunks) ? __expf(m1 - M) : 0.f;
+  const float inv = 1.f / wave_reduce_sum(w0 * l0 + w1 * l1);
 
   if (d >= head_size) return;
   const float* po = part_o + (size_t)h * n_chunks * head_size + d;
   float acc = 0.f;
 #pragma unroll 4
   for (int c = 0; c < n_chunks; ++c) {
-    const float w = __shfl(wgt, c, 64);
+    const float w = (c < 64) ? __shfl(w0, c, 64) : __shfl(w1, c - 64, 64);
     acc += w * po[(size_t)c * head_size];
   }
   out[(size_t)h * head_size + d] = f2b(acc * inv);
@@ -1584,13 +1802,32 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 }
 
 template <int QPK, int HS>
-static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
-                           void* kpool, void* vpool, const float* cos_t,
-                           const float* sin_t, int rope_ne, const int* pos,
-                           const int* slot, int layer, int n_layers_pool,
-                           int n_kv_heads, int max_seq, int n_chunks,
-                           float scale, int n_batch, int blocks,
-                           hipStream_t stream) {
+static void attn_dispatch2(void* out, float* part_o, float* part_ml,
+                           const void* qkv, void* kpool, void* vpool,
+                           const float* cos_t, const float* sin_t,
+                           int rope_ne, const int* pos, const int* slot,
+                           int layer, int n_layers_pool, int n_kv_heads,
+                           int max_seq, int n_chunks, float scale,
+                           int n_batch, int blocks, hipStream_t stream) {
+  // short/medium contexts: one block per (batch, kv_head), no global
+  // partials, no combine kernel — the whole attention step is ONE launch
+  if (max_seq <= ATTN_BLOCK_MAX_SEQ) {
+    const int nb = (n_batch > 0 ? n_batch : 1) * n_kv_heads;
+    if (n_batch > 0) {
+      hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, 1>), dim3(nb),
+                         dim3(256), 0, stream, (bf16*)out, (const bf16*)qkv,
+                         (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne,
+                         pos, slot, layer, n_layers_pool, n_kv_heads,
+                         max_seq, scale, n_batch);
+    } else {
+      hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, 0>), dim3(nb),
+                         dim3(256), 0, stream, (bf16*)out, (const bf16*)qkv,
+                         (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne,
+                         pos, slot, layer, n_layers_pool, n_kv_heads,
+                         max_seq, scale, 1);
+    }
+    return;
+  }
   if (n_batch > 0) {
     hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 1>), dim3(blocks),
                        dim3(256), 0, stream, part_o, part_ml,
@@ -1607,18 +1844,19 @@ static void attn_dispatch2(float* part_o, float* part_ml, const void* qkv,
 }
 
 template <int QPK>
-static int attn_dispatch1(int head_size, float* part_o, float* part_ml,
-                          const void* qkv, void* kpool, void* vpool,
-                          const float* cos_t, const float* sin_t,
-                          int rope_ne, const int* pos, const int* slot,
-                          int layer, int n_layers_pool, int n_kv_heads,
-                          int max_seq, int n_chunks, float scale,
-                          int n_batch, int blocks, hipStream_t stream) {
+static int attn_dispatch1(int head_size, void* out, float* part_o,
+                          float* part_ml, const void* qkv, void* kpool,
+                          void* vpool, const float* cos_t,
+                          const float* sin_t, int rope_ne, const int* pos,
+                          const int* slot, int layer, int n_layers_pool,
+                          int n_kv_heads, int max_seq, int n_chunks,
+                          float scale, int n_batch, int blocks,
+                          hipStream_t stream) {
 #define CASE_HS(H)                                                          \
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
-      attn_dispatch2<QPK, H>(part_o, part_ml, qkv, kpool, vpool, cos_t,     \
-                             sin_t, rope_ne, pos, slot, layer,              \
+      attn_dispatch2<QPK, H>(out, part_o, part_ml, qkv, kpool, vpool,       \
+                             cos_t, sin_t, rope_ne, pos, slot, layer,       \
                              n_layers_pool, n_kv_heads, max_seq, n_chunks,  \
                              scale, n_batch, blocks, stream);               \
       return 0;                                                             \
@@ -1646,31 +1884,31 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
   switch (qpk) {
     case 1:
       rc = attn_dispatch1<1>(
-          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 2:
       rc = attn_dispatch1<2>(
-          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 4:
       rc = attn_dispatch1<4>(
-          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 8:
       rc = attn_dispatch1<8>(
-          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 16:
       rc = attn_dispatch1<16>(
-          head_size, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
@@ -1678,6 +1916,7 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
       return -1;
   }
   if (rc != 0) return rc;
+  if (max_seq <= ATTN_BLOCK_MAX_SEQ) return 0;  // block-local: no combine
   // the combine kernel is batch-agnostic: [B, n_head, chunks, hs] is just
   // B*n_head heads
   const int n_head_eff = n_kv_heads * qpk * (n_batch > 0 ? n_batch : 1);

@@ -298,25 +298,28 @@ def test_rope_kv_append(ops, qpk, n_kv, hs, ne):
 
 
 # ---------------------------------------------------------------------------
-@pytest.mark.parametrize("qpk,n_kv,hs,S,ne", [
-    (4, 8, 128, 1, 128),
-    (4, 8, 128, 500, 128),
-    (4, 8, 128, 2048, 128),
-    (8, 8, 128, 333, 128),
-    (1, 8, 64, 100, 64),
-    (2, 4, 128, 77, 0),      # no rope (learned-pos models)
-    (16, 2, 64, 129, 16),    # partial rotary
+# max_seq <= 4096 exercises the one-launch block-local variant;
+# max_seq = 8192 exercises the global split-S + combine path
+@pytest.mark.parametrize("qpk,n_kv,hs,S,ne,max_seq,n_chunks", [
+    (4, 8, 128, 1, 128, 2048, 32),
+    (4, 8, 128, 500, 128, 2048, 32),
+    (4, 8, 128, 2048, 128, 2048, 32),
+    (8, 8, 128, 333, 128, 2048, 32),
+    (1, 8, 64, 100, 64, 2048, 32),
+    (2, 4, 128, 77, 0, 2048, 32),      # no rope (learned-pos models)
+    (16, 2, 64, 129, 16, 2048, 32),    # partial rotary
+    (4, 8, 128, 500, 128, 8192, 32),   # split-S path, short S
+    (4, 8, 128, 2048, 128, 8192, 32),  # split-S path
+    (4, 8, 128, 2048, 128, 8192, 128), # >64 active chunks in the combine
 ])
-def test_attn_decode(ops, qpk, n_kv, hs, S, ne):
+def test_attn_decode(ops, qpk, n_kv, hs, S, ne, max_seq, n_chunks):
     """Fused kernel: ropes q and the current k from the RAW qkv buffer,
     attends over pool[0..S-2] + current, and appends k/v at pos."""
     from mdi_llm_amd.models.model import build_rope_cache
 
     torch.manual_seed(22)
     n_head = n_kv * qpk
-    max_seq = 2048
     n_layers, n_slots, layer, slot_i = 2, 2, 1, 1
-    n_chunks = 32
     pos_i = S - 1
 
     kpool = mk(n_slots, n_layers, n_kv, max_seq, hs, seed=23)
