@@ -17,6 +17,7 @@ MoE stages fall back to the torch path.
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -190,10 +191,11 @@ class DecodeEngine:
         )
         self.pos = torch.zeros(1, device=dev, dtype=torch.int32)
 
-        self._r_qkv = self._rows(cfg.qkv_dim, E)
-        self._r_proj = self._rows(E, cfg.n_head * cfg.head_size)
-        self._r_down = self._rows(E, I)
-        self._r_head = self._rows(cfg.padded_vocab_size, E)
+        self._r_qkv = self._rows(cfg.qkv_dim, E, "MDI_ROWS_QKV")
+        self._r_proj = self._rows(E, cfg.n_head * cfg.head_size,
+                                  "MDI_ROWS_PROJ")
+        self._r_down = self._rows(E, I, "MDI_ROWS_DOWN")
+        self._r_head = self._rows(cfg.padded_vocab_size, E, "MDI_ROWS_HEAD")
 
         if self.is_starter:
             # per-slot current-token table for fully-fused step graphs
@@ -229,16 +231,19 @@ class DecodeEngine:
         return 1 if self.config.norm_class_name == "RMSNorm" else 2
 
     @staticmethod
-    def _rows(M: int, K: int) -> int:
-        """Output rows per wave for the decode GEMV, chosen from the MI355X
-        micro-bench sweep (tools/bench_kernels.py): long-K shapes want 4-row
-        ILP; small-M and huge-M shapes want max workgroup count."""
+    def _rows(M: int, K: int, env: str = "") -> int:
+        """Output rows per wave for the decode GEMV, from an in-graph A/B
+        sweep on MI355X (Llama-3-8B shapes): long-K shapes want 4-row
+        ILP; for everything else MORE blocks wins — many blocks per CU
+        pipeline the per-kernel memory ramp, so rows=1 beat rows=2/4 on
+        the qkv (6144x4096) and proj/down shapes by 3-15%. Env override
+        for tuning."""
+        if env and os.environ.get(env):
+            return int(os.environ[env])
         if K >= 16384:
             return 4
-        if M <= 4096 or M >= 65536:
+        if M <= 16384 or M >= 65536:
             return 1
-        if M <= 7168:
-            return 4
         return 2
 
     def sample_into_token(self, temperature: float, top_k, seed: int,

@@ -699,15 +699,17 @@ __global__ void attn_decode_block_kernel(
   const bf16* vrow_cur = krow_cur + head_size;
 
   // ---- block-cooperative staging: q rows (roped), current k row ----------
-  for (int i = threadIdx.x; i < 16 * head_size; i += 256) {
+  // the swizzle permutes 16-byte units WITHIN a row, so the pad rows
+  // (>= QPK) can be zero-filled with vector stores disjoint from the
+  // roped real rows — no barrier needed between the two loops
+  for (int i = threadIdx.x; i < (16 - QPK) * head_size / 8; i += 256)
+    reinterpret_cast<bf16x8_t*>(q_lds + QPK * head_size)[i] = bf16x8_t{};
+  for (int i = threadIdx.x; i < QPK * head_size; i += 256) {
     const int r = i / head_size;
     const int d = i % head_size;
-    bf16 val = f2b(0.f);
-    if (r < QPK) {
-      const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
-      val = f2b(rope_elem(qrow, d, rope_ne, cos_t, sin_t, pos));
-    }
-    q_lds[q_swz<HS>(r, d)] = val;
+    const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
+    q_lds[q_swz<HS>(r, d)] =
+        f2b(rope_elem(qrow, d, rope_ne, cos_t, sin_t, pos));
   }
   for (int d = threadIdx.x; d < head_size; d += 256)
     k_cur[d] = f2b(rope_elem(krow_cur, d, rope_ne, cos_t, sin_t, pos));
