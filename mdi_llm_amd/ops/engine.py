@@ -185,7 +185,6 @@ class DecodeEngine:
 
         # device-side slot/pos scalars (graph-replayable)
         self.slot = torch.zeros(1, device=dev, dtype=torch.int32)
-        self.slot_long = torch.zeros(1, device=dev, dtype=torch.int64)
         self.pos_table = torch.zeros(
             kv_pool.n_slots, device=dev, dtype=torch.int32
         )
@@ -201,7 +200,6 @@ class DecodeEngine:
             # per-slot current-token table for fully-fused step graphs
             self.token_table = torch.zeros(kv_pool.n_slots, device=dev,
                                            dtype=torch.int32)
-            self._ones_i32 = torch.ones(1, device=dev, dtype=torch.int32)
 
         self._graph_blocks: Optional[torch.cuda.CUDAGraph] = None
         self._graph_tail: Optional[torch.cuda.CUDAGraph] = None
@@ -218,9 +216,17 @@ class DecodeEngine:
         self.pos_table[slot] = pos
 
     def _stage_pos(self) -> None:
-        # pos <- pos_table[slot]   (inside the graph: slot is a device value)
-        self.slot_long.copy_(self.slot)
-        torch.index_select(self.pos_table, 0, self.slot_long, out=self.pos)
+        # pos <- pos_table[slot]   (inside the graph: slot is a device
+        # value; one tiny HIP launch instead of a copy + index_select)
+        self.ops.stage_slot(self.slot, pos_out=self.pos,
+                            pos_table=self.pos_table)
+
+    def _stage_pos_token(self) -> None:
+        # pos <- pos_table[slot]; token <- token_table[slot] in ONE launch
+        self.ops.stage_slot(self.slot, pos_out=self.pos,
+                            token_out=self.token,
+                            pos_table=self.pos_table,
+                            token_table=self.token_table)
 
     # ---------------------------------------------------------------------
     # kernel sequence (eager; also what gets captured)
@@ -459,20 +465,25 @@ class DecodeEngine:
     # fully-fused per-token step graphs (bench hot path)
     # ------------------------------------------------------------------
     def _sample_seq(self, temperature: float, top_k: int, seed: int,
-                    top_p: float = 1.0) -> None:
-        self.ops.sample(self.sample_out, self.logits, self.sample_scratch,
+                    top_p: float = 1.0, out=None, advance: str = "") -> None:
+        """Fused sampling; `advance` folds the step bookkeeping into the
+        sampler's unpack launch: "token" writes token_table[slot],
+        "token+pos" additionally advances pos_table[slot]."""
+        self.ops.sample(self.sample_out if out is None else out,
+                        self.logits, self.sample_scratch,
                         float(temperature), int(top_k or 0),
                         temperature > 0.0, int(seed) & 0x7FFFFFFF,
-                        self.pos, self.slot, 0, float(top_p))
+                        self.pos, self.slot, 0, float(top_p),
+                        token_table=self.token_table if advance else None,
+                        pos_table=self.pos_table
+                        if advance == "token+pos" else None,
+                        adv_slot=self.slot if advance else None,
+                        adv_pos=1 if advance == "token+pos" else 0)
 
-    def _advance(self) -> None:
-        # token_table[slot] = sample_out ; pos_table[slot] += 1 (in-graph)
-        self.token_table.index_copy_(0, self.slot_long, self.sample_out)
-        self.pos_table.index_add_(0, self.slot_long, self._ones_i32)
-
-    def _stage_token(self) -> None:
-        torch.index_select(self.token_table, 0, self.slot_long,
-                           out=self.token)
+    def _advance_pos(self) -> None:
+        # pos_table[slot] += 1 (in-graph, one tiny launch)
+        self.ops.stage_slot(self.slot, pos_table_mut=self.pos_table,
+                            adv_pos=1)
 
     def ensure_fused_graphs(self, temperature: float, top_k, seed: int) -> None:
         """Capture the three starter step graphs for fixed sampling params:
@@ -490,13 +501,11 @@ class DecodeEngine:
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                self._stage_pos()
-                self._stage_token()
+                self._stage_pos_token()
                 self._embed()
                 self._run_blocks()
                 self._tail_seq()
-                self._sample_seq(t, k, sd)
-                self._advance()
+                self._sample_seq(t, k, sd, advance="token+pos")
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self.kv_pool.k.zero_()
@@ -506,39 +515,34 @@ class DecodeEngine:
 
         g1 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g1):
-            self._stage_pos()
-            self._stage_token()
+            self._stage_pos_token()
             self._embed()
             self._run_blocks()
             self._tail_seq()
-            self._sample_seq(t, k, sd)
-            self._advance()
+            self._sample_seq(t, k, sd, advance="token+pos")
         self._graph_standalone = g1
 
         g2 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g2):
             self._stage_pos()
             self._tail_seq()
-            self._sample_seq(t, k, sd)
-            self._advance_token_only()
-            self._stage_token()
+            # the sampler writes the drawn token into BOTH token_table
+            # and the embed input scalar, and the blocks run with the
+            # pre-advance pos; pos advances at the end of the graph
+            self._sample_seq(t, k, sd, out=self.token, advance="token")
             self._embed()
             self._run_blocks()
-            self.pos_table.index_add_(0, self.slot_long, self._ones_i32)
+            self._advance_pos()
         self._graph_starter = g2
 
         g3 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g3):
             self._stage_pos()
             self._tail_seq()
-            self._sample_seq(t, k, sd)
-            self._advance_token_only()
+            self._sample_seq(t, k, sd, advance="token")
         self._graph_tail_sample = g3
         self._fused_params = params
         torch.cuda.synchronize()
-
-    def _advance_token_only(self) -> None:
-        self.token_table.index_copy_(0, self.slot_long, self.sample_out)
 
     def standalone_step(self, slot: int) -> None:
         """One full decode token for `slot` (token_table-chained)."""

@@ -282,7 +282,9 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
             torch::Tensor scratch, double temperature, int64_t top_k,
             bool noise, int64_t seed, c10::optional<torch::Tensor> pos,
             c10::optional<torch::Tensor> slot, int64_t n_batch,
-            double top_p) {
+            double top_p, c10::optional<torch::Tensor> token_table,
+            c10::optional<torch::Tensor> pos_table,
+            c10::optional<torch::Tensor> adv_slot, int64_t adv_pos) {
   check_i32(out_token, "out_token");
   check_bf16(logits, "logits");
   const int nb = n_batch > 0 ? (int)n_batch : 1;
@@ -302,11 +304,45 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
     TORCH_CHECK(slot->numel() >= nb, "slot too small");
     sp = slot->data_ptr<int>();
   }
+  int* ttp = nullptr;
+  int* ptp = nullptr;
+  const int* asp = nullptr;
+  if (token_table.has_value()) {
+    check_i32(*token_table, "token_table");
+    ttp = token_table->data_ptr<int>();
+  }
+  if (pos_table.has_value()) {
+    check_i32(*pos_table, "pos_table");
+    ptp = pos_table->data_ptr<int>();
+  }
+  if (adv_slot.has_value()) {
+    check_i32(*adv_slot, "adv_slot");
+    asp = adv_slot->data_ptr<int>();
+  }
   const int V = (int)(logits.numel() / nb);
   launch_sample(out_token.data_ptr(), logits.data_ptr(), V,
                 scratch.data_ptr(), (float)temperature, (int)top_k,
                 (float)top_p, noise ? 1 : 0, (unsigned)(int64_t)seed, pp, sp,
-                (int)n_batch, cur_stream());
+                (int)n_batch, ttp, ptp, asp, (int)adv_pos, cur_stream());
+}
+
+void stage_slot(torch::Tensor slot, c10::optional<torch::Tensor> pos_out,
+                c10::optional<torch::Tensor> token_out,
+                c10::optional<torch::Tensor> pos_table,
+                c10::optional<torch::Tensor> token_table,
+                c10::optional<torch::Tensor> pos_table_mut,
+                int64_t adv_pos) {
+  check_i32(slot, "slot");
+  int* po = pos_out.has_value() ? pos_out->data_ptr<int>() : nullptr;
+  int* to = token_out.has_value() ? token_out->data_ptr<int>() : nullptr;
+  const int* pt =
+      pos_table.has_value() ? pos_table->data_ptr<int>() : nullptr;
+  const int* tt =
+      token_table.has_value() ? token_table->data_ptr<int>() : nullptr;
+  int* ptm =
+      pos_table_mut.has_value() ? pos_table_mut->data_ptr<int>() : nullptr;
+  launch_stage_slot(po, to, pt, tt, ptm, slot.data_ptr<int>(),
+                    (int)adv_pos, cur_stream());
 }
 
 void rope_prefill_append(torch::Tensor qkv, torch::Tensor kpool,
@@ -415,5 +451,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("temperature"), py::arg("top_k"), py::arg("noise"),
         py::arg("seed"), py::arg("pos") = c10::nullopt,
         py::arg("slot") = c10::nullopt, py::arg("n_batch") = 0,
-        py::arg("top_p") = 1.0);
+        py::arg("top_p") = 1.0, py::arg("token_table") = c10::nullopt,
+        py::arg("pos_table") = c10::nullopt,
+        py::arg("adv_slot") = c10::nullopt, py::arg("adv_pos") = 0);
+  m.def("stage_slot", &stage_slot,
+        "one-launch step staging/bookkeeping on device scalars",
+        py::arg("slot"), py::arg("pos_out") = c10::nullopt,
+        py::arg("token_out") = c10::nullopt,
+        py::arg("pos_table") = c10::nullopt,
+        py::arg("token_table") = c10::nullopt,
+        py::arg("pos_table_mut") = c10::nullopt, py::arg("adv_pos") = 0);
 }

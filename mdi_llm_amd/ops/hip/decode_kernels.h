@@ -68,11 +68,22 @@ void launch_add(void* out, const void* a, const void* b, int n,
 // fused temperature/top-k/top-p/gumbel sampling; scratch: >=520 u32 PER
 // SAMPLE, zeroed initially (self-cleaning); n_batch draws from
 // [n_batch, V] logits.  The gumbel stream is keyed by (seed, slot, pos) —
-// reproducible and schedule-independent.
+// reproducible and schedule-independent.  When token_table/pos_table and
+// adv_slot are given (single-sample mode), the unpack step also writes
+// token_table[slot] and advances pos_table[slot] by adv_pos in the same
+// launch (graph bookkeeping without extra kernels).
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, float top_p, int noise_on,
                    unsigned seed, const int* pos, const int* slot,
-                   int n_batch, hipStream_t stream);
+                   int n_batch, int* token_table, int* pos_table,
+                   const int* adv_slot, int adv_pos, hipStream_t stream);
+
+// one-launch step staging: pos_out = pos_table[slot], token_out =
+// token_table[slot], optional pos_table_mut[slot] += 1; any output may be
+// null
+void launch_stage_slot(int* pos_out, int* token_out, const int* pos_table,
+                       const int* token_table, int* pos_table_mut,
+                       const int* slot, int adv_pos, hipStream_t stream);
 
 // prefill: rope+append all T positions (grid covers T)
 void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,

@@ -1588,18 +1588,50 @@ __global__ void sample_gumbel_argmax_kernel(
   }
 }
 
+// Besides unpacking the winning token and self-cleaning the scratch, this
+// kernel optionally performs the decode step's bookkeeping in the same
+// launch (single-sample mode): token_table[slot] = token, and
+// pos_table[slot] += adv_pos — replacing two torch index kernels in the
+// captured step graph.
 __global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
-                                     int* __restrict__ out) {
+                                     int* __restrict__ out,
+                                     int* __restrict__ token_table,
+                                     int* __restrict__ pos_table,
+                                     const int* __restrict__ slot,
+                                     int adv_pos) {
   scratch += (size_t)blockIdx.y * 520;
   out += blockIdx.y;
   if (threadIdx.x == 0) {
     unsigned long long packed =
         *reinterpret_cast<const unsigned long long*>(&scratch[516]);
-    out[0] = (int)(unsigned)(packed & 0xFFFFFFFFull);
+    const int tok = (int)(unsigned)(packed & 0xFFFFFFFFull);
+    out[0] = tok;
+    if (slot) {
+      const int sl = slot[0];
+      if (token_table) token_table[sl] = tok;
+      if (pos_table && adv_pos) pos_table[sl] += 1;
+    }
   }
   __syncthreads();  // t0's read precedes the clean
   // self-clean the scratch for the next call (single block, runs last)
   for (int i = threadIdx.x; i < 518; i += blockDim.x) scratch[i] = 0;
+}
+
+// Step staging in one tiny launch: pos_out = pos_table[slot],
+// token_out = token_table[slot], optional pos_table[slot] += 1
+// (replaces a copy + two index_selects / an index_add in the graphs).
+__global__ void stage_slot_kernel(int* __restrict__ pos_out,
+                                  int* __restrict__ token_out,
+                                  const int* __restrict__ pos_table,
+                                  const int* __restrict__ token_table,
+                                  int* __restrict__ pos_table_mut,
+                                  const int* __restrict__ slot,
+                                  int adv_pos) {
+  if (threadIdx.x != 0) return;
+  const int sl = slot[0];
+  if (pos_out) pos_out[0] = pos_table[sl];
+  if (token_out) token_out[0] = token_table[sl];
+  if (pos_table_mut && adv_pos) pos_table_mut[sl] += 1;
 }
 
 // ---------------------------------------------------------------------------
@@ -1609,7 +1641,8 @@ __global__ void sample_unpack_kernel(unsigned* __restrict__ scratch,
 void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, float top_p, int noise_on,
                    unsigned seed, const int* pos, const int* slot,
-                   int n_batch, hipStream_t stream) {
+                   int n_batch, int* token_table, int* pos_table,
+                   const int* adv_slot, int adv_pos, hipStream_t stream) {
   unsigned* sc = (unsigned*)scratch;
   const int B = n_batch > 0 ? n_batch : 1;
   const int blocks = B > 1 ? 32 : 128;
@@ -1645,7 +1678,16 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                      0, stream, (const bf16*)logits, V, sc, inv_t,
                      use_k || use_p, noise_on, seed, pos, slot);
   hipLaunchKernelGGL(sample_unpack_kernel, dim3(1, B), dim3(64), 0, stream,
-                     sc, (int*)out_token);
+                     sc, (int*)out_token, token_table, pos_table, adv_slot,
+                     adv_pos);
+}
+
+void launch_stage_slot(int* pos_out, int* token_out, const int* pos_table,
+                       const int* token_table, int* pos_table_mut,
+                       const int* slot, int adv_pos, hipStream_t stream) {
+  hipLaunchKernelGGL(stage_slot_kernel, dim3(1), dim3(64), 0, stream,
+                     pos_out, token_out, pos_table, token_table,
+                     pos_table_mut, slot, adv_pos);
 }
 
 static inline int gemv_grid(int M, int rows_per_block) {
