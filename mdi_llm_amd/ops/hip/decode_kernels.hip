@@ -327,6 +327,87 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
   }
 }
 
+// Direct-x variant (NORM 0/1 only): no LDS staging, no barrier — the W
+// non-temporal stream starts at instruction one.  Each lane reads its x
+// (and RMS weight) chunks straight from L1/L2 (x is KB-sized and hot),
+// and because a wave's lanes partition the full K range, the RMSNorm
+// mean-square reduces per-wave in registers (wave_reduce) with the exact
+// post-dot scale trick — no cross-wave exchange at all.  Measured: the
+// staged kernel spent ~1-4 us/launch filling LDS before W streaming
+// ramped (proj/down ran at 4.4 TB/s in-graph vs 6.1 for the long swiglu
+// kernel); this variant removes that ramp.  LayerNorm (NORM==2) keeps the
+// staged two-pass form.
+template <int EPI, int NORM, int ROWS>
+__global__ void gemv_direct_kernel(bf16* __restrict__ out,
+                                   const bf16* __restrict__ W,
+                                   const bf16* __restrict__ x,
+                                   const bf16* __restrict__ bias,
+                                   const bf16* __restrict__ res,
+                                   const bf16* __restrict__ nw, float eps,
+                                   int M, int K) {
+  static_assert(NORM == 0 || NORM == 1, "direct gemv: no LayerNorm");
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * ROWS;
+  const int row0 = (blockIdx.x * (blockDim.x >> 6) + wave) * ROWS;
+
+  float nscale = 1.f;
+  bool have_scale = false;
+  for (int row = row0; row < M; row += rows_per_grid) {
+    const bf16* wrow[ROWS];
+    float acc[ROWS];
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      wrow[r] = W + (size_t)min(row + r, M - 1) * K;
+      acc[r] = 0.f;
+    }
+    float s2 = 0.f;
+    for (int i = lane * 8; i < K; i += 64 * 8) {
+      bf16x8 xv = load8(x + i);
+      float xm[8];
+      if (NORM == 1) {
+        bf16x8 gv = load8(nw + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float f = b2f(xv.v[j]);
+          s2 += f * f;
+          xm[j] = f * b2f(gv.v[j]);
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xm[j] = b2f(xv.v[j]);
+      }
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+        bf16x8 wv = load8_nt(wrow[r] + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[r] += b2f(wv.v[j]) * xm[j];
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) acc[r] = wave_reduce_sum(acc[r]);
+    if (NORM == 1 && !have_scale) {
+      // lanes of this wave partitioned all of K: full mean-square
+      nscale = rsqrtf(wave_reduce_sum(s2) / K + eps);
+      have_scale = true;  // grid-stride re-entry reuses it
+    }
+    if (lane == 0) {
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+        const int rw = row + r;
+        if (rw >= M) break;
+        float a = acc[r] * nscale;
+        if (bias != nullptr) a += b2f(bias[rw]);
+        if (EPI == 1 && res != nullptr) a += b2f(res[rw]);
+        if (EPI == 2) a = gelu_tanh(a);
+        if (EPI == 3) a = a / (1.f + expf(-a));
+        out[rw] = f2b(a);
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // FP8 (OCP e4m3) weight variants: W stored as fp8 with one fp32 scale per
 // output row (absmax/448 quantization).  Halves the decode weight stream;
@@ -1716,11 +1797,21 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
   const int smem = K * sizeof(bf16);
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
+  // NORM 0/1: the direct-x kernel (no staging barrier); 2: staged LDS form
 #define GEMV_CASE1(E, N, R)                                                 \
-  hipLaunchKernelGGL((gemv_kernel<E, N, R>), grid, block, smem, stream,     \
-                     (bf16*)out, (const bf16*)W, (const bf16*)x,            \
-                     (const bf16*)bias, (const bf16*)res,                   \
-                     (const bf16*)norm_w, (const bf16*)norm_b, eps, M, K)
+  do {                                                                      \
+    if (N != 2)                                                             \
+      hipLaunchKernelGGL((gemv_direct_kernel<E, N == 2 ? 0 : N, R>), grid,  \
+                         block, 0, stream, (bf16*)out, (const bf16*)W,      \
+                         (const bf16*)x, (const bf16*)bias,                 \
+                         (const bf16*)res, (const bf16*)norm_w, eps, M, K); \
+    else                                                                    \
+      hipLaunchKernelGGL((gemv_kernel<E, N, R>), grid, block, smem, stream, \
+                         (bf16*)out, (const bf16*)W, (const bf16*)x,        \
+                         (const bf16*)bias, (const bf16*)res,               \
+                         (const bf16*)norm_w, (const bf16*)norm_b, eps, M,  \
+                         K);                                                \
+  } while (0)
 #define GEMV_CASE(E, N)                                                     \
   do {                                                                      \
     if (rows == 4) GEMV_CASE1(E, N, 4);                                     \
