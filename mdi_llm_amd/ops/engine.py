@@ -248,7 +248,9 @@ class DecodeEngine:
             return int(os.environ[env])
         if K >= 16384:
             return 4
-        if M <= 16384 or M >= 65536:
+        if M >= 65536:
+            return 2  # lm_head: grid caps at 4096 blocks; 2-row ILP wins
+        if M <= 16384:
             return 1
         return 2
 

@@ -363,6 +363,8 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
       acc[r] = 0.f;
     }
     float s2 = 0.f;
+    // unroll 2: two independent W chunks in flight per lane
+#pragma unroll 2
     for (int i = lane * 8; i < K; i += 64 * 8) {
       bf16x8 xv = load8(x + i);
       float xm[8];
