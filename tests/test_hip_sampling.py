@@ -122,3 +122,37 @@ def test_small_topk(ops):
         # allow bf16 ties at the threshold
         kth = torch.topk(logits.float(), 2).values[-1]
         assert logits.float()[tok] >= kth
+
+
+def test_stage_slot_op(ops):
+    """stage_slot: pos/token staging + pos advance in one launch."""
+    pos_table = torch.tensor([5, 9, 2], device=DEV, dtype=torch.int32)
+    token_table = torch.tensor([11, 22, 33], device=DEV, dtype=torch.int32)
+    slot = torch.tensor([1], device=DEV, dtype=torch.int32)
+    pos = torch.zeros(1, device=DEV, dtype=torch.int32)
+    tok = torch.zeros(1, device=DEV, dtype=torch.int32)
+    ops.stage_slot(slot, pos_out=pos, token_out=tok, pos_table=pos_table,
+                   token_table=token_table)
+    assert int(pos) == 9 and int(tok) == 22
+    ops.stage_slot(slot, pos_table_mut=pos_table, adv_pos=1)
+    assert pos_table.tolist() == [5, 10, 2]
+
+
+def test_sample_unpack_bookkeeping(ops):
+    """The sampler's unpack launch writes token_table[slot] and advances
+    pos_table[slot] when asked (the in-graph step bookkeeping)."""
+    torch.manual_seed(7)
+    logits = torch.randn(1000, device=DEV).to(torch.bfloat16)
+    scratch = torch.zeros(520, device=DEV, dtype=torch.int32)
+    out = torch.zeros(1, device=DEV, dtype=torch.int32)
+    pos = torch.tensor([4], device=DEV, dtype=torch.int32)
+    slot = torch.tensor([2], device=DEV, dtype=torch.int32)
+    pos_table = torch.tensor([0, 0, 4, 0], device=DEV, dtype=torch.int32)
+    token_table = torch.zeros(4, device=DEV, dtype=torch.int32)
+    ops.sample(out, logits, scratch, 0.0, 0, False, 1,
+               pos=pos, slot=slot, token_table=token_table,
+               pos_table=pos_table, adv_slot=slot, adv_pos=1)
+    argmax = int(logits.float().argmax())
+    assert int(out) == argmax
+    assert token_table.tolist() == [0, 0, argmax, 0]
+    assert pos_table.tolist() == [0, 0, 5, 0]
