@@ -502,101 +502,6 @@ __global__ void gemv_fp8_kernel(bf16* __restrict__ out,
   }
 }
 
-// fp8 dot against an fp32 x fragment (direct-x variant)
-DEVINL float dot16_fp8_f(const unsigned* wq, const float* xm) {
-  float acc = 0.f;
-#pragma unroll
-  for (int q = 0; q < 4; ++q) {
-    const f32x2 lo = __builtin_amdgcn_cvt_pk_f32_fp8(wq[q], false);
-    const f32x2 hi = __builtin_amdgcn_cvt_pk_f32_fp8(wq[q], true);
-    const int b = q * 4;
-    acc += lo[0] * xm[b + 0] + lo[1] * xm[b + 1] + hi[0] * xm[b + 2] +
-           hi[1] * xm[b + 3];
-  }
-  return acc;
-}
-
-// Direct-x fp8 variant (NORM 0/1): same rationale as gemv_direct_kernel —
-// no staging barrier, the fp8 W non-temporal stream starts immediately,
-// RMS mean-square reduced per-wave in registers.
-template <int EPI, int NORM, int ROWS>
-__global__ void gemv_fp8_direct_kernel(bf16* __restrict__ out,
-                                       const unsigned char* __restrict__ W,
-                                       const float* __restrict__ wscale,
-                                       const bf16* __restrict__ x,
-                                       const bf16* __restrict__ bias,
-                                       const bf16* __restrict__ res,
-                                       const bf16* __restrict__ nw,
-                                       float eps, int M, int K) {
-  static_assert(NORM == 0 || NORM == 1, "direct fp8 gemv: no LayerNorm");
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int rows_per_grid = gridDim.x * (blockDim.x >> 6) * ROWS;
-  const int row0 = (blockIdx.x * (blockDim.x >> 6) + wave) * ROWS;
-
-  float nscale = 1.f;
-  bool have_scale = false;
-  for (int row = row0; row < M; row += rows_per_grid) {
-    const unsigned char* wrow[ROWS];
-    float acc[ROWS];
-#pragma unroll
-    for (int r = 0; r < ROWS; ++r) {
-      wrow[r] = W + (size_t)min(row + r, M - 1) * K;
-      acc[r] = 0.f;
-    }
-    float s2 = 0.f;
-#pragma unroll 2
-    for (int i = lane * 16; i < K; i += 64 * 16) {
-      bf16x8 xa = load8(x + i);
-      bf16x8 xb = load8(x + i + 8);
-      float xm[16];
-      if (NORM == 1) {
-        bf16x8 g1 = load8(nw + i);
-        bf16x8 g2 = load8(nw + i + 8);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float fa = b2f(xa.v[j]);
-          const float fb = b2f(xb.v[j]);
-          s2 += fa * fa + fb * fb;
-          xm[j] = fa * b2f(g1.v[j]);
-          xm[j + 8] = fb * b2f(g2.v[j]);
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          xm[j] = b2f(xa.v[j]);
-          xm[j + 8] = b2f(xb.v[j]);
-        }
-      }
-#pragma unroll
-      for (int r = 0; r < ROWS; ++r) {
-        const i32x4_t wq = load16_nt_u8(wrow[r] + i);
-        acc[r] += dot16_fp8_f(reinterpret_cast<const unsigned*>(&wq), xm);
-      }
-    }
-#pragma unroll
-    for (int r = 0; r < ROWS; ++r) acc[r] = wave_reduce_sum(acc[r]);
-    if (NORM == 1 && !have_scale) {
-      nscale = rsqrtf(wave_reduce_sum(s2) / K + eps);
-      have_scale = true;
-    }
-    if (lane == 0) {
-#pragma unroll
-      for (int r = 0; r < ROWS; ++r) {
-        const int rw = row + r;
-        if (rw >= M) break;
-        float a = acc[r] * nscale * wscale[rw];
-        if (bias != nullptr) a += b2f(bias[rw]);
-        if (EPI == 1 && res != nullptr) a += b2f(res[rw]);
-        if (EPI == 2) a = gelu_tanh(a);
-        if (EPI == 3) a = a / (1.f + expf(-a));
-        out[rw] = f2b(a);
-      }
-    }
-  }
-}
-
 template <int NORM>
 __global__ void gemv_swiglu_fp8_kernel(
     bf16* __restrict__ out, const unsigned char* __restrict__ Wg,
@@ -1943,20 +1848,10 @@ void launch_gemv_fp8(void* out, const void* W, const float* wscale,
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
 #define GEMV8_CASE1(E, N, R)                                                \
-  do {                                                                      \
-    if (N != 2)                                                             \
-      hipLaunchKernelGGL((gemv_fp8_direct_kernel<E, N == 2 ? 0 : N, R>),    \
-                         grid, block, 0, stream, (bf16*)out,                \
-                         (const unsigned char*)W, wscale, (const bf16*)x,   \
-                         (const bf16*)bias, (const bf16*)res,               \
-                         (const bf16*)norm_w, eps, M, K);                   \
-    else                                                                    \
-      hipLaunchKernelGGL((gemv_fp8_kernel<E, N, R>), grid, block, smem,     \
-                         stream, (bf16*)out, (const unsigned char*)W,       \
-                         wscale, (const bf16*)x, (const bf16*)bias,         \
-                         (const bf16*)res, (const bf16*)norm_w,             \
-                         (const bf16*)norm_b, eps, M, K);                   \
-  } while (0)
+  hipLaunchKernelGGL((gemv_fp8_kernel<E, N, R>), grid, block, smem, stream, \
+                     (bf16*)out, (const unsigned char*)W, wscale,           \
+                     (const bf16*)x, (const bf16*)bias, (const bf16*)res,   \
+                     (const bf16*)norm_w, (const bf16*)norm_b, eps, M, K)
 #define GEMV8_CASE(E, N)                                                    \
   do {                                                                      \
     if (rows == 4) GEMV8_CASE1(E, N, 4);                                    \
