@@ -277,6 +277,26 @@ configs.extend(
             mlp_class_name="GptNeoxMLP",
         ),
         dict(
+            # MoE variant big enough for the HIP engine (hs=64)
+            name="nano-moe-gpu",
+            block_size=256,
+            vocab_size=512,
+            padding_multiple=64,
+            n_layer=2,
+            n_head=4,
+            n_embd=256,
+            n_query_groups=4,
+            rotary_percentage=1.0,
+            parallel_residual=False,
+            bias=False,
+            norm_class_name="RMSNorm",
+            mlp_class_name="LLaMAMoE",
+            intermediate_size=224,
+            n_expert=4,
+            n_expert_per_token=2,
+            norm_eps=1e-5,
+        ),
+        dict(
             name="nano-test-moe",
             block_size=128,
             vocab_size=256,

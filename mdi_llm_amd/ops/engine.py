@@ -30,7 +30,10 @@ __all__ = ["DecodeEngine", "engine_supported"]
 
 
 def engine_supported(config: ModelConfig) -> bool:
-    if config.mlp_class_name == "LLaMAMoE":
+    if config.mlp_class_name == "LLaMAMoE" and (
+        config.n_expert > 64 or config.n_expert_per_token > 8
+        or config.norm_class_name != "RMSNorm"
+    ):
         return False
     if config.head_size % 32 != 0 or config.head_size * config.q_per_kv < 64:
         return False
@@ -72,7 +75,18 @@ class _BlockWeights:
         else:
             self.norm2_w = self.norm2_b = None
         mlp = block.mlp
-        if config.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
+        if config.mlp_class_name == "LLaMAMoE":
+            # stacked expert slabs: the router index (device memory)
+            # offsets into these at graph-replay time
+            self.gate_w = p(mlp.gate.weight)
+            self.fc1_w = torch.stack(
+                [p(e.fc_1.weight) for e in mlp.experts]).contiguous()
+            self.fc2_w = torch.stack(
+                [p(e.fc_2.weight) for e in mlp.experts]).contiguous()
+            self.mlp_proj_w = torch.stack(
+                [p(e.proj.weight) for e in mlp.experts]).contiguous()
+            self.mlp_proj_b = None
+        elif config.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
             self.fc1_w = p(mlp.fc_1.weight)
             self.fc2_w = p(mlp.fc_2.weight)
             self.mlp_proj_w = p(mlp.proj.weight)
@@ -85,6 +99,8 @@ class _BlockWeights:
 
     def quantize_fp8(self, config) -> None:
         """Attach fp8 copies + per-row scales for the decode GEMVs."""
+        if config.mlp_class_name == "LLaMAMoE":
+            raise ValueError("fp8 weights are not supported for MoE stages")
         self.attn_w8, self.attn_s = quantize_fp8_rowwise(self.attn_w)
         self.proj_w8, self.proj_s = quantize_fp8_rowwise(self.proj_w)
         self.mlp_proj_w8, self.mlp_proj_s = quantize_fp8_rowwise(
@@ -166,6 +182,12 @@ class DecodeEngine:
         self.a = torch.zeros(E, **bf)        # proj(attn) (+x)
         self.act = torch.zeros(I, **bf)
         self.m_out = torch.zeros(E, **bf)
+        if cfg.mlp_class_name == "LLaMAMoE":
+            self.gate_logits = torch.zeros(cfg.n_expert, **bf)
+            self.moe_eidx = torch.zeros(cfg.n_expert_per_token, device=dev,
+                                        dtype=torch.int32)
+            self.moe_escale = torch.zeros(cfg.n_expert_per_token, device=dev,
+                                          dtype=torch.float32)
         self.part_o = torch.zeros(
             n_head * n_chunks * hs, device=dev, dtype=torch.float32
         )
@@ -326,6 +348,30 @@ class DecodeEngine:
         cfg = self.config
         eps = cfg.norm_eps
         nk = self._nk
+        if cfg.mlp_class_name == "LLaMAMoE":
+            # router: gate logits -> (top-k experts, softmax weights),
+            # then k expert swiglu+down passes selected by DEVICE-side
+            # indices (the whole sequence is graph-capturable; reference
+            # model.py:823-853 semantics)
+            ops = self.ops
+            ops.gemv(self.gate_logits, w.gate_w, inp, None, None, 0,
+                     norm_w, norm_b, nk, eps, 1)
+            ops.moe_gate_topk(self.moe_eidx, self.moe_escale,
+                              self.gate_logits, cfg.n_expert_per_token)
+            E, I = cfg.n_embd, cfg.intermediate_size
+            out = self.m_out if cfg.parallel_residual else self.x
+            res_j = res
+            for j in range(cfg.n_expert_per_token):
+                ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp, False,
+                                norm_w, norm_b, nk, eps,
+                                eidx=self.moe_eidx[j:j + 1],
+                                estride=I * E,
+                                escale=self.moe_escale[j:j + 1])
+                ops.gemv(out, w.mlp_proj_w, self.act, None, res_j, 1,
+                         None, None, 0, eps, self._r_down,
+                         eidx=self.moe_eidx[j:j + 1], estride=E * I)
+                res_j = out
+            return
         if cfg.mlp_class_name in ("LLaMAMLP", "GemmaMLP"):
             gelu_gate = cfg.mlp_class_name == "GemmaMLP"
             if self.fp8:

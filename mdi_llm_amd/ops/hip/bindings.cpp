@@ -65,13 +65,22 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
           c10::optional<torch::Tensor> bias, c10::optional<torch::Tensor> res,
           int64_t epilogue, c10::optional<torch::Tensor> norm_w,
           c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
-          double eps, int64_t rows) {
+          double eps, int64_t rows, c10::optional<torch::Tensor> eidx,
+          int64_t estride) {
   check_bf16(out, "out");
   check_bf16(W, "W");
   check_bf16(x, "x");
   const int K = (int)x.numel();
   const int M = (int)out.numel();
-  TORCH_CHECK(W.numel() == (int64_t)M * K, "W shape mismatch");
+  const int* eip = nullptr;
+  if (eidx.has_value()) {
+    check_i32(*eidx, "eidx");
+    TORCH_CHECK(norm_kind != 2, "expert indirection: no LayerNorm");
+    TORCH_CHECK(W.numel() % ((int64_t)M * K) == 0, "stacked W shape");
+    eip = eidx->data_ptr<int>();
+  } else {
+    TORCH_CHECK(W.numel() == (int64_t)M * K, "W shape mismatch");
+  }
   TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
   const void* bp = nullptr;
   const void* rp = nullptr;
@@ -98,7 +107,7 @@ void gemv(torch::Tensor out, torch::Tensor W, torch::Tensor x,
   }
   launch_gemv(out.data_ptr(), W.data_ptr(), x.data_ptr(), bp, rp, nwp, nbp,
               (float)eps, M, K, (int)epilogue, (int)norm_kind, (int)rows,
-              cur_stream());
+              eip, (long long)estride, cur_stream());
 }
 
 void gemv_fp8(torch::Tensor out, torch::Tensor W, torch::Tensor wscale,
@@ -166,15 +175,30 @@ void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
                  torch::Tensor x, bool gelu_gate,
                  c10::optional<torch::Tensor> norm_w,
                  c10::optional<torch::Tensor> norm_b, int64_t norm_kind,
-                 double eps) {
+                 double eps, c10::optional<torch::Tensor> eidx,
+                 int64_t estride, c10::optional<torch::Tensor> escale) {
   check_bf16(out, "out");
   check_bf16(Wg, "Wg");
   check_bf16(Wu, "Wu");
   check_bf16(x, "x");
   const int K = (int)x.numel();
   const int M = (int)out.numel();
-  TORCH_CHECK(Wg.numel() == (int64_t)M * K && Wu.numel() == (int64_t)M * K,
-              "weight shape mismatch");
+  const int* eip = nullptr;
+  const float* esp = nullptr;
+  if (eidx.has_value()) {
+    check_i32(*eidx, "eidx");
+    TORCH_CHECK(Wg.numel() % ((int64_t)M * K) == 0 &&
+                    Wu.numel() % ((int64_t)M * K) == 0,
+                "stacked weight shape");
+    eip = eidx->data_ptr<int>();
+  } else {
+    TORCH_CHECK(Wg.numel() == (int64_t)M * K && Wu.numel() == (int64_t)M * K,
+                "weight shape mismatch");
+  }
+  if (escale.has_value()) {
+    check_f32(*escale, "escale");
+    esp = escale->data_ptr<float>();
+  }
   TORCH_CHECK(K % 8 == 0, "K must be a multiple of 8");
   const void* nwp = nullptr;
   const void* nbp = nullptr;
@@ -189,7 +213,21 @@ void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
   }
   launch_gemv_swiglu(out.data_ptr(), Wg.data_ptr(), Wu.data_ptr(),
                      x.data_ptr(), nwp, nbp, (float)eps, M, K,
-                     gelu_gate ? 1 : 0, (int)norm_kind, cur_stream());
+                     gelu_gate ? 1 : 0, (int)norm_kind, eip,
+                     (long long)estride, esp, cur_stream());
+}
+
+void moe_gate_topk(torch::Tensor eidx, torch::Tensor escale,
+                   torch::Tensor logits, int64_t k) {
+  check_i32(eidx, "eidx");
+  check_f32(escale, "escale");
+  check_bf16(logits, "logits");
+  const int n_e = (int)logits.numel();
+  TORCH_CHECK(n_e <= 64 && k <= 8 && k >= 1 && eidx.numel() >= k &&
+                  escale.numel() >= k,
+              "moe_gate_topk: n_e <= 64, 1 <= k <= 8");
+  launch_moe_gate_topk(eidx.data_ptr<int>(), escale.data_ptr<float>(),
+                       logits.data_ptr(), n_e, (int)k, cur_stream());
 }
 
 void embed(torch::Tensor out, torch::Tensor wte, torch::Tensor token,
@@ -412,7 +450,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("res"), py::arg("epilogue"),
         py::arg("norm_w") = c10::nullopt, py::arg("norm_b") = c10::nullopt,
         py::arg("norm_kind") = 0, py::arg("eps") = 1e-5,
-        py::arg("rows") = 0);
+        py::arg("rows") = 0, py::arg("eidx") = c10::nullopt,
+        py::arg("estride") = 0);
   m.def("gemv_fp8", &gemv_fp8,
         "decode GEMV with fp8(e4m3) weights + per-row scales",
         py::arg("out"), py::arg("W"), py::arg("wscale"), py::arg("x"),
@@ -430,7 +469,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("Wg"), py::arg("Wu"), py::arg("x"),
         py::arg("gelu_gate"), py::arg("norm_w") = c10::nullopt,
         py::arg("norm_b") = c10::nullopt, py::arg("norm_kind") = 0,
-        py::arg("eps") = 1e-5);
+        py::arg("eps") = 1e-5, py::arg("eidx") = c10::nullopt,
+        py::arg("estride") = 0, py::arg("escale") = c10::nullopt);
+  m.def("moe_gate_topk", &moe_gate_topk,
+        "MoE router: top-k experts + softmax weights over the k",
+        py::arg("eidx"), py::arg("escale"), py::arg("logits"), py::arg("k"));
   m.def("embed", &embed, "embedding row gather");
   m.def("rope_kv_append", &rope_kv_append,
         "RoPE on interleaved qkv + KV cache append");

@@ -15,10 +15,14 @@ void launch_layernorm(void* out, const void* x, const void* w, const void* b,
 // epilogue: 0 none, 1 +res, 2 gelu(tanh), 3 silu
 // norm_kind: 0 none, 1 fused RMSNorm on x, 2 fused LayerNorm on x
 // rows: output rows per wave (1/2/4); 0 = auto by M
+// eidx/estride (optional): W += eidx[0]*estride at replay time — selects
+// an expert slab from a stacked [n_expert, M, K] weight (MoE routing
+// stays graph-capturable); norm_kind 2 does not support indirection
 void launch_gemv(void* out, const void* W, const void* x, const void* bias,
                  const void* res, const void* norm_w, const void* norm_b,
                  float eps, int M, int K, int epilogue, int norm_kind,
-                 int rows, hipStream_t stream);
+                 int rows, const int* eidx, long long estride,
+                 hipStream_t stream);
 
 // fp8(e4m3)-weight variants: W is bytes, wscale fp32 per output row
 void launch_gemv_fp8(void* out, const void* W, const float* wscale,
@@ -37,7 +41,13 @@ void launch_gemv_swiglu_fp8(void* out, const void* Wg, const float* gscale,
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const void* x, const void* norm_w, const void* norm_b,
                         float eps, int M, int K, int gelu_gate, int norm_kind,
-                        hipStream_t stream);
+                        const int* eidx, long long estride,
+                        const float* escale, hipStream_t stream);
+
+// MoE router: eidx/escale[k] = top-k experts of gate logits + softmax
+// weights over the selected k (<= 8 of <= 64 experts)
+void launch_moe_gate_topk(int* eidx, float* escale, const void* logits,
+                          int n_e, int k, hipStream_t stream);
 
 void launch_embed(void* out, const void* wte, const int* token, int n_embd,
                   float scale, hipStream_t stream);

@@ -344,8 +344,11 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
                                    const bf16* __restrict__ bias,
                                    const bf16* __restrict__ res,
                                    const bf16* __restrict__ nw, float eps,
-                                   int M, int K) {
+                                   int M, int K,
+                                   const int* __restrict__ eidx,
+                                   long long estride) {
   static_assert(NORM == 0 || NORM == 1, "direct gemv: no LayerNorm");
+  if (eidx != nullptr) W += (size_t)eidx[0] * (size_t)estride;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -405,6 +408,55 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
         if (EPI == 2) a = gelu_tanh(a);
         if (EPI == 3) a = a / (1.f + expf(-a));
         out[rw] = f2b(a);
+      }
+    }
+  }
+}
+
+// MoE router: top-k of the gate logits (<= 64 experts) + softmax over
+// the selected k (reference model.py:823-853 semantics: topk first, then
+// softmax over the k logits).  One wave; outputs device-side so a
+// captured graph replays the routing data-dependently.
+__global__ void moe_gate_topk_kernel(int* __restrict__ eidx,
+                                     float* __restrict__ escale,
+                                     const bf16* __restrict__ logits,
+                                     int n_e, int k) {
+  const int lane = threadIdx.x;
+  float cur = (lane < n_e) ? b2f(logits[lane]) : -1e30f;
+  float sel[8];
+  int seli[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    if (j >= k) continue;
+    float m = cur;
+    int mi = lane;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const float om = __shfl_xor(m, off, 64);
+      const int oi = __shfl_xor(mi, off, 64);
+      if (om > m || (om == m && oi < mi)) {
+        m = om;
+        mi = oi;
+      }
+    }
+    sel[j] = m;
+    seli[j] = mi;
+    if (lane == mi) cur = -1e30f;
+  }
+  if (lane == 0) {
+    const float mx = sel[0];
+    float ssum = 0.f;
+    float e[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      e[j] = (j < k) ? __expf(sel[j] - mx) : 0.f;
+      ssum += e[j];
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (j < k) {
+        eidx[j] = seli[j];
+        escale[j] = e[j] / ssum;
       }
     }
   }
@@ -543,7 +595,10 @@ __global__ void gemv_swiglu_fp8_kernel(
 }
 
 // SwiGLU pair GEMV: out[i] = silu(Wg_i . xn) * (Wu_i . xn), optional fused
-// pre-norm like gemv_kernel.
+// pre-norm like gemv_kernel.  eidx/estride select an expert weight slab
+// from stacked [n_expert, M, K] tensors AT LAUNCH-REPLAY TIME (the index
+// lives in device memory, so MoE routing stays inside a captured graph);
+// escale multiplies the output by the routing weight.
 template <int NORM>
 __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
                                    const bf16* __restrict__ Wg,
@@ -551,10 +606,19 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
                                    const bf16* __restrict__ x,
                                    const bf16* __restrict__ nw,
                                    const bf16* __restrict__ nb, float eps,
-                                   int M, int K, int gelu_gate) {
+                                   int M, int K, int gelu_gate,
+                                   const int* __restrict__ eidx,
+                                   long long estride,
+                                   const float* __restrict__ escale) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   __shared__ float red[8];
   bf16* xs = reinterpret_cast<bf16*>(smem);
+  if (eidx != nullptr) {
+    const size_t off = (size_t)eidx[0] * (size_t)estride;
+    Wg += off;
+    Wu += off;
+  }
+  const float oscale = (escale != nullptr) ? escale[0] : 1.f;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -1795,7 +1859,8 @@ void launch_layernorm(void* out, const void* x, const void* w, const void* b,
 void launch_gemv(void* out, const void* W, const void* x, const void* bias,
                  const void* res, const void* norm_w, const void* norm_b,
                  float eps, int M, int K, int epilogue, int norm_kind,
-                 int rows, hipStream_t stream) {
+                 int rows, const int* eidx, long long estride,
+                 hipStream_t stream) {
   const int smem = K * sizeof(bf16);
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
@@ -1806,7 +1871,8 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
       hipLaunchKernelGGL((gemv_direct_kernel<E, N == 2 ? 0 : N, R>), grid,  \
                          block, 0, stream, (bf16*)out, (const bf16*)W,      \
                          (const bf16*)x, (const bf16*)bias,                 \
-                         (const bf16*)res, (const bf16*)norm_w, eps, M, K); \
+                         (const bf16*)res, (const bf16*)norm_w, eps, M, K,  \
+                         eidx, estride);                                    \
     else                                                                    \
       hipLaunchKernelGGL((gemv_kernel<E, N, R>), grid, block, smem, stream, \
                          (bf16*)out, (const bf16*)W, (const bf16*)x,        \
@@ -1902,20 +1968,28 @@ void launch_gemv_swiglu_fp8(void* out, const void* Wg, const float* gscale,
 void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const void* x, const void* norm_w, const void* norm_b,
                         float eps, int M, int K, int gelu_gate, int norm_kind,
-                        hipStream_t stream) {
+                        const int* eidx, long long estride,
+                        const float* escale, hipStream_t stream) {
   const int smem = K * sizeof(bf16);
   dim3 grid(gemv_grid(M, 4)), block(256);
 #define SW_CASE(N)                                                          \
   hipLaunchKernelGGL((gemv_swiglu_kernel<N>), grid, block, smem, stream,    \
                      (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,          \
                      (const bf16*)x, (const bf16*)norm_w,                   \
-                     (const bf16*)norm_b, eps, M, K, gelu_gate)
+                     (const bf16*)norm_b, eps, M, K, gelu_gate, eidx,       \
+                     estride, escale)
   switch (norm_kind) {
     case 1: SW_CASE(1); break;
     case 2: SW_CASE(2); break;
     default: SW_CASE(0);
   }
 #undef SW_CASE
+}
+
+void launch_moe_gate_topk(int* eidx, float* escale, const void* logits,
+                          int n_e, int k, hipStream_t stream) {
+  hipLaunchKernelGGL(moe_gate_topk_kernel, dim3(1), dim3(64), 0, stream,
+                     eidx, escale, (const bf16*)logits, n_e, k);
 }
 
 void launch_embed(void* out, const void* wte, const int* token, int n_embd,
