@@ -12,7 +12,7 @@ into the modules instead — gptserver.py:975-978).
 Supported natively: RMSNorm/LayerNorm, rope/learned positions, GQA
 attention (head_size multiple of 32, head_size*q_per_kv >= 64), LLaMA
 (SwiGLU) / Gemma / GPT-NeoX MLPs, sequential and parallel residual.
-MoE stages fall back to the torch path.
+MoE decodes natively (device-side routing); fp8+MoE is refused.
 """
 
 from __future__ import annotations

@@ -324,16 +324,11 @@ def test_fused_generate_reproducible_and_stops():
 def test_moe_engine_matches_torch():
     """HIP MoE decode (device-side routing, stacked expert slabs) against
     the torch LLaMAMoE module, greedy token-exact over several steps."""
+    from mdi_llm_amd.models.stages import StarterStage
     from mdi_llm_amd.ops.engine import DecodeEngine, engine_supported
 
-    cfg = ModelConfig.from_name("nano-moe-gpu")
+    cfg, m = _build("nano-moe-gpu", seed=17)
     assert engine_supported(cfg)
-    torch.manual_seed(17)
-    m = GPT(cfg)
-    m.apply_init()
-    m = m.to(device=DEV, dtype=torch.bfloat16)
-    m.eval()
-    from mdi_llm_amd.models.stages import StarterStage
 
     stage = StarterStage(cfg, cfg.n_layer).to(device=DEV,
                                               dtype=torch.bfloat16)
