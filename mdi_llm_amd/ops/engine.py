@@ -183,6 +183,7 @@ class DecodeEngine:
         self.act = torch.zeros(I, **bf)
         self.m_out = torch.zeros(E, **bf)
         if cfg.mlp_class_name == "LLaMAMoE":
+            self.xn = torch.zeros(E, **bf)
             self.gate_logits = torch.zeros(cfg.n_expert, **bf)
             self.moe_eidx = torch.zeros(cfg.n_expert_per_token, device=dev,
                                         dtype=torch.int32)
@@ -352,18 +353,22 @@ class DecodeEngine:
             # router: gate logits -> (top-k experts, softmax weights),
             # then k expert swiglu+down passes selected by DEVICE-side
             # indices (the whole sequence is graph-capturable; reference
-            # model.py:823-853 semantics)
+            # model.py:823-853 semantics).  The norm runs ONCE as its own
+            # kernel — the bf16-rounded x_norm must feed the router
+            # exactly as in the torch reference, or near-tied experts
+            # get routed differently.
             ops = self.ops
-            ops.gemv(self.gate_logits, w.gate_w, inp, None, None, 0,
-                     norm_w, norm_b, nk, eps, 1)
+            ops.rmsnorm(self.xn, inp, norm_w, eps)
+            ops.gemv(self.gate_logits, w.gate_w, self.xn, None, None, 0,
+                     None, None, 0, eps, 1)
             ops.moe_gate_topk(self.moe_eidx, self.moe_escale,
                               self.gate_logits, cfg.n_expert_per_token)
             E, I = cfg.n_embd, cfg.intermediate_size
             out = self.m_out if cfg.parallel_residual else self.x
             res_j = res
             for j in range(cfg.n_expert_per_token):
-                ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, inp, False,
-                                norm_w, norm_b, nk, eps,
+                ops.gemv_swiglu(self.act, w.fc1_w, w.fc2_w, self.xn, False,
+                                None, None, 0, eps,
                                 eidx=self.moe_eidx[j:j + 1],
                                 estride=I * E,
                                 escale=self.moe_escale[j:j + 1])
