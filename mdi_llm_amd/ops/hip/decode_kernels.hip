@@ -667,7 +667,7 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
     ua = wave_reduce_sum(ua) * nscale;
     if (lane == 0) {
       float act = gelu_gate ? gelu_tanh(ga) : ga / (1.f + expf(-ga));
-      out[row] = f2b(act * ua);
+      out[row] = f2b(act * ua * oscale);
     }
   }
 }
