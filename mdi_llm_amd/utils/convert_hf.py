@@ -63,6 +63,15 @@ LLAMA_MAP = {
     "model.layers.{}.mlp.gate_proj.weight": "transformer.h.{}.mlp.fc_1.weight",
     "model.layers.{}.mlp.up_proj.weight": "transformer.h.{}.mlp.fc_2.weight",
     "model.layers.{}.mlp.down_proj.weight": "transformer.h.{}.mlp.proj.weight",
+    # MoE (Mixtral) — reference convert_hf_checkpoint.py:139-142
+    "model.layers.{}.block_sparse_moe.gate.weight":
+        "transformer.h.{}.mlp.gate.weight",
+    "model.layers.{}.block_sparse_moe.experts.{}.w1.weight":
+        "transformer.h.{}.mlp.experts.{}.fc_1.weight",
+    "model.layers.{}.block_sparse_moe.experts.{}.w3.weight":
+        "transformer.h.{}.mlp.experts.{}.fc_2.weight",
+    "model.layers.{}.block_sparse_moe.experts.{}.w2.weight":
+        "transformer.h.{}.mlp.experts.{}.proj.weight",
     "model.norm.weight": "transformer.ln_f.weight",
     "lm_head.weight": "lm_head.weight",
 }
@@ -167,6 +176,12 @@ def _map_key(template_map: Dict[str, str], key: str):
     if key in template_map:
         return template_map[key]
     parts = key.split(".")
+    digits = [p for p in parts if p.isdigit()]
+    if digits:
+        # all-digit template (handles multi-index keys like MoE experts)
+        templ = ".".join("{}" if p.isdigit() else p for p in parts)
+        if templ in template_map:
+            return template_map[templ].format(*digits)
     for i, p in enumerate(parts):
         if p.isdigit():
             templ = ".".join(parts[:i] + ["{}"] + parts[i + 1:])

@@ -230,3 +230,44 @@ def test_falcon_registry_matches_reference():
     cfg40 = ModelConfig.from_name("falcon-40b-instruct")
     assert cfg40.n_query_groups == 8 and not cfg40.shared_attention_norm
     assert ModelConfig.from_name("falcon-180B-chat").n_layer == 80
+
+
+def test_convert_mixtral_moe_synthetic(tmp_path):
+    """Synthetic Mixtral-layout (block_sparse_moe) dict converts and loads
+    into the local MoE model (reference convert_hf_checkpoint.py:139-142)."""
+    from mdi_llm_amd.utils.convert_hf import convert_hf_checkpoint
+
+    torch.manual_seed(9)
+    cfg = ModelConfig.from_name("nano-test-moe")
+    E, I, L = cfg.n_embd, cfg.intermediate_size, cfg.n_layer
+    hf = {"model.embed_tokens.weight": torch.randn(cfg.vocab_size, E),
+          "model.norm.weight": torch.randn(E),
+          "lm_head.weight": torch.randn(cfg.vocab_size, E)}
+    for l in range(L):
+        p = f"model.layers.{l}"
+        hf[f"{p}.input_layernorm.weight"] = torch.randn(E)
+        hf[f"{p}.post_attention_layernorm.weight"] = torch.randn(E)
+        for w in ("q_proj", "k_proj", "v_proj"):
+            hf[f"{p}.self_attn.{w}.weight"] = torch.randn(E, E)
+        hf[f"{p}.self_attn.o_proj.weight"] = torch.randn(E, E)
+        hf[f"{p}.block_sparse_moe.gate.weight"] = \
+            torch.randn(cfg.n_expert, E)
+        for e in range(cfg.n_expert):
+            q = f"{p}.block_sparse_moe.experts.{e}"
+            hf[f"{q}.w1.weight"] = torch.randn(I, E)
+            hf[f"{q}.w3.weight"] = torch.randn(I, E)
+            hf[f"{q}.w2.weight"] = torch.randn(E, I)
+    src = tmp_path / "mixtral"
+    src.mkdir()
+    torch.save(hf, src / "pytorch_model.bin")
+    out = tmp_path / "mixtral_lit"
+    convert_hf_checkpoint(src, out, model_name="nano-test-moe")
+    config, sd = load_from_pt(out)
+    assert torch.equal(sd["transformer.h.0.mlp.experts.2.fc_1.weight"],
+                       hf["model.layers.0.block_sparse_moe.experts.2.w1.weight"])
+    m = GPT(config)
+    m.load_state_dict(sd)
+    m.eval()
+    with torch.inference_mode():
+        logits = m(torch.randint(0, 255, (1, 8)))
+    assert torch.isfinite(logits).all()
