@@ -352,3 +352,40 @@ def test_moe_engine_matches_torch():
         assert int(logits.float().argmax()) == int(
             ref[0, -1].float().argmax()), i
         tok = ref[0, -1].float().argmax()
+
+
+@torch.inference_mode()
+def test_engine_long_context_split_s():
+    """max_seq > 4096 selects the global split-S + combine attention path
+    (vs the block-local kernel); decode must still match torch."""
+    from mdi_llm_amd import GPT, ModelConfig
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine
+
+    torch.manual_seed(31)
+    cfg = ModelConfig.from_name("nano-gpu", block_size=8192)
+    m = GPT(cfg)
+    m.apply_init()
+    m = m.to(device=DEV, dtype=torch.bfloat16)
+    m.eval()
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV,
+                                              dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.eval()
+    stage.set_kv_cache(1)
+    m.set_kv_cache(1)
+    assert stage.kv_pool.max_seq == 8192
+
+    eng = DecodeEngine(stage, stage.kv_pool, use_graphs=False)
+    prompt = torch.randint(0, 511, (40,), device=DEV)
+    ref = m(prompt.view(1, -1), input_pos=0, slot=0)
+    stage.forward_head(prompt.view(1, -1), slot=0, input_pos=0)
+    eng.set_slot_pos(0, 40)
+    tok = ref[0, -1].float().argmax()
+    for i in range(5):
+        ref = m(tok.view(1, 1), input_pos=40 + i, slot=0)
+        x = eng.decode_step_head(tok.to(torch.int32), slot=0)
+        logits = eng.tail(x)
+        assert int(logits.float().argmax()) == int(
+            ref[0, -1].float().argmax()), i
+        tok = ref[0, -1].float().argmax()
