@@ -672,78 +672,6 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
   }
 }
 
-// Direct-x SwiGLU variant (NORM 0/1): no LDS staging barrier — the two W
-// non-temporal streams start immediately; x (and the RMS weight) read
-// from L1 in the dot loop, mean-square reduced per-wave (same rationale
-// and measured win as gemv_direct_kernel).
-template <int NORM>
-__global__ void gemv_swiglu_direct_kernel(bf16* __restrict__ out,
-                                          const bf16* __restrict__ Wg,
-                                          const bf16* __restrict__ Wu,
-                                          const bf16* __restrict__ x,
-                                          const bf16* __restrict__ nw,
-                                          float eps, int M, int K,
-                                          int gelu_gate,
-                                          const int* __restrict__ eidx,
-                                          long long estride,
-                                          const float* __restrict__ escale) {
-  static_assert(NORM == 0 || NORM == 1, "direct swiglu: no LayerNorm");
-  if (eidx != nullptr) {
-    const size_t off = (size_t)eidx[0] * (size_t)estride;
-    Wg += off;
-    Wu += off;
-  }
-  const float oscale = (escale != nullptr) ? escale[0] : 1.f;
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int waves_per_grid = gridDim.x * (blockDim.x >> 6);
-  const int row0 = blockIdx.x * (blockDim.x >> 6) + wave;
-
-  float nscale = 1.f;
-  bool have_scale = false;
-  for (int row = row0; row < M; row += waves_per_grid) {
-    const bf16* grow = Wg + (size_t)row * K;
-    const bf16* urow = Wu + (size_t)row * K;
-    float ga = 0.f, ua = 0.f, s2 = 0.f;
-#pragma unroll 2
-    for (int i = lane * 8; i < K; i += 64 * 8) {
-      bf16x8 xv = load8(x + i);
-      float xm[8];
-      if (NORM == 1) {
-        bf16x8 gv = load8(nw + i);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float f = b2f(xv.v[j]);
-          s2 += f * f;
-          xm[j] = f * b2f(gv.v[j]);
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) xm[j] = b2f(xv.v[j]);
-      }
-      bf16x8 wg = load8_nt(grow + i);
-      bf16x8 wu = load8_nt(urow + i);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        ga += b2f(wg.v[j]) * xm[j];
-        ua += b2f(wu.v[j]) * xm[j];
-      }
-    }
-    ga = wave_reduce_sum(ga);
-    ua = wave_reduce_sum(ua);
-    if (NORM == 1 && !have_scale) {
-      nscale = rsqrtf(wave_reduce_sum(s2) / K + eps);
-      have_scale = true;
-    }
-    ga *= nscale;
-    ua *= nscale;
-    if (lane == 0) {
-      float act = gelu_gate ? gelu_tanh(ga) : ga / (1.f + expf(-ga));
-      out[row] = f2b(act * ua * oscale);
-    }
-  }
-}
-
 // ---------------------------------------------------------------------------
 // Embedding row gather: out[n_embd] = wte[token] (* scale)
 // token id read from device memory (graph-replayable).
@@ -2044,24 +1972,18 @@ void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const float* escale, hipStream_t stream) {
   const int smem = K * sizeof(bf16);
   dim3 grid(gemv_grid(M, 4)), block(256);
-  // NORM 0/1: direct-x form (no staging barrier); 2: staged LDS form
-#define SWD_CASE(N)                                                         \
-  hipLaunchKernelGGL((gemv_swiglu_direct_kernel<N>), grid, block, 0,        \
-                     stream, (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,  \
-                     (const bf16*)x, (const bf16*)norm_w, eps, M, K,        \
-                     gelu_gate, eidx, estride, escale)
+#define SW_CASE(N)                                                          \
+  hipLaunchKernelGGL((gemv_swiglu_kernel<N>), grid, block, smem, stream,    \
+                     (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,          \
+                     (const bf16*)x, (const bf16*)norm_w,                   \
+                     (const bf16*)norm_b, eps, M, K, gelu_gate, eidx,       \
+                     estride, escale)
   switch (norm_kind) {
-    case 1: SWD_CASE(1); break;
-    case 2:
-      hipLaunchKernelGGL((gemv_swiglu_kernel<2>), grid, block, smem,
-                         stream, (bf16*)out, (const bf16*)Wg,
-                         (const bf16*)Wu, (const bf16*)x,
-                         (const bf16*)norm_w, (const bf16*)norm_b, eps, M,
-                         K, gelu_gate, eidx, estride, escale);
-      break;
-    default: SWD_CASE(0);
+    case 1: SW_CASE(1); break;
+    case 2: SW_CASE(2); break;
+    default: SW_CASE(0);
   }
-#undef SWD_CASE
+#undef SW_CASE
 }
 
 void launch_moe_gate_topk(int* eidx, float* escale, const void* logits,
