@@ -185,8 +185,9 @@ class GroupDecodeEngine:
             else:
                 gate = F.linear(hn, w.fc1_w)
                 up = F.linear(hn, w.fc2_w)
-            act = (F.gelu(gate, approximate="tanh") if gelu_gate
-                   else F.silu(gate)) * up
+            # one fused HIP launch instead of the silu/gelu + mul pair
+            self.ops.swiglu_mul(up, gate, up, gelu_gate)
+            act = up
             if fp8:
                 X = a + self._mm(act, w.mlp_proj_w8, w.mlp_proj_s,
                                  w.mlp_proj_w)

@@ -217,6 +217,18 @@ void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
                      (long long)estride, esp, cur_stream());
 }
 
+void swiglu_mul(torch::Tensor out, torch::Tensor g, torch::Tensor u,
+                bool gelu_gate) {
+  check_bf16(out, "out");
+  check_bf16(g, "g");
+  check_bf16(u, "u");
+  TORCH_CHECK(out.numel() == g.numel() && g.numel() == u.numel() &&
+                  g.numel() % 8 == 0,
+              "swiglu_mul: equal sizes, multiple of 8");
+  launch_swiglu_mul(out.data_ptr(), g.data_ptr(), u.data_ptr(),
+                    (long long)g.numel(), gelu_gate ? 1 : 0, cur_stream());
+}
+
 void moe_gate_topk(torch::Tensor eidx, torch::Tensor escale,
                    torch::Tensor logits, int64_t k) {
   check_i32(eidx, "eidx");
@@ -471,6 +483,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("norm_b") = c10::nullopt, py::arg("norm_kind") = 0,
         py::arg("eps") = 1e-5, py::arg("eidx") = c10::nullopt,
         py::arg("estride") = 0, py::arg("escale") = c10::nullopt);
+  m.def("swiglu_mul", &swiglu_mul,
+        "batched act(gate)*up elementwise (out may alias up)",
+        py::arg("out"), py::arg("g"), py::arg("u"), py::arg("gelu_gate"));
   m.def("moe_gate_topk", &moe_gate_topk,
         "MoE router: top-k experts + softmax weights over the k",
         py::arg("eidx"), py::arg("escale"), py::arg("logits"), py::arg("k"));

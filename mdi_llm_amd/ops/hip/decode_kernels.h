@@ -44,6 +44,10 @@ void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const int* eidx, long long estride,
                         const float* escale, hipStream_t stream);
 
+// batched elementwise act(g)*u (out may alias u); n multiple of 8
+void launch_swiglu_mul(void* out, const void* g, const void* u,
+                       long long n, int gelu_gate, hipStream_t stream);
+
 // MoE router: eidx/escale[k] = top-k experts of gate logits + softmax
 // weights over the selected k (<= 8 of <= 64 experts)
 void launch_moe_gate_topk(int* eidx, float* escale, const void* logits,

@@ -672,6 +672,30 @@ __global__ void gemv_swiglu_kernel(bf16* __restrict__ out,
   }
 }
 
+// Batched SwiGLU glue for the grouped engine: out = act(g) * u over n
+// elements (replaces a silu + mul torch pair = two launches per layer).
+// out may alias u (pure elementwise).
+__global__ void swiglu_mul_kernel(bf16* __restrict__ out,
+                                  const bf16* __restrict__ g,
+                                  const bf16* __restrict__ u,
+                                  long long n, int gelu_gate) {
+  const long long i0 = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  const long long stride = (long long)gridDim.x * blockDim.x * 8;
+  for (long long i = i0; i < n; i += stride) {
+    bf16x8 gv = load8(g + i);
+    bf16x8 uv = load8(u + i);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = b2f(gv.v[j]);
+      const float act = gelu_gate ? gelu_tanh(gf)
+                                  : gf / (1.f + __expf(-gf));
+      o.v[j] = f2b(act * b2f(uv.v[j]));
+    }
+    *reinterpret_cast<int4*>(out + i) = *reinterpret_cast<const int4*>(o.v);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // Embedding row gather: out[n_embd] = wte[token] (* scale)
 // token id read from device memory (graph-replayable).
@@ -1984,6 +2008,16 @@ void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
     default: SW_CASE(0);
   }
 #undef SW_CASE
+}
+
+void launch_swiglu_mul(void* out, const void* g, const void* u,
+                       long long n, int gelu_gate, hipStream_t stream) {
+  long long blocks = (n / 8 + 255) / 256;
+  if (blocks > 4096) blocks = 4096;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(swiglu_mul_kernel, dim3((unsigned)blocks), dim3(256),
+                     0, stream, (bf16*)out, (const bf16*)g, (const bf16*)u,
+                     n, gelu_gate);
 }
 
 void launch_moe_gate_topk(int* eidx, float* escale, const void* logits,
