@@ -79,7 +79,7 @@ class GroupDecodeEngine:
 
         self.ops = require_hip_ops()
         cfg: ModelConfig = stage.config
-        _maybe_enable_tunableop(B)
+        _maybe_enable_tunableop(group_size)
         if not group_engine_supported(cfg):
             raise ValueError(f"{cfg.name!r} unsupported by GroupDecodeEngine")
         # fp8 grouped GEMMs measured SLOWER than bf16 hipBLASLt on ROCm 7
