@@ -39,6 +39,9 @@ def engine_supported(config: ModelConfig) -> bool:
         return False
     if config.head_size not in (64, 128, 256):
         return False
+    if config.q_per_kv not in (1, 2, 4, 8, 16):
+        return False  # attention kernel instantiations (e.g. falcon-7b's
+        # 71 q-heads per kv head run on the torch path)
     if config.n_embd % 8 != 0 or (config.intermediate_size or 0) % 8 != 0:
         return False
     return True
