@@ -362,3 +362,21 @@ def test_attn_decode(ops, qpk, n_kv, hs, S, ne, max_seq, n_chunks):
     got = out.view(n_head, hs).float()
     assert torch.allclose(got, ref, atol=3e-2, rtol=3e-2), \
         (got - ref).abs().max()
+
+
+@pytest.mark.parametrize("gelu", [False, True])
+def test_swiglu_mul(ops, gelu):
+    """Batched act(gate)*up elementwise (grouped-engine glue) vs torch."""
+    torch.manual_seed(33)
+    g = torch.randn(64, 224, device=DEV).to(torch.bfloat16)
+    u = torch.randn(64, 224, device=DEV).to(torch.bfloat16)
+    out = torch.empty_like(u)
+    ops.swiglu_mul(out, g, u, gelu)
+    act = (torch.nn.functional.gelu(g.float(), approximate="tanh") if gelu
+           else torch.nn.functional.silu(g.float()))
+    ref = act * u.float()
+    assert torch.allclose(out.float(), ref, atol=2e-2, rtol=2e-2)
+    # aliasing: out may be u
+    u2 = u.clone()
+    ops.swiglu_mul(u2, g, u2, gelu)
+    assert torch.equal(u2, out)
