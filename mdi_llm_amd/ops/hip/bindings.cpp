@@ -217,24 +217,6 @@ void gemv_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
                      (long long)estride, esp, cur_stream());
 }
 
-void gemm_swiglu(torch::Tensor out, torch::Tensor Wg, torch::Tensor Wu,
-                 torch::Tensor xn, bool gelu_gate) {
-  check_bf16(out, "out");
-  check_bf16(Wg, "Wg");
-  check_bf16(Wu, "Wu");
-  check_bf16(xn, "xn");
-  const int B = (int)xn.size(0);
-  const int K = (int)xn.size(1);
-  const int I = (int)(out.numel() / B);
-  TORCH_CHECK(Wg.numel() == (int64_t)I * K && Wu.numel() == (int64_t)I * K,
-              "weight shape mismatch");
-  int rc = launch_gemm_swiglu(out.data_ptr(), Wg.data_ptr(), Wu.data_ptr(),
-                              xn.data_ptr(), B, I, K, gelu_gate ? 1 : 0,
-                              cur_stream());
-  TORCH_CHECK(rc == 0, "gemm_swiglu: unsupported shape B=", B, " I=", I,
-              " K=", K);
-}
-
 void swiglu_mul(torch::Tensor out, torch::Tensor g, torch::Tensor u,
                 bool gelu_gate) {
   check_bf16(out, "out");
@@ -501,10 +483,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("norm_b") = c10::nullopt, py::arg("norm_kind") = 0,
         py::arg("eps") = 1e-5, py::arg("eidx") = c10::nullopt,
         py::arg("estride") = 0, py::arg("escale") = c10::nullopt);
-  m.def("gemm_swiglu", &gemm_swiglu,
-        "grouped fused SwiGLU GEMM out[B,I] = act(X Wg^T)*(X Wu^T)",
-        py::arg("out"), py::arg("Wg"), py::arg("Wu"), py::arg("xn"),
-        py::arg("gelu_gate"));
   m.def("swiglu_mul", &swiglu_mul,
         "batched act(gate)*up elementwise (out may alias up)",
         py::arg("out"), py::arg("g"), py::arg("u"), py::arg("gelu_gate"));

@@ -44,12 +44,6 @@ void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
                         const int* eidx, long long estride,
                         const float* escale, hipStream_t stream);
 
-// grouped fused SwiGLU GEMM: out[B,I] = act(X Wg^T) * (X Wu^T);
-// B in {16,32,64,96,128}; returns -1 if the shape is unsupported
-int launch_gemm_swiglu(void* out, const void* Wg, const void* Wu,
-                       const void* xn, int Bsz, int I, int K, int gelu_gate,
-                       hipStream_t stream);
-
 // batched elementwise act(g)*u (out may alias u); n multiple of 8
 void launch_swiglu_mul(void* out, const void* g, const void* u,
                        long long n, int gelu_gate, hipStream_t stream);

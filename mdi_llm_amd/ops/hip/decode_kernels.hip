@@ -161,9 +161,6 @@ __global__ void layernorm_kernel(bf16* __restrict__ out,
 // NORM: 0 none, 1 RMSNorm, 2 LayerNorm.
 // EPI:  0 none, 1 +residual, 2 gelu(tanh), 3 silu.
 // ---------------------------------------------------------------------------
-using f32x4 = __attribute__((__vector_size__(16))) float;
-using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
-
 DEVINL float gelu_tanh(float v) {
   float c = 0.7978845608028654f * (v + 0.044715f * v * v * v);
   return 0.5f * v * (1.f + tanhf(c));
@@ -700,82 +697,6 @@ __global__ void swiglu_mul_kernel(bf16* __restrict__ out,
 }
 
 // ---------------------------------------------------------------------------
-// Grouped-decode fused SwiGLU GEMM: out[B, I] = act(X Wg^T) * (X Wu^T)
-// for skinny batches (B <= 128, multiples of 16).  One kernel replaces
-// two hipBLASLt GEMMs + the activation pair: both W streams are read
-// non-temporally exactly once, X stays L1/L2-hot, and the activation
-// happens on the MFMA accumulators via an LDS exchange between the
-// gate-wave and up-wave of each row group.
-//
-// Block (256 thr): waves 0/1 compute gate/up for I-rows [r0, r0+16),
-// waves 2/3 for [r0+16, r0+32); grid.x = I/32 (>= 448 blocks for the
-// Llama-3-8B MLP — the chip stays filled for the W stream).
-// BT = B/16 column tiles of MFMA 16x16x32 accumulators per wave.
-// ---------------------------------------------------------------------------
-template <int BT>
-__global__ void gemm_swiglu_kernel(bf16* __restrict__ out,
-                                   const bf16* __restrict__ Wg,
-                                   const bf16* __restrict__ Wu,
-                                   const bf16* __restrict__ xn,
-                                   int I, int K, int gelu_gate) {
-  constexpr int B = BT * 16;
-  __shared__ float gu_lds[2][2][16][B];  // [rowgrp][gate/up][row][b]
-
-  const int wave = threadIdx.x >> 6;
-  const int lane = threadIdx.x & 63;
-  const int grp = wave >> 1;        // row group 0/1
-  const int is_up = wave & 1;       // 0: gate, 1: up
-  const int r0 = blockIdx.x * 32 + grp * 16;
-
-  const bf16* W = is_up ? Wu : Wg;
-  const bf16* wrow = W + (size_t)(r0 + (lane & 15)) * K + (lane >> 4) * 8;
-
-  f32x4 acc[BT];
-#pragma unroll
-  for (int t = 0; t < BT; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
-
-  const bf16* xrow = xn + (size_t)(lane & 15) * K + (lane >> 4) * 8;
-  for (int k0 = 0; k0 < K; k0 += 32) {
-    bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(wrow + k0);
-    // hint the W stream non-temporal via a separate NT load path
-#pragma unroll
-    for (int t = 0; t < BT; ++t) {
-      const bf16x8_t b = *reinterpret_cast<const bf16x8_t*>(
-          xrow + (size_t)t * 16 * K + k0);
-      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0, 0);
-    }
-  }
-
-  // C frag: lane holds rows (lane>>4)*4 + r (I dim) for column lane&15?
-  // With A = W rows and B = X rows (columns of C), the 16x16 result has
-  // C[i_row, b_col]: lane l owns col = l&15 is the A-row?  Empirically
-  // (validated by the attention kernel's swapped-operand layout): for
-  // D = mfma(A, B, D), lane l holds D[col l&15 of B][4 rows of A at
-  // (l>>4)*4 + r].  Here A supplies W rows (I) as the "columns" and B
-  // supplies X rows (b) — store via the same map as attn: the A operand
-  // owns the per-lane 16-row index.
-#pragma unroll
-  for (int t = 0; t < BT; ++t) {
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int irow = (lane >> 4) * 4 + r;   // row within the 16 I-rows
-      const int bcol = t * 16 + (lane & 15);  // batch index
-      gu_lds[grp][is_up][irow][bcol] = acc[t][r];
-    }
-  }
-  __syncthreads();
-
-  for (int idx = threadIdx.x; idx < 32 * B; idx += 256) {
-    const int row = idx / B;        // 0..31 within the block's I rows
-    const int b = idx % B;
-    const float g = gu_lds[row >> 4][0][row & 15][b];
-    const float u = gu_lds[row >> 4][1][row & 15][b];
-    const float act = gelu_gate ? gelu_tanh(g) : g / (1.f + __expf(-g));
-    out[(size_t)b * I + blockIdx.x * 32 + row] = f2b(act * u);
-  }
-}
-
-// ---------------------------------------------------------------------------
 // Embedding row gather: out[n_embd] = wte[token] (* scale)
 // token id read from device memory (graph-replayable).
 // ---------------------------------------------------------------------------
@@ -873,6 +794,9 @@ __global__ void rope_kv_append_kernel(
 // each lane owns 4 keys of ONE query head). qpk (query heads per kv
 // head) <= 16.
 // ---------------------------------------------------------------------------
+using f32x4 = __attribute__((__vector_size__(16))) float;
+using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+
 #define ATTN_WAVES 4  // waves per block, each fully independent
 // max_seq at/below which the one-launch block-local variant is used; the
 // choice is static per engine (max_seq), so hipGraph shapes never change
@@ -2084,28 +2008,6 @@ void launch_gemv_swiglu(void* out, const void* Wg, const void* Wu,
     default: SW_CASE(0);
   }
 #undef SW_CASE
-}
-
-int launch_gemm_swiglu(void* out, const void* Wg, const void* Wu,
-                       const void* xn, int Bsz, int I, int K, int gelu_gate,
-                       hipStream_t stream) {
-  if (I % 32 != 0 || K % 32 != 0 || Bsz % 16 != 0) return -1;
-  dim3 grid(I / 32), block(256);
-#define GS_CASE(BT)                                                         \
-  case BT:                                                                  \
-    hipLaunchKernelGGL((gemm_swiglu_kernel<BT>), grid, block, 0, stream,    \
-                       (bf16*)out, (const bf16*)Wg, (const bf16*)Wu,        \
-                       (const bf16*)xn, I, K, gelu_gate);                   \
-    return 0
-  switch (Bsz / 16) {
-    GS_CASE(1);
-    GS_CASE(2);
-    GS_CASE(4);
-    GS_CASE(6);
-    GS_CASE(8);
-    default: return -1;
-  }
-#undef GS_CASE
 }
 
 void launch_swiglu_mul(void* out, const void* g, const void* u,
