@@ -15,9 +15,9 @@ N_SAMPLES = 3
 MAX_NEW = 10
 
 
-def _build_and_save(tmp):
+def _build_and_save(tmp, model="nano-test"):
     torch.manual_seed(0)
-    cfg = ModelConfig.from_name("nano-test")
+    cfg = ModelConfig.from_name(model)
     m = GPT(cfg)
     m.apply_init()
     m.eval()
@@ -28,13 +28,13 @@ def _build_and_save(tmp):
     return cfg, m, prompts
 
 
-def _standalone_reference(tmp):
+def _standalone_reference(tmp, model="nano-test"):
     """Standalone (1-node) generation via the same runtime."""
     from mdi_llm_amd.models.stages import StarterStage
     from mdi_llm_amd.parallel.runner import TorchRunner
     from mdi_llm_amd.parallel.runtime import PipelineRuntime, SamplingParams
 
-    cfg = ModelConfig.from_name("nano-test")
+    cfg = ModelConfig.from_name(model)
     sd = torch.load(os.path.join(tmp, "model.pt"), weights_only=True)
     prompts = torch.load(os.path.join(tmp, "prompts.pt"), weights_only=True)
     stage = StarterStage(cfg, cfg.n_layer)
@@ -47,7 +47,7 @@ def _standalone_reference(tmp):
     return [s.tolist() for s in res.sequences]
 
 
-def _worker(rank, world, tmp, port, out_file):
+def _worker(rank, world, tmp, port, out_file, model="nano-test"):
     import torch.distributed as dist
 
     from mdi_llm_amd.models.stages import build_stage
@@ -60,7 +60,7 @@ def _worker(rank, world, tmp, port, out_file):
         "gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank,
         world_size=world,
     )
-    cfg = ModelConfig.from_name("nano-test")
+    cfg = ModelConfig.from_name(model)
     sd = torch.load(os.path.join(tmp, "model.pt"), weights_only=True)
     split = layer_split(cfg.n_layer, world)
     chunks = split_parameters(sd, world)
@@ -96,6 +96,20 @@ def test_pipeline_matches_standalone(world, tmp_path):
     ctx = mp.spawn(
         _worker, args=(world, tmp, port, out_file), nprocs=world, join=True
     )
+    got = torch.load(out_file, weights_only=True)
+    assert got == ref
+
+
+def test_pipeline_moe_matches_standalone(tmp_path):
+    """A mixture-of-experts model through the 2-stage pipeline must
+    reproduce its standalone token streams exactly (MoE routing is
+    per-token local state — the ring must not perturb it)."""
+    tmp = str(tmp_path)
+    _build_and_save(tmp, model="nano-test-moe")
+    ref = _standalone_reference(tmp, model="nano-test-moe")
+    out_file = os.path.join(tmp, "out.pt")
+    mp.spawn(_worker, args=(2, tmp, 29640, out_file, "nano-test-moe"),
+             nprocs=2, join=True)
     got = torch.load(out_file, weights_only=True)
     assert got == ref
 
