@@ -186,7 +186,7 @@ def main():
         runner.reset()
     if rank == 0:
         if geng is None:
-            rt.prepare_bench(sampling)
+            rt.prepare_bench(sampling, n_samples)
         toks = rt.bench_prefill(prompts)
         if geng is not None:
             for si in range(n_samples):

@@ -541,17 +541,41 @@ class DecodeEngine:
         self.ops.stage_slot(self.slot, pos_table_mut=self.pos_table,
                             adv_pos=1)
 
-    def ensure_fused_graphs(self, temperature: float, top_k, seed: int) -> None:
+    # activation/workspace tensors that are private to one in-flight
+    # sample step; everything else (weights, KV pool, pos/token tables,
+    # rope caches) is shared across lanes
+    _LANE_ATTRS = ("x", "qkv", "y", "a", "act", "m_out", "part_o",
+                   "part_ml", "logits", "token", "pos_emb",
+                   "sample_scratch", "sample_out", "slot", "pos", "xn",
+                   "gate_logits", "moe_eidx", "moe_escale")
+
+    def _set_lane(self, lane: int) -> None:
+        for a, t_ in self._lane_bufs[lane].items():
+            setattr(self, a, t_)
+
+    @property
+    def n_lanes(self) -> int:
+        return len(getattr(self, "_lane_graphs", None) or [1])
+
+    def ensure_fused_graphs(self, temperature: float, top_k, seed: int,
+                            n_lanes: int = 1) -> None:
         """Capture the three starter step graphs for fixed sampling params:
         standalone (embed->blocks->tail->sample), pipeline-starter
         (tail->sample->embed->blocks), and tail+sample only (drain rounds).
-        """
-        params = (float(temperature), int(top_k or 0), int(seed))
+
+        n_lanes > 1 captures the standalone step N times over N private
+        workspace sets, each replayed on its own HIP stream: different
+        in-flight samples (different KV slots) then overlap on the GPU,
+        hiding the per-kernel launch floor that dominates small models.
+        Token streams are unchanged — sampling is (seed, slot, pos)-keyed
+        and samples share no per-step state."""
+        params = (float(temperature), int(top_k or 0), int(seed),
+                  int(n_lanes))
         if not self.use_graphs or not self.is_starter:
             return
         if self._fused_params == params and self._graph_standalone is not None:
             return
-        t, k, sd = params
+        t, k, sd, n_lanes = params
         torch.cuda.synchronize()
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -569,14 +593,27 @@ class DecodeEngine:
         self.pos_table.zero_()
         self.token_table.zero_()
 
-        g1 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g1):
-            self._stage_pos_token()
-            self._embed()
-            self._run_blocks()
-            self._tail_seq()
-            self._sample_seq(t, k, sd, advance="token+pos")
-        self._graph_standalone = g1
+        base = {a: getattr(self, a) for a in self._LANE_ATTRS
+                if getattr(self, a, None) is not None}
+        self._lane_bufs = [dict(base)]
+        for _l in range(1, n_lanes):
+            self._lane_bufs.append(
+                {a: torch.zeros_like(t_) for a, t_ in base.items()})
+        self._lane_graphs = []
+        for l in range(n_lanes):
+            self._set_lane(l)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._stage_pos_token()
+                self._embed()
+                self._run_blocks()
+                self._tail_seq()
+                self._sample_seq(t, k, sd, advance="token+pos")
+            self._lane_graphs.append(g)
+        self._set_lane(0)
+        self._lane_streams = [torch.cuda.Stream()
+                              for _ in range(n_lanes)]
+        self._graph_standalone = self._lane_graphs[0]
 
         g2 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g2):
@@ -604,6 +641,29 @@ class DecodeEngine:
         """One full decode token for `slot` (token_table-chained)."""
         self.slot.fill_(slot)
         self._graph_standalone.replay()
+
+    # ---- multi-stream lanes (standalone multi-sample overlap) ----------
+    def lanes_begin(self) -> None:
+        """Order every lane stream after the current stream (prefill and
+        host-side staging complete before lane replays start)."""
+        cur = torch.cuda.current_stream()
+        for st in self._lane_streams:
+            st.wait_stream(cur)
+
+    def lanes_join(self) -> None:
+        """Order the current stream after every lane stream (no host
+        sync; D2H readers on the current stream see the lane writes)."""
+        cur = torch.cuda.current_stream()
+        for st in self._lane_streams:
+            cur.wait_stream(st)
+
+    def standalone_lane_step(self, lane: int, slot: int) -> None:
+        """standalone_step on lane `lane`'s private stream/workspace;
+        steps on different lanes run concurrently on the GPU."""
+        st = self._lane_streams[lane]
+        with torch.cuda.stream(st):
+            self._lane_bufs[lane]["slot"].fill_(slot)
+            self._lane_graphs[lane].replay()
 
     def starter_step(self, x_in: torch.Tensor, slot: int) -> torch.Tensor:
         """Pipeline starter: tail(x_in)+sample+next head; returns self.x."""

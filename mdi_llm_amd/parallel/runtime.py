@@ -16,6 +16,7 @@ the reference's queue aliasing (gptserver.py:276-278).
 
 from __future__ import annotations
 
+import os
 import time
 from dataclasses import dataclass, field
 from typing import List, Optional, Sequence, Tuple
@@ -206,8 +207,9 @@ class PipelineRuntime:
         eng = runner.engine
         n_samples = len(prompts)
         # graphs must be captured before prefill (capture scribbles caches)
+        L = self._lane_count(n_samples)
         eng.ensure_fused_graphs(sampling.temperature, sampling.top_k,
-                                sampling.seed or 0)
+                                sampling.seed or 0, n_lanes=L)
         runner.reset()
 
         res = GenerationResult()
@@ -222,8 +224,12 @@ class PipelineRuntime:
         active = set(range(n_samples))
         total_new = 0
         while active:
-            for s in list(active):
-                tok = int(eng.token_table[s])  # 4-byte D2H sync
+            # consume the current token of every active sample (one D2H
+            # sync point per round), then launch the next steps lane-
+            # parallel: samples on different lanes overlap on the GPU
+            stepping = []
+            for s in sorted(active):
+                tok = int(eng.token_table[s])
                 seqs[s].append(tok)
                 new_counts[s] += 1
                 total_new += 1
@@ -237,8 +243,12 @@ class PipelineRuntime:
                         or runner.pos[s] + 1 >= runner.stage.max_seq_length):
                     active.discard(s)
                     continue
-                eng.standalone_step(s)
+                stepping.append(s)
+            eng.lanes_begin()
+            for s in stepping:
+                eng.standalone_lane_step(s % eng.n_lanes, s)
                 runner.pos[s] += 1
+            eng.lanes_join()
         res.gen_time = time.perf_counter() - t_start
         res.total_new_tokens = total_new
         res.sequences = [torch.tensor(q, dtype=torch.int64) for q in seqs]
@@ -273,14 +283,29 @@ class PipelineRuntime:
     # in the decode loop; each phase leaves the ring drained so callers can
     # barrier/synchronize between phases)
     # ------------------------------------------------------------------
-    def prepare_bench(self, sampling: SamplingParams) -> None:
+    @staticmethod
+    def _lane_count(n_samples: int) -> int:
+        """Concurrent HIP-stream lanes for standalone multi-sample decode
+        (MDI_LANES; 0/unset = auto: up to 4, capped by the sample count).
+        Small models are launch-floor-bound, so overlapping samples on
+        separate streams raises aggregate throughput without changing
+        any token (sampling is (seed, slot, pos)-keyed)."""
+        v = int(os.environ.get("MDI_LANES", "0") or 0)
+        if v <= 0:
+            v = 4
+        return max(1, min(v, n_samples))
+
+    def prepare_bench(self, sampling: SamplingParams,
+                      n_samples: int = 1) -> None:
         """Capture the fused step graphs BEFORE prefill (capture warm-up
         scribbles on the KV pool, so it must precede cache filling)."""
         r = self.runner
         if (self.is_starter and getattr(r, "backend", "") == "hip"
                 and sampling.top_p >= 1.0 and r.engine.use_graphs):
+            lanes = self._lane_count(n_samples) if self.world == 1 else 1
             r.engine.ensure_fused_graphs(
-                sampling.temperature, sampling.top_k, sampling.seed or 0
+                sampling.temperature, sampling.top_k, sampling.seed or 0,
+                n_lanes=lanes,
             )
             r.reset()
 
@@ -329,17 +354,21 @@ class PipelineRuntime:
         )
         eng = runner.engine if fused else None
         if fused:
-            eng.ensure_fused_graphs(sampling.temperature, sampling.top_k,
-                                    sampling.seed or 0)
+            eng.ensure_fused_graphs(
+                sampling.temperature, sampling.top_k, sampling.seed or 0,
+                n_lanes=self._lane_count(n) if self.world == 1 else 1)
             for s in range(n):
                 eng.token_table[s] = toks[s].view(())
 
         if self.world == 1:
             if fused:
+                L = eng.n_lanes
+                eng.lanes_begin()
                 for _ in range(n_rounds):
                     for s in range(n):
-                        eng.standalone_step(s)
+                        eng.standalone_lane_step(s % L, s)
                         runner.pos[s] += 1
+                eng.lanes_join()
                 return [eng.token_table[s: s + 1] for s in range(n)]
             for _ in range(n_rounds):
                 for s in range(n):
