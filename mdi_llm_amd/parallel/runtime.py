@@ -244,11 +244,16 @@ class PipelineRuntime:
                     active.discard(s)
                     continue
                 stepping.append(s)
-            eng.lanes_begin()
-            for s in stepping:
-                eng.standalone_lane_step(s % eng.n_lanes, s)
-                runner.pos[s] += 1
-            eng.lanes_join()
+            if eng.n_lanes == 1:
+                for s in stepping:
+                    eng.standalone_step(s)
+                    runner.pos[s] += 1
+            else:
+                eng.lanes_begin()
+                for s in stepping:
+                    eng.standalone_lane_step(s % eng.n_lanes, s)
+                    runner.pos[s] += 1
+                eng.lanes_join()
         res.gen_time = time.perf_counter() - t_start
         res.total_new_tokens = total_new
         res.sequences = [torch.tensor(q, dtype=torch.int64) for q in seqs]
@@ -363,6 +368,14 @@ class PipelineRuntime:
         if self.world == 1:
             if fused:
                 L = eng.n_lanes
+                if L == 1:
+                    # single lane: stay on the current stream (no stream
+                    # round-trip per replay)
+                    for _ in range(n_rounds):
+                        for s in range(n):
+                            eng.standalone_step(s)
+                            runner.pos[s] += 1
+                    return [eng.token_table[s: s + 1] for s in range(n)]
                 eng.lanes_begin()
                 for _ in range(n_rounds):
                     for s in range(n):
