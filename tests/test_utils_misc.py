@@ -103,3 +103,16 @@ def test_console_utils(capsys):
     ev.set()
     t.join(timeout=1)
     assert not t.is_alive()
+
+
+def test_lane_count_env(monkeypatch):
+    from mdi_llm_amd.parallel.runtime import PipelineRuntime
+
+    monkeypatch.delenv("MDI_LANES", raising=False)
+    assert PipelineRuntime._lane_count(1) == 1
+    assert PipelineRuntime._lane_count(3) == 3
+    assert PipelineRuntime._lane_count(10) == 4  # auto cap
+    monkeypatch.setenv("MDI_LANES", "2")
+    assert PipelineRuntime._lane_count(8) == 2
+    monkeypatch.setenv("MDI_LANES", "16")
+    assert PipelineRuntime._lane_count(3) == 3  # capped by samples
