@@ -272,10 +272,10 @@ class DecodeEngine:
         for tuning."""
         if env and os.environ.get(env):
             return int(os.environ[env])
-        if K >= 16384:
-            return 4
         if M >= 65536:
-            return 2  # lm_head: grid caps at 4096 blocks; 2-row ILP wins
+            return 4  # lm_head: grid caps at 4096 blocks; row ILP wins
+        if K >= 16384:
+            return 1  # direct-x long-K (70B down): max blocks wins
         if M <= 16384:
             return 1
         return 2
