@@ -181,8 +181,14 @@ def main():
     ]
     t0 = time.time()
     if geng is not None:
+        n_groups = max(1, n_samples // B)
+        lanes = 1
+        if world == 1:
+            import os as _os
+            lanes = max(1, min(int(_os.environ.get("MDI_LANES", "0") or 0)
+                               or 2, n_groups))
         geng.ensure_graphs(sampling.temperature, sampling.top_k,
-                           sampling.seed or 0)
+                           sampling.seed or 0, n_lanes=lanes)
         runner.reset()
     if rank == 0:
         if geng is None:

@@ -257,13 +257,34 @@ class GroupDecodeEngine:
         self.token_table.index_copy_(0, self.slots_long, self.tokens)
 
     # ------------------------------------------------------------------
-    def ensure_graphs(self, temperature, top_k, seed) -> None:
-        params = (float(temperature), int(top_k or 0), int(seed))
+    # per-lane private workspaces (everything per-group-step; the
+    # pos/token tables and KV pool are per-slot and stay shared)
+    _LANE_ATTRS = ("X", "XN", "HN", "QKV", "Y", "part_o", "part_ml",
+                   "LOGITS", "sample_scratch", "tokens", "slots",
+                   "slots_long", "pos")
+
+    def _set_lane(self, lane: int) -> None:
+        for a, t_ in self._lane_bufs[lane].items():
+            setattr(self, a, t_)
+
+    @property
+    def n_lanes(self) -> int:
+        return len(getattr(self, "_lane_graphs", None) or [1])
+
+    def ensure_graphs(self, temperature, top_k, seed,
+                      n_lanes: int = 1) -> None:
+        """n_lanes > 1 (standalone only): capture the group step over N
+        private workspace sets replayed on N HIP streams, so consecutive
+        GROUPS overlap — one group's GEMM weight streams fill the other
+        group's attention/sampler phases.  Tokens are unchanged (groups
+        own disjoint slots; sampling is (seed, slot, pos)-keyed)."""
+        params = (float(temperature), int(top_k or 0), int(seed),
+                  int(n_lanes))
         if not self.use_graphs:
             return
         if self._fused_params == params:
             return
-        t, k, sd = params
+        t, k, sd, n_lanes = params
         torch.cuda.synchronize()
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
@@ -285,16 +306,29 @@ class GroupDecodeEngine:
             self.token_table.zero_()
 
         if self.is_starter:
-            g1 = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g1):
-                self._stage_idx()
-                self._embed_seq()
-                self._run_blocks()
-                self._tail_seq()
-                self._sample_seq(t, k, sd)
-                self._write_tokens()
-                self._advance()
-            self._g_standalone = g1
+            base = {a: getattr(self, a) for a in self._LANE_ATTRS
+                    if getattr(self, a, None) is not None}
+            self._lane_bufs = [dict(base)]
+            for _l in range(1, n_lanes):
+                self._lane_bufs.append(
+                    {a: torch.zeros_like(t_) for a, t_ in base.items()})
+            self._lane_graphs = []
+            for l in range(n_lanes):
+                self._set_lane(l)
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    self._stage_idx()
+                    self._embed_seq()
+                    self._run_blocks()
+                    self._tail_seq()
+                    self._sample_seq(t, k, sd)
+                    self._write_tokens()
+                    self._advance()
+                self._lane_graphs.append(g)
+            self._set_lane(0)
+            self._lane_streams = [torch.cuda.Stream()
+                                  for _ in range(n_lanes)]
+            self._g_standalone = self._lane_graphs[0]
             g2 = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g2):
                 self._stage_idx()
@@ -330,6 +364,25 @@ class GroupDecodeEngine:
 
     def standalone_step(self) -> None:
         self._g_standalone.replay()
+
+    def lanes_begin(self) -> None:
+        cur = torch.cuda.current_stream()
+        for st in self._lane_streams:
+            st.wait_stream(cur)
+
+    def lanes_join(self) -> None:
+        cur = torch.cuda.current_stream()
+        for st in self._lane_streams:
+            cur.wait_stream(st)
+
+    def standalone_lane_step(self, lane: int,
+                             slot_tensor: torch.Tensor) -> None:
+        """set_group + standalone_step on lane `lane`'s stream."""
+        st = self._lane_streams[lane]
+        with torch.cuda.stream(st):
+            self._lane_bufs[lane]["slots"].copy_(slot_tensor,
+                                                 non_blocking=True)
+            self._lane_graphs[lane].replay()
 
     def starter_step(self, X_in: torch.Tensor) -> torch.Tensor:
         if X_in.data_ptr() != self.X.data_ptr():

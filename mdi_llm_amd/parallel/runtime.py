@@ -428,10 +428,18 @@ class PipelineRuntime:
         geng.ensure_graphs beforehand)."""
         G = len(group_slots)
         if self.world == 1:
+            L = geng.n_lanes
+            if L == 1:
+                for _ in range(n_rounds):
+                    for g in range(G):
+                        geng.set_group(group_slots[g])
+                        geng.standalone_step()
+                return
+            geng.lanes_begin()
             for _ in range(n_rounds):
                 for g in range(G):
-                    geng.set_group(group_slots[g])
-                    geng.standalone_step()
+                    geng.standalone_lane_step(g % L, group_slots[g])
+            geng.lanes_join()
             return
         for g in range(G):
             geng.set_group(group_slots[g])
