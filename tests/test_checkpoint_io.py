@@ -271,3 +271,62 @@ def test_convert_mixtral_moe_synthetic(tmp_path):
     with torch.inference_mode():
         logits = m(torch.randint(0, 255, (1, 8)))
     assert torch.isfinite(logits).all()
+
+
+def test_convert_falcon_split_norm_synthetic(tmp_path):
+    """40b/180B-style falcon (separate ln_attn/ln_mlp, GQA) conversion
+    and reverse map (reference convert_hf_checkpoint.py:86-94)."""
+    from mdi_llm_amd.config import name_to_config
+    from mdi_llm_amd.utils.convert_hf import (
+        convert_hf_checkpoint,
+        convert_lit_checkpoint,
+    )
+
+    name_to_config.setdefault(
+        "falcon-nano40",
+        dict(
+            name="falcon-nano40", block_size=128, vocab_size=256,
+            padding_multiple=64, n_layer=2, n_head=8, n_embd=64,
+            rotary_percentage=1.0, n_query_groups=2, bias=False,
+            parallel_residual=True, norm_class_name="LayerNorm",
+            mlp_class_name="GptNeoxMLP",
+        ),
+    )
+    torch.manual_seed(12)
+    cfg = ModelConfig.from_name("falcon-nano40")
+    E, L = cfg.n_embd, cfg.n_layer
+    hf = {"transformer.word_embeddings.weight":
+          torch.randn(cfg.vocab_size, E),
+          "transformer.ln_f.weight": torch.randn(E),
+          "transformer.ln_f.bias": torch.randn(E),
+          "lm_head.weight": torch.randn(cfg.vocab_size, E)}
+    for l in range(L):
+        p = f"transformer.h.{l}"
+        hf[f"{p}.ln_attn.weight"] = torch.randn(E)
+        hf[f"{p}.ln_attn.bias"] = torch.randn(E)
+        hf[f"{p}.ln_mlp.weight"] = torch.randn(E)
+        hf[f"{p}.ln_mlp.bias"] = torch.randn(E)
+        hf[f"{p}.self_attention.query_key_value.weight"] = \
+            torch.randn(cfg.qkv_dim, E)
+        hf[f"{p}.self_attention.dense.weight"] = torch.randn(E, E)
+        hf[f"{p}.mlp.dense_h_to_4h.weight"] = torch.randn(4 * E, E)
+        hf[f"{p}.mlp.dense_4h_to_h.weight"] = torch.randn(E, 4 * E)
+    src = tmp_path / "f40"
+    src.mkdir()
+    torch.save(hf, src / "pytorch_model.bin")
+    out = tmp_path / "f40_lit"
+    convert_hf_checkpoint(src, out, model_name="falcon-nano40")
+    config, sd = load_from_pt(out)
+    assert "transformer.h.0.norm_2.weight" in sd  # split norms mapped
+    m = GPT(config)
+    m.load_state_dict(sd)
+    m.eval()
+    with torch.inference_mode():
+        logits = m(torch.randint(0, 255, (1, 6)))
+    assert torch.isfinite(logits).all()
+    back = tmp_path / "f40_back" / "pytorch_model.bin"
+    convert_lit_checkpoint(out, back, model_name="falcon-nano40")
+    hf2 = torch.load(back, weights_only=True)
+    assert set(hf2) == set(hf)
+    for k in hf:
+        assert torch.equal(hf[k], hf2[k]), k
