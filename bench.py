@@ -67,8 +67,13 @@ def main():
         log("WARNING: --gpus>1 but WORLD_SIZE=1; launch via torchrun")
     n_stages = world
     on_gpu = torch.cuda.is_available()
+    share_gpu = on_gpu and world > torch.cuda.device_count()
     if on_gpu:
-        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        # more ranks than GPUs (single-box pipeline validation): ranks
+        # share devices round-robin and the ring stages via gloo (RCCL
+        # refuses two ranks on one device)
+        local = int(os.environ.get("LOCAL_RANK", 0))
+        device = torch.device(f"cuda:{local % torch.cuda.device_count()}")
         torch.cuda.set_device(device)
     else:
         # CPU fallback so the exact driver invocation is testable without a
@@ -81,7 +86,9 @@ def main():
 
     cpu_group = None
     if world > 1:
-        dist.init_process_group("nccl" if on_gpu else "gloo")
+        backend = "nccl" if (on_gpu and not share_gpu) else "gloo"
+        backend = os.environ.get("MDI_PP_BACKEND", backend)
+        dist.init_process_group(backend)
         cpu_group = dist.new_group(backend="gloo")
 
     B = max(args.group_size, 1)

@@ -521,10 +521,17 @@ class DecodeEngine:
     # fully-fused per-token step graphs (bench hot path)
     # ------------------------------------------------------------------
     def _sample_seq(self, temperature: float, top_k: int, seed: int,
-                    top_p: float = 1.0, out=None, advance: str = "") -> None:
+                    top_p: float = 1.0, out=None, advance: str = "",
+                    pos_bias: int = 0) -> None:
         """Fused sampling; `advance` folds the step bookkeeping into the
         sampler's unpack launch: "token" writes token_table[slot],
-        "token+pos" additionally advances pos_table[slot]."""
+        "token+pos" additionally advances pos_table[slot].
+
+        pos_bias keys the draw by the DRAWN token's sequence index: the
+        standalone step graph (embed->blocks->tail->sample) reads pos
+        before its advance, so it passes 1; the pipeline tail-first
+        graphs read the already-advanced pos and pass 0.  With matching
+        keys the standalone and pipeline token streams are identical."""
         self.ops.sample(self.sample_out if out is None else out,
                         self.logits, self.sample_scratch,
                         float(temperature), int(top_k or 0),
@@ -534,7 +541,8 @@ class DecodeEngine:
                         pos_table=self.pos_table
                         if advance == "token+pos" else None,
                         adv_slot=self.slot if advance else None,
-                        adv_pos=1 if advance == "token+pos" else 0)
+                        adv_pos=1 if advance == "token+pos" else 0,
+                        pos_bias=pos_bias)
 
     def _advance_pos(self) -> None:
         # pos_table[slot] += 1 (in-graph, one tiny launch)
@@ -585,7 +593,7 @@ class DecodeEngine:
                 self._embed()
                 self._run_blocks()
                 self._tail_seq()
-                self._sample_seq(t, k, sd, advance="token+pos")
+                self._sample_seq(t, k, sd, advance="token+pos", pos_bias=1)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self.kv_pool.k.zero_()
@@ -608,7 +616,7 @@ class DecodeEngine:
                 self._embed()
                 self._run_blocks()
                 self._tail_seq()
-                self._sample_seq(t, k, sd, advance="token+pos")
+                self._sample_seq(t, k, sd, advance="token+pos", pos_bias=1)
             self._lane_graphs.append(g)
         self._set_lane(0)
         self._lane_streams = [torch.cuda.Stream()

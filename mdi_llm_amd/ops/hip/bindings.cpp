@@ -334,7 +334,8 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
             c10::optional<torch::Tensor> slot, int64_t n_batch,
             double top_p, c10::optional<torch::Tensor> token_table,
             c10::optional<torch::Tensor> pos_table,
-            c10::optional<torch::Tensor> adv_slot, int64_t adv_pos) {
+            c10::optional<torch::Tensor> adv_slot, int64_t adv_pos,
+            int64_t pos_bias) {
   check_i32(out_token, "out_token");
   check_bf16(logits, "logits");
   const int nb = n_batch > 0 ? (int)n_batch : 1;
@@ -373,7 +374,8 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
   launch_sample(out_token.data_ptr(), logits.data_ptr(), V,
                 scratch.data_ptr(), (float)temperature, (int)top_k,
                 (float)top_p, noise ? 1 : 0, (unsigned)(int64_t)seed, pp, sp,
-                (int)n_batch, ttp, ptp, asp, (int)adv_pos, cur_stream());
+                (int)n_batch, ttp, ptp, asp, (int)adv_pos, (int)pos_bias,
+                cur_stream());
 }
 
 void stage_slot(torch::Tensor slot, c10::optional<torch::Tensor> pos_out,
@@ -511,7 +513,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("slot") = c10::nullopt, py::arg("n_batch") = 0,
         py::arg("top_p") = 1.0, py::arg("token_table") = c10::nullopt,
         py::arg("pos_table") = c10::nullopt,
-        py::arg("adv_slot") = c10::nullopt, py::arg("adv_pos") = 0);
+        py::arg("adv_slot") = c10::nullopt, py::arg("adv_pos") = 0,
+        py::arg("pos_bias") = 0);
   m.def("stage_slot", &stage_slot,
         "one-launch step staging/bookkeeping on device scalars",
         py::arg("slot"), py::arg("pos_out") = c10::nullopt,

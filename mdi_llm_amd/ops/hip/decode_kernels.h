@@ -90,7 +90,8 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, float top_p, int noise_on,
                    unsigned seed, const int* pos, const int* slot,
                    int n_batch, int* token_table, int* pos_table,
-                   const int* adv_slot, int adv_pos, hipStream_t stream);
+                   const int* adv_slot, int adv_pos, int pos_bias,
+                   hipStream_t stream);
 
 // one-launch step staging: pos_out = pos_table[slot], token_out =
 // token_table[slot], optional pos_table_mut[slot] += 1; any output may be

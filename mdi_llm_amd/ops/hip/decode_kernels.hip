@@ -1578,10 +1578,15 @@ __global__ void sample_mass_hist_hi_kernel(const bf16* __restrict__ logits,
   if (tid < 256) h[tid] = 0.f;
   __syncthreads();
   const float m = sortable_to_float(scratch[515]) * inv_temp;
+  // restrict the nucleus mass to codes kept by the preceding top-k pass
+  // (scratch[514]): torch composes top-p ON the top-k-masked
+  // distribution (models/sampling.py), so the kept sets match
+  const unsigned kt = scratch[514];
   for (int i = blockIdx.x * blockDim.x + tid; i < V;
        i += gridDim.x * blockDim.x) {
     unsigned short bits = reinterpret_cast<const unsigned short*>(logits)[i];
     unsigned u = bf16_sortable(bits);
+    if (u < kt) continue;
     float l = b2f(*reinterpret_cast<bf16*>(&bits)) * inv_temp;
     atomicAdd(&h[u >> 8], __expf(l - m));
   }
@@ -1622,10 +1627,12 @@ __global__ void sample_mass_hist_lo_kernel(const bf16* __restrict__ logits,
   __syncthreads();
   const unsigned bucket = scratch[512];
   const float m = sortable_to_float(scratch[515]) * inv_temp;
+  const unsigned kt = scratch[514];  // top-k threshold (see hi pass)
   for (int i = blockIdx.x * blockDim.x + tid; i < V;
        i += gridDim.x * blockDim.x) {
     unsigned short bits = reinterpret_cast<const unsigned short*>(logits)[i];
     unsigned u = bf16_sortable(bits);
+    if (u < kt) continue;
     if ((u >> 8) != bucket) continue;
     float l = b2f(*reinterpret_cast<bf16*>(&bits)) * inv_temp;
     atomicAdd(&h[u & 255], __expf(l - m));
@@ -1712,13 +1719,17 @@ __global__ void sample_select_lo_kernel(unsigned* __restrict__ scratch,
 __global__ void sample_gumbel_argmax_kernel(
     const bf16* __restrict__ logits, int V, unsigned* __restrict__ scratch,
     float inv_temp, int use_threshold, int noise, unsigned seed,
-    const int* __restrict__ pos_p, const int* __restrict__ slot_p) {
+    const int* __restrict__ pos_p, const int* __restrict__ slot_p,
+    int pos_bias) {
   logits += (size_t)blockIdx.y * V;
   scratch += (size_t)blockIdx.y * 520;
   const unsigned t = use_threshold ? scratch[514] : 0u;
-  // counter-based RNG keyed by (seed, slot, position): reproducible and
-  // independent of the scheduling order across samples
-  const unsigned pos_v = pos_p ? (unsigned)pos_p[blockIdx.y] : 0u;
+  // counter-based RNG keyed by (seed, slot, position of the DRAWN token):
+  // reproducible and independent of the scheduling order across samples.
+  // pos_bias aligns graphs that read pos before vs after the advance
+  // (standalone embed-first step vs pipeline tail-first step).
+  const unsigned pos_v =
+      pos_p ? (unsigned)(pos_p[blockIdx.y] + pos_bias) : 0u;
   const unsigned slot_v = slot_p ? (unsigned)slot_p[blockIdx.y] : blockIdx.y;
   const unsigned salt = seed ^ (pos_v * 0x9E3779B9u) ^
                         (slot_v * 0x85EBCA6Bu);
@@ -1813,7 +1824,8 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    float temperature, int top_k, float top_p, int noise_on,
                    unsigned seed, const int* pos, const int* slot,
                    int n_batch, int* token_table, int* pos_table,
-                   const int* adv_slot, int adv_pos, hipStream_t stream) {
+                   const int* adv_slot, int adv_pos, int pos_bias,
+                   hipStream_t stream) {
   unsigned* sc = (unsigned*)scratch;
   const int B = n_batch > 0 ? n_batch : 1;
   const int blocks = B > 1 ? 32 : 128;
@@ -1847,7 +1859,7 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
   }
   hipLaunchKernelGGL(sample_gumbel_argmax_kernel, dim3(blocks, B), dim3(256),
                      0, stream, (const bf16*)logits, V, sc, inv_t,
-                     use_k || use_p, noise_on, seed, pos, slot);
+                     use_k || use_p, noise_on, seed, pos, slot, pos_bias);
   hipLaunchKernelGGL(sample_unpack_kernel, dim3(1, B), dim3(64), 0, stream,
                      sc, (int*)out_token, token_table, pos_table, adv_slot,
                      adv_pos);

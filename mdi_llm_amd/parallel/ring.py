@@ -12,6 +12,13 @@ A headerless "scheduled" mode skips the header entirely when the message
 sequence is deterministic (the bench decode loop): payloads are fixed-size
 (1, n_embd) and arrival order is the seeding order, so no host readback of
 headers is needed on the critical path.
+
+A "staged" mode bridges backends that cannot move device tensors (gloo):
+device payloads are staged through pinned host mirrors around each
+send/recv.  This is how multiple pipeline ranks can SHARE one physical
+GPU (RCCL refuses two ranks on one device), so the full HIP pipeline —
+graph replays interleaved with ring traffic, per-slot send-buffer reuse —
+is testable on a single-GPU box.
 """
 
 from __future__ import annotations
@@ -35,6 +42,7 @@ class RingComm:
         n_slots: int,
         dtype: torch.dtype = torch.bfloat16,
         group: Optional[dist.ProcessGroup] = None,
+        staged: Optional[bool] = None,
     ) -> None:
         assert dist.is_initialized(), "torch.distributed must be initialized"
         self.group = group
@@ -45,6 +53,11 @@ class RingComm:
         self.device = device
         self.n_embd = n_embd
         self.max_seq = max_seq
+        if staged is None:
+            # gloo cannot carry CUDA tensors point-to-point: stage via host
+            backend = str(dist.get_backend(group))
+            staged = device.type == "cuda" and "gloo" in backend
+        self.staged = bool(staged)
 
         dev = device
         # per-slot single-token send buffers (decode) + one shared prefill
@@ -60,6 +73,36 @@ class RingComm:
                                     device=dev)
         self._pending: dict = {}
         self._PF = "prefill"
+        self._mirrors: dict = {}
+
+    # -- staged-mode helpers ----------------------------------------------
+    def _mirror(self, key, t: torch.Tensor) -> torch.Tensor:
+        m = self._mirrors.get(key)
+        if m is None or m.shape != t.shape:
+            m = torch.empty_like(t, device="cpu").pin_memory()
+            self._mirrors[key] = m
+        return m
+
+    def _isend(self, t: torch.Tensor, key) -> "dist.Work":
+        """isend of a device tensor; staged mode bounces through a pinned
+        per-key host mirror (blocking D2H copy orders after the producing
+        stream work, then the wire send is async)."""
+        if not self.staged:
+            return dist.isend(t, self.next_rank, group=self.group)
+        m = self._mirror(("s", key), t)
+        m.copy_(t)  # syncs the current stream for this copy only
+        return dist.isend(m, self.next_rank, group=self.group)
+
+    def _recv_into(self, t: torch.Tensor, key) -> None:
+        if not self.staged:
+            dist.recv(t, self.prev_rank, group=self.group)
+            return
+        m = self._mirror(("r", key), t)
+        dist.recv(m, self.prev_rank, group=self.group)
+        # blocking H2D: the mirror is reused by the next recv, so the copy
+        # must complete before this returns (staged mode is a correctness
+        # bridge, not the RCCL fast path)
+        t.copy_(m)
 
     # -- headered path (general generate loop) ----------------------------
     def send(self, sample: int, x: Optional[torch.Tensor],
@@ -82,9 +125,9 @@ class RingComm:
             payload.copy_(x2)
         hdr = self.send_hdr[slot]
         hdr[0], hdr[1], hdr[2], hdr[3] = sample, ntok, int(stop), 0
-        works = [dist.isend(hdr, self.next_rank, group=self.group)]
+        works = [self._isend(hdr, ("h", slot))]
         if payload is not None:
-            works.append(dist.isend(payload, self.next_rank, group=self.group))
+            works.append(self._isend(payload, key))
         self._pending[key] = works
 
     def recv(self) -> Tuple[int, Optional[torch.Tensor], bool]:
@@ -92,13 +135,17 @@ class RingComm:
 
         Returns (sample, activations (ntok, n_embd) view or None, stop).
         """
-        dist.recv(self.recv_hdr, self.prev_rank, group=self.group)
-        hdr = self.recv_hdr.cpu()  # host sync: 16 bytes
+        if self.staged:
+            hdr = self._mirror(("r", "hdr"), self.recv_hdr)
+            dist.recv(hdr, self.prev_rank, group=self.group)
+        else:
+            dist.recv(self.recv_hdr, self.prev_rank, group=self.group)
+            hdr = self.recv_hdr.cpu()  # host sync: 16 bytes
         sample, ntok, stop = int(hdr[0]), int(hdr[1]), bool(hdr[2])
         x = None
         if ntok > 0:
             x = self.recv_buf[:ntok]
-            dist.recv(x, self.prev_rank, group=self.group)
+            self._recv_into(x, "payload")
         return sample, x, stop
 
     # -- scheduled (headerless) path: fixed T=1 payloads -------------------
@@ -108,12 +155,11 @@ class RingComm:
                 w.wait()
         self.send_buf[slot, 0].copy_(x.view(-1))
         self._pending[slot] = [
-            dist.isend(self.send_buf[slot, :1], self.next_rank,
-                       group=self.group)
+            self._isend(self.send_buf[slot, :1], slot)
         ]
 
     def recv_sched(self) -> torch.Tensor:
-        dist.recv(self.recv_buf[:1], self.prev_rank, group=self.group)
+        self._recv_into(self.recv_buf[:1], "sched")
         return self.recv_buf[0]
 
     # -- grouped (batched) scheduled path: fixed (B, n_embd) payloads ------
@@ -132,11 +178,11 @@ class RingComm:
         buf = self.gsend[gi]
         buf.copy_(X.view(buf.shape))
         self._pending[key] = [
-            dist.isend(buf, self.next_rank, group=self.group)
+            self._isend(buf, key)
         ]
 
     def recv_group(self) -> torch.Tensor:
-        dist.recv(self.grecv, self.prev_rank, group=self.group)
+        self._recv_into(self.grecv, "group")
         return self.grecv
 
     def drain(self) -> None:

@@ -480,11 +480,13 @@ class PipelineRuntime:
         r = self.runner
         if getattr(r, "backend", "") == "hip":
             # fused on-GPU sampler (radix top-k + mass-radix top-p +
-            # gumbel), no host sync
+            # gumbel), no host sync.  Clone: sample_out is a single device
+            # scalar reused by every draw — callers keep per-sample tokens
+            # across rounds, so each must own its value.
             return r.engine.sample_into_token(
                 sampling.temperature, sampling.top_k, sampling.seed or 0,
                 sampling.top_p,
-            )
+            ).clone()
         tok = sample_token(
             logits,
             temperature=sampling.temperature,

@@ -54,7 +54,7 @@ class Tokenizer:
         if cfg_path.is_file():
             with open(cfg_path, encoding="utf-8") as fp:
                 cfg = json.load(fp)
-        self.use_bos = self._check_use_bos(cfg)
+        self.use_bos = self._check_use_bos(cfg, checkpoint_dir)
         bos_token = cfg.get("bos_token")
         if isinstance(bos_token, dict):
             bos_token = bos_token.get("content")
@@ -73,11 +73,27 @@ class Tokenizer:
             self.eos_id = eos[0] if isinstance(eos, list) else eos
 
     @staticmethod
-    def _check_use_bos(cfg: dict) -> bool:
+    def _check_use_bos(cfg: dict, checkpoint_dir: Path) -> bool:
         if "add_bos_token" in cfg:
             return bool(cfg["add_bos_token"])
-        # LLaMA-style tokenizers default to prepending BOS
-        return cfg.get("tokenizer_class") in ("LlamaTokenizer", "PreTrainedTokenizerFast")
+        # prefer the tokenizer.json post-processor signal: a template that
+        # emits the bos token means the tokenizer itself prepends BOS
+        tok_json = checkpoint_dir / "tokenizer.json"
+        bos_token = cfg.get("bos_token")
+        if isinstance(bos_token, dict):
+            bos_token = bos_token.get("content")
+        if tok_json.is_file() and bos_token:
+            try:
+                with open(tok_json, encoding="utf-8") as fp:
+                    post = json.load(fp).get("post_processor") or {}
+                blob = json.dumps(post)
+                return f'"{bos_token}"' in blob
+            except (OSError, ValueError):
+                pass
+        # LLaMA-family tokenizers default to prepending BOS (a bare
+        # PreTrainedTokenizerFast does not — reference tokenizer.py:80-85
+        # defaults True only for LlamaTokenizer)
+        return cfg.get("tokenizer_class") == "LlamaTokenizer"
 
     # -- API ---------------------------------------------------------------
     @property
