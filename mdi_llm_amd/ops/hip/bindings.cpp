@@ -378,6 +378,15 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
                 cur_stream());
 }
 
+void route_env(torch::Tensor hdr, torch::Tensor slot_out,
+               torch::Tensor pos_table, int64_t dummy_slot) {
+  check_i32(hdr, "hdr");
+  check_i32(slot_out, "slot_out");
+  check_i32(pos_table, "pos_table");
+  launch_route_env(hdr.data_ptr<int>(), slot_out.data_ptr<int>(),
+                   pos_table.data_ptr<int>(), (int)dummy_slot, cur_stream());
+}
+
 void stage_slot(torch::Tensor slot, c10::optional<torch::Tensor> pos_out,
                 c10::optional<torch::Tensor> token_out,
                 c10::optional<torch::Tensor> pos_table,
@@ -515,6 +524,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("pos_table") = c10::nullopt,
         py::arg("adv_slot") = c10::nullopt, py::arg("adv_pos") = 0,
         py::arg("pos_bias") = 0);
+  m.def("route_env", &route_env,
+        "device-side envelope routing (slot from header, dummy on stop)",
+        py::arg("hdr"), py::arg("slot_out"), py::arg("pos_table"),
+        py::arg("dummy_slot"));
   m.def("stage_slot", &stage_slot,
         "one-launch step staging/bookkeeping on device scalars",
         py::arg("slot"), py::arg("pos_out") = c10::nullopt,

@@ -93,6 +93,11 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    const int* adv_slot, int adv_pos, int pos_bias,
                    hipStream_t stream);
 
+// envelope routing for the pipelined secondary serve: slot_out <-
+// hdr[0] (data) or dummy_slot (stop/flush, with pos reset)
+void launch_route_env(const int* hdr, int* slot_out, int* pos_table,
+                      int dummy_slot, hipStream_t stream);
+
 // one-launch step staging: pos_out = pos_table[slot], token_out =
 // token_table[slot], optional pos_table_mut[slot] += 1; any output may be
 // null

@@ -1865,6 +1865,30 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                      adv_pos);
 }
 
+// ---- envelope routing: slot <- hdr[0] on device (pipelined serve) ------
+// kind = hdr[2]: 0 data, 1 stop, 2 flush.  Non-data envelopes route to the
+// dummy KV slot (scratch compute, nothing real is touched) and reset its
+// position so dummy KV writes stay in-bounds.
+__global__ void route_env_kernel(const int* __restrict__ hdr,
+                                 int* __restrict__ slot_out,
+                                 int* __restrict__ pos_table,
+                                 int dummy_slot) {
+  if (threadIdx.x == 0) {
+    int s = hdr[0];
+    if (hdr[2] != 0 || s < 0 || s >= dummy_slot) {
+      s = dummy_slot;
+      pos_table[dummy_slot] = 0;
+    }
+    slot_out[0] = s;
+  }
+}
+
+void launch_route_env(const int* hdr, int* slot_out, int* pos_table,
+                      int dummy_slot, hipStream_t stream) {
+  hipLaunchKernelGGL(route_env_kernel, dim3(1), dim3(64), 0, stream, hdr,
+                     slot_out, pos_table, dummy_slot);
+}
+
 void launch_stage_slot(int* pos_out, int* token_out, const int* pos_table,
                        const int* token_table, int* pos_table_mut,
                        const int* slot, int adv_pos, hipStream_t stream) {

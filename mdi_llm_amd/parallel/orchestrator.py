@@ -148,6 +148,11 @@ class MDIRuntime:
 
         # 1. configure secondaries over HTTP (reference model_dist.py:402-484)
         client = ControlClient()
+        # envelope (pipelined) serve: device-side header routing on GPU
+        # secondaries, no per-hop host syncs; MDI_ENV_SERVE=0/1 overrides
+        env_serve = self.device.type == "cuda"
+        if os.environ.get("MDI_ENV_SERVE"):
+            env_serve = os.environ["MDI_ENV_SERVE"] not in ("0", "false")
         for r in range(1, self.world):
             addr, port = self.topology.http_endpoint(r)
             msg = {
@@ -161,6 +166,7 @@ class MDIRuntime:
                 "max_seq_length": seq_len,
                 "master_addr": self.topology.master_addr,
                 "master_port": self.topology.master_port,
+                "env_serve": env_serve,
             }
             # path hint works on shared/local filesystems; params are pushed
             # when the chunks were split just now (the secondary host may not
@@ -232,6 +238,7 @@ class MDIRuntime:
             tokens_per_sample,
             SamplingParams(temperature, top_k, top_p, seed),
             stop_tokens=stop_tokens,
+            env=env_serve and self.world > 1,
         )
 
         # 4. teardown (PUT /stop; reference model_dist.py:486-497)
@@ -285,13 +292,18 @@ class MDIRuntime:
         self.stage.max_seq_length = seq_len
         self.stage.eval()
 
-        self.runner = make_runner(self.stage, n_samples, self.device)
+        env_serve = bool(msg.get("env_serve"))
+        # +1 KV slot: the envelope serve routes stop/flush envelopes to a
+        # scratch slot on device
+        self.runner = make_runner(self.stage,
+                                  n_samples + (1 if env_serve else 0),
+                                  self.device)
         self.comm = RingComm(config.n_embd, seq_len, self.device, n_samples,
                              dtype=self.dtype)
         self.runtime = PipelineRuntime(self.runner, msg["rank"],
                                        msg["world"], self.comm, self.device)
-        self._log("serving")
-        self.runtime.serve()
+        self._log(f"serving (env={env_serve})")
+        self.runtime.serve(env=env_serve, n_samples=n_samples)
         dist.barrier()
         dist.destroy_process_group()
         # wait for the control-plane stop before exiting

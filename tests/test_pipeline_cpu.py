@@ -47,7 +47,8 @@ def _standalone_reference(tmp, model="nano-test"):
     return [s.tolist() for s in res.sequences]
 
 
-def _worker(rank, world, tmp, port, out_file, model="nano-test"):
+def _worker(rank, world, tmp, port, out_file, model="nano-test",
+            env=False):
     import torch.distributed as dist
 
     from mdi_llm_amd.models.stages import build_stage
@@ -77,10 +78,11 @@ def _worker(rank, world, tmp, port, out_file, model="nano-test"):
         prompts = torch.load(os.path.join(tmp, "prompts.pt"),
                              weights_only=True)
         res = rt.generate(prompts, MAX_NEW,
-                          SamplingParams(temperature=0.8, top_k=50, seed=42))
+                          SamplingParams(temperature=0.8, top_k=50, seed=42),
+                          env=env)
         torch.save([s.tolist() for s in res.sequences], out_file)
     else:
-        rt.serve()
+        rt.serve(env=env, n_samples=N_SAMPLES)
     dist.barrier()
     dist.destroy_process_group()
 
@@ -96,6 +98,22 @@ def test_pipeline_matches_standalone(world, tmp_path):
     ctx = mp.spawn(
         _worker, args=(world, tmp, port, out_file), nprocs=world, join=True
     )
+    got = torch.load(out_file, weights_only=True)
+    assert got == ref
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_pipeline_env_mode_matches_standalone(world, tmp_path):
+    """Envelope (pipelined-serve) protocol: fixed hdr+payload pairs,
+    stop/flush envelopes, pre-posted recv window — same tokens as
+    standalone (host-routed fallback on CPU)."""
+    tmp = str(tmp_path)
+    _build_and_save(tmp)
+    ref = _standalone_reference(tmp)
+    port = 29651 + world
+    out_file = os.path.join(tmp, "out.pt")
+    mp.spawn(_worker, args=(world, tmp, port, out_file, "nano-test", True),
+             nprocs=world, join=True)
     got = torch.load(out_file, weights_only=True)
     assert got == ref
 

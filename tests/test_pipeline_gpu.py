@@ -66,7 +66,8 @@ def _standalone_hip(tmp, model=MODEL, top_p=1.0):
     return [s.tolist() for s in res.sequences]
 
 
-def _worker(rank, world, tmp, port, out_file, top_p, model=MODEL):
+def _worker(rank, world, tmp, port, out_file, top_p, model=MODEL,
+            env=False):
     import torch.distributed as dist
 
     from mdi_llm_amd import ModelConfig
@@ -91,7 +92,8 @@ def _worker(rank, world, tmp, port, out_file, top_p, model=MODEL):
     stage.load_state_dict(chunks[rank])
     stage.eval()
 
-    runner = make_runner(stage, N_SAMPLES, device)
+    n_slots = N_SAMPLES + (1 if (env and rank > 0) else 0)
+    runner = make_runner(stage, n_slots, device)
     assert runner.backend == "hip", "HIP engine must drive the GPU pipeline"
     comm = RingComm(cfg.n_embd, stage.max_seq_length, device, N_SAMPLES,
                     dtype=torch.bfloat16)
@@ -105,10 +107,11 @@ def _worker(rank, world, tmp, port, out_file, top_p, model=MODEL):
         res = rt.generate(
             [p.to(device) for p in prompts], MAX_NEW,
             SamplingParams(temperature=0.8, top_k=50, top_p=top_p, seed=42),
+            env=env,
         )
         torch.save([s.tolist() for s in res.sequences], out_file)
     else:
-        rt.serve()
+        rt.serve(env=env, n_samples=N_SAMPLES)
     dist.barrier()
     dist.destroy_process_group()
 
@@ -123,6 +126,22 @@ def test_gpu_pipeline_matches_standalone(world, tmp_path):
     port = 29711 + world
     mp.spawn(_worker, args=(world, tmp, port, out_file, 1.0), nprocs=world,
              join=True)
+    got = torch.load(out_file, weights_only=True)
+    assert got == ref
+
+
+@pytest.mark.parametrize("world", [2, 3])
+def test_gpu_pipeline_env_matches_standalone(world, tmp_path):
+    """Envelope serve on the HIP engine: device-side header routing
+    (route_env graph), pre-posted recv pairs, stop/flush protocol —
+    token-exact vs standalone on the GPU."""
+    tmp = str(tmp_path)
+    _build_and_save(tmp)
+    ref = _standalone_hip(tmp)
+    out_file = os.path.join(tmp, "out.pt")
+    port = 29721 + world
+    mp.spawn(_worker, args=(world, tmp, port, out_file, 1.0, MODEL, True),
+             nprocs=world, join=True)
     got = torch.load(out_file, weights_only=True)
     assert got == ref
 
