@@ -216,6 +216,24 @@ class DecodeEngine:
         )
         self.pos = torch.zeros(1, device=dev, dtype=torch.int32)
 
+        # fused attention+proj (one launch, in-launch granule hand-off):
+        # the proj weight stream overlaps the attention compute instead of
+        # serializing behind it (see attn_proj_kernel in decode_kernels.hip)
+        K_attn = cfg.n_head * cfg.head_size
+        self._fuse_attn_proj = (
+            not self.fp8
+            and not cfg.parallel_residual
+            and stage.max_seq_length <= 4096
+            and K_attn % 128 == 0
+            and K_attn // 2 <= 4096
+            and os.environ.get("MDI_FUSE_ATTN_PROJ", "1") not in ("0",)
+        )
+        if self._fuse_attn_proj:
+            self.y_gran = torch.zeros(
+                len(self.blocks), K_attn // 2, device=dev,
+                dtype=torch.int64,
+            )
+
         self._r_qkv = self._rows(cfg.qkv_dim, E, "MDI_ROWS_QKV")
         self._r_proj = self._rows(E, cfg.n_head * cfg.head_size,
                                   "MDI_ROWS_PROJ")
@@ -241,6 +259,16 @@ class DecodeEngine:
     def set_slot_pos(self, slot: int, pos: int) -> None:
         """Host-side bookkeeping (prefill / sample init)."""
         self.pos_table[slot] = pos
+
+    def clear_scratch(self) -> None:
+        """Zero granule hand-off buffers (stale (slot, pos) tags from a
+        previous run would otherwise satisfy a consumer sweep early)."""
+        if getattr(self, "y_gran", None) is not None:
+            self.y_gran.zero_()
+        for bufs in getattr(self, "_lane_bufs", None) or []:
+            g = bufs.get("y_gran")
+            if g is not None:
+                g.zero_()
 
     def _stage_pos(self) -> None:
         # pos <- pos_table[slot]   (inside the graph: slot is a device
@@ -314,6 +342,15 @@ class DecodeEngine:
             else:
                 ops.gemv(self.qkv, w.attn_w, self.x, w.attn_b, None, 0,
                          w.norm1_w, w.norm1_b, nk, eps, self._r_qkv)
+            if self._fuse_attn_proj:
+                # attention + proj + residual add in ONE launch; the proj
+                # rows stream into LDS while the attention blocks run
+                ops.attn_proj(self.a, self.qkv, self.kv_pool.k,
+                              self.kv_pool.v, self.cos, self.sin, self.pos,
+                              self.slot, li, scale, w.proj_w, w.proj_b,
+                              self.x, self.y_gran[li])
+                self._mlp(self.a, w, self.a, w.norm2_w, w.norm2_b)
+                continue
             ops.attn_decode(
                 self.y, self.part_o, self.part_ml, self.qkv, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slot, li,
@@ -449,6 +486,7 @@ class DecodeEngine:
         # zero any cache rows the warm-up touched
         self.kv_pool.k.zero_()
         self.kv_pool.v.zero_()
+        self.clear_scratch()
 
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
@@ -556,7 +594,7 @@ class DecodeEngine:
     _LANE_ATTRS = ("x", "qkv", "y", "a", "act", "m_out", "part_o",
                    "part_ml", "logits", "token", "pos_emb",
                    "sample_scratch", "sample_out", "slot", "pos", "xn",
-                   "gate_logits", "moe_eidx", "moe_escale")
+                   "gate_logits", "moe_eidx", "moe_escale", "y_gran")
 
     def _set_lane(self, lane: int) -> None:
         for a, t_ in self._lane_bufs[lane].items():
@@ -601,6 +639,7 @@ class DecodeEngine:
         self.kv_pool.v.zero_()
         self.pos_table.zero_()
         self.token_table.zero_()
+        self.clear_scratch()
 
         base = {a: getattr(self, a) for a in self._LANE_ATTRS
                 if getattr(self, a, None) is not None}
@@ -743,6 +782,7 @@ class DecodeEngine:
         self.kv_pool.k.zero_()
         self.kv_pool.v.zero_()
         self.pos_table.zero_()
+        self.clear_scratch()
         g = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g):
             seq()

@@ -328,6 +328,50 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
               " head_size=", hs);
 }
 
+void attn_proj(torch::Tensor out, torch::Tensor qkv, torch::Tensor kpool,
+               torch::Tensor vpool, torch::Tensor cos_t,
+               torch::Tensor sin_t, torch::Tensor pos, torch::Tensor slot,
+               int64_t layer, double scale, torch::Tensor W,
+               c10::optional<torch::Tensor> bias,
+               c10::optional<torch::Tensor> res, torch::Tensor gran) {
+  check_bf16(out, "out");
+  check_bf16(qkv, "qkv");
+  check_bf16(kpool, "kpool");
+  check_bf16(vpool, "vpool");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
+  check_i32(pos, "pos");
+  check_i32(slot, "slot");
+  check_bf16(W, "W");
+  TORCH_CHECK(gran.is_cuda() && gran.scalar_type() == torch::kInt64,
+              "gran must be int64 on GPU");
+  const int n_layers_pool = (int)kpool.size(1);
+  const int n_kv = (int)kpool.size(2);
+  const int max_seq = (int)kpool.size(3);
+  const int hs = (int)kpool.size(4);
+  const int rope_ne = cos_t.dim() > 1 ? (int)cos_t.size(1) : 0;
+  const int qkv_dim = (int)qkv.numel();
+  const int qpk = qkv_dim / (n_kv * hs) - 2;
+  const int K = n_kv * qpk * hs;
+  const int M = (int)W.size(0);
+  TORCH_CHECK((int)W.size(1) == K, "proj W inner dim mismatch");
+  TORCH_CHECK(out.numel() >= M, "out too small");
+  TORCH_CHECK(gran.numel() >= K / 2, "granule buffer too small");
+  TORCH_CHECK(K % 128 == 0 && K / 2 <= 4096,
+              "attn_proj: unsupported K for granule sweep");
+  int rc = launch_attn_proj(
+      out.data_ptr(), qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      rope_ne ? cos_t.data_ptr<float>() : nullptr,
+      rope_ne ? sin_t.data_ptr<float>() : nullptr, rope_ne,
+      pos.data_ptr<int>(), slot.data_ptr<int>(), (int)layer, n_layers_pool,
+      n_kv, max_seq, hs, qpk, (float)scale, W.data_ptr(),
+      bias.has_value() ? bias->data_ptr() : nullptr,
+      res.has_value() ? res->data_ptr() : nullptr, gran.data_ptr(), M,
+      cur_stream());
+  TORCH_CHECK(rc == 0, "attn_proj: unsupported geometry qpk=", qpk,
+              " head_size=", hs, " max_seq=", max_seq);
+}
+
 void sample(torch::Tensor out_token, torch::Tensor logits,
             torch::Tensor scratch, double temperature, int64_t top_k,
             bool noise, int64_t seed, c10::optional<torch::Tensor> pos,
@@ -514,6 +558,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "prefill: rope q/k for T positions + append k/v to the pool");
   m.def("prefill_attn", &prefill_attn,
         "causal GQA prefill flash attention (MFMA, online softmax)");
+  m.def("attn_proj", &attn_proj,
+        "fused GQA flash-decode attention + output projection (one "
+        "launch; in-launch granule hand-off overlaps the proj weight "
+        "stream with attention)",
+        py::arg("out"), py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
+        py::arg("cos"), py::arg("sin"), py::arg("pos"), py::arg("slot"),
+        py::arg("layer"), py::arg("scale"), py::arg("W"),
+        py::arg("bias"), py::arg("res"), py::arg("gran"));
   m.def("sample", &sample,
         "fused temperature/top-k/gumbel token sampling (128k vocab ~15us)",
         py::arg("out_token"), py::arg("logits"), py::arg("scratch"),

@@ -82,6 +82,16 @@ void launch_add(void* out, const void* a, const void* b, int n,
 // fused temperature/top-k/top-p/gumbel sampling; scratch: >=520 u32 PER
 // SAMPLE, zeroed initially (self-cleaning); n_batch draws from
 // [n_batch, V] logits.  The gumbel stream is keyed by (seed, slot, pos) —
+// fused attention + output-projection (one launch; granule hand-off
+// inside).  Returns -1 when the geometry/max_seq has no instantiation.
+int launch_attn_proj(void* out, const void* qkv, void* kpool, void* vpool,
+                     const float* cos_t, const float* sin_t, int rope_ne,
+                     const int* pos, const int* slot, int layer,
+                     int n_layers_pool, int n_kv_heads, int max_seq,
+                     int head_size, int qpk, float scale, const void* W,
+                     const void* bias, const void* res, void* gran, int M,
+                     hipStream_t stream);
+
 // reproducible and schedule-independent.  When token_table/pos_table and
 // adv_slot are given (single-sample mode), the unpack step also writes
 // token_table[slot] and advances pos_table[slot] by adv_pos in the same

@@ -1023,6 +1023,292 @@ __global__ void attn_decode_block_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// Fused attention + output-projection (B=1 decode, max_seq <= 4096).
+//
+// Why fuse: the proj GEMV's weight stream (M*K bf16, e.g. 33.6 MB for
+// Llama-3-8B) has NO data dependency on the attention output — only the
+// dot phase does.  As two launches, the stream cannot start until
+// attention finishes, so the layer pays attention latency (~10 us, the
+// 8-block kernel is latency-bound at decode S) PLUS the proj kernel's
+// cold-start ramp (~11 us measured: profiles/decode_token_profile_r01.md).
+// In ONE launch, blocks [0, n_kv) run the attention (identical to
+// attn_decode_block_kernel) and publish y as 8-byte {tag, 2xbf16}
+// granules (guide Guideline 16 R2: the data IS the flag — relaxed
+// agent-scope atomics, no fences); blocks [n_kv, n_kv+PB) stage their
+// proj W rows into LDS at full HBM rate MEANWHILE, sweep the granules
+// once attention lands, and dot from LDS.  The hop price sits where the
+// price table says it belongs: in work that overlaps streaming a launch
+// boundary would serialize.
+//
+// Tagging: tag = slot*max_seq + (pos+1) — unique per (slot, position),
+// never 0, identical across a hipGraph replay's captured arguments (the
+// kernel reads pos/slot from device scalars), so granule buffers need
+// zeroing only at engine reset, not per launch.
+// Spin bound: consumers give up after ~2e6 sweep passes and poison y
+// with NaN (loud wrong, never a hung GPU).
+// ---------------------------------------------------------------------------
+typedef __attribute__((address_space(1))) unsigned long long gu64_t;
+
+template <int QPK, int HS>
+__global__ void attn_proj_kernel(
+    bf16* __restrict__ out,       // [M] = res + bias + W @ y
+    const bf16* __restrict__ qkv, // [qkv_dim] interleaved, RAW
+    bf16* __restrict__ kpool, bf16* __restrict__ vpool,
+    const float* __restrict__ cos_t, const float* __restrict__ sin_t,
+    int rope_ne, const int* __restrict__ pos_p,
+    const int* __restrict__ slot_p, int layer, int n_layers_pool,
+    int n_kv_heads, int max_seq, float scale,
+    const bf16* __restrict__ W,    // [M, K] proj weight, K = n_head*HS
+    const bf16* __restrict__ bias, // [M] or null
+    const bf16* __restrict__ res,  // [M] residual (x)
+    unsigned long long* __restrict__ gran,  // [K/2] per-layer granules
+    int M, int R) {
+  constexpr int head_size = HS;
+  constexpr int ODIM = HS * QPK / 64;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const int K = n_kv_heads * QPK * head_size;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int S = pos_p[0] + 1;
+  const int pos = S - 1;
+  const int slot = slot_p[0];
+  const unsigned tag = (unsigned)(slot * max_seq + S);
+  gu64_t* g64 = (gu64_t*)gran;
+
+  if ((int)blockIdx.x >= n_kv_heads) {
+    // ------------------------- projection role -------------------------
+    const int pb = (int)blockIdx.x - n_kv_heads;
+    const int BR = 4 * R;
+    const int row0 = pb * BR;
+    bf16* wlds = reinterpret_cast<bf16*>(smem);       // [BR, K]
+    bf16* ylds = wlds + (size_t)BR * K;               // [K]
+    if (wave > 0) {
+      // waves 1..3: stage BR rows of W (nt: streamed once)
+      const int chunks = BR * (K / 8);
+      for (int i = (wave - 1) * 64 + lane; i < chunks; i += 192) {
+        const int r = i / (K / 8);
+        const int c = i - r * (K / 8);
+        const int grow = min(row0 + r, M - 1);
+        bf16x8 wv = load8_nt(W + (size_t)grow * K + c * 8);
+        *reinterpret_cast<int4*>(wlds + (size_t)r * K + c * 8) =
+            *reinterpret_cast<int4*>(wv.v);
+      }
+    } else {
+      // wave 0: sweep the y granules into LDS as they land
+      const int n_own = (K / 2) / 64;  // granules per lane (<= 64)
+      unsigned long long done = 0;
+      int remaining = n_own;
+      unsigned spins = 0;
+      while (remaining > 0) {
+        for (int kk = 0; kk < n_own; ++kk) {
+          if (done & (1ull << kk)) continue;
+          const int gi = kk * 64 + lane;
+          const unsigned long long v = __hip_atomic_load(
+              &g64[gi], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+          if ((unsigned)(v >> 32) == tag) {
+            *reinterpret_cast<unsigned*>(ylds + (size_t)gi * 2) =
+                (unsigned)v;
+            done |= 1ull << kk;
+            --remaining;
+          }
+        }
+        if (remaining > 0) {
+          if (++spins > 2000000u) {  // give up: poison, never hang
+            for (int kk = 0; kk < n_own; ++kk)
+              if (!(done & (1ull << kk)))
+                *reinterpret_cast<unsigned*>(
+                    ylds + (size_t)(kk * 64 + lane) * 2) = 0x7FC07FC0u;
+            remaining = 0;
+          }
+          __builtin_amdgcn_s_sleep(1);
+        }
+      }
+    }
+    __syncthreads();
+    // dot phase: wave w computes rows [w*R, w*R+R) from LDS
+    for (int r = wave * R; r < wave * R + R; ++r) {
+      const int grow = row0 + r;
+      float acc = 0.f;
+      for (int i = lane * 8; i < K; i += 64 * 8) {
+        bf16x8 wv = load8(wlds + (size_t)r * K + i);
+        bf16x8 yv = load8(ylds + i);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc += b2f(wv.v[j]) * b2f(yv.v[j]);
+      }
+      acc = wave_reduce_sum(acc);
+      if (lane == 0 && grow < M) {
+        float a = acc;
+        if (bias != nullptr) a += b2f(bias[grow]);
+        if (res != nullptr) a += b2f(res[grow]);
+        out[grow] = f2b(a);
+      }
+    }
+    return;
+  }
+
+  // --------------------------- attention role ---------------------------
+  // identical computation to attn_decode_block_kernel<QPK, HS, 0>, with
+  // the combine epilogue publishing granules instead of storing y
+  bf16* q_lds = reinterpret_cast<bf16*>(smem);      // [16 * HS] (swizzled)
+  bf16* k_cur = q_lds + 16 * head_size;             // [HS]
+  float* o_part = reinterpret_cast<float*>(
+      smem + ((17 * head_size * 2 + 15) & ~15));    // [4][QPK][HS]
+  float* ml_part = o_part + ATTN_WAVES * QPK * head_size;  // [4][QPK][2]
+
+  const int g = (int)blockIdx.x;
+
+  const size_t cache_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq * head_size;
+  const bf16* krow_cur = qkv + ((size_t)g * (QPK + 2) + QPK) * head_size;
+  const bf16* vrow_cur = krow_cur + head_size;
+
+  for (int i = tid; i < (16 - QPK) * head_size / 8; i += 256)
+    reinterpret_cast<bf16x8_t*>(q_lds + QPK * head_size)[i] = bf16x8_t{};
+  for (int i = tid; i < QPK * head_size; i += 256) {
+    const int r = i / head_size;
+    const int d = i % head_size;
+    const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
+    q_lds[q_swz<HS>(r, d)] =
+        f2b(rope_elem(qrow, d, rope_ne, cos_t, sin_t, pos));
+  }
+  for (int d = tid; d < head_size; d += 256)
+    k_cur[d] = f2b(rope_elem(krow_cur, d, rope_ne, cos_t, sin_t, pos));
+  __syncthreads();
+
+  for (int d = tid; d < head_size; d += 256) {
+    kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
+    vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
+  }
+
+  const int qb = lane % QPK;
+  const int sub = lane >> 4;
+  const int d0 = (lane / QPK) * ODIM;
+
+  float m_st = -1e30f, l_st = 0.f;
+  float o_acc[ODIM];
+#pragma unroll
+  for (int i = 0; i < ODIM; ++i) o_acc[i] = 0.f;
+
+  const int n_tiles = (S + 15) / 16;
+  for (int t = wave; t < n_tiles; t += ATTN_WAVES) {
+    const int key0 = t * 16;
+    f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15;
+    const int koff = (lane >> 4) * 8;
+    const int akey = key0 + arow;
+    const bf16* krow = kpool + cache_base + (size_t)akey * head_size;
+    const bool row_valid = akey < S;
+    const bool row_cur = akey == pos;
+#pragma unroll
+    for (int c = 0; c < HS / 32; ++c) {
+      bf16x8_t af = {};
+      if (row_cur)
+        af = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
+      else if (row_valid)
+        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      const bf16x8_t bq = *reinterpret_cast<const bf16x8_t*>(
+          &q_lds[q_swz<HS>(arow, c * 32 + koff)]);
+      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bq, acc4, 0, 0, 0);
+    }
+    float sc[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = key0 + sub * 4 + r;
+      sc[r] = (key < S) ? acc4[r] * scale : -1e30f;
+    }
+    float tmax = fmaxf(fmaxf(sc[0], sc[1]), fmaxf(sc[2], sc[3]));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float m_new = fmaxf(m_st, tmax);
+    const float alpha = __expf(m_st - m_new);
+    m_st = m_new;
+    float e[4];
+    float psum = 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int key = key0 + sub * 4 + r;
+      e[r] = (key < S) ? __expf(sc[r] - m_new) : 0.f;
+      psum += e[r];
+    }
+    psum += __shfl_xor(psum, 16, 64);
+    psum += __shfl_xor(psum, 32, 64);
+    l_st = l_st * alpha + psum;
+    const float alphaB = __shfl(alpha, qb, 64);
+#pragma unroll
+    for (int i = 0; i < ODIM; ++i) o_acc[i] *= alphaB;
+#pragma unroll
+    for (int sI = 0; sI < 16; ++sI) {
+      const float pv = __shfl(e[sI & 3], ((sI >> 2) << 4) | qb, 64);
+      const int skey = min(key0 + sI, pos);
+      const bf16* vrow =
+          (skey == pos)
+              ? vrow_cur + d0
+              : vpool + cache_base + (size_t)skey * head_size + d0;
+      if constexpr (ODIM >= 8) {
+#pragma unroll
+        for (int i = 0; i < ODIM; i += 8) {
+          bf16x8 vv = load8(vrow + i);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) o_acc[i + j] += pv * b2f(vv.v[j]);
+        }
+      } else if constexpr (ODIM == 4) {
+        int2 raw = *reinterpret_cast<const int2*>(vrow);
+        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o_acc[j] += pv * b2f(vv[j]);
+      } else if constexpr (ODIM == 2) {
+        int raw = *reinterpret_cast<const int*>(vrow);
+        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+        o_acc[0] += pv * b2f(vv[0]);
+        o_acc[1] += pv * b2f(vv[1]);
+      } else {
+        o_acc[0] += pv * b2f(vrow[0]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < ODIM; ++i)
+    o_part[(wave * QPK + qb) * head_size + d0 + i] = o_acc[i];
+  if (lane < QPK) {
+    ml_part[(wave * QPK + lane) * 2 + 0] = m_st;
+    ml_part[(wave * QPK + lane) * 2 + 1] = l_st;
+  }
+  __syncthreads();
+
+  // combine + publish granule pairs {tag, y[2i], y[2i+1]}
+  for (int i = tid; i < QPK * head_size / 2; i += 256) {
+    const int e0 = 2 * i;
+    const int h = e0 / head_size;
+    const int d = e0 % head_size;
+    float M_ = -1e30f;
+#pragma unroll
+    for (int w = 0; w < ATTN_WAVES; ++w)
+      M_ = fmaxf(M_, ml_part[(w * QPK + h) * 2 + 0]);
+    float den = 0.f, num0 = 0.f, num1 = 0.f;
+#pragma unroll
+    for (int w = 0; w < ATTN_WAVES; ++w) {
+      const float wgt = __expf(ml_part[(w * QPK + h) * 2 + 0] - M_);
+      den += wgt * ml_part[(w * QPK + h) * 2 + 1];
+      num0 += wgt * o_part[(w * QPK + h) * head_size + d];
+      num1 += wgt * o_part[(w * QPK + h) * head_size + d + 1];
+    }
+    const bf16 y0 = f2b(num0 / den);
+    const bf16 y1 = f2b(num1 / den);
+    const unsigned pack =
+        (unsigned)*reinterpret_cast<const unsigned short*>(&y0) |
+        ((unsigned)*reinterpret_cast<const unsigned short*>(&y1) << 16);
+    const int gi = (g * QPK * head_size) / 2 + i;
+    __hip_atomic_store(&g64[gi],
+                       ((unsigned long long)tag << 32) | pack,
+                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  }
+}
+
 // BATCH: 0 -> single-token mode (pos_p/slot_p are device scalars);
 //        1 -> batched mode: pos_p/slot_p are arrays [n_batch], and
 //             qkv/part_o/part_ml/out get a leading batch dimension.
@@ -1895,6 +2181,107 @@ void launch_stage_slot(int* pos_out, int* token_out, const int* pos_table,
   hipLaunchKernelGGL(stage_slot_kernel, dim3(1), dim3(64), 0, stream,
                      pos_out, token_out, pos_table, token_table,
                      pos_table_mut, slot, adv_pos);
+}
+
+// fused attention+proj launcher.  Returns 0 on success, -1 when the
+// geometry has no instantiation (caller falls back to the split path).
+template <int QPK, int HS>
+static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
+                                void* vpool, const float* cos_t,
+                                const float* sin_t, int rope_ne,
+                                const int* pos, const int* slot, int layer,
+                                int n_layers_pool, int n_kv_heads,
+                                int max_seq, float scale, const void* W,
+                                const void* bias, const void* res,
+                                void* gran, int M, hipStream_t stream) {
+  const int K = n_kv_heads * QPK * HS;
+  int R = 2;
+  if ((size_t)(4 * R + 1) * K * 2 > 79 * 1024) R = 1;
+  const int BR = 4 * R;
+  const int PB = (M + BR - 1) / BR;
+  size_t smem_proj = (size_t)(BR + 1) * K * 2;
+  size_t smem_attn = (size_t)((17 * HS * 2 + 15) & ~15) +
+                     (size_t)ATTN_WAVES * QPK * HS * 4 +
+                     (size_t)ATTN_WAVES * QPK * 2 * 4;
+  size_t smem = smem_proj > smem_attn ? smem_proj : smem_attn;
+  static bool attr_set = false;
+  if (!attr_set) {
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&attn_proj_kernel<QPK, HS>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    attr_set = true;
+  }
+  hipLaunchKernelGGL((attn_proj_kernel<QPK, HS>), dim3(n_kv_heads + PB),
+                     dim3(256), smem, stream, (bf16*)out, (const bf16*)qkv,
+                     (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
+                     slot, layer, n_layers_pool, n_kv_heads, max_seq, scale,
+                     (const bf16*)W, (const bf16*)bias, (const bf16*)res,
+                     (unsigned long long*)gran, M, R);
+}
+
+template <int QPK>
+static int attn_proj_dispatch1(int head_size, void* out, const void* qkv,
+                               void* kpool, void* vpool, const float* cos_t,
+                               const float* sin_t, int rope_ne,
+                               const int* pos, const int* slot, int layer,
+                               int n_layers_pool, int n_kv_heads,
+                               int max_seq, float scale, const void* W,
+                               const void* bias, const void* res, void* gran,
+                               int M, hipStream_t stream) {
+#define CASE_HS_P(H)                                                        \
+  if (head_size == H) {                                                     \
+    if constexpr (QPK * H >= 64) {                                          \
+      attn_proj_dispatch2<QPK, H>(out, qkv, kpool, vpool, cos_t, sin_t,     \
+                                  rope_ne, pos, slot, layer, n_layers_pool, \
+                                  n_kv_heads, max_seq, scale, W, bias, res, \
+                                  gran, M, stream);                         \
+      return 0;                                                             \
+    }                                                                       \
+  }
+  CASE_HS_P(64)
+  CASE_HS_P(128)
+  CASE_HS_P(256)
+#undef CASE_HS_P
+  return -1;
+}
+
+int launch_attn_proj(void* out, const void* qkv, void* kpool, void* vpool,
+                     const float* cos_t, const float* sin_t, int rope_ne,
+                     const int* pos, const int* slot, int layer,
+                     int n_layers_pool, int n_kv_heads, int max_seq,
+                     int head_size, int qpk, float scale, const void* W,
+                     const void* bias, const void* res, void* gran, int M,
+                     hipStream_t stream) {
+  if (max_seq > ATTN_BLOCK_MAX_SEQ) return -1;
+  switch (qpk) {
+    case 1:
+      return attn_proj_dispatch1<1>(head_size, out, qkv, kpool, vpool,
+                                    cos_t, sin_t, rope_ne, pos, slot, layer,
+                                    n_layers_pool, n_kv_heads, max_seq,
+                                    scale, W, bias, res, gran, M, stream);
+    case 2:
+      return attn_proj_dispatch1<2>(head_size, out, qkv, kpool, vpool,
+                                    cos_t, sin_t, rope_ne, pos, slot, layer,
+                                    n_layers_pool, n_kv_heads, max_seq,
+                                    scale, W, bias, res, gran, M, stream);
+    case 4:
+      return attn_proj_dispatch1<4>(head_size, out, qkv, kpool, vpool,
+                                    cos_t, sin_t, rope_ne, pos, slot, layer,
+                                    n_layers_pool, n_kv_heads, max_seq,
+                                    scale, W, bias, res, gran, M, stream);
+    case 8:
+      return attn_proj_dispatch1<8>(head_size, out, qkv, kpool, vpool,
+                                    cos_t, sin_t, rope_ne, pos, slot, layer,
+                                    n_layers_pool, n_kv_heads, max_seq,
+                                    scale, W, bias, res, gran, M, stream);
+    case 16:
+      return attn_proj_dispatch1<16>(head_size, out, qkv, kpool, vpool,
+                                     cos_t, sin_t, rope_ne, pos, slot,
+                                     layer, n_layers_pool, n_kv_heads,
+                                     max_seq, scale, W, bias, res, gran, M,
+                                     stream);
+  }
+  return -1;
 }
 
 static inline int gemv_grid(int M, int rows_per_block) {

@@ -90,6 +90,7 @@ class HipRunner(TorchRunner):
     def reset(self) -> None:
         super().reset()
         self.engine.pos_table.zero_()
+        self.engine.clear_scratch()
 
     @torch.inference_mode()
     def prefill_head(self, tokens: torch.Tensor, slot: int) -> torch.Tensor:
