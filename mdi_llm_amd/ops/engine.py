@@ -229,9 +229,10 @@ class DecodeEngine:
             and os.environ.get("MDI_FUSE_ATTN_PROJ", "1") not in ("0",)
         )
         if self._fuse_attn_proj:
+            # per-layer hand-off scratch: [K/2 y pairs][16 flag slots]
             self.y_gran = torch.zeros(
-                len(self.blocks), K_attn // 2, device=dev,
-                dtype=torch.int64,
+                len(self.blocks), K_attn // 2 + 16, device=dev,
+                dtype=torch.int32,
             )
 
         self._r_qkv = self._rows(cfg.qkv_dim, E, "MDI_ROWS_QKV")

@@ -343,8 +343,8 @@ void attn_proj(torch::Tensor out, torch::Tensor qkv, torch::Tensor kpool,
   check_i32(pos, "pos");
   check_i32(slot, "slot");
   check_bf16(W, "W");
-  TORCH_CHECK(gran.is_cuda() && gran.scalar_type() == torch::kInt64,
-              "gran must be int64 on GPU");
+  TORCH_CHECK(gran.is_cuda() && gran.scalar_type() == torch::kInt32,
+              "gran must be int32 on GPU");
   const int n_layers_pool = (int)kpool.size(1);
   const int n_kv = (int)kpool.size(2);
   const int max_seq = (int)kpool.size(3);
@@ -356,7 +356,7 @@ void attn_proj(torch::Tensor out, torch::Tensor qkv, torch::Tensor kpool,
   const int M = (int)W.size(0);
   TORCH_CHECK((int)W.size(1) == K, "proj W inner dim mismatch");
   TORCH_CHECK(out.numel() >= M, "out too small");
-  TORCH_CHECK(gran.numel() >= K / 2, "granule buffer too small");
+  TORCH_CHECK(gran.numel() >= K / 2 + 16, "y/flag scratch too small");
   TORCH_CHECK(K % 128 == 0 && K / 2 <= 4096,
               "attn_proj: unsupported K for granule sweep");
   int rc = launch_attn_proj(

@@ -1033,22 +1033,28 @@ __global__ void attn_decode_block_kernel(
 // 8-block kernel is latency-bound at decode S) PLUS the proj kernel's
 // cold-start ramp (~11 us measured: profiles/decode_token_profile_r01.md).
 // In ONE launch, blocks [0, n_kv) run the attention (identical to
-// attn_decode_block_kernel) and publish y as 8-byte {tag, 2xbf16}
-// granules (guide Guideline 16 R2: the data IS the flag — relaxed
-// agent-scope atomics, no fences); blocks [n_kv, n_kv+PB) stage their
-// proj W rows into LDS at full HBM rate MEANWHILE, sweep the granules
-// once attention lands, and dot from LDS.  The hop price sits where the
-// price table says it belongs: in work that overlaps streaming a launch
-// boundary would serialize.
+// attn_decode_block_kernel) and publish y with write-through 4-byte
+// agent-scope stores + ONE tag-valued flag word per kv head (guide
+// Guideline 16 R1: sc1 payload, vmcnt(0) drain per storing wave, relaxed
+// flag); blocks [n_kv, n_kv+PB) stage their proj W rows into LDS at full
+// HBM rate MEANWHILE, then ONE LANE polls the n_kv flag words, one
+// agent-scope acquire fence covers the block, and y is bulk-loaded with
+// plain coalesced reads.  (A first version swept 8-byte data-tagged
+// granules from every consumer block — measured 285 vs 320 tok/s on
+// 8B: 512 sweepers re-reading 16 KB all phase starve the weight stream,
+// the price table's polling-cost row.  One lane polling n_kv words is
+// ~1000x less poll traffic.)
 //
 // Tagging: tag = slot*max_seq + (pos+1) — unique per (slot, position),
 // never 0, identical across a hipGraph replay's captured arguments (the
-// kernel reads pos/slot from device scalars), so granule buffers need
+// kernel reads pos/slot from device scalars), so the scratch needs
 // zeroing only at engine reset, not per launch.
-// Spin bound: consumers give up after ~2e6 sweep passes and poison y
-// with NaN (loud wrong, never a hung GPU).
+// Spin bound: consumers give up after ~5e6 polls and poison y with NaN
+// (loud wrong, never a hung GPU).
+//
+// Per-layer scratch layout (int32): [K/2 y-pairs][16 flag slots].
 // ---------------------------------------------------------------------------
-typedef __attribute__((address_space(1))) unsigned long long gu64_t;
+typedef __attribute__((address_space(1))) unsigned int gu32_t;
 
 template <int QPK, int HS>
 __global__ void attn_proj_kernel(
@@ -1062,7 +1068,7 @@ __global__ void attn_proj_kernel(
     const bf16* __restrict__ W,    // [M, K] proj weight, K = n_head*HS
     const bf16* __restrict__ bias, // [M] or null
     const bf16* __restrict__ res,  // [M] residual (x)
-    unsigned long long* __restrict__ gran,  // [K/2] per-layer granules
+    unsigned int* __restrict__ gran,  // [K/2 + 16] per-layer y + flags
     int M, int R) {
   constexpr int head_size = HS;
   constexpr int ODIM = HS * QPK / 64;
@@ -1076,7 +1082,8 @@ __global__ void attn_proj_kernel(
   const int pos = S - 1;
   const int slot = slot_p[0];
   const unsigned tag = (unsigned)(slot * max_seq + S);
-  gu64_t* g64 = (gu64_t*)gran;
+  gu32_t* g32 = (gu32_t*)gran;
+  gu32_t* flags = g32 + K / 2;  // [n_kv_heads] tag-valued flag words
 
   if ((int)blockIdx.x >= n_kv_heads) {
     // ------------------------- projection role -------------------------
@@ -1085,10 +1092,11 @@ __global__ void attn_proj_kernel(
     const int row0 = pb * BR;
     bf16* wlds = reinterpret_cast<bf16*>(smem);       // [BR, K]
     bf16* ylds = wlds + (size_t)BR * K;               // [K]
-    if (wave > 0) {
-      // waves 1..3: stage BR rows of W (nt: streamed once)
+    // all 4 waves stage BR rows of W (nt: streamed once); zero poll
+    // traffic while the weight stream runs
+    {
       const int chunks = BR * (K / 8);
-      for (int i = (wave - 1) * 64 + lane; i < chunks; i += 192) {
+      for (int i = tid; i < chunks; i += 256) {
         const int r = i / (K / 8);
         const int c = i - r * (K / 8);
         const int grow = min(row0 + r, M - 1);
@@ -1096,35 +1104,34 @@ __global__ void attn_proj_kernel(
         *reinterpret_cast<int4*>(wlds + (size_t)r * K + c * 8) =
             *reinterpret_cast<int4*>(wv.v);
       }
-    } else {
-      // wave 0: sweep the y granules into LDS as they land
-      const int n_own = (K / 2) / 64;  // granules per lane (<= 64)
-      unsigned long long done = 0;
-      int remaining = n_own;
+    }
+    // ONE lane polls the flags; one acquire fence + barrier covers the
+    // whole block (guide Guideline 16)
+    if (tid == 0) {
       unsigned spins = 0;
-      while (remaining > 0) {
-        for (int kk = 0; kk < n_own; ++kk) {
-          if (done & (1ull << kk)) continue;
-          const int gi = kk * 64 + lane;
-          const unsigned long long v = __hip_atomic_load(
-              &g64[gi], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
-          if ((unsigned)(v >> 32) == tag) {
-            *reinterpret_cast<unsigned*>(ylds + (size_t)gi * 2) =
-                (unsigned)v;
-            done |= 1ull << kk;
-            --remaining;
-          }
+      bool ok = true;
+      for (int h = 0; h < n_kv_heads; ++h) {
+        while (__hip_atomic_load(&flags[h], __ATOMIC_RELAXED,
+                                 __HIP_MEMORY_SCOPE_AGENT) != tag) {
+          if (++spins > 5000000u) { ok = false; break; }
+          __builtin_amdgcn_s_sleep(2);
         }
-        if (remaining > 0) {
-          if (++spins > 2000000u) {  // give up: poison, never hang
-            for (int kk = 0; kk < n_own; ++kk)
-              if (!(done & (1ull << kk)))
-                *reinterpret_cast<unsigned*>(
-                    ylds + (size_t)(kk * 64 + lane) * 2) = 0x7FC07FC0u;
-            remaining = 0;
-          }
-          __builtin_amdgcn_s_sleep(1);
-        }
+        if (!ok) break;
+      }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+      // give-up verdict for the whole block (word just past ylds)
+      *reinterpret_cast<int*>(ylds + K) = ok ? 1 : 0;
+    }
+    __syncthreads();
+    const bool y_ok = *reinterpret_cast<const int*>(ylds + K) != 0;
+    // bulk-load y: plain coalesced reads (the acquire dropped stale lines)
+    for (int i = tid * 2; i < K / 2; i += 512) {
+      if (y_ok) {
+        *reinterpret_cast<int2*>(ylds + (size_t)i * 2) =
+            *reinterpret_cast<const int2*>(&gran[i]);
+      } else {
+        *reinterpret_cast<int2*>(ylds + (size_t)i * 2) =
+            int2{0x7FC07FC0, 0x7FC07FC0};  // poison: NaN bf16
       }
     }
     __syncthreads();
@@ -1280,7 +1287,8 @@ __global__ void attn_proj_kernel(
   }
   __syncthreads();
 
-  // combine + publish granule pairs {tag, y[2i], y[2i+1]}
+  // combine + publish: write-through 4-byte pair stores (R1 payload),
+  // per-wave vmcnt(0) drain, then ONE lane stores this head's flag
   for (int i = tid; i < QPK * head_size / 2; i += 256) {
     const int e0 = 2 * i;
     const int h = e0 / head_size;
@@ -1303,10 +1311,15 @@ __global__ void attn_proj_kernel(
         (unsigned)*reinterpret_cast<const unsigned short*>(&y0) |
         ((unsigned)*reinterpret_cast<const unsigned short*>(&y1) << 16);
     const int gi = (g * QPK * head_size) / 2 + i;
-    __hip_atomic_store(&g64[gi],
-                       ((unsigned long long)tag << 32) | pack,
-                       __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    __hip_atomic_store(&g32[gi], pack, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
   }
+  // EVERY storing wave drains its write-through stores before the flag
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  if (tid == 0)
+    __hip_atomic_store(&flags[g], tag, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
 }
 
 // BATCH: 0 -> single-token mode (pos_p/slot_p are device scalars);
@@ -2199,7 +2212,7 @@ static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
   if ((size_t)(4 * R + 1) * K * 2 > 79 * 1024) R = 1;
   const int BR = 4 * R;
   const int PB = (M + BR - 1) / BR;
-  size_t smem_proj = (size_t)(BR + 1) * K * 2;
+  size_t smem_proj = (size_t)(BR + 1) * K * 2 + 16;  // +verdict word
   size_t smem_attn = (size_t)((17 * HS * 2 + 15) & ~15) +
                      (size_t)ATTN_WAVES * QPK * HS * 4 +
                      (size_t)ATTN_WAVES * QPK * 2 * 4;
@@ -2216,7 +2229,7 @@ static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
                      (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
                      slot, layer, n_layers_pool, n_kv_heads, max_seq, scale,
                      (const bf16*)W, (const bf16*)bias, (const bf16*)res,
-                     (unsigned long long*)gran, M, R);
+                     (unsigned int*)gran, M, R);
 }
 
 template <int QPK>
