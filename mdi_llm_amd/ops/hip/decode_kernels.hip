@@ -1069,7 +1069,7 @@ __global__ void attn_proj_kernel(
     const bf16* __restrict__ bias, // [M] or null
     const bf16* __restrict__ res,  // [M] residual (x)
     unsigned int* __restrict__ gran,  // [K/2 + 16] per-layer y + flags
-    int M, int R) {
+    int M, int BR) {
   constexpr int head_size = HS;
   constexpr int ODIM = HS * QPK / 64;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -1087,70 +1087,68 @@ __global__ void attn_proj_kernel(
 
   if ((int)blockIdx.x >= n_kv_heads) {
     // ------------------------- projection role -------------------------
+    // Grid-stride over row groups: the grid is capped at full residency
+    // (2 blocks/CU: 8 waves/CU is the hard wave limit), so no block ever
+    // queues behind a first wave of blocks — a non-resident straggler
+    // staging its tile alone after everyone else cost ~6 us/layer in the
+    // first version of this kernel.
     const int pb = (int)blockIdx.x - n_kv_heads;
-    const int BR = 4 * R;
-    const int row0 = pb * BR;
-    bf16* wlds = reinterpret_cast<bf16*>(smem);       // [BR, K]
-    bf16* ylds = wlds + (size_t)BR * K;               // [K]
-    // all 4 waves stage BR rows of W (nt: streamed once); zero poll
-    // traffic while the weight stream runs
-    {
-      const int chunks = BR * (K / 8);
-      for (int i = tid; i < chunks; i += 256) {
+    const int PBe = (int)gridDim.x - n_kv_heads;
+    bf16* wlds = reinterpret_cast<bf16*>(smem);            // [BR, K]
+    int* vflag = reinterpret_cast<int*>(smem + (size_t)BR * K * 2);
+    const bf16* ygl = reinterpret_cast<const bf16*>(gran); // y after fence
+    bool polled = false;
+    bool y_ok = true;
+    for (int row0 = pb * BR; row0 < M; row0 += PBe * BR) {
+      if (row0 != pb * BR) __syncthreads();  // prev dot read wlds
+      const int nr = min(BR, M - row0);
+      // all 4 waves stage nr rows of W (nt: streamed once)
+      for (int i = tid; i < nr * (K / 8); i += 256) {
         const int r = i / (K / 8);
         const int c = i - r * (K / 8);
-        const int grow = min(row0 + r, M - 1);
-        bf16x8 wv = load8_nt(W + (size_t)grow * K + c * 8);
+        bf16x8 wv = load8_nt(W + (size_t)(row0 + r) * K + c * 8);
         *reinterpret_cast<int4*>(wlds + (size_t)r * K + c * 8) =
             *reinterpret_cast<int4*>(wv.v);
       }
-    }
-    // ONE lane polls the flags; one acquire fence + barrier covers the
-    // whole block (guide Guideline 16)
-    if (tid == 0) {
-      unsigned spins = 0;
-      bool ok = true;
-      for (int h = 0; h < n_kv_heads; ++h) {
-        while (__hip_atomic_load(&flags[h], __ATOMIC_RELAXED,
-                                 __HIP_MEMORY_SCOPE_AGENT) != tag) {
-          if (++spins > 5000000u) { ok = false; break; }
-          __builtin_amdgcn_s_sleep(2);
+      if (!polled && tid == 0) {
+        // ONE lane polls the n_kv flag words; one agent-scope acquire
+        // covers the block after the barrier (guide Guideline 16)
+        unsigned spins = 0;
+        bool ok = true;
+        for (int h = 0; h < n_kv_heads; ++h) {
+          while (__hip_atomic_load(&flags[h], __ATOMIC_RELAXED,
+                                   __HIP_MEMORY_SCOPE_AGENT) != tag) {
+            if (++spins > 5000000u) { ok = false; break; }
+            __builtin_amdgcn_s_sleep(2);
+          }
+          if (!ok) break;
         }
-        if (!ok) break;
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        *vflag = ok ? 1 : 0;
       }
-      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
-      // give-up verdict for the whole block (word just past ylds)
-      *reinterpret_cast<int*>(ylds + K) = ok ? 1 : 0;
-    }
-    __syncthreads();
-    const bool y_ok = *reinterpret_cast<const int*>(ylds + K) != 0;
-    // bulk-load y: plain coalesced reads (the acquire dropped stale lines)
-    for (int i = tid * 2; i < K / 2; i += 512) {
-      if (y_ok) {
-        *reinterpret_cast<int2*>(ylds + (size_t)i * 2) =
-            *reinterpret_cast<const int2*>(&gran[i]);
-      } else {
-        *reinterpret_cast<int2*>(ylds + (size_t)i * 2) =
-            int2{0x7FC07FC0, 0x7FC07FC0};  // poison: NaN bf16
+      __syncthreads();
+      if (!polled) {
+        y_ok = *vflag != 0;
+        polled = true;
       }
-    }
-    __syncthreads();
-    // dot phase: wave w computes rows [w*R, w*R+R) from LDS
-    for (int r = wave * R; r < wave * R + R; ++r) {
-      const int grow = row0 + r;
-      float acc = 0.f;
-      for (int i = lane * 8; i < K; i += 64 * 8) {
-        bf16x8 wv = load8(wlds + (size_t)r * K + i);
-        bf16x8 yv = load8(ylds + i);
+      // dot phase: wave w computes rows r = w, w+4, ... from LDS W and
+      // L2-hot y (8-16 KB, read by every block after the acquire)
+      for (int r = wave; r < nr; r += 4) {
+        float acc = 0.f;
+        for (int i = lane * 8; i < K; i += 64 * 8) {
+          bf16x8 wv = load8(wlds + (size_t)r * K + i);
+          bf16x8 yv = load8(ygl + i);
 #pragma unroll
-        for (int j = 0; j < 8; ++j) acc += b2f(wv.v[j]) * b2f(yv.v[j]);
-      }
-      acc = wave_reduce_sum(acc);
-      if (lane == 0 && grow < M) {
-        float a = acc;
-        if (bias != nullptr) a += b2f(bias[grow]);
-        if (res != nullptr) a += b2f(res[grow]);
-        out[grow] = f2b(a);
+          for (int j = 0; j < 8; ++j) acc += b2f(wv.v[j]) * b2f(yv.v[j]);
+        }
+        acc = wave_reduce_sum(acc);
+        if (lane == 0) {
+          const int grow = row0 + r;
+          float a = y_ok ? acc : __int_as_float(0x7FC00000);  // poison
+          if (bias != nullptr) a += b2f(bias[grow]);
+          if (res != nullptr) a += b2f(res[grow]);
+          out[grow] = f2b(a);
+        }
       }
     }
     return;
@@ -2208,11 +2206,18 @@ static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
                                 const void* bias, const void* res,
                                 void* gran, int M, hipStream_t stream) {
   const int K = n_kv_heads * QPK * HS;
-  int R = 2;
-  if ((size_t)(4 * R + 1) * K * 2 > 79 * 1024) R = 1;
-  const int BR = 4 * R;
-  const int PB = (M + BR - 1) / BR;
-  size_t smem_proj = (size_t)(BR + 1) * K * 2 + 16;  // +verdict word
+  // rows per block group: fit every block resident at once (2 blocks/CU
+  // of 4 waves = the 8-wave/CU limit) AND the W tile in <=79 KB of LDS;
+  // grid-stride covers any remainder
+  const int cap_blocks = 2 * 256 - n_kv_heads;
+  int BR = (M + cap_blocks - 1) / cap_blocks;
+  const int br_lds = (79 * 1024 - 16) / (K * 2);
+  if (br_lds < 1) return;  // K too large for the LDS tile (no instantiation)
+  if (BR > br_lds) BR = br_lds;
+  if (BR < 1) BR = 1;
+  int PB = (M + BR - 1) / BR;
+  if (PB > cap_blocks) PB = cap_blocks;
+  size_t smem_proj = (size_t)BR * K * 2 + 16;  // +verdict word
   size_t smem_attn = (size_t)((17 * HS * 2 + 15) & ~15) +
                      (size_t)ATTN_WAVES * QPK * HS * 4 +
                      (size_t)ATTN_WAVES * QPK * 2 * 4;
@@ -2229,7 +2234,7 @@ static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
                      (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
                      slot, layer, n_layers_pool, n_kv_heads, max_seq, scale,
                      (const bf16*)W, (const bf16*)bias, (const bf16*)res,
-                     (unsigned int*)gran, M, R);
+                     (unsigned int*)gran, M, BR);
 }
 
 template <int QPK>
