@@ -1102,13 +1102,43 @@ __global__ void attn_proj_kernel(
     for (int row0 = pb * BR; row0 < M; row0 += PBe * BR) {
       if (row0 != pb * BR) __syncthreads();  // prev dot read wlds
       const int nr = min(BR, M - row0);
-      // all 4 waves stage nr rows of W (nt: streamed once)
-      for (int i = tid; i < nr * (K / 8); i += 256) {
-        const int r = i / (K / 8);
-        const int c = i - r * (K / 8);
-        bf16x8 wv = load8_nt(W + (size_t)(row0 + r) * K + c * 8);
-        *reinterpret_cast<int4*>(wlds + (size_t)r * K + c * 8) =
-            *reinterpret_cast<int4*>(wv.v);
+      // all 4 waves stage nr rows of W (nt: streamed once).  4 loads in
+      // flight per lane: a serial load->ds_write chain is latency-bound
+      // (one outstanding 16 B load/lane ~ 2 TB/s chip-wide, the very
+      // ramp this kernel exists to remove)
+      {
+        const int chunks = nr * (K / 8);
+        int i = tid;
+        for (; i + 3 * 256 < chunks; i += 4 * 256) {
+          bf16x8 w0, w1, w2, w3;
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const int ii = i + u * 256;
+            const int r = ii / (K / 8);
+            const int c = ii - r * (K / 8);
+            bf16x8 wv = load8_nt(W + (size_t)(row0 + r) * K + c * 8);
+            if (u == 0) w0 = wv;
+            else if (u == 1) w1 = wv;
+            else if (u == 2) w2 = wv;
+            else w3 = wv;
+          }
+#pragma unroll
+          for (int u = 0; u < 4; ++u) {
+            const int ii = i + u * 256;
+            const int r = ii / (K / 8);
+            const int c = ii - r * (K / 8);
+            const bf16x8& wv = u == 0 ? w0 : u == 1 ? w1 : u == 2 ? w2 : w3;
+            *reinterpret_cast<int4*>(wlds + (size_t)r * K + c * 8) =
+                *reinterpret_cast<const int4*>(wv.v);
+          }
+        }
+        for (; i < chunks; i += 256) {
+          const int r = i / (K / 8);
+          const int c = i - r * (K / 8);
+          bf16x8 wv = load8_nt(W + (size_t)(row0 + r) * K + c * 8);
+          *reinterpret_cast<int4*>(wlds + (size_t)r * K + c * 8) =
+              *reinterpret_cast<int4*>(wv.v);
+        }
       }
       if (!polled && tid == 0) {
         // ONE lane polls the n_kv flag words; one agent-scope acquire
