@@ -366,8 +366,11 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
       acc[r] = 0.f;
     }
     float s2 = 0.f;
-    // unroll 2: two independent W chunks in flight per lane
-#pragma unroll 2
+    // unroll 4: four independent W chunks in flight per lane (2 measured
+    // only ~3.5 TB/s on the short-K qkv/proj shapes — not enough bytes
+    // in flight per CU; 4 puts ~32 KB/CU on the wire, the guide's
+    // 'streaming' level)
+#pragma unroll 4
     for (int i = lane * 8; i < K; i += 64 * 8) {
       bf16x8 xv = load8(x + i);
       float xm[8];
