@@ -337,7 +337,7 @@ __global__ void gemv_kernel(bf16* __restrict__ out,
 // ramped (proj/down ran at 4.4 TB/s in-graph vs 6.1 for the long swiglu
 // kernel); this variant removes that ramp.  LayerNorm (NORM==2) keeps the
 // staged two-pass form.
-template <int EPI, int NORM, int ROWS>
+template <int EPI, int NORM, int ROWS, int UNR>
 __global__ void gemv_direct_kernel(bf16* __restrict__ out,
                                    const bf16* __restrict__ W,
                                    const bf16* __restrict__ x,
@@ -366,12 +366,11 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
       acc[r] = 0.f;
     }
     float s2 = 0.f;
-    // unroll 4: four independent W chunks in flight per lane (2 measured
-    // only ~3.5 TB/s on the short-K qkv/proj shapes — not enough bytes
-    // in flight per CU; 4 puts ~32 KB/CU on the wire, the guide's
-    // 'streaming' level)
-#pragma unroll 4
-    for (int i = lane * 8; i < K; i += 64 * 8) {
+    // UNR independent W chunks in flight per lane: short-K shapes
+    // (qkv/proj, ~8 iterations total) need 4 to keep ~32 KB/CU on the
+    // wire (measured 3.5 TB/s at 2); long-K shapes (down, 28 iters)
+    // measured FASTER at 2 (deeper unroll costs registers/tail there)
+    auto dot_chunk = [&](int i) {
       bf16x8 xv = load8(x + i);
       float xm[8];
       if (NORM == 1) {
@@ -392,6 +391,13 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
 #pragma unroll
         for (int j = 0; j < 8; ++j) acc[r] += b2f(wv.v[j]) * xm[j];
       }
+    };
+    if constexpr (UNR >= 4) {
+#pragma unroll 4
+      for (int i = lane * 8; i < K; i += 64 * 8) dot_chunk(i);
+    } else {
+#pragma unroll 2
+      for (int i = lane * 8; i < K; i += 64 * 8) dot_chunk(i);
     }
 #pragma unroll
     for (int r = 0; r < ROWS; ++r) acc[r] = wave_reduce_sum(acc[r]);
@@ -2365,10 +2371,16 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
   // NORM 0/1: the direct-x kernel (no staging barrier); 2: staged LDS form
 #define GEMV_CASE1(E, N, R)                                                 \
   do {                                                                      \
-    if (N != 2)                                                             \
-      hipLaunchKernelGGL((gemv_direct_kernel<E, N == 2 ? 0 : N, R>), grid,  \
-                         block, 0, stream, (bf16*)out, (const bf16*)W,      \
-                         (const bf16*)x, (const bf16*)bias,                 \
+    if (N != 2 && K <= 6144 && R == 1)                                      \
+      hipLaunchKernelGGL((gemv_direct_kernel<E, N == 2 ? 0 : N, R, 4>),     \
+                         grid, block, 0, stream, (bf16*)out,                \
+                         (const bf16*)W, (const bf16*)x, (const bf16*)bias, \
+                         (const bf16*)res, (const bf16*)norm_w, eps, M, K,  \
+                         eidx, estride);                                    \
+    else if (N != 2)                                                        \
+      hipLaunchKernelGGL((gemv_direct_kernel<E, N == 2 ? 0 : N, R, 2>),     \
+                         grid, block, 0, stream, (bf16*)out,                \
+                         (const bf16*)W, (const bf16*)x, (const bf16*)bias, \
                          (const bf16*)res, (const bf16*)norm_w, eps, M, K,  \
                          eidx, estride);                                    \
     else                                                                    \
