@@ -218,7 +218,11 @@ class DecodeEngine:
 
         # fused attention+proj (one launch, in-launch granule hand-off):
         # the proj weight stream overlaps the attention compute instead of
-        # serializing behind it (see attn_proj_kernel in decode_kernels.hip)
+        # serializing behind it (attn_proj_kernel in decode_kernels.hip).
+        # MEASURED OFF by default: on Llama-3-8B the attention role
+        # stretches ~2x under the co-resident staging traffic (23 us fused
+        # vs 9.4 + 8.7 split once the split proj got its 4-deep load
+        # pipeline); opt in with MDI_FUSE_ATTN_PROJ=1.
         K_attn = cfg.n_head * cfg.head_size
         self._fuse_attn_proj = (
             not self.fp8
@@ -226,7 +230,7 @@ class DecodeEngine:
             and stage.max_seq_length <= 4096
             and K_attn % 128 == 0
             and K_attn // 2 <= 4096
-            and os.environ.get("MDI_FUSE_ATTN_PROJ", "1") not in ("0",)
+            and os.environ.get("MDI_FUSE_ATTN_PROJ", "0") == "1"
         )
         if self._fuse_attn_proj:
             # per-layer hand-off scratch: [K/2 y pairs][16 flag slots]
