@@ -366,10 +366,13 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
       acc[r] = 0.f;
     }
     float s2 = 0.f;
-    // UNR independent W chunks in flight per lane: short-K shapes
-    // (qkv/proj, ~8 iterations total) need 4 to keep ~32 KB/CU on the
-    // wire (measured 3.5 TB/s at 2); long-K shapes (down, 28 iters)
-    // measured FASTER at 2 (deeper unroll costs registers/tail there)
+    // UNR independent W chunks in flight per lane.  Measured verdicts
+    // (Llama-3-8B, rocprof per-kernel): NORM==0 short-K (proj 4096x4096)
+    // 16.3 -> 8.7 us at UNR=4; NORM==1 (qkv, fused-RMS x*nw staging eats
+    // the register headroom) and long-K (down, 28 iters) both REGRESS at
+    // 4 — they stay at 2.  A register-staged fused-RMS variant (stage
+    // x*nw once, stream W pure) also measured WORSE (22 us): its 1-row
+    // waves pay the 16-load prologue without amortization.
     auto dot_chunk = [&](int i) {
       bf16x8 xv = load8(x + i);
       float xm[8];
@@ -418,66 +421,6 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
         if (EPI == 3) a = a / (1.f + expf(-a));
         out[rw] = f2b(a);
       }
-    }
-  }
-}
-
-// Register-staged fused-RMSNorm GEMV (qkv shape): the direct-x NORM==1
-// form interleaves x and norm-weight loads with the W stream every
-// chunk, which keeps the W pipeline at ~3.5 TB/s regardless of unroll
-// depth (measured).  Here the normalized x (bf16, one extra rounding —
-// matches the torch path, which also materializes the norm in bf16) is
-// staged into registers ONCE, and the row loop issues nothing but
-// non-temporal W loads: MAXCH*16 B in flight per lane.
-// K must be a multiple of 512 and <= MAXCH*512.
-template <int MAXCH>
-__global__ void gemv_direct_pre_kernel(bf16* __restrict__ out,
-                                       const bf16* __restrict__ W,
-                                       const bf16* __restrict__ x,
-                                       const bf16* __restrict__ bias,
-                                       const bf16* __restrict__ nw,
-                                       float eps, int M, int K) {
-  const int tid = threadIdx.x;
-  const int lane = tid & 63;
-  const int wave = tid >> 6;
-  const int rows_per_grid = gridDim.x * (blockDim.x >> 6);
-  const int row0 = blockIdx.x * (blockDim.x >> 6) + wave;
-  const int nch = K >> 9;
-
-  bf16x8 xm[MAXCH];
-  float s2 = 0.f;
-#pragma unroll
-  for (int cI = 0; cI < MAXCH; ++cI) {
-    if (cI < nch) {
-      const int i = cI * 512 + lane * 8;
-      bf16x8 xv = load8(x + i);
-      bf16x8 gv = load8(nw + i);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float f = b2f(xv.v[j]);
-        s2 += f * f;
-        xm[cI].v[j] = f2b(f * b2f(gv.v[j]));
-      }
-    }
-  }
-  const float nscale = rsqrtf(wave_reduce_sum(s2) / K + eps);
-
-  for (int row = row0; row < M; row += rows_per_grid) {
-    const bf16* wr = W + (size_t)row * K + lane * 8;
-    float acc = 0.f;
-#pragma unroll
-    for (int cI = 0; cI < MAXCH; ++cI) {
-      if (cI < nch) {
-        bf16x8 wv = load8_nt(wr + cI * 512);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          acc += b2f(wv.v[j]) * b2f(xm[cI].v[j]);
-      }
-    }
-    acc = wave_reduce_sum(acc) * nscale;
-    if (lane == 0) {
-      if (bias != nullptr) acc += b2f(bias[row]);
-      out[row] = f2b(acc);
     }
   }
 }
@@ -2428,25 +2371,10 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
   const int smem = K * sizeof(bf16);
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
-  // register-staged fused-RMS form for the qkv shape (see kernel comment)
-  if (epilogue == 0 && norm_kind == 1 && rows == 1 && eidx == nullptr &&
-      res == nullptr && K >= 512 && K % 512 == 0 && K <= 8192) {
-    if (K <= 4096)
-      hipLaunchKernelGGL((gemv_direct_pre_kernel<8>), grid, block, 0,
-                         stream, (bf16*)out, (const bf16*)W,
-                         (const bf16*)x, (const bf16*)bias,
-                         (const bf16*)norm_w, eps, M, K);
-    else
-      hipLaunchKernelGGL((gemv_direct_pre_kernel<16>), grid, block, 0,
-                         stream, (bf16*)out, (const bf16*)W,
-                         (const bf16*)x, (const bf16*)bias,
-                         (const bf16*)norm_w, eps, M, K);
-    return;
-  }
   // NORM 0/1: the direct-x kernel (no staging barrier); 2: staged LDS form
 #define GEMV_CASE1(E, N, R)                                                 \
   do {                                                                      \
-    if (N != 2 && K <= 16384 && R == 1)                                     \
+    if (N == 0 && K <= 6144 && R == 1)                                      \
       hipLaunchKernelGGL((gemv_direct_kernel<E, N == 2 ? 0 : N, R, 4>),     \
                          grid, block, 0, stream, (bf16*)out,                \
                          (const bf16*)W, (const bf16*)x, (const bf16*)bias, \
