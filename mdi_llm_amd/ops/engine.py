@@ -128,8 +128,12 @@ class DecodeEngine:
         import os
 
         # adaptive split-S: 16 chunks up to 4k contexts (measured best at
-        # short S), scale up for longer KV budgets so waves cover the chip
-        default_chunks = max(16, min(128, stage.max_seq_length // 64))
+        # short S; the one-launch block-local kernel is used there anyway),
+        # then one wave per 32 keys capped at 256 chunks — 8 kv-heads x
+        # 256 chunks = 2048 waves = 8/CU, the full wave occupancy (128
+        # chunks measured only ~0.6 TB/s of KV streaming at S=8192: half
+        # the waves, 4 serial tiles each)
+        default_chunks = max(16, min(256, stage.max_seq_length // 32))
         if stage.max_seq_length <= 4096:
             default_chunks = 16
         if n_chunks == 16:

@@ -1598,32 +1598,42 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
   const int d = (wid % DS) * 64 + lane;
   if (h >= n_head) return;
 
-  // chunk weights lane-parallel in TWO registers (lane c holds chunks c
-  // and c+64), so up to 128 chunks combine correctly
-  float m0 = -1e30f, m1 = -1e30f, l0 = 0.f, l1 = 0.f;
-  if (lane < n_chunks) {
-    m0 = part_ml[((size_t)h * n_chunks + lane) * 2];
-    l0 = part_ml[((size_t)h * n_chunks + lane) * 2 + 1];
+  // chunk weights lane-parallel in FOUR registers (lane c holds chunks
+  // c, c+64, c+128, c+192), so up to 256 chunks combine correctly
+  float mr[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
+  float lr[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int c = lane + r * 64;
+    if (c < n_chunks) {
+      mr[r] = part_ml[((size_t)h * n_chunks + c) * 2];
+      lr[r] = part_ml[((size_t)h * n_chunks + c) * 2 + 1];
+    }
   }
-  if (lane + 64 < n_chunks) {
-    m1 = part_ml[((size_t)h * n_chunks + lane + 64) * 2];
-    l1 = part_ml[((size_t)h * n_chunks + lane + 64) * 2 + 1];
-  }
-  float M = fmaxf(m0, m1);
+  float M = fmaxf(fmaxf(mr[0], mr[1]), fmaxf(mr[2], mr[3]));
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1)
     M = fmaxf(M, __shfl_xor(M, off, 64));
-  const float w0 = (lane < n_chunks) ? __expf(m0 - M) : 0.f;
-  const float w1 = (lane + 64 < n_chunks) ? __expf(m1 - M) : 0.f;
-  const float inv = 1.f / wave_reduce_sum(w0 * l0 + w1 * l1);
+  float wr[4];
+  float wl = 0.f;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    wr[r] = (lane + r * 64 < n_chunks) ? __expf(mr[r] - M) : 0.f;
+    wl += wr[r] * lr[r];
+  }
+  const float inv = 1.f / wave_reduce_sum(wl);
 
   if (d >= head_size) return;
   const float* po = part_o + (size_t)h * n_chunks * head_size + d;
   float acc = 0.f;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {  // fixed register index (no waterfall)
+    const int c0 = r * 64;
+    if (c0 >= n_chunks) break;
+    const int ce = n_chunks < c0 + 64 ? n_chunks : c0 + 64;
 #pragma unroll 4
-  for (int c = 0; c < n_chunks; ++c) {
-    const float w = (c < 64) ? __shfl(w0, c, 64) : __shfl(w1, c - 64, 64);
-    acc += w * po[(size_t)c * head_size];
+    for (int c = c0; c < ce; ++c)
+      acc += __shfl(wr[r], c - c0, 64) * po[(size_t)c * head_size];
   }
   out[(size_t)h * head_size + d] = f2b(acc * inv);
 }
