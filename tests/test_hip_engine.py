@@ -416,3 +416,54 @@ def test_lane_overlap_token_equality(monkeypatch):
         r = rt.generate(prompts, 15, sp)
         outs.append([s.tolist() for s in r.sequences])
     assert outs[0] == outs[1]
+
+
+@torch.inference_mode()
+def test_engine_long_context_split_s():
+    """Long-context decode (max_seq 8192 -> split-S + 256-chunk combine
+    path) vs the torch model at S >= 4096.  Llama-3's block_size is 8192;
+    the short-context block-local kernel is bypassed above 4096."""
+    import dataclasses
+
+    from mdi_llm_amd import GPT, ModelConfig
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine
+
+    cfg = dataclasses.replace(ModelConfig.from_name("nano-gpu"),
+                              block_size=8192)
+    torch.manual_seed(5)
+    m = GPT(cfg)
+    m.apply_init()
+    m = m.to(device=DEV, dtype=torch.bfloat16)
+    m.eval()
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV,
+                                              dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.max_seq_length = 8192
+    m.max_seq_length = 8192
+    stage.eval()
+    stage.set_kv_cache(1)
+    m.set_kv_cache(1)
+
+    torch.manual_seed(6)
+    S0 = 4100
+    prompt = torch.randint(0, 511, (S0,), device=DEV)
+    ref_logits = m(prompt.view(1, -1), input_pos=0, slot=0)
+
+    eng = DecodeEngine(stage, stage.kv_pool, use_graphs=False)
+    assert eng.n_chunks >= 256, eng.n_chunks  # the scaled split-S config
+    eng.prefill_prompt(prompt, 0, 0)
+    eng.set_slot_pos(0, S0)
+
+    tok = ref_logits[0, -1].float().argmax()
+    pos = S0
+    for i in range(6):
+        ref_logits = m(tok.view(1, 1), input_pos=pos, slot=0)
+        x = eng.decode_step_head(tok.to(torch.int32), slot=0)
+        logits = eng.tail(x)
+        diff = (logits.float() - ref_logits[0, -1].float()).abs().max()
+        assert diff < 0.5, (i, float(diff))
+        assert int(logits.float().argmax()) == int(
+            ref_logits[0, -1].float().argmax()), i
+        tok = ref_logits[0, -1].float().argmax()
+        pos += 1
