@@ -139,6 +139,7 @@ class DecodeEngine:
         if n_chunks == 16:
             n_chunks = default_chunks
         n_chunks = int(os.environ.get("MDI_ATTN_CHUNKS", n_chunks))
+        n_chunks = max(4, (n_chunks // 4) * 4)  # block-shared q staging
         self.weight_dtype = os.environ.get("MDI_WEIGHT_DTYPE", "bf16")
         self.ops = require_hip_ops()
         self.config: ModelConfig = stage.config

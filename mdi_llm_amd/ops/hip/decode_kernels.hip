@@ -1376,10 +1376,14 @@ __global__ void attn_decode_kernel(
     const int* __restrict__ slot_p, int layer, int n_layers_pool,
     int n_kv_heads, int max_seq, int n_chunks, float scale, int n_batch) {
   constexpr int head_size = HS;
-  // LDS per wave: q tile (16xHS bf16), current-token k row, p tile,
-  // m/l/alpha
-  __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
-  __shared__ __attribute__((aligned(16))) bf16 k_cur[ATTN_WAVES][HS];
+  // q tile + current-token k row are BLOCK-shared: n_chunks is a
+  // multiple of ATTN_WAVES, so all 4 waves of a block carry the same
+  // (batch, kv-head) and the roped q staging runs once per block, not
+  // once per wave (the per-wave form paid ~2-4 us of scalar rope loads
+  // per wave — the dominant cost at 256 chunks).  p/m/l/alpha stay
+  // per-wave (per-chunk online-softmax state).
+  __shared__ __attribute__((aligned(16))) bf16 q_lds[16 * HS];
+  __shared__ __attribute__((aligned(16))) bf16 k_cur[HS];
   __shared__ float p_lds[ATTN_WAVES][16][16];
   __shared__ float m_lds[ATTN_WAVES][16];
   __shared__ float l_lds[ATTN_WAVES][16];
@@ -1401,6 +1405,10 @@ __global__ void attn_decode_kernel(
   const int slot = slot_p[b];
   const int keys_per_chunk = ((S + n_chunks - 1) / n_chunks + 15) & ~15;
   const int k_begin = chunk * keys_per_chunk;
+  // whole-block early exit: this block's 4 chunks are all past S (the
+  // combine only reads the first ceil(S / keys_per_chunk) chunks)
+  const int c0 = chunk - wave;  // block's first chunk (uniform)
+  if ((long long)c0 * keys_per_chunk >= (long long)S) return;
   const int n_head_all = n_kv_heads * QPK;
   if (BATCH) {
     const int qkv_dim = n_kv_heads * (QPK + 2) * head_size;
@@ -1416,8 +1424,8 @@ __global__ void attn_decode_kernel(
   const bf16* vrow_cur = krow_cur + head_size;
 
   // ---- stage q rows (QPK real, rest zero) into LDS, rope fused ----------
-  // q row j of group g lives at qkv[(g*(QPK+2)+j)*head_size]
-  for (int i = lane; i < 16 * head_size; i += 64) {
+  // block-cooperative: 256 threads, once per block
+  for (int i = threadIdx.x; i < 16 * head_size; i += 256) {
     const int r = i / head_size;
     const int d = i % head_size;
     bf16 val = f2b(0.f);
@@ -1425,11 +1433,11 @@ __global__ void attn_decode_kernel(
       const bf16* qrow = qkv + ((size_t)g * (QPK + 2) + r) * head_size;
       val = f2b(rope_elem(qrow, d, rope_ne, cos_t, sin_t, pos));
     }
-    q_lds[wave][q_swz<HS>(r, d)] = val;
+    q_lds[q_swz<HS>(r, d)] = val;
   }
   // current-token k row, roped (the pool does not hold it yet)
-  for (int d = lane; d < head_size; d += 64)
-    k_cur[wave][d] = f2b(rope_elem(krow_cur, d, rope_ne, cos_t, sin_t, pos));
+  for (int d = threadIdx.x; d < head_size; d += 256)
+    k_cur[d] = f2b(rope_elem(krow_cur, d, rope_ne, cos_t, sin_t, pos));
   if (lane < 16) {
     m_lds[wave][lane] = -1e30f;
     l_lds[wave][lane] = 0.f;
@@ -1469,11 +1477,11 @@ __global__ void attn_decode_kernel(
     for (int c = 0; c < HS / 32; ++c) {  // K=32 chunks cover head_size
       bf16x8_t af = {};
       if (row_cur)
-        af = *reinterpret_cast<const bf16x8_t*>(&k_cur[wave][c * 32 + koff]);
+        af = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
       else if (row_valid)
         af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
       const bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
-          &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
+          &q_lds[q_swz<HS>(arow, c * 32 + koff)]);
       acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc4, 0, 0, 0);
     }
     // C[row=key, col=qhead]: lane holds keys (sub*4 + r) for qhead qa
@@ -1551,7 +1559,7 @@ __global__ void attn_decode_kernel(
   // other wave reads pool row pos this step — they use k_cur/vrow_cur)
   if (k_begin <= pos && pos < k_begin + keys_per_chunk) {
     for (int d = lane; d < head_size; d += 64) {
-      kpool[cache_base + (size_t)pos * head_size + d] = k_cur[wave][d];
+      kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
       vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
     }
   }
@@ -1568,14 +1576,9 @@ __global__ void attn_decode_kernel(
       part_ml[((size_t)hA * n_chunks + chunk) * 2 + 0] = m_lds[wave][qa];
       part_ml[((size_t)hA * n_chunks + chunk) * 2 + 1] = l_lds[wave][qa];
     }
-  } else if (lane == 0) {
-    // empty chunk: mark invalid via l = 0, m = -inf
-    for (int q = 0; q < QPK; ++q) {
-      const int h = g * QPK + q;
-      part_ml[((size_t)h * n_chunks + chunk) * 2 + 0] = -1e30f;
-      part_ml[((size_t)h * n_chunks + chunk) * 2 + 1] = 0.f;
-    }
   }
+  // chunks with k_begin >= S write nothing: the S-aware combine reads
+  // only the first ceil(S / keys_per_chunk) chunk slots
 }
 
 // Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L
@@ -1586,7 +1589,9 @@ __global__ void attn_decode_kernel(
 __global__ void attn_combine_kernel(bf16* __restrict__ out,
                                     const float* __restrict__ part_o,
                                     const float* __restrict__ part_ml,
-                                    int n_chunks, int head_size, int n_head) {
+                                    int n_chunks, int head_size, int n_head,
+                                    const int* __restrict__ pos_p,
+                                    int n_head_per_b) {
   // one WAVE per (head, 64-dim slice); every lane recomputes the chunk
   // weights lane-parallel (lane == chunk, shfl broadcast), so blocks are
   // fully independent and the chunk-accumulate loop is the only serial
@@ -1597,6 +1602,12 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
   const int h = wid / DS;
   const int d = (wid % DS) * 64 + lane;
   if (h >= n_head) return;
+  // only the first ceil(S / keys_per_chunk) chunk slots were written
+  // (same formula as attn_decode_kernel); the rest are stale
+  const int S = pos_p[h / n_head_per_b] + 1;
+  const int kpc = ((S + n_chunks - 1) / n_chunks + 15) & ~15;
+  int n_act = (S + kpc - 1) / kpc;
+  if (n_act > n_chunks) n_act = n_chunks;
 
   // chunk weights lane-parallel in FOUR registers (lane c holds chunks
   // c, c+64, c+128, c+192), so up to 256 chunks combine correctly
@@ -1605,7 +1616,7 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int c = lane + r * 64;
-    if (c < n_chunks) {
+    if (c < n_act) {
       mr[r] = part_ml[((size_t)h * n_chunks + c) * 2];
       lr[r] = part_ml[((size_t)h * n_chunks + c) * 2 + 1];
     }
@@ -1618,7 +1629,7 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
   float wl = 0.f;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    wr[r] = (lane + r * 64 < n_chunks) ? __expf(mr[r] - M) : 0.f;
+    wr[r] = (lane + r * 64 < n_act) ? __expf(mr[r] - M) : 0.f;
     wl += wr[r] * lr[r];
   }
   const float inv = 1.f / wave_reduce_sum(wl);
@@ -1629,8 +1640,8 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {  // fixed register index (no waterfall)
     const int c0 = r * 64;
-    if (c0 >= n_chunks) break;
-    const int ce = n_chunks < c0 + 64 ? n_chunks : c0 + 64;
+    if (c0 >= n_act) break;
+    const int ce = n_act < c0 + 64 ? n_act : c0 + 64;
 #pragma unroll 4
     for (int c = c0; c < ce; ++c)
       acc += __shfl(wr[r], c - c0, 64) * po[(size_t)c * head_size];
@@ -2669,7 +2680,7 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
   const int cblocks = (n_waves * 64 + 255) / 256;
   hipLaunchKernelGGL(attn_combine_kernel, dim3(cblocks), dim3(256), 0,
                      stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
-                     n_head_eff);
+                     n_head_eff, pos, n_kv_heads * qpk);
   return 0;
 }
 
