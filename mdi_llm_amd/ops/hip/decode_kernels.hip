@@ -1581,36 +1581,35 @@ __global__ void attn_decode_kernel(
   // only the first ceil(S / keys_per_chunk) chunk slots
 }
 
-// Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L
-// grid = n_head blocks, 256 threads; wave 0 computes chunk weights lane-
-// parallel (lane == chunk), then all threads accumulate dims with the
-// chunk loop's loads fully independent (ILP) — the serial-per-lane version
-// of this kernel was 19 us (latency-bound); this one is ~2 us.
+// Combine split-S partials: out[h][d] = sum_c w_c * part_o[h][c][d] / L.
+// One BLOCK per (head, 64-dim slice); each of the 4 waves reduces a
+// 64-chunk strip (lane <-> dim, coalesced 256-B loads per chunk, unroll-8
+// ILP), partials summed through LDS.  The previous one-wave-per-(h,d)
+// form walked all 256 chunks serially from 64 waves total: ~45 us/layer
+// at S >= 4k, 10x the split kernel itself.  Only the first
+// ceil(S / keys_per_chunk) chunk slots are read (the split kernel writes
+// nothing for later chunks).
 __global__ void attn_combine_kernel(bf16* __restrict__ out,
                                     const float* __restrict__ part_o,
                                     const float* __restrict__ part_ml,
                                     int n_chunks, int head_size, int n_head,
                                     const int* __restrict__ pos_p,
                                     int n_head_per_b) {
-  // one WAVE per (head, 64-dim slice); every lane recomputes the chunk
-  // weights lane-parallel (lane == chunk, shfl broadcast), so blocks are
-  // fully independent and the chunk-accumulate loop is the only serial
-  // part (unrolled for load ILP).
   const int DS = (head_size + 63) / 64;
-  const int wid = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int h = (int)blockIdx.x / DS;
   const int lane = threadIdx.x & 63;
-  const int h = wid / DS;
-  const int d = (wid % DS) * 64 + lane;
+  const int wave = threadIdx.x >> 6;
+  const int d = ((int)blockIdx.x % DS) * 64 + lane;
   if (h >= n_head) return;
-  // only the first ceil(S / keys_per_chunk) chunk slots were written
-  // (same formula as attn_decode_kernel); the rest are stale
   const int S = pos_p[h / n_head_per_b] + 1;
   const int kpc = ((S + n_chunks - 1) / n_chunks + 15) & ~15;
   int n_act = (S + kpc - 1) / kpc;
   if (n_act > n_chunks) n_act = n_chunks;
 
-  // chunk weights lane-parallel in FOUR registers (lane c holds chunks
-  // c, c+64, c+128, c+192), so up to 256 chunks combine correctly
+  __shared__ float partial[ATTN_WAVES][64];
+
+  // every wave computes the identical chunk weights (lane c holds chunks
+  // c, c+64, c+128, c+192 in four registers)
   float mr[4] = {-1e30f, -1e30f, -1e30f, -1e30f};
   float lr[4] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
@@ -1634,19 +1633,25 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
   }
   const float inv = 1.f / wave_reduce_sum(wl);
 
-  if (d >= head_size) return;
-  const float* po = part_o + (size_t)h * n_chunks * head_size + d;
+  // wave w reduces chunks [w*64, w*64+64) for its dim
+  const float wrw = wave == 0 ? wr[0] : wave == 1 ? wr[1]
+                  : wave == 2 ? wr[2] : wr[3];
   float acc = 0.f;
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {  // fixed register index (no waterfall)
-    const int c0 = r * 64;
-    if (c0 >= n_act) break;
+  if (d < head_size) {
+    const int c0 = wave * 64;
     const int ce = n_act < c0 + 64 ? n_act : c0 + 64;
-#pragma unroll 4
+    const float* po = part_o + (size_t)h * n_chunks * head_size + d;
+#pragma unroll 8
     for (int c = c0; c < ce; ++c)
-      acc += __shfl(wr[r], c - c0, 64) * po[(size_t)c * head_size];
+      acc += __shfl(wrw, c - c0, 64) * po[(size_t)c * head_size];
   }
-  out[(size_t)h * head_size + d] = f2b(acc * inv);
+  partial[wave][lane] = acc;
+  __syncthreads();
+  if (wave == 0 && d < head_size) {
+    const float a = partial[0][lane] + partial[1][lane] +
+                    partial[2][lane] + partial[3][lane];
+    out[(size_t)h * head_size + d] = f2b(a * inv);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -2676,8 +2681,7 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
   // B*n_head heads
   const int n_head_eff = n_kv_heads * qpk * (n_batch > 0 ? n_batch : 1);
   const int ds = (head_size + 63) / 64;
-  const int n_waves = n_head_eff * ds;
-  const int cblocks = (n_waves * 64 + 255) / 256;
+  const int cblocks = n_head_eff * ds;  // one block per (head, dim slice)
   hipLaunchKernelGGL(attn_combine_kernel, dim3(cblocks), dim3(256), 0,
                      stream, (bf16*)out, part_o, part_ml, n_chunks, head_size,
                      n_head_eff, pos, n_kv_heads * qpk);
