@@ -466,15 +466,15 @@ for name, n_layer, n_head, n_embd in (
     )
 
 # -- Pythia / GPT-NeoX (reference: config.py pythia block) ---------------
-for size, n_layer, n_head, n_embd in (
-    ("70m", 6, 8, 512),
-    ("160m", 12, 12, 768),
-    ("410m", 24, 16, 1024),
-    ("1b", 16, 8, 2048),
-    ("1.4b", 24, 16, 2048),
-    ("2.8b", 32, 32, 2560),
-    ("6.9b", 32, 32, 4096),
-    ("12b", 36, 40, 5120),
+for size, n_layer, n_head, n_embd, pad in (
+    ("70m", 6, 8, 512, 128),
+    ("160m", 12, 12, 768, 128),
+    ("410m", 24, 16, 1024, 128),
+    ("1b", 16, 8, 2048, 128),
+    ("1.4b", 24, 16, 2048, 128),
+    ("2.8b", 32, 32, 2560, 128),
+    ("6.9b", 32, 32, 4096, 256),
+    ("12b", 36, 40, 5120, 128),
 ):
     configs.append(
         dict(
@@ -482,7 +482,7 @@ for size, n_layer, n_head, n_embd in (
             hf_config=dict(org="EleutherAI", name=f"pythia-{size}"),
             block_size=2048,
             vocab_size=50254,
-            padding_multiple=128,
+            padding_multiple=pad,
             n_layer=n_layer,
             n_head=n_head,
             n_embd=n_embd,
@@ -535,11 +535,13 @@ configs.append(
         mlp_class_name="LLaMAMLP",
         intermediate_size=14336,
         norm_eps=1e-5,
+        padded_vocab_size=32000,
     )
 )
 configs.append(
     {**configs[-1], "name": "Mistral-7B-Instruct-v0.2",
-     "hf_config": dict(org="mistralai", name="Mistral-7B-Instruct-v0.2")}
+     "hf_config": dict(org="mistralai", name="Mistral-7B-Instruct-v0.2"),
+     "block_size": 32768}
 )
 
 # -- Mixtral (LLaMAMoE, reference: model.py:823-853 local MoE) -----------
@@ -549,6 +551,7 @@ configs.append(
         hf_config=dict(org="mistralai", name="Mixtral-8x7B-v0.1"),
         block_size=32768,
         vocab_size=32000,
+        padded_vocab_size=32000,
         padding_multiple=512,
         n_layer=32,
         n_head=32,
@@ -649,5 +652,1358 @@ for kind in ("", "-chat"):
             bias=False,
         )
     )
+
+
+# -- reference-parity registry completion (round 2): every remaining
+# named config of the reference registry (same hyperparameters for
+# checkpoint compatibility; /root/reference/src/sub/config.py:180-1667)
+# -- stablelm ----------------------------------------------------
+configs.extend([
+    dict(
+        name='stablelm-base-alpha-3b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablelm-base-alpha-3b'},
+    ),
+    dict(
+        name='stablelm-base-alpha-7b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablelm-base-alpha-7b'},
+        n_head=48,
+        n_embd=6144,
+        padding_multiple=256,
+    ),
+    dict(
+        name='stablelm-tuned-alpha-3b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablelm-tuned-alpha-3b'},
+        n_head=32,
+    ),
+    dict(
+        name='stablelm-tuned-alpha-7b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablelm-tuned-alpha-7b'},
+        n_head=48,
+        n_embd=6144,
+        padding_multiple=256,
+    ),
+    dict(
+        name='stablelm-3b-4e1t',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablelm-3b-4e1t'},
+        padded_vocab_size=50304,
+        n_layer=32,
+        n_head=32,
+        n_embd=2560,
+        parallel_residual=False,
+        bias=False,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=6912,
+    ),
+    dict(
+        name='stablelm-zephyr-3b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablelm-zephyr-3b'},
+        padded_vocab_size=50304,
+        n_layer=32,
+        n_head=32,
+        n_embd=2560,
+        parallel_residual=False,
+        bias=False,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=6912,
+    ),
+])
+# -- stablecode --------------------------------------------------
+configs.extend([
+    dict(
+        name='stablecode-completion-alpha-3b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablecode-completion-alpha-3b'},
+        block_size=16384,
+        vocab_size=49152,
+        n_layer=32,
+        n_embd=2560,
+    ),
+    dict(
+        name='stablecode-completion-alpha-3b-4k',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablecode-completion-alpha-3b-4k'},
+        vocab_size=49152,
+        n_layer=32,
+        n_embd=2560,
+    ),
+    dict(
+        name='stablecode-instruct-alpha-3b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stablecode-instruct-alpha-3b'},
+        vocab_size=49152,
+        n_layer=32,
+        n_embd=2560,
+    ),
+])
+# -- stable-code -------------------------------------------------
+configs.extend([
+    dict(
+        name='stable-code-3b',
+        hf_config={'org': 'stabilityai',
+        'name': 'stable-code-3b'},
+        padded_vocab_size=50304,
+        n_layer=32,
+        n_embd=2560,
+        block_size=16384,
+        parallel_residual=False,
+        bias=False,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=6912,
+    ),
+])
+# -- pythia ------------------------------------------------------
+configs.extend([
+    dict(
+        name='pythia-14m',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-14m'},
+        block_size=512,
+        n_layer=6,
+        n_embd=128,
+        n_head=4,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-31m',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-31m'},
+        block_size=1024,
+        n_layer=6,
+        n_embd=256,
+        n_head=8,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-70m-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-70m-deduped'},
+        block_size=2048,
+        n_layer=6,
+        n_embd=512,
+        n_head=8,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-160m-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-160m-deduped'},
+        block_size=2048,
+        n_layer=12,
+        n_embd=768,
+        n_head=12,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-410m-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-410m-deduped'},
+        block_size=2048,
+        n_layer=24,
+        n_embd=1024,
+        n_head=16,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-1b-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-1b-deduped'},
+        block_size=2048,
+        n_embd=2048,
+        n_head=8,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-1.4b-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-1.4b-deduped'},
+        block_size=2048,
+        n_layer=24,
+        n_embd=2048,
+        n_head=16,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-2.8b-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-2.8b-deduped'},
+        block_size=2048,
+        n_layer=32,
+        n_embd=2560,
+        padding_multiple=128,
+    ),
+    dict(
+        name='pythia-6.9b-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-6.9b-deduped'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+    ),
+    dict(
+        name='pythia-12b-deduped',
+        hf_config={'org': 'EleutherAI',
+        'name': 'pythia-12b-deduped'},
+        block_size=2048,
+        n_layer=36,
+        n_embd=5120,
+        n_head=40,
+    ),
+])
+# -- dolly -------------------------------------------------------
+configs.extend([
+    dict(
+        name='dolly-v2-3b',
+        hf_config={'org': 'databricks',
+        'name': 'dolly-v2-3b'},
+        block_size=2048,
+        n_layer=32,
+        n_embd=2560,
+        padded_vocab_size=50280,
+    ),
+    dict(
+        name='dolly-v2-7b',
+        hf_config={'org': 'databricks',
+        'name': 'dolly-v2-7b'},
+        block_size=2048,
+        n_layer=32,
+        padded_vocab_size=50280,
+    ),
+    dict(
+        name='dolly-v2-12b',
+        hf_config={'org': 'databricks',
+        'name': 'dolly-v2-12b'},
+        block_size=2048,
+        n_layer=36,
+        n_embd=5120,
+        n_head=40,
+        padded_vocab_size=50280,
+    ),
+])
+# -- RedPajama ---------------------------------------------------
+configs.extend([
+    dict(
+        name='RedPajama-INCITE-Base-3B-v1',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-Base-3B-v1'},
+        block_size=2048,
+        n_layer=32,
+        n_embd=2560,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-Chat-3B-v1',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-Chat-3B-v1'},
+        block_size=2048,
+        n_layer=32,
+        n_embd=2560,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-Instruct-3B-v1',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-Instruct-3B-v1'},
+        block_size=2048,
+        n_layer=32,
+        n_embd=2560,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-7B-Base',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-7B-Base'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-7B-Chat',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-7B-Chat'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-7B-Instruct',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-7B-Instruct'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-Base-7B-v0.1',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-Base-7B-v0.1'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-Chat-7B-v0.1',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-Chat-7B-v0.1'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+    dict(
+        name='RedPajama-INCITE-Instruct-7B-v0.1',
+        hf_config={'org': 'togethercomputer',
+        'name': 'RedPajama-INCITE-Instruct-7B-v0.1'},
+        block_size=2048,
+        n_layer=32,
+        padding_multiple=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+    ),
+])
+# -- open_llama --------------------------------------------------
+configs.extend([
+    dict(
+        name='open_llama_3b',
+        hf_config={'org': 'openlm-research',
+        'name': 'open_llama_3b'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=26,
+        n_embd=3200,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=8640,
+    ),
+    dict(
+        name='open_llama_7b',
+        hf_config={'org': 'openlm-research',
+        'name': 'open_llama_7b'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+    ),
+    dict(
+        name='open_llama_13b',
+        hf_config={'org': 'openlm-research',
+        'name': 'open_llama_13b'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+])
+# -- vicuna ------------------------------------------------------
+configs.extend([
+    dict(
+        name='vicuna-7b-v1.3',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-7b-v1.3'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+    ),
+    dict(
+        name='vicuna-13b-v1.3',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-13b-v1.3'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+    dict(
+        name='vicuna-33b-v1.3',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-33b-v1.3'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=60,
+        n_head=52,
+        n_embd=6656,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=17920,
+    ),
+    dict(
+        name='vicuna-7b-v1.5',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-7b-v1.5'},
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+    ),
+    dict(
+        name='vicuna-7b-v1.5-16k',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-7b-v1.5-16k'},
+        block_size=16384,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        rope_condense_ratio=4,
+    ),
+    dict(
+        name='vicuna-13b-v1.5',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-13b-v1.5'},
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+    dict(
+        name='vicuna-13b-v1.5-16k',
+        hf_config={'org': 'lmsys',
+        'name': 'vicuna-13b-v1.5-16k'},
+        block_size=16384,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+        rope_condense_ratio=4,
+    ),
+])
+# -- longchat ----------------------------------------------------
+configs.extend([
+    dict(
+        name='longchat-7b-16k',
+        hf_config={'org': 'lmsys',
+        'name': 'longchat-7b-16k'},
+        block_size=16384,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        rope_condense_ratio=8,
+    ),
+    dict(
+        name='longchat-13b-16k',
+        hf_config={'org': 'lmsys',
+        'name': 'longchat-13b-16k'},
+        block_size=16384,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+        rope_condense_ratio=8,
+    ),
+])
+# -- Nous --------------------------------------------------------
+configs.extend([
+    dict(
+        name='Nous-Hermes-llama-2-7b',
+        hf_config={'org': 'NousResearch',
+        'name': 'Nous-Hermes-llama-2-7b'},
+        padded_vocab_size=32000,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+    ),
+    dict(
+        name='Nous-Hermes-13b',
+        hf_config={'org': 'NousResearch',
+        'name': 'Nous-Hermes-13b'},
+        block_size=2048,
+        vocab_size=32000,
+        padded_vocab_size=32001,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+    dict(
+        name='Nous-Hermes-Llama2-13b',
+        hf_config={'org': 'NousResearch',
+        'name': 'Nous-Hermes-Llama2-13b'},
+        vocab_size=32000,
+        padded_vocab_size=32032,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+])
+# -- Llama-3 -----------------------------------------------------
+configs.extend([
+    dict(
+        name='Llama-3-8B',
+        hf_config={'org': 'meta-llama',
+        'name': 'Meta-Llama-3-8B'},
+        block_size=8192,
+        vocab_size=128000,
+        padded_vocab_size=128256,
+        n_layer=32,
+        n_head=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=14336,
+        rope_base=500000,
+    ),
+    dict(
+        name='Llama-3-8B-Instruct',
+        hf_config={'org': 'meta-llama',
+        'name': 'Meta-Llama-3-8B-Instruct'},
+        block_size=8192,
+        vocab_size=128000,
+        padded_vocab_size=128256,
+        n_layer=32,
+        n_head=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=14336,
+        rope_base=500000,
+    ),
+    dict(
+        name='Llama-3-70B',
+        hf_config={'org': 'meta-llama',
+        'name': 'Meta-Llama-3-70B'},
+        block_size=8192,
+        vocab_size=128000,
+        padded_vocab_size=128256,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+        rope_base=500000,
+    ),
+    dict(
+        name='Llama-3-70B-Instruct',
+        hf_config={'org': 'meta-llama',
+        'name': 'Meta-Llama-3-70B-Instruct'},
+        block_size=8192,
+        vocab_size=128000,
+        padded_vocab_size=128256,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+        rope_base=500000,
+    ),
+])
+# -- Gemma -------------------------------------------------------
+configs.extend([
+    dict(
+        name='Gemma-2b',
+        hf_config={'org': 'google',
+        'name': 'gemma-2b'},
+        scale_embeddings=True,
+        vocab_size=256000,
+        padding_multiple=64,
+        n_embd=2048,
+        n_layer=18,
+        n_head=8,
+        n_query_groups=1,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='GemmaMLP',
+        gelu_approximate='tanh',
+        intermediate_size=16384,
+    ),
+    dict(
+        name='Gemma-7b',
+        hf_config={'org': 'google',
+        'name': 'gemma-7b'},
+        scale_embeddings=True,
+        vocab_size=256000,
+        padding_multiple=64,
+        n_embd=3072,
+        n_layer=28,
+        n_head=16,
+        head_size=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='GemmaMLP',
+        gelu_approximate='tanh',
+        intermediate_size=24576,
+    ),
+    dict(
+        name='Gemma-2b-it',
+        hf_config={'org': 'google',
+        'name': 'gemma-2b-it'},
+        scale_embeddings=True,
+        vocab_size=256000,
+        padding_multiple=64,
+        n_embd=2048,
+        n_layer=18,
+        n_head=8,
+        n_query_groups=1,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='GemmaMLP',
+        gelu_approximate='tanh',
+        intermediate_size=16384,
+    ),
+    dict(
+        name='Gemma-7b-it',
+        hf_config={'org': 'google',
+        'name': 'gemma-7b-it'},
+        scale_embeddings=True,
+        vocab_size=256000,
+        padding_multiple=64,
+        n_embd=3072,
+        n_layer=28,
+        n_head=16,
+        head_size=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='GemmaMLP',
+        gelu_approximate='tanh',
+        intermediate_size=24576,
+    ),
+])
+# -- CodeGemma ---------------------------------------------------
+configs.extend([
+    dict(
+        name='CodeGemma-7b-it',
+        hf_config={'org': 'google',
+        'name': 'codegemma-7b-it'},
+        scale_embeddings=True,
+        vocab_size=256000,
+        padding_multiple=64,
+        n_embd=3072,
+        n_layer=28,
+        n_head=16,
+        head_size=256,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='GemmaMLP',
+        gelu_approximate='tanh',
+        intermediate_size=24576,
+    ),
+])
+# -- Danube ------------------------------------------------------
+configs.extend([
+    dict(
+        name='Danube2-1.8b-chat',
+        hf_config={'org': 'h2oai',
+        'name': 'h2o-danube2-1.8b-chat'},
+        vocab_size=32000,
+        n_layer=24,
+        n_head=32,
+        n_embd=2560,
+        block_size=4096,
+        intermediate_size=6912,
+        padding_multiple=64,
+        norm_eps=1e-05,
+        rope_base=10000,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+    ),
+])
+# -- FreeWilly ---------------------------------------------------
+configs.extend([
+    dict(
+        name='FreeWilly2',
+        hf_config={'org': 'stabilityai',
+        'name': 'FreeWilly2'},
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+    ),
+])
+# -- CodeLlama ---------------------------------------------------
+configs.extend([
+    dict(
+        name='CodeLlama-7b-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-7b-hf'},
+        block_size=16384,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-13b-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-13b-hf'},
+        block_size=16384,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-34b-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-34b-hf'},
+        block_size=16384,
+        vocab_size=32000,
+        padded_vocab_size=32000,
+        n_layer=48,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=22016,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-70b-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-70b-hf'},
+        block_size=16384,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-7b-Python-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-7b-Python-hf'},
+        block_size=16384,
+        vocab_size=32000,
+        padded_vocab_size=32000,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-13b-Python-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-13b-Python-hf'},
+        block_size=16384,
+        vocab_size=32000,
+        padded_vocab_size=32000,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-34b-Python-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-34b-Python-hf'},
+        block_size=16384,
+        vocab_size=32000,
+        padded_vocab_size=32000,
+        n_layer=48,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=22016,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-70b-Python-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-70b-Python-hf'},
+        block_size=16384,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-7b-Instruct-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-7b-Instruct-hf'},
+        block_size=16384,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-13b-Instruct-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-13b-Instruct-hf'},
+        block_size=2048,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-34b-Instruct-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-34b-Instruct-hf'},
+        block_size=16384,
+        vocab_size=32000,
+        padded_vocab_size=32000,
+        n_layer=48,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=22016,
+        rope_base=1000000,
+    ),
+    dict(
+        name='CodeLlama-70b-Instruct-hf',
+        hf_config={'org': 'codellama',
+        'name': 'CodeLlama-70b-Instruct-hf'},
+        block_size=16384,
+        vocab_size=32016,
+        padding_multiple=16,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+        rope_base=1000000,
+    ),
+])
+# -- Platypus ----------------------------------------------------
+configs.extend([
+    dict(
+        name='Platypus-30B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Platypus-30B'},
+        block_size=2048,
+        padded_vocab_size=32000,
+        n_layer=60,
+        n_head=52,
+        n_embd=6656,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-06,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=17920,
+    ),
+    dict(
+        name='Platypus2-7B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Platypus2-7B'},
+        padded_vocab_size=32000,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+    ),
+    dict(
+        name='Platypus2-13B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Platypus2-13B'},
+        padded_vocab_size=32000,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+    dict(
+        name='Platypus2-70B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Platypus2-70B'},
+        padded_vocab_size=32000,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+    ),
+])
+# -- Camel -------------------------------------------------------
+configs.extend([
+    dict(
+        name='Camel-Platypus2-13B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Camel-Platypus2-13B'},
+        padded_vocab_size=32000,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+    dict(
+        name='Camel-Platypus2-70B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Camel-Platypus2-70B'},
+        padded_vocab_size=32000,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+    ),
+])
+# -- Stable- -----------------------------------------------------
+configs.extend([
+    dict(
+        name='Stable-Platypus2-13B',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Stable-Platypus2-13B'},
+        padded_vocab_size=32000,
+        n_layer=40,
+        n_head=40,
+        n_embd=5120,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=13824,
+    ),
+])
+# -- Platypus ----------------------------------------------------
+configs.extend([
+    dict(
+        name='Platypus2-70B-instruct',
+        hf_config={'org': 'garage-bAInd',
+        'name': 'Platypus2-70B-instruct'},
+        padded_vocab_size=32000,
+        n_layer=80,
+        n_head=64,
+        n_embd=8192,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=28672,
+    ),
+])
+# -- LLaMA -------------------------------------------------------
+configs.extend([
+    dict(
+        name='LLaMA-2-7B-32K',
+        hf_config={'org': 'togethercomputer',
+        'name': 'LLaMA-2-7B-32K'},
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        rope_condense_ratio=8,
+    ),
+])
+# -- phi ---------------------------------------------------------
+configs.extend([
+    dict(
+        name='phi-1_5',
+        hf_config={'org': 'microsoft',
+        'name': 'phi-1_5'},
+        vocab_size=50257,
+        padded_vocab_size=51200,
+        block_size=2048,
+        n_embd=2048,
+        n_layer=24,
+        rotary_percentage=0.5,
+        shared_attention_norm=True,
+        lm_head_bias=True,
+        gelu_approximate='tanh',
+    ),
+])
+# -- Mistral -----------------------------------------------------
+configs.extend([
+    dict(
+        name='Mistral-7B-Instruct-v0.1',
+        hf_config={'org': 'mistralai',
+        'name': 'Mistral-7B-Instruct-v0.1'},
+        padded_vocab_size=32000,
+        block_size=4096,
+        n_layer=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=14336,
+    ),
+])
+# -- Mixtral -----------------------------------------------------
+configs.extend([
+    dict(
+        name='Mixtral-8x7B-Instruct-v0.1',
+        hf_config={'org': 'mistralai',
+        'name': 'Mixtral-8x7B-Instruct-v0.1'},
+        padded_vocab_size=32000,
+        block_size=32768,
+        n_layer=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMoE',
+        intermediate_size=14336,
+        rope_base=1000000,
+        n_expert=8,
+        n_expert_per_token=2,
+    ),
+])
+# -- Mistral -----------------------------------------------------
+configs.extend([
+    dict(
+        name='Mistral-7B-v0.2',
+        hf_config={'org': 'unsloth',
+        'name': 'Mistral-7B-v0.2'},
+        padded_vocab_size=32000,
+        block_size=32768,
+        n_layer=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=14336,
+    ),
+    dict(
+        name='Mistral-7B-v0.3',
+        hf_config={'org': 'mistralai',
+        'name': 'Mistral-7B-v0.3'},
+        padded_vocab_size=32768,
+        block_size=32768,
+        n_layer=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=14336,
+    ),
+    dict(
+        name='Mistral-7B-Instruct-v0.3',
+        hf_config={'org': 'mistralai',
+        'name': 'Mistral-7B-Instruct-v0.3'},
+        padded_vocab_size=32768,
+        block_size=32768,
+        n_layer=32,
+        n_query_groups=8,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=14336,
+    ),
+])
+# -- tiny-llama --------------------------------------------------
+configs.extend([
+    dict(
+        name='tiny-llama-1.1b-chat',
+        hf_config={'org': 'TinyLlama',
+        'name': 'TinyLlama-1.1B-Chat-v1.0'},
+        block_size=2048,
+        vocab_size=32000,
+        padding_multiple=64,
+        n_layer=22,
+        n_head=32,
+        n_embd=2048,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        norm_eps=1e-05,
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=5632,
+        n_query_groups=4,
+    ),
+])
+# -- Llama-2 -----------------------------------------------------
+configs.extend([
+    dict(
+        name='Llama-2-7b-chat-hf-function-calling-v2',
+        hf_config={'org': 'Trelis',
+        'name': 'Llama-2-7b-chat-hf-function-calling-v2'},
+        padding_multiple=64,
+        n_layer=32,
+        rotary_percentage=1.0,
+        parallel_residual=False,
+        bias=False,
+        norm_class_name='RMSNorm',
+        mlp_class_name='LLaMAMLP',
+        intermediate_size=11008,
+        norm_eps=1e-06,
+        block_size=4096,
+        vocab_size=32000,
+        n_head=32,
+        n_embd=4096,
+        rope_base=10000,
+    ),
+])
+
 
 name_to_config: dict[str, dict] = {c["name"]: c for c in configs}

@@ -143,35 +143,250 @@ class MistralInstruct(PromptStyle):
         return f"<s>[INST] {prompt} [/INST]"
 
 
+def _ids(tokenizer: Tokenizer, *seqs) -> Tuple[List[int], ...]:
+    """Build stop sequences, dropping any whose tokens don't exist in
+    this tokenizer (the reference indexes blindly and would crash)."""
+    out: list = []
+    if tokenizer.eos_id is not None:
+        out.append([tokenizer.eos_id])
+    for seq in seqs:
+        ids = []
+        ok = True
+        for t in seq:
+            tid = t if isinstance(t, int) else tokenizer.token_to_id(t)
+            if tid is None:
+                ok = False
+                break
+            ids.append(tid)
+        if ok and ids:
+            out.append(ids)
+    return tuple(out)
+
+
+# ---- remaining reference families (ref prompts.py:59-300) --------------
+class FLAN(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return (
+            "Below is an instruction that describes a task. Write a response "
+            "that appropriately completes the request.\n\n"
+            f"### Instruction:\n{prompt}\n\n### Response:\n"
+        )
+
+
+class Longform(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return (
+            "Below is an instruction that describes a task, paired with an "
+            "input that provides further context. Write a response that "
+            "appropriately completes the request.\n\n"
+            f"### Instruction:\n{prompt}\n\n### Response:\n"
+        )
+
+
+class StableLMAlpha(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return (
+            "<|SYSTEM|># StableLM Tuned (Alpha version)\n- StableLM is a "
+            "helpful and harmless open-source AI language model developed by "
+            "StabilityAI.\n- StableLM is excited to be able to help the "
+            "user, but will refuse to do anything that could be considered "
+            "harmful to the user.\n- StableLM is more than just an "
+            "information source, StableLM is also able to write poetry, "
+            "short stories, and make jokes.\n- StableLM will refuse to "
+            f"participate in anything that could harm a human.<|USER|>"
+            f"{prompt}<|ASSISTANT|>"
+        )
+
+    def stop_tokens(self, tokenizer: Tokenizer) -> Tuple[List[int], ...]:
+        return _ids(tokenizer, ["<|SYSTEM|>"], ["<|ASSISTANT|>"],
+                    ["<|USER|>"])
+
+
+class StableLMZephyr(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"<|user|>\n{prompt}<|endoftext|>\n<|assistant|>\n"
+
+
+class TogetherComputerChat(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"<human>: {prompt}\n<bot>:"
+
+    def stop_tokens(self, tokenizer: Tokenizer) -> Tuple[List[int], ...]:
+        return _ids(tokenizer, ["<", "human", ">:"], ["<", "bot", ">:"])
+
+
+class TogetherComputerInstruct(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"Q: {prompt}\nA:"
+
+    def stop_tokens(self, tokenizer: Tokenizer) -> Tuple[List[int], ...]:
+        # NeoX tokenizer ids 187/535/2756 are '\n', '\n\n', '\n\n\n'
+        return _ids(tokenizer, ["Q", ":"], ["Question"], ["A", ":"],
+                    ["Label", ":"], [187, 187], [535], [2756])
+
+
+class Falcon(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"Do not prefix your replies with 'Bot: '\nUser: {prompt}\n"
+
+    def stop_tokens(self, tokenizer: Tokenizer) -> Tuple[List[int], ...]:
+        # 193 is '\n' in the falcon tokenizer
+        return _ids(tokenizer, ["User", ":"], [193, "User"])
+
+
+class Vicuna(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return (
+            "A chat between a curious user and an artificial intelligence "
+            "assistant. The assistant gives helpful, detailed, and polite "
+            f"answers to the user's questions. USER: {prompt} ASSISTANT:"
+        )
+
+
+class Llama2FunctionCalling(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        # single search_bing example function, as the published adapter
+        # expects (curly braces doubled to stay format-safe)
+        function_list = json.dumps({
+            "function": "search_bing",
+            "description": (
+                "Search the web for content on Bing. This allows users to "
+                "search online/the internet/the web for content."
+            ),
+            "arguments": [{
+                "name": "query",
+                "type": "string",
+                "description": "The search query string",
+            }],
+        }).replace("{", "{{").replace("}", "}}")
+        sys_prompt = (
+            "You are a helpful, respectful and honest assistant. Always "
+            "answer as helpfully aspossible. Your only response should be "
+            "JSON formatted functions"
+        )
+        return (
+            f"<FUNCTIONS>{function_list.strip()}</FUNCTIONS>\n\n"
+            f"[INST]<<SYS>>\n{sys_prompt}\n<</SYS>>\n\n{prompt}[/INST]\n\n"
+        )
+
+
+class FreeWilly2(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return (
+            "### System:\nThis is a system prompt, please behave and help "
+            "the user.\n\n"
+            f"### User:\n{prompt}\n\n### Assistant:\n"
+        )
+
+
+class Platypus(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"### Instruction:\n\n{prompt}\n\n### Response:\n"
+
+
+class NousResearch(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"### Instruction:\n{prompt}\n\n### Response:\n"
+
+
+class StableCode(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"###Instruction\n{prompt}###Response\n"
+
+
+class CodeLlama(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        # no default system prompt (HF conversational-instructions doc)
+        return f"<s>[INST] {prompt} [/INST]"
+
+
+class Phi1(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"{prompt}\n\nAnswer:"
+
+    def stop_tokens(self, tokenizer: Tokenizer) -> Tuple[List[int], ...]:
+        # 198 is '\n' in the codegen tokenizer
+        return _ids(tokenizer, ["Answer", ":"], [198, "Answer", ":"])
+
+
+class H2Oai(PromptStyle):
+    def apply(self, prompt: str, **kwargs) -> str:
+        return f"<|prompt|>{prompt}</s><|answer|>"
+
+
 prompt_styles: dict[str, Type[PromptStyle]] = {
     "default": Default,
     "noprompt": NoPrompt,
     "alpaca": Alpaca,
+    "flan": FLAN,
+    "longform": Longform,
+    "stablelm-alpha": StableLMAlpha,
+    "stablelm-zephyr": StableLMZephyr,
+    "togethercomputer-chat": TogetherComputerChat,
+    "togethercomputer-instruct": TogetherComputerInstruct,
+    "falcon": Falcon,
+    "vicuna": Vicuna,
+    "llama2-function-calling": Llama2FunctionCalling,
     "llama2": Llama2,
     "llama3": Llama3,
+    "freewilly2": FreeWilly2,
+    "platypus": Platypus,
+    "nous-research": NousResearch,
+    "stablecode": StableCode,
+    "codellama": CodeLlama,
+    "phi-1": Phi1,
+    "phi-2": Phi2,
     "tinyllama": TinyLlama,
     "chatml": ChatML,
-    "phi-2": Phi2,
     "gemma": Gemma,
+    "h2oai": H2Oai,
     "mistral": MistralInstruct,
 }
 
 
 def model_name_to_prompt_style(model_name: str) -> PromptStyle:
     """Map a model/config name to its chat style
-    (reference prompts.py:325-366)."""
-    if re.search(r"Llama-2.*-chat", model_name):
+    (reference prompts.py:325-366, same precedence order)."""
+    if re.search(r"stablelm-tuned-alpha", model_name):
+        return StableLMAlpha()
+    if re.search(r"stablelm-zephyr-3b", model_name):
+        return StableLMZephyr()
+    if re.search("stablecode-instruct", model_name):
+        return StableCode()
+    if re.search(r"RedPajama-INCITE.*-Chat", model_name):
+        return TogetherComputerChat()
+    if re.search(r"RedPajama-INCITE.*-Instruct", model_name):
+        return TogetherComputerInstruct()
+    if re.search(r"falcon.*-instruct", model_name):
+        return Falcon()
+    if re.search(r"vicuna|longchat", model_name):
+        return Vicuna()
+    if re.search("Llama-2-7b-chat-hf-function-calling-v2", model_name):
+        return Llama2FunctionCalling()
+    if re.search("Llama-2.*-chat", model_name):
         return Llama2()
     if re.search(r"Llama-3.*-Instruct", model_name, re.IGNORECASE):
         return Llama3()
-    if re.search(r"TinyLlama.*Chat", model_name):
-        return TinyLlama()
+    if re.search("FreeWilly2", model_name):
+        return FreeWilly2()
+    if re.search("Platypus", model_name):
+        return Platypus()
+    if re.search("Nous-Hermes", model_name):
+        return NousResearch()
+    if re.search("CodeLlama", model_name):
+        return CodeLlama()
+    if re.search("Mistral.*Instruct", model_name):
+        return MistralInstruct()
+    if re.search("phi-1", model_name):
+        return Phi1()
     if re.search("phi-2", model_name):
         return Phi2()
-    if re.search(r"gemma.*-it", model_name):
+    if re.search(r"TinyLlama.*Chat|tiny-llama.*chat", model_name):
+        return TinyLlama()
+    if re.search(r"(Code)?[Gg]emma.*-it", model_name):
         return Gemma()
-    if re.search(r"Mistral.*Instruct", model_name):
-        return MistralInstruct()
+    if re.search(r"Danube2.*-chat", model_name):
+        return H2Oai()
     return Default()
 
 
