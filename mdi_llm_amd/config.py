@@ -212,6 +212,26 @@ configs.extend(
             mlp_class_name="GptNeoxMLP",
         ),
         dict(
+            # phi-2-style tiny config (parallel residual, shared attn
+            # norm, LayerNorm + biases, partial rotary) for converter
+            # round-trip tests
+            name="nano-phi-test",
+            block_size=128,
+            vocab_size=256,
+            padding_multiple=64,
+            n_layer=2,
+            n_head=4,
+            n_embd=64,
+            rotary_percentage=0.5,
+            parallel_residual=True,
+            shared_attention_norm=True,
+            bias=True,
+            lm_head_bias=True,
+            norm_class_name="LayerNorm",
+            mlp_class_name="GptNeoxMLP",
+            gelu_approximate="tanh",
+        ),
+        dict(
             name="nano-test-gpt2",
             block_size=128,
             vocab_size=256,

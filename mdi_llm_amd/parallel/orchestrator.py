@@ -98,6 +98,9 @@ class MDIRuntime:
 
     # ------------------------------------------------------------------
     def _log(self, msg: str) -> None:
+        from ..utils.console import get_logger
+
+        get_logger().debug("[node %s] %s", self.rank, msg)
         if self.verb:
             print(f"[node {self.rank}] {msg}", file=sys.stderr, flush=True)
 

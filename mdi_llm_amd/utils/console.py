@@ -8,7 +8,7 @@ import sys
 import threading
 import time
 
-__all__ = ["loading_bar", "waiting_animation"]
+__all__ = ["loading_bar", "waiting_animation", "setup_debug_logging", "get_logger"]
 
 
 def loading_bar(current: int, total: int, width: int = 30,
@@ -41,3 +41,29 @@ def waiting_animation(stop_event: threading.Event, message: str = "working",
     t = threading.Thread(target=_spin, daemon=True)
     t.start()
     return t
+
+
+def setup_debug_logging(role: str, log_dir="logs") -> "logging.Logger":
+    """Wire the named framework logger to a per-role debug log file
+    (reference starter.py:36-44 / secondary.py:29-38 parity: logger
+    "model_dist", DEBUG level, timestamped file handler under logs/).
+
+    Returns the logger; CLIs call this under ``--debug``."""
+    import logging
+    from pathlib import Path
+
+    log = logging.getLogger("model_dist")
+    Path(log_dir).mkdir(parents=True, exist_ok=True)
+    handler = logging.FileHandler(
+        Path(log_dir) / f"logs_{role}.log", mode="w")
+    handler.setFormatter(
+        logging.Formatter("[%(asctime)s] -> %(levelname)s: %(message)s"))
+    log.setLevel(logging.DEBUG)
+    log.addHandler(handler)
+    return log
+
+
+def get_logger() -> "logging.Logger":
+    import logging
+
+    return logging.getLogger("model_dist")

@@ -20,6 +20,7 @@ import torch
 
 from ..config import ModelConfig
 from .checkpoint import save_checkpoint
+from .incremental import IncrementalSaver
 
 __all__ = ["convert_hf_checkpoint", "convert_lit_checkpoint"]
 
@@ -220,72 +221,113 @@ def _load_shard(path: Path) -> dict:
     return torch.load(path, map_location="cpu", weights_only=True)
 
 
+def _iter_shard_tensors(path: Path):
+    """Yield (key, tensor) one at a time without materializing the whole
+    shard: safetensors via lazy per-tensor reads, .bin via mmap."""
+    if path.suffix == ".safetensors":
+        from safetensors import safe_open
+
+        with safe_open(str(path), framework="pt", device="cpu") as f:
+            for k in f.keys():
+                yield k, f.get_tensor(k)
+        return
+    try:
+        d = torch.load(path, map_location="cpu", weights_only=True,
+                       mmap=True)
+    except (RuntimeError, ValueError):
+        d = torch.load(path, map_location="cpu", weights_only=True)
+    yield from d.items()
+
+
 def convert_hf_checkpoint(
     hf_dir: PathLike,
     out_dir: PathLike = None,
     model_name: str = None,
     dtype: torch.dtype = None,
+    config_overrides: dict = None,
 ) -> Path:
     """Convert an HF checkpoint dir to the litGPT layout
-    (``model_config.yaml`` + ``lit_model.pth``), shard by shard."""
+    (``model_config.yaml`` + ``lit_model.pth``), shard by shard with
+    bounded RAM (each tensor streams through transform -> zip)."""
     hf_dir = Path(hf_dir)
     out_dir = Path(out_dir) if out_dir else hf_dir
-    config = ModelConfig.from_name(model_name or hf_dir.name)
+    config = ModelConfig.from_name(model_name or hf_dir.name,
+                                   **(config_overrides or {}))
     fam = _family(config)
     tmap = {"llama": LLAMA_MAP, "neox": NEOX_MAP, "gpt2": GPT2_MAP,
             "phi": PHI_MAP, "falcon": FALCON_MAP}[fam]
 
-    sd: dict = {}
+    # streamed conversion (bounded RAM — reference achieves this with
+    # lazy tensors + an incremental pickler, litgpt_utils.py:14-343; here
+    # each tensor goes shard->transform->zip and is freed, so peak RSS is
+    # one tensor + any q/k/v pieces awaiting their weave)
+    out_dir.mkdir(parents=True, exist_ok=True)
     pending_qkv: Dict[int, dict] = {}
-    for shard in _iter_hf_weight_files(hf_dir):
-        weights = _load_shard(shard)
-        for key, t in weights.items():
-            key = key.removeprefix("transformer.") if fam == "gpt2" else key
-            if dtype is not None and t.is_floating_point():
-                t = t.to(dtype)
-            if fam in ("llama", "phi") and ".self_attn." in key and (
-                "q_proj" in key or "k_proj" in key or "v_proj" in key
-            ):
-                layer = int(key.split(".")[2])
-                which = key.split(".")[4][0]  # q/k/v
-                suffix = "b" if key.endswith("bias") else ""
-                pending_qkv.setdefault(layer, {})[which + suffix] = t
-                continue
-            if fam == "gpt2" and ".attn.c_attn." in key:
-                layer = int(key.split(".")[1])
-                d = pending_qkv.setdefault(layer, {})
-                if key.endswith("weight"):
-                    w = t.t().contiguous()  # Conv1D -> Linear
-                    d["q"], d["k"], d["v"] = w.chunk(3, dim=0)
-                else:
-                    d["qb"], d["kb"], d["vb"] = t.chunk(3, dim=0)
-                continue
-            lit = _map_key(tmap, key)
-            if lit is None:
-                continue
-            if fam == "gpt2" and any(key.endswith(s) for s in GPT2_TRANSPOSE):
-                t = t.t().contiguous()
-            if lit in ("transformer.wte.weight", "lm_head.weight"):
-                t = _pad_vocab(t, config)
-            sd[lit] = t
-        del weights
-        gc.collect()
+    wte = None
+    seen_lm_head = False
+    with IncrementalSaver(out_dir / "lit_model.pth") as saver:
 
-    for layer, d in pending_qkv.items():
-        if "q" in d:
-            sd[f"transformer.h.{layer}.attn.attn.weight"] = weave_qkv(
-                d["q"], d["k"], d["v"], config
-            )
-        if "qb" in d and "kb" in d and "vb" in d:
-            sd[f"transformer.h.{layer}.attn.attn.bias"] = weave_qkv(
-                d["qb"].unsqueeze(1), d["kb"].unsqueeze(1),
-                d["vb"].unsqueeze(1), config
-            ).squeeze(1)
+        def emit(name: str, t: torch.Tensor) -> None:
+            nonlocal wte, seen_lm_head
+            if name == "transformer.wte.weight":
+                wte = t  # kept (mmap-backed where possible) for tying
+            if name == "lm_head.weight":
+                seen_lm_head = True
+            saver.add(name, t)
 
-    if "lm_head.weight" not in sd and "transformer.wte.weight" in sd:
-        sd["lm_head.weight"] = sd["transformer.wte.weight"].clone()
+        def flush_qkv(layer: int, d: dict) -> None:
+            if "q" in d and "k" in d and "v" in d:
+                emit(f"transformer.h.{layer}.attn.attn.weight",
+                     weave_qkv(d.pop("q"), d.pop("k"), d.pop("v"), config))
+            if "qb" in d and "kb" in d and "vb" in d:
+                emit(f"transformer.h.{layer}.attn.attn.bias",
+                     weave_qkv(d.pop("qb").unsqueeze(1),
+                               d.pop("kb").unsqueeze(1),
+                               d.pop("vb").unsqueeze(1),
+                               config).squeeze(1))
 
-    save_checkpoint(out_dir, config, sd)
+        for shard in _iter_hf_weight_files(hf_dir):
+            for key, t in _iter_shard_tensors(shard):
+                key = (key.removeprefix("transformer.")
+                       if fam == "gpt2" else key)
+                if dtype is not None and t.is_floating_point():
+                    t = t.to(dtype)
+                if fam in ("llama", "phi") and ".self_attn." in key and (
+                    "q_proj" in key or "k_proj" in key or "v_proj" in key
+                ):
+                    layer = int(key.split(".")[2])
+                    which = key.split(".")[4][0]  # q/k/v
+                    suffix = "b" if key.endswith("bias") else ""
+                    d = pending_qkv.setdefault(layer, {})
+                    d[which + suffix] = t
+                    flush_qkv(layer, d)
+                    continue
+                if fam == "gpt2" and ".attn.c_attn." in key:
+                    layer = int(key.split(".")[1])
+                    d = pending_qkv.setdefault(layer, {})
+                    if key.endswith("weight"):
+                        w = t.t().contiguous()  # Conv1D -> Linear
+                        d["q"], d["k"], d["v"] = w.chunk(3, dim=0)
+                    else:
+                        d["qb"], d["kb"], d["vb"] = t.chunk(3, dim=0)
+                    flush_qkv(layer, d)
+                    continue
+                lit = _map_key(tmap, key)
+                if lit is None:
+                    continue
+                if fam == "gpt2" and any(
+                        key.endswith(sfx) for sfx in GPT2_TRANSPOSE):
+                    t = t.t().contiguous()
+                if lit in ("transformer.wte.weight", "lm_head.weight"):
+                    t = _pad_vocab(t, config)
+                emit(lit, t)
+                del t
+            gc.collect()
+
+        if not seen_lm_head and wte is not None:
+            saver.add("lm_head.weight", wte)
+
+    config.save(out_dir / "model_config.yaml")
     return out_dir / "lit_model.pth"
 
 
@@ -322,23 +364,56 @@ def convert_lit_checkpoint(ckpt_dir: PathLike, out_path: PathLike,
                 "transformer.h.{}.ln_mlp.bias"
     elif fam == "llama":
         rev = {v: k for k, v in LLAMA_MAP.items()}
+    elif fam == "neox":
+        # NeoX keeps its fused query_key_value layout in both formats
+        rev = {v: k for k, v in NEOX_MAP.items()}
+    elif fam == "phi":
+        rev = {v: k for k, v in PHI_MAP.items()}
+    elif fam == "gpt2":
+        # re-add the HF "transformer." prefix; Conv1D transposes and the
+        # c_attn merge are handled in the loop below
+        rev = {v: ("transformer." + k if not k.startswith("lm_head") else k)
+               for k, v in GPT2_MAP.items()}
     else:
-        raise NotImplementedError(
-            "lit->HF export currently supports llama and falcon")
+        raise NotImplementedError(f"lit->HF export: unknown family {fam}")
     inv = {}
     for key, t in sd.items():
-        if fam == "llama" and key.endswith(".attn.attn.weight"):
+        if fam in ("llama", "phi") and key.endswith(".attn.attn.weight"):
             layer = key.split(".")[2]
             q, k, v = unweave_qkv(t, config)
             inv[f"model.layers.{layer}.self_attn.q_proj.weight"] = q
             inv[f"model.layers.{layer}.self_attn.k_proj.weight"] = k
             inv[f"model.layers.{layer}.self_attn.v_proj.weight"] = v
             continue
+        if fam == "phi" and key.endswith(".attn.attn.bias"):
+            layer = key.split(".")[2]
+            q, k, v = unweave_qkv(t.unsqueeze(1), config)
+            inv[f"model.layers.{layer}.self_attn.q_proj.bias"] = q.squeeze(1)
+            inv[f"model.layers.{layer}.self_attn.k_proj.bias"] = k.squeeze(1)
+            inv[f"model.layers.{layer}.self_attn.v_proj.bias"] = v.squeeze(1)
+            continue
+        if fam == "gpt2" and ".attn.attn." in key:
+            # lit grouped-interleave -> HF [q|k|v] c_attn (Conv1D layout)
+            layer = key.split(".")[2]
+            if key.endswith("weight"):
+                q, k, v = unweave_qkv(t, config)
+                inv[f"transformer.h.{layer}.attn.c_attn.weight"] = (
+                    torch.cat([q, k, v], dim=0).t().contiguous())
+            else:
+                q, k, v = unweave_qkv(t.unsqueeze(1), config)
+                inv[f"transformer.h.{layer}.attn.c_attn.bias"] = (
+                    torch.cat([q, k, v], dim=0).squeeze(1))
+            continue
+        if fam == "gpt2" and key == "lm_head.weight":
+            continue  # tied to wte in HF GPT-2 checkpoints
         hf_key = _map_key(rev, key)
         if hf_key is None:
             continue
         if key in ("transformer.wte.weight", "lm_head.weight"):
             t = t[: config.vocab_size]
+        if fam == "gpt2" and any(hf_key.endswith(sfx)
+                                 for sfx in GPT2_TRANSPOSE):
+            t = t.t().contiguous()
         inv[hf_key] = t
     out_path = Path(out_path)
     out_path.parent.mkdir(parents=True, exist_ok=True)

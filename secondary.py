@@ -21,6 +21,10 @@ def main(args):
 
     torch.manual_seed(args.seed)
     idx = int(args.nodes_config[1])
+    if getattr(args, "debug", False):
+        from mdi_llm_amd.utils.console import setup_debug_logging
+
+        setup_debug_logging(f"secondary{idx}", SCRIPT_DIR / "logs")
     rt = MDIRuntime(
         f"secondary:{idx}",
         config_file=Path(args.nodes_config[0]),
@@ -35,6 +39,8 @@ def main(args):
 if __name__ == "__main__":
     p = argparse.ArgumentParser(description="Secondary node - MDI (MI355X)")
     p.add_argument("-v", "--verb", action="store_true")
+    p.add_argument("-d", "--debug", action="store_true",
+                   help="write debug logs to logs/logs_secondary<i>.log")
     p.add_argument("--chunk", type=Path, default=None)
     p.add_argument(
         "--nodes-config",

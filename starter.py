@@ -113,6 +113,9 @@ def build_parser():
 if __name__ == "__main__":
     args = build_parser().parse_args()
     if args.debug:
+        from mdi_llm_amd.utils.console import setup_debug_logging
+
+        setup_debug_logging("starter", SCRIPT_DIR / "logs")
         prof = cProfile.Profile()
         prof.enable()
         main(args)
