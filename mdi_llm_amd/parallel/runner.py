@@ -142,4 +142,12 @@ def make_runner(stage, n_slots: int, device: torch.device,
     if engine_supported(stage.config):
         return HipRunner(stage, n_slots, n_chunks=n_chunks,
                          use_graphs=use_graphs)
+    import sys
+
+    print(
+        f"[mdi_llm_amd] WARNING: config {stage.config.name!r} has no HIP "
+        "decode-engine instantiation (unsupported attention geometry or "
+        "MoE shape) — running the much slower torch fallback on this GPU",
+        file=sys.stderr, flush=True,
+    )
     return TorchRunner(stage, n_slots)
