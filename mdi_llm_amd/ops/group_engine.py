@@ -127,6 +127,19 @@ class GroupDecodeEngine:
         self.HN = torch.zeros(B, E, **bf)
         self.QKV = torch.zeros(B, cfg.qkv_dim, **bf)
         self.Y = torch.zeros(B, n_head * hs, **bf)
+        # hand-written M-tile MFMA GEMM path (MDI_MTILE=1 opts in; see
+        # ops/hip/decode_kernels.hip mtile_gemm_kernel): fixed buffers so
+        # the whole rotation stays hipGraph-capturable
+        self.use_mtile = (
+            os.environ.get("MDI_MTILE", "0") == "1"
+            and B in (16, 32, 64, 128)
+            and all(self._mtile_k_ok(B, k)
+                    for k in (E, n_head * hs, I))
+        )
+        if self.use_mtile:
+            self.A2 = torch.zeros(B, E, **bf)
+            self.G = torch.zeros(B, I, **bf)
+            self.U2 = torch.zeros(B, I, **bf)
         self.part_o = torch.zeros(B * n_head * n_chunks * hs, device=dev,
                                   dtype=torch.float32)
         self.part_ml = torch.zeros(B * n_head * n_chunks * 2, device=dev,
@@ -151,6 +164,11 @@ class GroupDecodeEngine:
         self._g_tail: Optional[torch.cuda.CUDAGraph] = None
         self._g_mid: Optional[torch.cuda.CUDAGraph] = None
         self._fused_params = None
+
+    @staticmethod
+    def _mtile_k_ok(B: int, K: int) -> bool:
+        kc = 512 if B <= 32 else (256 if B <= 64 else 128)
+        return K % kc == 0 and K >= 2 * kc
 
     # ------------------------------------------------------------------
     def set_slot_pos(self, slot: int, pos: int) -> None:
@@ -179,7 +197,33 @@ class GroupDecodeEngine:
                                 scale_b=ws[None, :],
                                 out_dtype=torch.bfloat16)
 
+    def _run_blocks_mtile(self) -> None:
+        """Block stack on the hand-written M-tile MFMA GEMMs (bias and
+        residual adds fused into the GEMM epilogue)."""
+        cfg = self.config
+        scale = 1.0 / (cfg.head_size ** 0.5)
+        gelu_gate = cfg.mlp_class_name == "GemmaMLP"
+        ops = self.ops
+        for li, w in enumerate(self.blocks):
+            xn = self._rms(self.X, w.norm1_w, self.XN)
+            ops.mtile_gemm(self.QKV, w.attn_w, xn, w.attn_b, None)
+            ops.attn_decode(
+                self.Y, self.part_o, self.part_ml, self.QKV, self.kv_pool.k,
+                self.kv_pool.v, self.cos, self.sin, self.pos, self.slots,
+                li, self.n_chunks, scale, self.B,
+            )
+            ops.mtile_gemm(self.A2, w.proj_w, self.Y, w.proj_b, self.X)
+            hn = self._rms(self.A2, w.norm2_w, self.HN)
+            ops.mtile_gemm(self.G, w.fc1_w, hn, None, None)
+            ops.mtile_gemm(self.U2, w.fc2_w, hn, None, None)
+            ops.swiglu_mul(self.U2, self.G, self.U2, gelu_gate)
+            ops.mtile_gemm(self.X, w.mlp_proj_w, self.U2, w.mlp_proj_b,
+                           self.A2)
+
     def _run_blocks(self) -> None:
+        if self.use_mtile:
+            self._run_blocks_mtile()
+            return
         cfg = self.config
         scale = 1.0 / (cfg.head_size ** 0.5)
         gelu_gate = cfg.mlp_class_name == "GemmaMLP"
@@ -228,7 +272,9 @@ class GroupDecodeEngine:
 
     def _tail_seq(self) -> None:
         xn = self._rms(self.X, self.lnf_w, self.XN)
-        if self.fp8:
+        if self.use_mtile and self.head_w.size(0) % 16 == 0:
+            self.ops.mtile_gemm(self.LOGITS, self.head_w, xn, None, None)
+        elif self.fp8:
             self.LOGITS.copy_(
                 self._mm(xn, self.head_w8, self.head_s, self.head_w))
         else:

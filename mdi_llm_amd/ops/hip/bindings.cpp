@@ -422,6 +422,26 @@ void sample(torch::Tensor out_token, torch::Tensor logits,
                 cur_stream());
 }
 
+void mtile_gemm(torch::Tensor Y, torch::Tensor W, torch::Tensor X,
+                c10::optional<torch::Tensor> bias,
+                c10::optional<torch::Tensor> res) {
+  check_bf16(Y, "Y");
+  check_bf16(W, "W");
+  check_bf16(X, "X");
+  const int Bsz = (int)X.size(0);
+  const int K = (int)X.size(1);
+  const int M = (int)W.size(0);
+  TORCH_CHECK((int)W.size(1) == K, "mtile_gemm: K mismatch");
+  TORCH_CHECK(Y.numel() >= (int64_t)Bsz * M, "mtile_gemm: Y too small");
+  int rc = launch_mtile_gemm(
+      Y.data_ptr(), W.data_ptr(), X.data_ptr(),
+      bias.has_value() ? bias->data_ptr() : nullptr,
+      res.has_value() ? res->data_ptr() : nullptr, Bsz, M, K,
+      cur_stream());
+  TORCH_CHECK(rc == 0, "mtile_gemm: unsupported shape B=", Bsz, " K=", K,
+              " M=", M);
+}
+
 void route_env(torch::Tensor hdr, torch::Tensor slot_out,
                torch::Tensor pos_table, int64_t dummy_slot) {
   check_i32(hdr, "hdr");
@@ -576,6 +596,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("pos_table") = c10::nullopt,
         py::arg("adv_slot") = c10::nullopt, py::arg("adv_pos") = 0,
         py::arg("pos_bias") = 0);
+  m.def("mtile_gemm", &mtile_gemm,
+        "grouped-decode M-tile MFMA GEMM (bulk LDS-staged X, nt W stream)",
+        py::arg("Y"), py::arg("W"), py::arg("X"), py::arg("bias"),
+        py::arg("res"));
   m.def("route_env", &route_env,
         "device-side envelope routing (slot from header, dummy on stop)",
         py::arg("hdr"), py::arg("slot_out"), py::arg("pos_table"),

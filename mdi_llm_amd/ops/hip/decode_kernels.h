@@ -103,6 +103,12 @@ void launch_sample(void* out_token, const void* logits, int V, void* scratch,
                    const int* adv_slot, int adv_pos, int pos_bias,
                    hipStream_t stream);
 
+// grouped M-tile MFMA GEMM: Y[B,M] = X[B,K] @ W[M,K]^T (+bias, +res);
+// returns -1 when (B, K) has no instantiation
+int launch_mtile_gemm(void* Y, const void* W, const void* X,
+                      const void* bias, const void* res, int Bsz, int M,
+                      int K, hipStream_t stream);
+
 // envelope routing for the pipelined secondary serve: slot_out <-
 // hdr[0] (data) or dummy_slot (stop/flush, with pos reset)
 void launch_route_env(const int* hdr, int* slot_out, int* pos_table,

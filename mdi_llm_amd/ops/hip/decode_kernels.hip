@@ -54,6 +54,9 @@ DEVINL i32x4_t load16_nt_u8(const unsigned char* p) {
   return __builtin_nontemporal_load(reinterpret_cast<const i32x4_t*>(p));
 }
 
+using f32x4 = __attribute__((__vector_size__(16))) float;
+using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+
 DEVINL float wave_reduce_sum(float v) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
@@ -422,6 +425,145 @@ __global__ void gemv_direct_kernel(bf16* __restrict__ out,
         out[rw] = f2b(a);
       }
     }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Grouped-decode M-tile MFMA GEMM:  Y[B, M] = X[B, K] @ W[M, K]^T
+// (+ bias / + residual), B in {16, 32, 64, 128}.
+//
+// The hipBLASLt skinny-M GEMMs run these shapes at ~68% of the weight-
+// stream roofline (round-1 measurement); this is the priced recipe from
+// the round-1 ROADMAP: X staged into LDS in BULK COALESCED chunks with
+// register double-buffering (the naive per-fragment L2 X loads measured
+// 1.4-1.6 TB/s — latency-bound), W streamed non-temporally straight into
+// MFMA A-fragments.
+//
+// Geometry: one block per 16 output rows; its 4 waves split each K-chunk
+// (intra-block split-K), partials reduced through LDS at the end.  MFMA
+// 16x16x32 bf16 with the same fragment convention as the attention
+// kernels: lane&15 = fragment row, (lane>>4)*8 = k offset; C[m, b] lives
+// in lane (sub*4+r rows of column b = lane&15).
+// X LDS tile is XOR-swizzled (row&7)<<4 bytes so ds_read_b128 at a
+// power-of-two row stride stays bank-conflict-free.
+// ---------------------------------------------------------------------------
+template <int KC>
+DEVINL int x_swz(int row, int k) {  // element index into a [B][KC] tile
+  int byte = (row * KC + k) * 2;
+  byte ^= (row & 7) << 4;
+  return byte >> 1;
+}
+
+template <int B>
+__global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
+                                  const bf16* __restrict__ W,
+                                  const bf16* __restrict__ X,
+                                  const bf16* __restrict__ bias,
+                                  const bf16* __restrict__ res,
+                                  int M, int K) {
+  constexpr int KC = (B <= 32) ? 512 : (B <= 64 ? 256 : 128);
+  constexpr int BT = B / 16;      // 16-wide b tiles
+  constexpr int PRE = B * KC / 8 / 256;  // staged int4 loads per thread (8)
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  bf16* xbuf0 = reinterpret_cast<bf16*>(smem);
+  bf16* xbuf1 = xbuf0 + B * KC;
+  // c_red ([4 waves][16 rows][B] fp32, 16*B*16 bytes <= 32 KB) OVERLAYS
+  // the X buffers: the tiles are dead once the last chunk is consumed,
+  // so total LDS stays 2*B*KC*2 = 64 KB -> 2 blocks/CU
+  float* c_red = reinterpret_cast<float*>(smem);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = (int)blockIdx.x * 16;
+  const int arow = lane & 15;
+  const int sub = lane >> 4;
+  const int koff = sub * 8;
+  const int nch = K / KC;  // launcher guarantees K % KC == 0
+
+  f32x4 acc[BT];
+#pragma unroll
+  for (int t = 0; t < BT; ++t) acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // ---- bulk-coalesced stage of chunk 0 ----------------------------------
+  // thread i covers X element block (b = i / (KC/8), k8 = i % (KC/8))
+  {
+#pragma unroll
+    for (int u = 0; u < PRE; ++u) {
+      const int i = tid + u * 256;
+      const int b = i / (KC / 8);
+      const int k8 = (i - b * (KC / 8)) * 8;
+      bf16x8 xv = load8(X + (size_t)b * K + k8);
+      // k8 is 16-byte aligned and the swizzle permutes whole 16-B units,
+      // so the 8 elements land contiguously: one ds_write_b128
+      *reinterpret_cast<int4*>(&xbuf0[x_swz<KC>(b, k8)]) =
+          *reinterpret_cast<int4*>(xv.v);
+    }
+  }
+  __syncthreads();
+
+  const bf16* wrow = W + (size_t)(row0 + arow) * K;
+  for (int ci = 0; ci < nch; ++ci) {
+    bf16* cur = (ci & 1) ? xbuf1 : xbuf0;
+    bf16* nxt = (ci & 1) ? xbuf0 : xbuf1;
+    // issue next chunk's X loads EARLY (registers), write to LDS late
+    bf16x8 pre[PRE];
+    if (ci + 1 < nch) {
+      const int kb = (ci + 1) * KC;
+#pragma unroll
+      for (int u = 0; u < PRE; ++u) {
+        const int i = tid + u * 256;
+        const int b = i / (KC / 8);
+        const int k8 = (i - b * (KC / 8)) * 8;
+        pre[u] = load8(X + (size_t)b * K + kb + k8);
+      }
+    }
+    // compute: wave w covers k quarter [w*KC/4, (w+1)*KC/4) of the chunk
+    const int q0 = wave * (KC / 4);
+#pragma unroll 2
+    for (int s = 0; s < KC / 4 / 32; ++s) {
+      const int kl = q0 + s * 32;           // k local to the chunk
+      bf16x8 wr_ = load8_nt(wrow + (size_t)ci * KC + kl + koff);
+      const bf16x8_t wf = *reinterpret_cast<const bf16x8_t*>(wr_.v);
+#pragma unroll
+      for (int t = 0; t < BT; ++t) {
+        const bf16x8_t xf = *reinterpret_cast<const bf16x8_t*>(
+            &cur[x_swz<KC>(t * 16 + arow, kl + koff)]);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(wf, xf, acc[t],
+                                                         0, 0, 0);
+      }
+    }
+    __syncthreads();  // chunk consumed by every wave
+    if (ci + 1 < nch) {
+#pragma unroll
+      for (int u = 0; u < PRE; ++u) {
+        const int i = tid + u * 256;
+        const int b = i / (KC / 8);
+        const int k8 = (i - b * (KC / 8)) * 8;
+        *reinterpret_cast<int4*>(&nxt[x_swz<KC>(b, k8)]) =
+            *reinterpret_cast<int4*>(pre[u].v);
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- intra-block K reduce through LDS + epilogue ----------------------
+#pragma unroll
+  for (int t = 0; t < BT; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r)
+      c_red[(wave * 16 + sub * 4 + r) * B + t * 16 + arow] = acc[t][r];
+  __syncthreads();
+  // threads cover (row, b) pairs: 16*B values
+  for (int i = tid; i < 16 * B; i += 256) {
+    const int r = i / B;
+    const int b = i - r * B;
+    float v = c_red[(0 * 16 + r) * B + b] + c_red[(1 * 16 + r) * B + b] +
+              c_red[(2 * 16 + r) * B + b] + c_red[(3 * 16 + r) * B + b];
+    const int m = row0 + r;
+    if (bias != nullptr) v += b2f(bias[m]);
+    if (res != nullptr) v += b2f(res[(size_t)b * M + m]);
+    Y[(size_t)b * M + m] = f2b(v);
   }
 }
 
@@ -806,9 +948,6 @@ __global__ void rope_kv_append_kernel(
 // each lane owns 4 keys of ONE query head). qpk (query heads per kv
 // head) <= 16.
 // ---------------------------------------------------------------------------
-using f32x4 = __attribute__((__vector_size__(16))) float;
-using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
-
 #define ATTN_WAVES 4  // waves per block, each fully independent
 // max_seq at/below which the one-launch block-local variant is used; the
 // choice is static per engine (max_seq), so hipGraph shapes never change
@@ -2367,6 +2506,37 @@ int launch_attn_proj(void* out, const void* qkv, void* kpool, void* vpool,
                                      max_seq, scale, W, bias, res, gran, M,
                                      stream);
   }
+  return -1;
+}
+
+// grouped M-tile GEMM launcher; returns -1 when the shape has no
+// instantiation (caller falls back to hipBLASLt)
+int launch_mtile_gemm(void* Y, const void* W, const void* X,
+                      const void* bias, const void* res, int Bsz, int M,
+                      int K, hipStream_t stream) {
+  if (M % 16 != 0) return -1;
+  const int grid = M / 16;
+#define MT_CASE(BB, KCC)                                                    \
+  if (Bsz == BB) {                                                          \
+    if (K % KCC != 0 || K < 2 * KCC) return -1;                             \
+    static bool attr_set_##BB = false;                                      \
+    if (!attr_set_##BB) {                                                   \
+      (void)hipFuncSetAttribute(                                            \
+          reinterpret_cast<const void*>(&mtile_gemm_kernel<BB>),            \
+          hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);          \
+      attr_set_##BB = true;                                                 \
+    }                                                                       \
+    hipLaunchKernelGGL((mtile_gemm_kernel<BB>), dim3(grid), dim3(256),      \
+                       (size_t)2 * BB * KCC * 2, stream, (bf16*)Y,          \
+                       (const bf16*)W, (const bf16*)X, (const bf16*)bias,   \
+                       (const bf16*)res, M, K);                             \
+    return 0;                                                               \
+  }
+  MT_CASE(16, 512)
+  MT_CASE(32, 512)
+  MT_CASE(64, 256)
+  MT_CASE(128, 128)
+#undef MT_CASE
   return -1;
 }
 
