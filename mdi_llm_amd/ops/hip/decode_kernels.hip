@@ -503,6 +503,23 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
   __syncthreads();
 
   const bf16* wrow = W + (size_t)(row0 + arow) * K;
+  const int q0 = wave * (KC / 4);
+  constexpr int SPC = KC / 4 / 32;  // mfma steps per chunk per wave
+
+  // W fragment ring, TWO chunks deep: global loads stay in flight across
+  // s_barrier, so the weight stream never drains at a chunk boundary
+  // (the v1 form with unroll-2 inside the chunk measured 2-3.3 TB/s —
+  // latency-bound exactly like the unpipelined GEMVs)
+  bf16x8 wfr[2][SPC];
+#pragma unroll
+  for (int s = 0; s < SPC; ++s)
+    wfr[0][s] = load8_nt(wrow + q0 + s * 32 + koff);
+  if (1 < nch) {
+#pragma unroll
+    for (int s = 0; s < SPC; ++s)
+      wfr[1][s] = load8_nt(wrow + KC + q0 + s * 32 + koff);
+  }
+
   for (int ci = 0; ci < nch; ++ci) {
     bf16* cur = (ci & 1) ? xbuf1 : xbuf0;
     bf16* nxt = (ci & 1) ? xbuf0 : xbuf1;
@@ -518,13 +535,12 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
         pre[u] = load8(X + (size_t)b * K + kb + k8);
       }
     }
-    // compute: wave w covers k quarter [w*KC/4, (w+1)*KC/4) of the chunk
-    const int q0 = wave * (KC / 4);
-#pragma unroll 2
-    for (int s = 0; s < KC / 4 / 32; ++s) {
-      const int kl = q0 + s * 32;           // k local to the chunk
-      bf16x8 wr_ = load8_nt(wrow + (size_t)ci * KC + kl + koff);
-      const bf16x8_t wf = *reinterpret_cast<const bf16x8_t*>(wr_.v);
+    // consume this chunk's W ring slot against the staged X tile
+    bf16x8(&wc)[SPC] = wfr[ci & 1];
+#pragma unroll
+    for (int s = 0; s < SPC; ++s) {
+      const int kl = q0 + s * 32;
+      const bf16x8_t wf = *reinterpret_cast<const bf16x8_t*>(wc[s].v);
 #pragma unroll
       for (int t = 0; t < BT; ++t) {
         const bf16x8_t xf = *reinterpret_cast<const bf16x8_t*>(
@@ -533,7 +549,15 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
                                                          0, 0, 0);
       }
     }
-    __syncthreads();  // chunk consumed by every wave
+    // refill the consumed ring slot from chunk ci+2
+    if (ci + 2 < nch) {
+      const size_t kb = (size_t)(ci + 2) * KC;
+#pragma unroll
+      for (int s = 0; s < SPC; ++s)
+        wc[s] = load8_nt(wrow + kb + q0 + s * 32 + koff);
+    }
+    // write the pre-loaded X tile into the other buffer; ONE barrier per
+    // chunk (cur is only overwritten two iterations later)
     if (ci + 1 < nch) {
 #pragma unroll
       for (int u = 0; u < PRE; ++u) {
