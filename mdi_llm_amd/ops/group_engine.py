@@ -127,17 +127,27 @@ class GroupDecodeEngine:
         self.HN = torch.zeros(B, E, **bf)
         self.QKV = torch.zeros(B, cfg.qkv_dim, **bf)
         self.Y = torch.zeros(B, n_head * hs, **bf)
-        # hand-written M-tile MFMA GEMM path (MDI_MTILE=1 opts in; see
-        # ops/hip/decode_kernels.hip mtile_gemm_kernel): fixed buffers so
-        # the whole rotation stays hipGraph-capturable
-        self.use_mtile = (
-            os.environ.get("MDI_MTILE", "0") == "1"
-            and B in (16, 32, 64, 128)
-            and all(self._mtile_k_ok(B, k)
-                    for k in (E, n_head * hs, I))
+        # hand-written M-tile MFMA GEMM (ops/hip mtile_gemm_kernel).
+        # Measured per-shape vs hipBLASLt (profiles/mtile_gemm_r02.md):
+        # it WINS on the skinny-M proj shape (1.49x at B=32, 1.04x at
+        # B=64) and loses on the wide-M shapes (the library streams those
+        # at ~5 TB/s) — so the default is a hybrid: mtile for proj at
+        # B <= 64, hipBLASLt elsewhere.  MDI_MTILE=0 disables, =all
+        # forces every projection onto the hand-written kernel.
+        mt_env = os.environ.get("MDI_MTILE", "proj")
+        shapes_ok = (B in (16, 32, 64, 128)
+                     and all(self._mtile_k_ok(B, k)
+                             for k in (E, n_head * hs, I)))
+        self.use_mtile = mt_env == "all" and shapes_ok
+        self._mtile_proj = (
+            mt_env in ("all", "proj", "1") and B <= 64
+            and B in (16, 32, 64)
+            and self._mtile_k_ok(B, n_head * hs)
+            and E % 16 == 0
         )
-        if self.use_mtile:
+        if self.use_mtile or self._mtile_proj:
             self.A2 = torch.zeros(B, E, **bf)
+        if self.use_mtile:
             self.G = torch.zeros(B, I, **bf)
             self.U2 = torch.zeros(B, I, **bf)
         self.part_o = torch.zeros(B * n_head * n_chunks * hs, device=dev,
@@ -244,8 +254,13 @@ class GroupDecodeEngine:
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slots, li,
                 self.n_chunks, scale, self.B,
             )
-            # residual adds fused into the GEMMs (addmm beta=1) on bf16
-            if fp8:
+            # residual adds fused into the GEMMs (addmm beta=1) on bf16;
+            # the proj shape goes to the hand-written M-tile MFMA kernel
+            # where it measures faster than the library (see __init__)
+            if self._mtile_proj and not fp8:
+                self.ops.mtile_gemm(self.A2, w.proj_w, self.Y, w.proj_b, X)
+                a = self.A2
+            elif fp8:
                 a = X + self._mm(self.Y, w.proj_w8, w.proj_s, w.proj_w)
             elif w.proj_b is None:
                 a = torch.addmm(X, self.Y, w.proj_w.t())

@@ -510,14 +510,14 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
   // s_barrier, so the weight stream never drains at a chunk boundary
   // (the v1 form with unroll-2 inside the chunk measured 2-3.3 TB/s —
   // latency-bound exactly like the unpipelined GEMVs)
-  bf16x8 wfr[4][SPC];
+  bf16x8 wfr[2][SPC];
 #pragma unroll
-  for (int r4 = 0; r4 < 4; ++r4) {
-    if (r4 < nch) {
+  for (int s = 0; s < SPC; ++s)
+    wfr[0][s] = load8_nt(wrow + q0 + s * 32 + koff);
+  if (1 < nch) {
 #pragma unroll
-      for (int s = 0; s < SPC; ++s)
-        wfr[r4][s] = load8_nt(wrow + (size_t)r4 * KC + q0 + s * 32 + koff);
-    }
+    for (int s = 0; s < SPC; ++s)
+      wfr[1][s] = load8_nt(wrow + KC + q0 + s * 32 + koff);
   }
 
   // chunk loop unrolled by TWO so the ring slot and LDS buffer selection
@@ -547,8 +547,8 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
                                                          0, 0, 0);
       }
     }
-    if (ci + 4 < nch) {
-      const size_t kb = (size_t)(ci + 4) * KC;
+    if (ci + 2 < nch) {
+      const size_t kb = (size_t)(ci + 2) * KC;
 #pragma unroll
       for (int sI = 0; sI < SPC; ++sI)
         wc[sI] = load8_nt(wrow + kb + q0 + sI * 32 + koff);
@@ -565,11 +565,9 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
     }
     __syncthreads();
   };
-  for (int ci = 0; ci < nch; ci += 4) {  // launcher: nch % 4 == 0
+  for (int ci = 0; ci < nch; ci += 2) {   // launcher guarantees nch even
     chunk_body(ci, wfr[0], xbuf0, xbuf1);
     chunk_body(ci + 1, wfr[1], xbuf1, xbuf0);
-    chunk_body(ci + 2, wfr[2], xbuf0, xbuf1);
-    chunk_body(ci + 3, wfr[3], xbuf1, xbuf0);
   }
 
   // ---- intra-block K reduce through LDS + epilogue ----------------------
@@ -2543,7 +2541,7 @@ int launch_mtile_gemm(void* Y, const void* W, const void* X,
   const int grid = M / 16;
 #define MT_CASE(BB, KCC)                                                    \
   if (Bsz == BB) {                                                          \
-    if (K % (4 * KCC) != 0 || K < 4 * KCC) return -1;                             \
+    if (K % (2 * KCC) != 0 || K < 2 * KCC) return -1;                             \
     static bool attr_set_##BB = false;                                      \
     if (!attr_set_##BB) {                                                   \
       (void)hipFuncSetAttribute(                                            \
