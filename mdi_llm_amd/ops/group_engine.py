@@ -129,12 +129,13 @@ class GroupDecodeEngine:
         self.Y = torch.zeros(B, n_head * hs, **bf)
         # hand-written M-tile MFMA GEMM (ops/hip mtile_gemm_kernel).
         # Measured per-shape vs hipBLASLt (profiles/mtile_gemm_r02.md):
-        # it WINS on the skinny-M proj shape (1.49x at B=32, 1.04x at
-        # B=64) and loses on the wide-M shapes (the library streams those
-        # at ~5 TB/s) — so the default is a hybrid: mtile for proj at
-        # B <= 64, hipBLASLt elsewhere.  MDI_MTILE=0 disables, =all
-        # forces every projection onto the hand-written kernel.
-        mt_env = os.environ.get("MDI_MTILE", "proj")
+        # standalone it WINS the skinny-M proj shape (1.49x at B=32) and
+        # loses the wide-M shapes (~0.3-0.7x; the library streams those
+        # at ~5 TB/s), and IN-ROTATION even the proj win evaporates
+        # (addmm fuses the residual and picks a better algorithm:
+        # B=32 +0.2% noise, B=64 -2%).  Default OFF — a documented
+        # measured refutation; MDI_MTILE=proj|all opts in.
+        mt_env = os.environ.get("MDI_MTILE", "0")
         shapes_ok = (B in (16, 32, 64, 128)
                      and all(self._mtile_k_ok(B, k)
                              for k in (E, n_head * hs, I)))
