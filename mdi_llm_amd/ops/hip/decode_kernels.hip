@@ -16,6 +16,7 @@
 //  * all sequence positions / sample slots arrive via device memory
 //    (int32 tensors) so the whole decode step is hipGraph-replayable.
 
+#include <cstdlib>
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -2565,7 +2566,15 @@ int launch_mtile_gemm(void* Y, const void* W, const void* X,
 
 static inline int gemv_grid(int M, int rows_per_block) {
   int blocks = (M + rows_per_block - 1) / rows_per_block;
-  return blocks < 4096 ? blocks : 4096;
+  // optional cap: fewer, longer-lived waves (each loops rows, letting the
+  // next row's loads pipeline over the current row's tail) — sweep knob
+  static int cap = -1;
+  if (cap < 0) {
+    const char* e = getenv("MDI_GEMV_GRID_CAP");
+    cap = e ? atoi(e) : 4096;
+    if (cap <= 0) cap = 4096;
+  }
+  return blocks < cap ? blocks : cap;
 }
 
 void launch_rmsnorm(void* out, const void* x, const void* w, int n, float eps,
