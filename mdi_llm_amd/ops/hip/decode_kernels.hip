@@ -2772,8 +2772,16 @@ static void attn_dispatch2(void* out, float* part_o, float* part_ml,
                            int max_seq, int n_chunks, float scale,
                            int n_batch, int blocks, hipStream_t stream) {
   // short/medium contexts: one block per (batch, kv_head), no global
-  // partials, no combine kernel — the whole attention step is ONE launch
-  if (max_seq <= ATTN_BLOCK_MAX_SEQ) {
+  // partials, no combine kernel — the whole attention step is ONE launch.
+  // MDI_ATTN_FORCE_SPLITS=1 forces the split-S + combine path instead
+  // (A/B knob: the round-2 split-S got block-cooperative staging and a
+  // parallel combine, so its short-S cost may have changed)
+  static int force_splits = -1;
+  if (force_splits < 0) {
+    const char* e = getenv("MDI_ATTN_FORCE_SPLITS");
+    force_splits = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits) {
     const int nb = (n_batch > 0 ? n_batch : 1) * n_kv_heads;
     if (n_batch > 0) {
       hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, 1>), dim3(nb),
@@ -2878,7 +2886,13 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
       return -1;
   }
   if (rc != 0) return rc;
-  if (max_seq <= ATTN_BLOCK_MAX_SEQ) return 0;  // block-local: no combine
+  static int force_splits2 = -1;
+  if (force_splits2 < 0) {
+    const char* e = getenv("MDI_ATTN_FORCE_SPLITS");
+    force_splits2 = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits2)
+    return 0;  // block-local: no combine
   // the combine kernel is batch-agnostic: [B, n_head, chunks, hs] is just
   // B*n_head heads
   const int n_head_eff = n_kv_heads * qpk * (n_batch > 0 ? n_batch : 1);
