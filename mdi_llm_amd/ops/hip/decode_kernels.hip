@@ -591,6 +591,70 @@ __global__ void mtile_gemm_kernel(bf16* __restrict__ Y,
   }
 }
 
+// Register-staged fused-RMSNorm GEMV (qkv shape): the direct-x NORM==1
+// form interleaves x and norm-weight loads with the W stream every
+// chunk, which keeps the W pipeline at ~3.5 TB/s regardless of unroll
+// depth (measured).  Here the normalized x (bf16, one extra rounding —
+// matches the torch path, which also materializes the norm in bf16) is
+// staged into registers ONCE, and the row loop issues nothing but
+// non-temporal W loads: MAXCH*16 B in flight per lane.
+// LAUNCHED WITH A REDUCED GRID so each wave runs SEVERAL rows: the
+// first version ran 1 row/wave and the 16-load staging prologue cost as
+// much as the row itself (22 us vs 14.5 direct — measured, reverted);
+// amortized over ~3 rows the prologue is ~15% overhead.
+// K must be a multiple of 512 and <= MAXCH*512.
+template <int MAXCH>
+__global__ void gemv_direct_pre_kernel(bf16* __restrict__ out,
+                                       const bf16* __restrict__ W,
+                                       const bf16* __restrict__ x,
+                                       const bf16* __restrict__ bias,
+                                       const bf16* __restrict__ nw,
+                                       float eps, int M, int K) {
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int rows_per_grid = gridDim.x * (blockDim.x >> 6);
+  const int row0 = blockIdx.x * (blockDim.x >> 6) + wave;
+  const int nch = K >> 9;
+
+  bf16x8 xm[MAXCH];
+  float s2 = 0.f;
+#pragma unroll
+  for (int cI = 0; cI < MAXCH; ++cI) {
+    if (cI < nch) {
+      const int i = cI * 512 + lane * 8;
+      bf16x8 xv = load8(x + i);
+      bf16x8 gv = load8(nw + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float f = b2f(xv.v[j]);
+        s2 += f * f;
+        xm[cI].v[j] = f2b(f * b2f(gv.v[j]));
+      }
+    }
+  }
+  const float nscale = rsqrtf(wave_reduce_sum(s2) / K + eps);
+
+  for (int row = row0; row < M; row += rows_per_grid) {
+    const bf16* wr = W + (size_t)row * K + lane * 8;
+    float acc = 0.f;
+#pragma unroll
+    for (int cI = 0; cI < MAXCH; ++cI) {
+      if (cI < nch) {
+        bf16x8 wv = load8_nt(wr + cI * 512);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc += b2f(wv.v[j]) * b2f(xm[cI].v[j]);
+      }
+    }
+    acc = wave_reduce_sum(acc) * nscale;
+    if (lane == 0) {
+      if (bias != nullptr) acc += b2f(bias[row]);
+      out[row] = f2b(acc);
+    }
+  }
+}
+
 // MoE router: top-k of the gate logits (<= 64 experts) + softmax over
 // the selected k (reference model.py:823-853 semantics: topk first, then
 // softmax over the k logits).  One wave; outputs device-side so a
@@ -2599,6 +2663,29 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
   const int smem = K * sizeof(bf16);
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
+  // register-staged fused-RMS form for the qkv shape, grid shrunk so the
+  // x*nw staging prologue amortizes over ~3 rows per wave (see kernel)
+  static int pre_on = -1;
+  if (pre_on < 0) {
+    const char* e = getenv("MDI_GEMV_PRE");
+    pre_on = e ? atoi(e) : 1;
+  }
+  if (pre_on && epilogue == 0 && norm_kind == 1 && rows == 1 &&
+      eidx == nullptr && res == nullptr && K >= 512 && K % 512 == 0 &&
+      K <= 8192 && M % 4 == 0) {
+    dim3 pgrid(gemv_grid(M, 4 * 3));
+    if (K <= 4096)
+      hipLaunchKernelGGL((gemv_direct_pre_kernel<8>), pgrid, block, 0,
+                         stream, (bf16*)out, (const bf16*)W,
+                         (const bf16*)x, (const bf16*)bias,
+                         (const bf16*)norm_w, eps, M, K);
+    else
+      hipLaunchKernelGGL((gemv_direct_pre_kernel<16>), pgrid, block, 0,
+                         stream, (bf16*)out, (const bf16*)W,
+                         (const bf16*)x, (const bf16*)bias,
+                         (const bf16*)norm_w, eps, M, K);
+    return;
+  }
   // NORM 0/1: the direct-x kernel (no staging barrier); 2: staged LDS form
 #define GEMV_CASE1(E, N, R)                                                 \
   do {                                                                      \
@@ -2654,6 +2741,29 @@ void launch_gemv_fp8(void* out, const void* W, const float* wscale,
   const int smem = K * sizeof(bf16);
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
+  // register-staged fused-RMS form for the qkv shape, grid shrunk so the
+  // x*nw staging prologue amortizes over ~3 rows per wave (see kernel)
+  static int pre_on = -1;
+  if (pre_on < 0) {
+    const char* e = getenv("MDI_GEMV_PRE");
+    pre_on = e ? atoi(e) : 1;
+  }
+  if (pre_on && epilogue == 0 && norm_kind == 1 && rows == 1 &&
+      eidx == nullptr && res == nullptr && K >= 512 && K % 512 == 0 &&
+      K <= 8192 && M % 4 == 0) {
+    dim3 pgrid(gemv_grid(M, 4 * 3));
+    if (K <= 4096)
+      hipLaunchKernelGGL((gemv_direct_pre_kernel<8>), pgrid, block, 0,
+                         stream, (bf16*)out, (const bf16*)W,
+                         (const bf16*)x, (const bf16*)bias,
+                         (const bf16*)norm_w, eps, M, K);
+    else
+      hipLaunchKernelGGL((gemv_direct_pre_kernel<16>), pgrid, block, 0,
+                         stream, (bf16*)out, (const bf16*)W,
+                         (const bf16*)x, (const bf16*)bias,
+                         (const bf16*)norm_w, eps, M, K);
+    return;
+  }
 #define GEMV8_CASE1(E, N, R)                                                \
   hipLaunchKernelGGL((gemv_fp8_kernel<E, N, R>), grid, block, smem, stream, \
                      (bf16*)out, (const unsigned char*)W, wscale,           \
