@@ -2668,7 +2668,7 @@ void launch_gemv(void* out, const void* W, const void* x, const void* bias,
   static int pre_on = -1;
   if (pre_on < 0) {
     const char* e = getenv("MDI_GEMV_PRE");
-    pre_on = e ? atoi(e) : 1;
+    pre_on = e ? atoi(e) : 0;
   }
   if (pre_on && epilogue == 0 && norm_kind == 1 && rows == 1 &&
       eidx == nullptr && res == nullptr && K >= 512 && K % 512 == 0 &&
@@ -2746,7 +2746,7 @@ void launch_gemv_fp8(void* out, const void* W, const float* wscale,
   static int pre_on = -1;
   if (pre_on < 0) {
     const char* e = getenv("MDI_GEMV_PRE");
-    pre_on = e ? atoi(e) : 1;
+    pre_on = e ? atoi(e) : 0;
   }
   if (pre_on && epilogue == 0 && norm_kind == 1 && rows == 1 &&
       eidx == nullptr && res == nullptr && K >= 512 && K % 512 == 0 &&
