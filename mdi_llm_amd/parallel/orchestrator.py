@@ -188,7 +188,11 @@ class MDIRuntime:
         if self.world > 1:
             os.environ.setdefault("MASTER_ADDR", self.topology.master_addr)
             os.environ.setdefault("MASTER_PORT", str(self.topology.master_port))
+            # MDI_DIST_BACKEND overrides (e.g. gloo to run several ranks
+            # on ONE GPU — RCCL refuses duplicate devices; the ring then
+            # stages through host mirrors automatically)
             backend = "nccl" if self.device.type == "cuda" else "gloo"
+            backend = os.environ.get("MDI_DIST_BACKEND", backend)
             # bounded timeout: a dead peer surfaces as an error instead of a
             # wedged ring (reference detects peer death via zero-byte recv,
             # connections.py:174-182)
@@ -272,6 +276,7 @@ class MDIRuntime:
         os.environ.setdefault("MASTER_ADDR", msg["master_addr"])
         os.environ.setdefault("MASTER_PORT", str(msg["master_port"]))
         backend = "nccl" if self.device.type == "cuda" else "gloo"
+        backend = os.environ.get("MDI_DIST_BACKEND", backend)
         dist.init_process_group(backend, rank=msg["rank"],
                                 world_size=msg["world"],
                                 timeout=timedelta(seconds=600))
