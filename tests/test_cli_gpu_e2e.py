@@ -82,3 +82,31 @@ def test_starter_secondary_cli_one_gpu(tmp_path):
     finally:
         if sec.poll() is None:
             sec.kill()
+
+
+def test_train_one_gpu(tmp_path):
+    """prepare_data + train.py (AMP bf16, AdamW, cosine LR) on cuda:0."""
+    sys.path.insert(0, str(ROOT))
+    from tests.helpers import make_toy_checkpoint
+
+    ckpt = tmp_path / "ckpt" / "nano-gpu"
+    make_toy_checkpoint(ckpt, name="nano-gpu")
+    data = tmp_path / "data"
+    text = tmp_path / "input.txt"
+    text.write_text("the quick brown fox jumps over the lazy dog. " * 400)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(ROOT)
+    r = subprocess.run(
+        [sys.executable, str(ROOT / "prepare_data.py"), "--input",
+         str(text), "--tokenizer-dir", str(ckpt), "--out-dir", str(data)],
+        env=env, capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-1000:]
+    r = subprocess.run(
+        [sys.executable, str(ROOT / "train.py"), "--ckpt", str(ckpt),
+         "--data-dir", str(data), "--max-iters", "10", "--batch-size", "4",
+         "--block-size", "64", "--grad-accum", "2", "--dtype", "bfloat16",
+         "--device", "cuda:0", "--eval-interval", "100",
+         "--log-interval", "2"],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "loss" in r.stdout
