@@ -112,4 +112,11 @@ if __name__ == "__main__":
     p.add_argument("--top-k", type=int, default=200)
     p.add_argument("--sequence-length", type=int, default=None)
     p.add_argument("--seed", type=int, default=10137)
-    main(p.parse_args())
+    p.add_argument("-d", "--debug", action="store_true",
+                   help="write debug logs to logs/logs_chat.log")
+    _args = p.parse_args()
+    if _args.debug:
+        from mdi_llm_amd.utils.console import setup_debug_logging
+
+        setup_debug_logging("chat", SCRIPT_DIR / "logs")
+    main(_args)

@@ -125,8 +125,6 @@ class DecodeEngine:
         n_chunks: int = 16,
         use_graphs: bool = True,
     ) -> None:
-        import os
-
         # adaptive split-S: 16 chunks up to 4k contexts (measured best at
         # short S; the one-launch block-local kernel is used there anyway),
         # then one wave per 32 keys capped at 256 chunks — 8 kv-heads x
