@@ -1687,29 +1687,39 @@ __global__ void attn_decode_kernel(
   for (int i = 0; i < ODIM; ++i) o_acc[i] = 0.f;
 
   const int n_tiles = keys_per_chunk / 16;
-  for (int t = 0; t < n_tiles; ++t) {
-    const int key0 = k_begin + t * 16;
-    // NOTE: no early break — every wave in the block must reach every
-    // __syncthreads(); empty tiles run fully masked.
+  const int arow = lane & 15;          // key row in tile (and B's qhead col)
+  const int koff = (lane >> 4) * 8;    // dim offset within 32-chunk
 
-    // ---- QK^T via MFMA: A = K tile (16 keys x 32 dims), B = q^T ----------
-    f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
-    const int arow = lane & 15;          // key row in tile (and B's qhead col)
-    const int koff = (lane >> 4) * 8;    // dim offset within 32-chunk
-    const int akey = key0 + arow;
+  // K-tile fragments double-buffered in registers: tile t+1's four
+  // 16-B loads are issued before tile t's MFMA/softmax/PV consume them
+  // (global loads stay in flight across the barriers), so the 2-4 serial
+  // tiles per wave at long S no longer stack their load latencies.
+  auto load_af = [&](int t, bf16x8_t (&af)[HS / 32]) {
+    const int akey = k_begin + t * 16 + arow;
     const bf16* krow = kpool + cache_base + (size_t)akey * head_size;
     const bool row_valid = akey < S;
     const bool row_cur = akey == pos;  // newest key: read from LDS k_cur
 #pragma unroll
-    for (int c = 0; c < HS / 32; ++c) {  // K=32 chunks cover head_size
-      bf16x8_t af = {};
+    for (int c = 0; c < HS / 32; ++c) {
+      af[c] = bf16x8_t{};
       if (row_cur)
-        af = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
+        af[c] = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
       else if (row_valid)
-        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+        af[c] = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+    }
+  };
+
+  auto tile_body = [&](int t, const bf16x8_t (&afr)[HS / 32]) {
+    const int key0 = k_begin + t * 16;
+    // NOTE: no early break — every wave in the block must reach every
+    // __syncthreads(); empty tiles run fully masked.
+    f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int c = 0; c < HS / 32; ++c) {
       const bf16x8_t bf = *reinterpret_cast<const bf16x8_t*>(
           &q_lds[q_swz<HS>(arow, c * 32 + koff)]);
-      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc4, 0, 0, 0);
+      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr[c], bf, acc4,
+                                                     0, 0, 0);
     }
     // C[row=key, col=qhead]: lane holds keys (sub*4 + r) for qhead qa
     float sc[4];
@@ -1718,7 +1728,6 @@ __global__ void attn_decode_kernel(
       const int key = key0 + sub * 4 + r;
       sc[r] = (key < S) ? acc4[r] * scale : -1e30f;
     }
-
     // ---- online softmax per qhead (4 lanes per qhead: strides 16,32) -----
     float tmax = fmaxf(fmaxf(sc[0], sc[1]), fmaxf(sc[2], sc[3]));
     tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
@@ -1779,6 +1788,15 @@ __global__ void attn_decode_kernel(
         o_acc[0] += p * b2f(vrow[0]);
       }
     }
+  };
+
+  bf16x8_t afA[HS / 32], afB[HS / 32];
+  if (n_tiles > 0) load_af(0, afA);
+  for (int t = 0; t < n_tiles; t += 2) {
+    if (t + 1 < n_tiles) load_af(t + 1, afB);
+    tile_body(t, afA);
+    if (t + 2 < n_tiles) load_af(t + 2, afA);
+    if (t + 1 < n_tiles) tile_body(t + 1, afB);
   }
 
   // ---- append the current token's k,v to the pool ------------------------
@@ -2741,29 +2759,6 @@ void launch_gemv_fp8(void* out, const void* W, const float* wscale,
   const int smem = K * sizeof(bf16);
   if (rows == 0) rows = M >= 32768 ? 4 : (M > 8192 ? 2 : 1);
   dim3 grid(gemv_grid(M, 4 * rows)), block(256);
-  // register-staged fused-RMS form for the qkv shape, grid shrunk so the
-  // x*nw staging prologue amortizes over ~3 rows per wave (see kernel)
-  static int pre_on = -1;
-  if (pre_on < 0) {
-    const char* e = getenv("MDI_GEMV_PRE");
-    pre_on = e ? atoi(e) : 0;
-  }
-  if (pre_on && epilogue == 0 && norm_kind == 1 && rows == 1 &&
-      eidx == nullptr && res == nullptr && K >= 512 && K % 512 == 0 &&
-      K <= 8192 && M % 4 == 0) {
-    dim3 pgrid(gemv_grid(M, 4 * 3));
-    if (K <= 4096)
-      hipLaunchKernelGGL((gemv_direct_pre_kernel<8>), pgrid, block, 0,
-                         stream, (bf16*)out, (const bf16*)W,
-                         (const bf16*)x, (const bf16*)bias,
-                         (const bf16*)norm_w, eps, M, K);
-    else
-      hipLaunchKernelGGL((gemv_direct_pre_kernel<16>), pgrid, block, 0,
-                         stream, (bf16*)out, (const bf16*)W,
-                         (const bf16*)x, (const bf16*)bias,
-                         (const bf16*)norm_w, eps, M, K);
-    return;
-  }
 #define GEMV8_CASE1(E, N, R)                                                \
   hipLaunchKernelGGL((gemv_fp8_kernel<E, N, R>), grid, block, smem, stream, \
                      (bf16*)out, (const unsigned char*)W, wscale,           \
