@@ -42,6 +42,9 @@ def parse_args():
     ap.add_argument("--weights", choices=["bf16", "fp8"], default="bf16",
                     help="decode weight dtype (fp8 = e4m3 per-row-scaled; "
                          "NOT the headline config — reported in dtype)")
+    ap.add_argument("--kv", choices=["bf16", "fp8"], default="bf16",
+                    help="KV-cache dtype (fp8 = e4m3 per-row-scaled cache; "
+                         "NOT the headline config — reported in dtype)")
     ap.add_argument("--no-graphs", action="store_true")
     return ap.parse_args()
 
@@ -54,6 +57,8 @@ def main():
     args = parse_args()
     if args.weights == "fp8":
         os.environ["MDI_WEIGHT_DTYPE"] = "fp8"
+    if args.kv == "fp8":
+        os.environ["MDI_KV_DTYPE"] = "fp8"
     from mdi_llm_amd.config import ModelConfig
     from mdi_llm_amd.models.stages import build_stage
     from mdi_llm_amd.parallel.ring import RingComm
@@ -261,7 +266,9 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": ("bf16_act_fp8_weights" if args.weights == "fp8" else ("bf16" if on_gpu else "float32")),
+            "dtype": (("bf16_act_fp8_weights" if args.weights == "fp8"
+                       else ("bf16" if on_gpu else "float32"))
+                      + ("_fp8kv" if args.kv == "fp8" else "")),
             "data": "synthetic",
             "config": {
                 "model": cfg.name,

@@ -141,6 +141,25 @@ class KVCachePool:
         self.seq_len = torch.zeros(n_slots, dtype=torch.int64)
         self.n_slots = n_slots
         self.max_seq = max_seq
+        self.fp8 = False
+        self.kscale = None
+        self.vscale = None
+
+    def to_fp8(self) -> None:
+        """Convert the arena to an fp8 (OCP e4m3) cache: uint8 payload +
+        one fp32 scale per cached row.  Only the HIP kernels read/write
+        this layout; the torch ``append`` path refuses (the engine gates
+        fp8-KV on the HIP prefill path)."""
+        if self.fp8:
+            return
+        shape = self.k.shape
+        dev = self.k.device
+        self.k = torch.zeros(shape, device=dev, dtype=torch.uint8)
+        self.v = torch.zeros(shape, device=dev, dtype=torch.uint8)
+        self.kscale = torch.zeros(shape[:-1], device=dev,
+                                  dtype=torch.float32)
+        self.vscale = torch.zeros_like(self.kscale)
+        self.fp8 = True
 
     def reset(self, slot: Optional[int] = None) -> None:
         if slot is None:
@@ -153,6 +172,12 @@ class KVCachePool:
     ) -> Tuple[torch.Tensor, torch.Tensor]:
         """Write k,v of shape (n_kv_heads, T, head_size) at ``pos``; return
         views of the full cache up to ``pos+T``."""
+        if self.fp8:
+            raise RuntimeError(
+                "fp8 KV cache is written only by the HIP kernels "
+                "(rope_prefill_append / fused attention append); the torch "
+                "path cannot append to it"
+            )
         T = k.size(-2)
         self.k[slot, layer, :, pos : pos + T] = k
         self.v[slot, layer, :, pos : pos + T] = v

@@ -139,6 +139,7 @@ class DecodeEngine:
         n_chunks = int(os.environ.get("MDI_ATTN_CHUNKS", n_chunks))
         n_chunks = max(4, (n_chunks // 4) * 4)  # block-shared q staging
         self.weight_dtype = os.environ.get("MDI_WEIGHT_DTYPE", "bf16")
+        self.kv_dtype = os.environ.get("MDI_KV_DTYPE", "bf16")
         self.ops = require_hip_ops()
         self.config: ModelConfig = stage.config
         cfg = self.config
@@ -219,6 +220,26 @@ class DecodeEngine:
         )
         self.pos = torch.zeros(1, device=dev, dtype=torch.int32)
 
+        # fp8 (OCP e4m3) KV cache: halves KV-read bandwidth at long
+        # context and halves cache memory per sample.  Per-row scales;
+        # written only by the HIP kernels, so it requires the HIP prefill
+        # path (torch prefill would need to append bf16 rows).
+        self.kv8 = (
+            self.kv_dtype == "fp8"
+            and self.device.type == "cuda"
+            and self.supports_hip_prefill
+        )
+        if self.kv_dtype == "fp8" and not self.kv8:
+            import warnings
+
+            warnings.warn(
+                f"MDI_KV_DTYPE=fp8 ignored for {cfg.name!r}: the fp8 KV "
+                "cache needs the HIP prefill path (RMSNorm + LLaMA/Gemma "
+                "MLP, sequential residual)"
+            )
+        if self.kv8:
+            kv_pool.to_fp8()
+
         # fused attention+proj (one launch, in-launch granule hand-off):
         # the proj weight stream overlaps the attention compute instead of
         # serializing behind it (attn_proj_kernel in decode_kernels.hip).
@@ -229,6 +250,7 @@ class DecodeEngine:
         K_attn = cfg.n_head * cfg.head_size
         self._fuse_attn_proj = (
             not self.fp8
+            and not self.kv8
             and not cfg.parallel_residual
             and stage.max_seq_length <= 4096
             and K_attn % 128 == 0
@@ -363,6 +385,7 @@ class DecodeEngine:
                 self.y, self.part_o, self.part_ml, self.qkv, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slot, li,
                 self.n_chunks, scale,
+                kscale=self.kv_pool.kscale, vscale=self.kv_pool.vscale,
             )
             if cfg.parallel_residual:
                 # x = x + proj(y) + mlp(norm2(x) or norm1(x))
@@ -539,9 +562,13 @@ class DecodeEngine:
             ops.rmsnorm(xn, x, w.norm1_w, cfg.norm_eps)
             qkv = F.linear(xn, w.attn_w, w.attn_b)
             ops.rope_prefill_append(qkv, self.kv_pool.k, self.kv_pool.v,
-                                    self.cos, self.sin, pos0, slot, li)
+                                    self.cos, self.sin, pos0, slot, li,
+                                    kscale=self.kv_pool.kscale,
+                                    vscale=self.kv_pool.vscale)
             ops.prefill_attn(Y, qkv, self.kv_pool.k, self.kv_pool.v, pos0,
-                             slot, li, scale)
+                             slot, li, scale,
+                             kscale=self.kv_pool.kscale,
+                             vscale=self.kv_pool.vscale)
             a = x + F.linear(Y, w.proj_w, w.proj_b)
             hn = xn
             ops.rmsnorm(hn, a, w.norm2_w, cfg.norm_eps)

@@ -222,6 +222,7 @@ class GroupDecodeEngine:
                 self.Y, self.part_o, self.part_ml, self.QKV, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slots,
                 li, self.n_chunks, scale, self.B,
+                kscale=self.kv_pool.kscale, vscale=self.kv_pool.vscale,
             )
             ops.mtile_gemm(self.A2, w.proj_w, self.Y, w.proj_b, self.X)
             hn = self._rms(self.A2, w.norm2_w, self.HN)
@@ -254,6 +255,7 @@ class GroupDecodeEngine:
                 self.Y, self.part_o, self.part_ml, qkv, self.kv_pool.k,
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slots, li,
                 self.n_chunks, scale, self.B,
+                kscale=self.kv_pool.kscale, vscale=self.kv_pool.vscale,
             )
             # residual adds fused into the GEMMs (addmm beta=1) on bf16;
             # the proj shape goes to the hand-written M-tile MFMA kernel

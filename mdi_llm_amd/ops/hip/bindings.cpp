@@ -286,13 +286,24 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
                  torch::Tensor kpool, torch::Tensor vpool,
                  torch::Tensor cos_t, torch::Tensor sin_t, torch::Tensor pos,
                  torch::Tensor slot, int64_t layer, int64_t n_chunks,
-                 double scale, int64_t n_batch) {
+                 double scale, int64_t n_batch,
+                 c10::optional<torch::Tensor> kscale,
+                 c10::optional<torch::Tensor> vscale) {
   check_bf16(out, "out");
   check_f32(part_o, "part_o");
   check_f32(part_ml, "part_ml");
   check_bf16(qkv, "qkv");
-  check_bf16(kpool, "kpool");
-  check_bf16(vpool, "vpool");
+  const bool kv8 = kscale.has_value();
+  if (kv8) {
+    TORCH_CHECK(kpool.scalar_type() == torch::kUInt8 &&
+                    vpool.scalar_type() == torch::kUInt8,
+                "fp8 KV cache must be uint8 pools");
+    check_f32(*kscale, "kscale");
+    check_f32(*vscale, "vscale");
+  } else {
+    check_bf16(kpool, "kpool");
+    check_bf16(vpool, "vpool");
+  }
   check_f32(cos_t, "cos");
   check_f32(sin_t, "sin");
   check_i32(pos, "pos");
@@ -319,6 +330,8 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
   int rc = launch_attn_decode(
       out.data_ptr(), part_o.data_ptr<float>(), part_ml.data_ptr<float>(),
       qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      kv8 ? kscale->data_ptr<float>() : nullptr,
+      kv8 ? vscale->data_ptr<float>() : nullptr,
       rope_ne ? cos_t.data_ptr<float>() : nullptr,
       rope_ne ? sin_t.data_ptr<float>() : nullptr, rope_ne,
       pos.data_ptr<int>(), slot.data_ptr<int>(), (int)layer, n_layers_pool,
@@ -473,10 +486,21 @@ void stage_slot(torch::Tensor slot, c10::optional<torch::Tensor> pos_out,
 void rope_prefill_append(torch::Tensor qkv, torch::Tensor kpool,
                          torch::Tensor vpool, torch::Tensor cos_t,
                          torch::Tensor sin_t, int64_t pos0, int64_t slot,
-                         int64_t layer) {
+                         int64_t layer,
+                         c10::optional<torch::Tensor> kscale,
+                         c10::optional<torch::Tensor> vscale) {
   check_bf16(qkv, "qkv");
-  check_bf16(kpool, "kpool");
-  check_bf16(vpool, "vpool");
+  const bool kv8 = kscale.has_value();
+  if (kv8) {
+    TORCH_CHECK(kpool.scalar_type() == torch::kUInt8 &&
+                    vpool.scalar_type() == torch::kUInt8,
+                "fp8 KV cache must be uint8 pools");
+    check_f32(*kscale, "kscale");
+    check_f32(*vscale, "vscale");
+  } else {
+    check_bf16(kpool, "kpool");
+    check_bf16(vpool, "vpool");
+  }
   check_f32(cos_t, "cos");
   check_f32(sin_t, "sin");
   const int n_layers_pool = (int)kpool.size(1);
@@ -490,6 +514,8 @@ void rope_prefill_append(torch::Tensor qkv, torch::Tensor kpool,
   TORCH_CHECK((int64_t)pos0 + T <= max_seq, "prefill overflows the pool");
   launch_rope_prefill_append(
       qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      kv8 ? kscale->data_ptr<float>() : nullptr,
+      kv8 ? vscale->data_ptr<float>() : nullptr,
       rope_ne ? cos_t.data_ptr<float>() : nullptr,
       rope_ne ? sin_t.data_ptr<float>() : nullptr, (int)pos0, (int)slot,
       (int)layer, n_layers_pool, n_kv, max_seq, hs, rope_ne, qpk, T,
@@ -498,11 +524,22 @@ void rope_prefill_append(torch::Tensor qkv, torch::Tensor kpool,
 
 void prefill_attn(torch::Tensor out, torch::Tensor qkv, torch::Tensor kpool,
                   torch::Tensor vpool, int64_t pos0, int64_t slot,
-                  int64_t layer, double scale) {
+                  int64_t layer, double scale,
+                  c10::optional<torch::Tensor> kscale,
+                  c10::optional<torch::Tensor> vscale) {
   check_bf16(out, "out");
   check_bf16(qkv, "qkv");
-  check_bf16(kpool, "kpool");
-  check_bf16(vpool, "vpool");
+  const bool kv8 = kscale.has_value();
+  if (kv8) {
+    TORCH_CHECK(kpool.scalar_type() == torch::kUInt8 &&
+                    vpool.scalar_type() == torch::kUInt8,
+                "fp8 KV cache must be uint8 pools");
+    check_f32(*kscale, "kscale");
+    check_f32(*vscale, "vscale");
+  } else {
+    check_bf16(kpool, "kpool");
+    check_bf16(vpool, "vpool");
+  }
   const int n_layers_pool = (int)kpool.size(1);
   const int n_kv = (int)kpool.size(2);
   const int max_seq = (int)kpool.size(3);
@@ -512,6 +549,8 @@ void prefill_attn(torch::Tensor out, torch::Tensor qkv, torch::Tensor kpool,
   TORCH_CHECK(out.numel() == (int64_t)T * n_kv * qpk * hs, "out size");
   int rc = launch_prefill_attn(
       out.data_ptr(), qkv.data_ptr(), kpool.data_ptr(), vpool.data_ptr(),
+      kv8 ? kscale->data_ptr<float>() : nullptr,
+      kv8 ? vscale->data_ptr<float>() : nullptr,
       (int)pos0, (int)slot, (int)layer, n_layers_pool, n_kv, max_seq, hs,
       qpk, T, (float)scale, cur_stream());
   TORCH_CHECK(rc == 0, "prefill_attn: unsupported head_size ", hs);
@@ -572,12 +611,20 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("part_o"), py::arg("part_ml"),
         py::arg("qkv"), py::arg("kpool"), py::arg("vpool"), py::arg("cos"),
         py::arg("sin"), py::arg("pos"), py::arg("slot"), py::arg("layer"),
-        py::arg("n_chunks"), py::arg("scale"), py::arg("n_batch") = 0);
+        py::arg("n_chunks"), py::arg("scale"), py::arg("n_batch") = 0,
+        py::arg("kscale") = c10::nullopt, py::arg("vscale") = c10::nullopt);
   m.def("add", &add, "bf16 residual add");
   m.def("rope_prefill_append", &rope_prefill_append,
-        "prefill: rope q/k for T positions + append k/v to the pool");
+        "prefill: rope q/k for T positions + append k/v to the pool",
+        py::arg("qkv"), py::arg("kpool"), py::arg("vpool"), py::arg("cos"),
+        py::arg("sin"), py::arg("pos0"), py::arg("slot"), py::arg("layer"),
+        py::arg("kscale") = c10::nullopt, py::arg("vscale") = c10::nullopt);
   m.def("prefill_attn", &prefill_attn,
-        "causal GQA prefill flash attention (MFMA, online softmax)");
+        "causal GQA prefill flash attention (MFMA, online softmax)",
+        py::arg("out"), py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
+        py::arg("pos0"), py::arg("slot"), py::arg("layer"),
+        py::arg("scale"), py::arg("kscale") = c10::nullopt,
+        py::arg("vscale") = c10::nullopt);
   m.def("attn_proj", &attn_proj,
         "fused GQA flash-decode attention + output projection (one "
         "launch; in-launch granule hand-off overlaps the proj weight "

@@ -70,6 +70,7 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 // on qkv/part_o/part_ml/out).
 int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        const void* qkv, void* kpool, void* vpool,
+                       float* kscale, float* vscale,
                        const float* cos_t, const float* sin_t, int rope_ne,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
@@ -123,6 +124,7 @@ void launch_stage_slot(int* pos_out, int* token_out, const int* pos_table,
 
 // prefill: rope+append all T positions (grid covers T)
 void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,
+                                float* kscale, float* vscale,
                                 const float* cos_t, const float* sin_t,
                                 int pos0, int slot, int layer,
                                 int n_layers_pool, int n_kv_heads,
@@ -131,7 +133,8 @@ void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,
 
 // prefill flash attention (causal, GQA); out [T, n_head*head_size]
 int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
-                        const void* vpool, int pos0, int slot, int layer,
+                        const void* vpool, const float* kscale,
+                        const float* vscale, int pos0, int slot, int layer,
                         int n_layers_pool, int n_kv_heads, int max_seq,
                         int head_size, int qpk, int T, float scale,
                         hipStream_t stream);

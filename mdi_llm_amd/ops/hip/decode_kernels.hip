@@ -17,6 +17,7 @@
 //    (int32 tensors) so the whole decode step is hipGraph-replayable.
 
 #include <cstdlib>
+#include <type_traits>
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 
@@ -57,6 +58,54 @@ DEVINL i32x4_t load16_nt_u8(const unsigned char* p) {
 
 using f32x4 = __attribute__((__vector_size__(16))) float;
 using bf16x8_t = __attribute__((ext_vector_type(8))) __bf16;
+
+DEVINL float wave_reduce_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+// ---- fp8 (OCP e4m3) KV cache -------------------------------------------
+// Rows are stored per-row-scaled (one fp32 scale per cached (slot, layer,
+// kv-head, position) row, absmax/448 — same scheme as the fp8 weights).
+// gfx950 cvt builtins convert packed pairs; KV8 kernel variants load 8
+// bytes instead of 16 per fragment and dequantize in registers.
+using u8kv = unsigned char;
+typedef float v2fkv __attribute__((ext_vector_type(2)));
+
+DEVINL u8kv f32_to_fp8(float f) {
+  const int packed = __builtin_amdgcn_cvt_pk_fp8_f32(f, 0.f, 0, false);
+  return (u8kv)(packed & 0xFF);
+}
+
+DEVINL float fp8_to_f32(u8kv v) {
+  return __builtin_amdgcn_cvt_f32_fp8((int)v, 0);
+}
+
+DEVINL bf16x8_t fp8x8_to_bf16(const u8kv* p, float scl) {
+  const int lo = *reinterpret_cast<const int*>(p);
+  const int hi = *reinterpret_cast<const int*>(p + 4);
+  const v2fkv a = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
+  const v2fkv b = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
+  const v2fkv c = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
+  const v2fkv d = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
+  bf16x8_t r;
+  r[0] = (__bf16)(a.x * scl);
+  r[1] = (__bf16)(a.y * scl);
+  r[2] = (__bf16)(b.x * scl);
+  r[3] = (__bf16)(b.y * scl);
+  r[4] = (__bf16)(c.x * scl);
+  r[5] = (__bf16)(c.y * scl);
+  r[6] = (__bf16)(d.x * scl);
+  r[7] = (__bf16)(d.y * scl);
+  return r;
+}
+
+// atomicMax for a non-negative float held in LDS as its uint bits
+DEVINL void lds_fmax_u(unsigned* addr, float v) {
+  atomicMax(addr, __float_as_uint(v));
+}
 
 DEVINL float wave_reduce_sum(float v) {
 #pragma unroll
@@ -1066,15 +1115,19 @@ DEVINL float rope_elem(const bf16* row, int d, int ne,
 
 // Block-local variant: grid.x = (n_batch) * n_kv_heads, block = 256.
 // BATCH semantics as in attn_decode_kernel below.
-template <int QPK, int HS, int BATCH>
+template <int QPK, int HS, int BATCH, int KV8>
 __global__ void attn_decode_block_kernel(
     bf16* __restrict__ out,       // [B?, n_head * head_size]
     const bf16* __restrict__ qkv, // [B?, qkv_dim] interleaved, RAW
-    bf16* __restrict__ kpool, bf16* __restrict__ vpool,
+    void* __restrict__ kpool_v, void* __restrict__ vpool_v,
+    float* __restrict__ kscale, float* __restrict__ vscale,
     const float* __restrict__ cos_t, const float* __restrict__ sin_t,
     int rope_ne, const int* __restrict__ pos_p,
     const int* __restrict__ slot_p, int layer, int n_layers_pool,
     int n_kv_heads, int max_seq, float scale, int n_batch) {
+  using kvt = std::conditional_t<KV8 != 0, u8kv, bf16>;
+  kvt* kpool = reinterpret_cast<kvt*>(kpool_v);
+  kvt* vpool = reinterpret_cast<kvt*>(vpool_v);
   constexpr int head_size = HS;
   static_assert(HS % 32 == 0 && HS * QPK >= 64, "unsupported attn geometry");
   constexpr int ODIM = HS * QPK / 64;  // output dims per lane (PV map)
@@ -1085,6 +1138,7 @@ __global__ void attn_decode_block_kernel(
   __shared__ __attribute__((aligned(16))) bf16 k_cur[HS];
   __shared__ float o_part[ATTN_WAVES][QPK][HS];
   __shared__ float ml_part[ATTN_WAVES][QPK][2];
+  __shared__ unsigned kvmax[2];  // KV8: row absmax bits (k, v)
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -1105,8 +1159,12 @@ __global__ void attn_decode_block_kernel(
   const size_t cache_base =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
       (size_t)max_seq * head_size;
+  const size_t scl_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq;
   const bf16* krow_cur = qkv + ((size_t)g * (QPK + 2) + QPK) * head_size;
   const bf16* vrow_cur = krow_cur + head_size;
+  if (KV8 && threadIdx.x < 2) kvmax[threadIdx.x] = 0;
 
   // ---- block-cooperative staging: q rows (roped), current k row ----------
   // the swizzle permutes 16-byte units WITHIN a row, so the pad rows
@@ -1126,9 +1184,32 @@ __global__ void attn_decode_block_kernel(
   __syncthreads();  // the only barrier before the combine
 
   // append the current token's k,v (no other block touches (slot, g))
-  for (int d = threadIdx.x; d < head_size; d += 256) {
-    kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
-    vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
+  if constexpr (KV8) {
+    float lk = 0.f, lv = 0.f;
+    for (int d = threadIdx.x; d < head_size; d += 256) {
+      lk = fmaxf(lk, fabsf(b2f(k_cur[d])));
+      lv = fmaxf(lv, fabsf(b2f(vrow_cur[d])));
+    }
+    lds_fmax_u(&kvmax[0], lk);
+    lds_fmax_u(&kvmax[1], lv);
+    __syncthreads();
+    const float ks = fmaxf(__uint_as_float(kvmax[0]), 1e-12f) / 448.f;
+    const float vs = fmaxf(__uint_as_float(kvmax[1]), 1e-12f) / 448.f;
+    for (int d = threadIdx.x; d < head_size; d += 256) {
+      kpool[cache_base + (size_t)pos * head_size + d] =
+          f32_to_fp8(b2f(k_cur[d]) / ks);
+      vpool[cache_base + (size_t)pos * head_size + d] =
+          f32_to_fp8(b2f(vrow_cur[d]) / vs);
+    }
+    if (threadIdx.x == 0) {
+      kscale[scl_base + pos] = ks;
+      vscale[scl_base + pos] = vs;
+    }
+  } else {
+    for (int d = threadIdx.x; d < head_size; d += 256) {
+      kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
+      vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
+    }
   }
 
   // ---- per-lane roles ----------------------------------------------------
@@ -1156,16 +1237,24 @@ __global__ void attn_decode_block_kernel(
     const int arow = lane & 15;
     const int koff = (lane >> 4) * 8;
     const int akey = key0 + arow;
-    const bf16* krow = kpool + cache_base + (size_t)akey * head_size;
+    const kvt* krow = kpool + cache_base + (size_t)akey * head_size;
     const bool row_valid = akey < S;
     const bool row_cur = akey == pos;
+    float kscl = 1.f;
+    if (KV8 && row_valid && !row_cur) kscl = kscale[scl_base + akey];
 #pragma unroll
     for (int c = 0; c < HS / 32; ++c) {
       bf16x8_t af = {};
       if (row_cur)
         af = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
-      else if (row_valid)
-        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      else if (row_valid) {
+        if constexpr (KV8)
+          af = fp8x8_to_bf16(
+              reinterpret_cast<const u8kv*>(krow) + c * 32 + koff, kscl);
+        else
+          af = *reinterpret_cast<const bf16x8_t*>(
+              reinterpret_cast<const bf16*>(krow) + c * 32 + koff);
+      }
       const bf16x8_t bq = *reinterpret_cast<const bf16x8_t*>(
           &q_lds[q_swz<HS>(arow, c * 32 + koff)]);
       acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bq, acc4, 0, 0, 0);
@@ -1207,29 +1296,40 @@ __global__ void attn_decode_block_kernel(
     for (int s = 0; s < 16; ++s) {
       const float p = __shfl(e[s & 3], ((s >> 2) << 4) | qb, 64);
       const int skey = min(key0 + s, pos);
-      const bf16* vrow =
-          (skey == pos)
-              ? vrow_cur + d0
-              : vpool + cache_base + (size_t)skey * head_size + d0;
+      const bool vcur = skey == pos;
+      const bf16* vrow_b = vrow_cur + d0;
+      const kvt* vrow_p = vpool + cache_base + (size_t)skey * head_size + d0;
+      float vscl = 1.f;
+      if (KV8 && !vcur) vscl = vscale[scl_base + skey];
       if constexpr (ODIM >= 8) {
 #pragma unroll
         for (int i = 0; i < ODIM; i += 8) {
-          bf16x8 vv = load8(vrow + i);
+          bf16x8_t vv;
+          if (vcur) {
+            vv = *reinterpret_cast<const bf16x8_t*>(vrow_b + i);
+          } else if constexpr (KV8) {
+            vv = fp8x8_to_bf16(
+                reinterpret_cast<const u8kv*>(vrow_p) + i, vscl);
+          } else {
+            vv = *reinterpret_cast<const bf16x8_t*>(
+                reinterpret_cast<const bf16*>(vrow_p) + i);
+          }
 #pragma unroll
-          for (int j = 0; j < 8; ++j) o_acc[i + j] += p * b2f(vv.v[j]);
+          for (int j = 0; j < 8; ++j) o_acc[i + j] += p * (float)vv[j];
         }
-      } else if constexpr (ODIM == 4) {
-        int2 raw = *reinterpret_cast<const int2*>(vrow);
-        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+      } else {
+        // narrow ODIM (small head geometries): per-element loads
 #pragma unroll
-        for (int j = 0; j < 4; ++j) o_acc[j] += p * b2f(vv[j]);
-      } else if constexpr (ODIM == 2) {
-        int raw = *reinterpret_cast<const int*>(vrow);
-        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
-        o_acc[0] += p * b2f(vv[0]);
-        o_acc[1] += p * b2f(vv[1]);
-      } else {  // ODIM == 1
-        o_acc[0] += p * b2f(vrow[0]);
+        for (int j = 0; j < ODIM; ++j) {
+          float v;
+          if (vcur)
+            v = b2f(vrow_b[j]);
+          else if constexpr (KV8)
+            v = fp8_to_f32(reinterpret_cast<const u8kv*>(vrow_p)[j]) * vscl;
+          else
+            v = b2f(reinterpret_cast<const bf16*>(vrow_p)[j]);
+          o_acc[j] += p * v;
+        }
       }
     }
   }
@@ -1592,16 +1692,20 @@ __global__ void attn_proj_kernel(
 // BATCH: 0 -> single-token mode (pos_p/slot_p are device scalars);
 //        1 -> batched mode: pos_p/slot_p are arrays [n_batch], and
 //             qkv/part_o/part_ml/out get a leading batch dimension.
-template <int QPK, int HS, int BATCH>
+template <int QPK, int HS, int BATCH, int KV8>
 __global__ void attn_decode_kernel(
     float* __restrict__ part_o,   // [B?, n_head, n_chunks, head_size]
     float* __restrict__ part_ml,  // [B?, n_head, n_chunks, 2]
     const bf16* __restrict__ qkv, // [B?, qkv_dim] interleaved, RAW
-    bf16* __restrict__ kpool, bf16* __restrict__ vpool,
+    void* __restrict__ kpool_v, void* __restrict__ vpool_v,
+    float* __restrict__ kscale, float* __restrict__ vscale,
     const float* __restrict__ cos_t, const float* __restrict__ sin_t,
     int rope_ne, const int* __restrict__ pos_p,
     const int* __restrict__ slot_p, int layer, int n_layers_pool,
     int n_kv_heads, int max_seq, int n_chunks, float scale, int n_batch) {
+  using kvt = std::conditional_t<KV8 != 0, u8kv, bf16>;
+  kvt* kpool = reinterpret_cast<kvt*>(kpool_v);
+  kvt* vpool = reinterpret_cast<kvt*>(vpool_v);
   constexpr int head_size = HS;
   // q tile + current-token k row are BLOCK-shared: n_chunks is a
   // multiple of ATTN_WAVES, so all 4 waves of a block carry the same
@@ -1647,6 +1751,9 @@ __global__ void attn_decode_kernel(
   const size_t cache_base =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
       (size_t)max_seq * head_size;
+  const size_t scl_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq;
   const bf16* krow_cur = qkv + ((size_t)g * (QPK + 2) + QPK) * head_size;
   const bf16* vrow_cur = krow_cur + head_size;
 
@@ -1696,16 +1803,24 @@ __global__ void attn_decode_kernel(
   // tiles per wave at long S no longer stack their load latencies.
   auto load_af = [&](int t, bf16x8_t (&af)[HS / 32]) {
     const int akey = k_begin + t * 16 + arow;
-    const bf16* krow = kpool + cache_base + (size_t)akey * head_size;
+    const kvt* krow = kpool + cache_base + (size_t)akey * head_size;
     const bool row_valid = akey < S;
     const bool row_cur = akey == pos;  // newest key: read from LDS k_cur
+    float kscl = 1.f;
+    if (KV8 && row_valid && !row_cur) kscl = kscale[scl_base + akey];
 #pragma unroll
     for (int c = 0; c < HS / 32; ++c) {
       af[c] = bf16x8_t{};
-      if (row_cur)
+      if (row_cur) {
         af[c] = *reinterpret_cast<const bf16x8_t*>(&k_cur[c * 32 + koff]);
-      else if (row_valid)
-        af[c] = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      } else if (row_valid) {
+        if constexpr (KV8)
+          af[c] = fp8x8_to_bf16(
+              reinterpret_cast<const u8kv*>(krow) + c * 32 + koff, kscl);
+        else
+          af[c] = *reinterpret_cast<const bf16x8_t*>(
+              reinterpret_cast<const bf16*>(krow) + c * 32 + koff);
+      }
     }
   };
 
@@ -1763,29 +1878,40 @@ __global__ void attn_decode_kernel(
     const int smax = min(16, S - key0);
     for (int s = 0; s < smax; ++s) {
       const float p = p_lds[wave][s][qb];
-      const bf16* vrow =
-          (key0 + s == pos)
-              ? vrow_cur + d0
-              : vpool + cache_base + (size_t)(key0 + s) * head_size + d0;
+      const int skey = key0 + s;
+      const bool vcur = skey == pos;
+      const bf16* vrow_b = vrow_cur + d0;
+      const kvt* vrow_p = vpool + cache_base + (size_t)skey * head_size + d0;
+      float vscl = 1.f;
+      if (KV8 && !vcur) vscl = vscale[scl_base + skey];
       if constexpr (ODIM >= 8) {
 #pragma unroll
         for (int i = 0; i < ODIM; i += 8) {
-          bf16x8 vv = load8(vrow + i);
+          bf16x8_t vv;
+          if (vcur) {
+            vv = *reinterpret_cast<const bf16x8_t*>(vrow_b + i);
+          } else if constexpr (KV8) {
+            vv = fp8x8_to_bf16(
+                reinterpret_cast<const u8kv*>(vrow_p) + i, vscl);
+          } else {
+            vv = *reinterpret_cast<const bf16x8_t*>(
+                reinterpret_cast<const bf16*>(vrow_p) + i);
+          }
 #pragma unroll
-          for (int j = 0; j < 8; ++j) o_acc[i + j] += p * b2f(vv.v[j]);
+          for (int j = 0; j < 8; ++j) o_acc[i + j] += p * (float)vv[j];
         }
-      } else if constexpr (ODIM == 4) {
-        int2 raw = *reinterpret_cast<const int2*>(vrow);
-        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
+      } else {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) o_acc[j] += p * b2f(vv[j]);
-      } else if constexpr (ODIM == 2) {
-        int raw = *reinterpret_cast<const int*>(vrow);
-        const bf16* vv = reinterpret_cast<const bf16*>(&raw);
-        o_acc[0] += p * b2f(vv[0]);
-        o_acc[1] += p * b2f(vv[1]);
-      } else {  // ODIM == 1
-        o_acc[0] += p * b2f(vrow[0]);
+        for (int j = 0; j < ODIM; ++j) {
+          float v;
+          if (vcur)
+            v = b2f(vrow_b[j]);
+          else if constexpr (KV8)
+            v = fp8_to_f32(reinterpret_cast<const u8kv*>(vrow_p)[j]) * vscl;
+          else
+            v = b2f(reinterpret_cast<const bf16*>(vrow_p)[j]);
+          o_acc[j] += p * v;
+        }
       }
     }
   };
@@ -1803,9 +1929,29 @@ __global__ void attn_decode_kernel(
   // exactly one chunk's range contains pos; its wave owns the append (no
   // other wave reads pool row pos this step — they use k_cur/vrow_cur)
   if (k_begin <= pos && pos < k_begin + keys_per_chunk) {
-    for (int d = lane; d < head_size; d += 64) {
-      kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
-      vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
+    if constexpr (KV8) {
+      float lk = 0.f, lv = 0.f;
+      for (int d = lane; d < head_size; d += 64) {
+        lk = fmaxf(lk, fabsf(b2f(k_cur[d])));
+        lv = fmaxf(lv, fabsf(b2f(vrow_cur[d])));
+      }
+      const float ks = fmaxf(wave_reduce_max(lk), 1e-12f) / 448.f;
+      const float vs = fmaxf(wave_reduce_max(lv), 1e-12f) / 448.f;
+      for (int d = lane; d < head_size; d += 64) {
+        kpool[cache_base + (size_t)pos * head_size + d] =
+            f32_to_fp8(b2f(k_cur[d]) / ks);
+        vpool[cache_base + (size_t)pos * head_size + d] =
+            f32_to_fp8(b2f(vrow_cur[d]) / vs);
+      }
+      if (lane == 0) {
+        kscale[scl_base + pos] = ks;
+        vscale[scl_base + pos] = vs;
+      }
+    } else {
+      for (int d = lane; d < head_size; d += 64) {
+        kpool[cache_base + (size_t)pos * head_size + d] = k_cur[d];
+        vpool[cache_base + (size_t)pos * head_size + d] = vrow_cur[d];
+      }
     }
   }
 
@@ -1903,12 +2049,18 @@ __global__ void attn_combine_kernel(bf16* __restrict__ out,
 // Prefill: rope q/k for ALL T prompt positions + append k/v to the pool.
 // qkv: [T, qkv_dim] (roped in place); grid (n_kv_heads, T).
 // ---------------------------------------------------------------------------
+template <int KV8>
 __global__ void rope_prefill_append_kernel(
-    bf16* __restrict__ qkv, bf16* __restrict__ kpool,
-    bf16* __restrict__ vpool, const float* __restrict__ cos_t,
+    bf16* __restrict__ qkv, void* __restrict__ kpool_v,
+    void* __restrict__ vpool_v, float* __restrict__ kscale,
+    float* __restrict__ vscale, const float* __restrict__ cos_t,
     const float* __restrict__ sin_t, int pos0, int slot, int layer,
     int n_layers_pool, int n_kv_heads, int max_seq, int head_size,
     int rope_ne, int qpk) {
+  using kvt = std::conditional_t<KV8 != 0, u8kv, bf16>;
+  kvt* kpool = reinterpret_cast<kvt*>(kpool_v);
+  kvt* vpool = reinterpret_cast<kvt*>(vpool_v);
+  __shared__ unsigned kvmax[2];
   const int g = blockIdx.x;
   const int t = blockIdx.y;
   const int pos = pos0 + t;
@@ -1931,6 +2083,7 @@ __global__ void rope_prefill_append_kernel(
     row[d] = f2b(x1 * c1 - x2 * s1);
     row[d + half] = f2b(x2 * c2 + x1 * s2);
   }
+  if (KV8 && tid < 2) kvmax[tid] = 0;
   __syncthreads();
   const size_t cache_off =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
@@ -1938,9 +2091,34 @@ __global__ void rope_prefill_append_kernel(
       (size_t)pos * head_size;
   const bf16* krow = base + (size_t)qpk * head_size;
   const bf16* vrow = base + (size_t)(qpk + 1) * head_size;
-  for (int d = tid; d < head_size; d += blockDim.x) {
-    kpool[cache_off + d] = krow[d];
-    vpool[cache_off + d] = vrow[d];
+  if constexpr (KV8) {
+    float lk = 0.f, lv = 0.f;
+    for (int d = tid; d < head_size; d += blockDim.x) {
+      lk = fmaxf(lk, fabsf(b2f(krow[d])));
+      lv = fmaxf(lv, fabsf(b2f(vrow[d])));
+    }
+    lds_fmax_u(&kvmax[0], lk);
+    lds_fmax_u(&kvmax[1], lv);
+    __syncthreads();
+    const float ks = fmaxf(__uint_as_float(kvmax[0]), 1e-12f) / 448.f;
+    const float vs = fmaxf(__uint_as_float(kvmax[1]), 1e-12f) / 448.f;
+    for (int d = tid; d < head_size; d += blockDim.x) {
+      kpool[cache_off + d] = f32_to_fp8(b2f(krow[d]) / ks);
+      vpool[cache_off + d] = f32_to_fp8(b2f(vrow[d]) / vs);
+    }
+    if (tid == 0) {
+      const size_t sb =
+          (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+              (size_t)max_seq +
+          (size_t)pos;
+      kscale[sb] = ks;
+      vscale[sb] = vs;
+    }
+  } else {
+    for (int d = tid; d < head_size; d += blockDim.x) {
+      kpool[cache_off + d] = krow[d];
+      vpool[cache_off + d] = vrow[d];
+    }
   }
 }
 
@@ -1953,13 +2131,17 @@ __global__ void rope_prefill_append_kernel(
 // directly — no S^2 score materialization, no split-S partials.
 // grid: ceil(n_head * n_qtiles / 4) blocks of 256 threads.
 // ---------------------------------------------------------------------------
-template <int HS>
+template <int HS, int KV8>
 __global__ void prefill_attn_kernel(
     bf16* __restrict__ out,        // [T, n_head*HS]
     const bf16* __restrict__ qkv,  // [T, qkv_dim], q already roped
-    const bf16* __restrict__ kpool, const bf16* __restrict__ vpool,
+    const void* __restrict__ kpool_v, const void* __restrict__ vpool_v,
+    const float* __restrict__ kscale, const float* __restrict__ vscale,
     int pos0, int slot, int layer, int n_layers_pool, int n_kv_heads,
     int max_seq, int qpk, int T, float scale) {
+  using kvt = std::conditional_t<KV8 != 0, u8kv, bf16>;
+  const kvt* kpool = reinterpret_cast<const kvt*>(kpool_v);
+  const kvt* vpool = reinterpret_cast<const kvt*>(vpool_v);
   __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
   __shared__ float p_lds[ATTN_WAVES][16][16];
   __shared__ float m_lds[ATTN_WAVES][16];
@@ -1982,6 +2164,9 @@ __global__ void prefill_attn_kernel(
   const size_t cache_base =
       (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
       (size_t)max_seq * HS;
+  const size_t scl_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq;
 
   // stage 16 q rows (t = q_base..q_base+15) of head h into LDS
   for (int i = lane; i < 16 * (HS / 8); i += 64) {
@@ -2022,13 +2207,21 @@ __global__ void prefill_attn_kernel(
     f32x4 acc4 = {0.f, 0.f, 0.f, 0.f};
     const int arow = lane & 15;
     const int koff = (lane >> 4) * 8;
-    const bf16* krow = kpool + cache_base + (size_t)(key0 + arow) * HS;
+    const kvt* krow = kpool + cache_base + (size_t)(key0 + arow) * HS;
     const bool row_valid = (key0 + arow) < k_last;
+    float kscl = 1.f;
+    if (KV8 && row_valid) kscl = kscale[scl_base + key0 + arow];
 #pragma unroll
     for (int c = 0; c < HS / 32; ++c) {
       bf16x8_t af = {};
-      if (row_valid)
-        af = *reinterpret_cast<const bf16x8_t*>(krow + c * 32 + koff);
+      if (row_valid) {
+        if constexpr (KV8)
+          af = fp8x8_to_bf16(
+              reinterpret_cast<const u8kv*>(krow) + c * 32 + koff, kscl);
+        else
+          af = *reinterpret_cast<const bf16x8_t*>(
+              reinterpret_cast<const bf16*>(krow) + c * 32 + koff);
+      }
       const bf16x8_t bfr = *reinterpret_cast<const bf16x8_t*>(
           &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
       acc4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, acc4, 0, 0, 0);
@@ -2074,13 +2267,21 @@ __global__ void prefill_attn_kernel(
     const int smax = min(16, k_last - key0);
     for (int s = 0; s < smax; ++s) {
       const float p = p_lds[wave][s][qb];
-      const bf16* vrow = vpool + cache_base + (size_t)(key0 + s) * HS + d0;
+      const kvt* vrow = vpool + cache_base + (size_t)(key0 + s) * HS + d0;
+      float vscl = 1.f;
+      if (KV8) vscl = vscale[scl_base + key0 + s];
 #pragma unroll
       for (int i = 0; i < ODIM; i += 8) {
-        bf16x8 vv = load8(vrow + i);
+        bf16x8_t vv;
+        if constexpr (KV8)
+          vv = fp8x8_to_bf16(reinterpret_cast<const u8kv*>(vrow) + i,
+                             vscl);
+        else
+          vv = *reinterpret_cast<const bf16x8_t*>(
+              reinterpret_cast<const bf16*>(vrow) + i);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          if (i + j < ODIM) o_acc[i + j] += p * b2f(vv.v[j]);
+          if (i + j < ODIM) o_acc[i + j] += p * (float)vv[j];
       }
     }
   }
@@ -2871,11 +3072,13 @@ void launch_rope_kv_append(void* qkv, void* kpool, void* vpool,
 template <int QPK, int HS>
 static void attn_dispatch2(void* out, float* part_o, float* part_ml,
                            const void* qkv, void* kpool, void* vpool,
+                           float* kscale, float* vscale,
                            const float* cos_t, const float* sin_t,
                            int rope_ne, const int* pos, const int* slot,
                            int layer, int n_layers_pool, int n_kv_heads,
                            int max_seq, int n_chunks, float scale,
                            int n_batch, int blocks, hipStream_t stream) {
+  const int kv8 = kscale != nullptr;
   // short/medium contexts: one block per (batch, kv_head), no global
   // partials, no combine kernel — the whole attention step is ONE launch.
   // MDI_ATTN_FORCE_SPLITS=1 forces the split-S + combine path instead
@@ -2888,40 +3091,43 @@ static void attn_dispatch2(void* out, float* part_o, float* part_ml,
   }
   if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits) {
     const int nb = (n_batch > 0 ? n_batch : 1) * n_kv_heads;
+#define ATTN_BLK(BATCHV, KV8V, NBATCH)                                      \
+    hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, BATCHV, KV8V>),   \
+                       dim3(nb), dim3(256), 0, stream, (bf16*)out,          \
+                       (const bf16*)qkv, kpool, vpool, kscale, vscale,      \
+                       cos_t, sin_t, rope_ne, pos, slot, layer,             \
+                       n_layers_pool, n_kv_heads, max_seq, scale, NBATCH)
     if (n_batch > 0) {
-      hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, 1>), dim3(nb),
-                         dim3(256), 0, stream, (bf16*)out, (const bf16*)qkv,
-                         (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne,
-                         pos, slot, layer, n_layers_pool, n_kv_heads,
-                         max_seq, scale, n_batch);
+      if (kv8) ATTN_BLK(1, 1, n_batch);
+      else ATTN_BLK(1, 0, n_batch);
     } else {
-      hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, 0>), dim3(nb),
-                         dim3(256), 0, stream, (bf16*)out, (const bf16*)qkv,
-                         (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne,
-                         pos, slot, layer, n_layers_pool, n_kv_heads,
-                         max_seq, scale, 1);
+      if (kv8) ATTN_BLK(0, 1, 1);
+      else ATTN_BLK(0, 0, 1);
     }
+#undef ATTN_BLK
     return;
   }
+#define ATTN_SPL(BATCHV, KV8V, NBATCH)                                      \
+  hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, BATCHV, KV8V>),           \
+                     dim3(blocks), dim3(256), 0, stream, part_o, part_ml,   \
+                     (const bf16*)qkv, kpool, vpool, kscale, vscale, cos_t, \
+                     sin_t, rope_ne, pos, slot, layer, n_layers_pool,       \
+                     n_kv_heads, max_seq, n_chunks, scale, NBATCH)
   if (n_batch > 0) {
-    hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 1>), dim3(blocks),
-                       dim3(256), 0, stream, part_o, part_ml,
-                       (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
-                       sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-                       n_kv_heads, max_seq, n_chunks, scale, n_batch);
+    if (kv8) ATTN_SPL(1, 1, n_batch);
+    else ATTN_SPL(1, 0, n_batch);
   } else {
-    hipLaunchKernelGGL((attn_decode_kernel<QPK, HS, 0>), dim3(blocks),
-                       dim3(256), 0, stream, part_o, part_ml,
-                       (const bf16*)qkv, (bf16*)kpool, (bf16*)vpool, cos_t,
-                       sin_t, rope_ne, pos, slot, layer, n_layers_pool,
-                       n_kv_heads, max_seq, n_chunks, scale, 1);
+    if (kv8) ATTN_SPL(0, 1, 1);
+    else ATTN_SPL(0, 0, 1);
   }
+#undef ATTN_SPL
 }
 
 template <int QPK>
 static int attn_dispatch1(int head_size, void* out, float* part_o,
                           float* part_ml, const void* qkv, void* kpool,
-                          void* vpool, const float* cos_t,
+                          void* vpool, float* kscale, float* vscale,
+                          const float* cos_t,
                           const float* sin_t, int rope_ne, const int* pos,
                           const int* slot, int layer, int n_layers_pool,
                           int n_kv_heads, int max_seq, int n_chunks,
@@ -2931,9 +3137,10 @@ static int attn_dispatch1(int head_size, void* out, float* part_o,
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
       attn_dispatch2<QPK, H>(out, part_o, part_ml, qkv, kpool, vpool,       \
-                             cos_t, sin_t, rope_ne, pos, slot, layer,       \
-                             n_layers_pool, n_kv_heads, max_seq, n_chunks,  \
-                             scale, n_batch, blocks, stream);               \
+                             kscale, vscale, cos_t, sin_t, rope_ne, pos,    \
+                             slot, layer, n_layers_pool, n_kv_heads,        \
+                             max_seq, n_chunks, scale, n_batch, blocks,     \
+                             stream);                                       \
       return 0;                                                             \
     }                                                                       \
   }
@@ -2948,6 +3155,7 @@ static int attn_dispatch1(int head_size, void* out, float* part_o,
 // n_batch == 0 -> single-token mode; > 0 -> batched (pos/slot arrays)
 int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        const void* qkv, void* kpool, void* vpool,
+                       float* kscale, float* vscale,
                        const float* cos_t, const float* sin_t, int rope_ne,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
@@ -2959,31 +3167,36 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
   switch (qpk) {
     case 1:
       rc = attn_dispatch1<1>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
+          vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 2:
       rc = attn_dispatch1<2>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
+          vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 4:
       rc = attn_dispatch1<4>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
+          vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 8:
       rc = attn_dispatch1<8>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
+          vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
     case 16:
       rc = attn_dispatch1<16>(
-          head_size, out, part_o, part_ml, qkv, kpool, vpool, cos_t, sin_t,
+          head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
+          vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
           n_chunks, scale, n_batch, blocks, stream);
       break;
@@ -3010,33 +3223,50 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
 }
 
 void launch_rope_prefill_append(void* qkv, void* kpool, void* vpool,
+                                float* kscale, float* vscale,
                                 const float* cos_t, const float* sin_t,
                                 int pos0, int slot, int layer,
                                 int n_layers_pool, int n_kv_heads,
                                 int max_seq, int head_size, int rope_ne,
                                 int qpk, int T, hipStream_t stream) {
-  hipLaunchKernelGGL(rope_prefill_append_kernel, dim3(n_kv_heads, T),
-                     dim3(256), 0, stream, (bf16*)qkv, (bf16*)kpool,
-                     (bf16*)vpool, cos_t, sin_t, pos0, slot, layer,
-                     n_layers_pool, n_kv_heads, max_seq, head_size, rope_ne,
-                     qpk);
+  if (kscale != nullptr)
+    hipLaunchKernelGGL((rope_prefill_append_kernel<1>),
+                       dim3(n_kv_heads, T), dim3(256), 0, stream,
+                       (bf16*)qkv, kpool, vpool, kscale, vscale, cos_t,
+                       sin_t, pos0, slot, layer, n_layers_pool, n_kv_heads,
+                       max_seq, head_size, rope_ne, qpk);
+  else
+    hipLaunchKernelGGL((rope_prefill_append_kernel<0>),
+                       dim3(n_kv_heads, T), dim3(256), 0, stream,
+                       (bf16*)qkv, kpool, vpool, kscale, vscale, cos_t,
+                       sin_t, pos0, slot, layer, n_layers_pool, n_kv_heads,
+                       max_seq, head_size, rope_ne, qpk);
 }
 
 int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
-                        const void* vpool, int pos0, int slot, int layer,
+                        const void* vpool, const float* kscale,
+                        const float* vscale, int pos0, int slot, int layer,
                         int n_layers_pool, int n_kv_heads, int max_seq,
                         int head_size, int qpk, int T, float scale,
                         hipStream_t stream) {
   const int n_head = n_kv_heads * qpk;
   const int n_qtiles = (T + 15) / 16;
   const int blocks = (n_head * n_qtiles + ATTN_WAVES - 1) / ATTN_WAVES;
+  const int kv8 = kscale != nullptr;
 #define PF_CASE(H)                                                          \
   if (head_size == H) {                                                     \
-    hipLaunchKernelGGL((prefill_attn_kernel<H>), dim3(blocks), dim3(256),   \
-                       0, stream, (bf16*)out, (const bf16*)qkv,             \
-                       (const bf16*)kpool, (const bf16*)vpool, pos0, slot,  \
-                       layer, n_layers_pool, n_kv_heads, max_seq, qpk, T,   \
-                       scale);                                              \
+    if (kv8)                                                                \
+      hipLaunchKernelGGL((prefill_attn_kernel<H, 1>), dim3(blocks),         \
+                         dim3(256), 0, stream, (bf16*)out,                  \
+                         (const bf16*)qkv, kpool, vpool, kscale, vscale,    \
+                         pos0, slot, layer, n_layers_pool, n_kv_heads,      \
+                         max_seq, qpk, T, scale);                           \
+    else                                                                    \
+      hipLaunchKernelGGL((prefill_attn_kernel<H, 0>), dim3(blocks),         \
+                         dim3(256), 0, stream, (bf16*)out,                  \
+                         (const bf16*)qkv, kpool, vpool, kscale, vscale,    \
+                         pos0, slot, layer, n_layers_pool, n_kv_heads,      \
+                         max_seq, qpk, T, scale);                           \
     return 0;                                                               \
   }
   PF_CASE(64)

@@ -467,3 +467,106 @@ def test_engine_long_context_split_s():
             ref_logits[0, -1].float().argmax()), i
         tok = ref_logits[0, -1].float().argmax()
         pos += 1
+
+
+@torch.inference_mode()
+def test_engine_fp8_kv_matches_torch(monkeypatch):
+    """fp8 (e4m3, per-row-scaled) KV cache: engine vs the bf16-KV torch
+    model — logits within fp8 tolerance, greedy decode stays on track."""
+    monkeypatch.setenv("MDI_KV_DTYPE", "fp8")
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine
+
+    cfg, m = _build(seed=31)
+    stage = StarterStage(cfg, cfg.n_layer).to(device=DEV,
+                                              dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.eval()
+    stage.set_kv_cache(2)
+    m.set_kv_cache(2)
+
+    torch.manual_seed(32)
+    prompt = torch.randint(0, 511, (12,), device=DEV)
+    ref_logits = m(prompt.view(1, -1), input_pos=0, slot=0)
+
+    eng = DecodeEngine(stage, stage.kv_pool, use_graphs=True)
+    assert eng.kv8 and stage.kv_pool.fp8
+    assert stage.kv_pool.k.dtype == torch.uint8
+    eng.capture_graphs()
+    eng.prefill_prompt(prompt, 0, 0)
+    eng.set_slot_pos(0, 12)
+    x = eng.prefill_prompt(prompt, 0, 0)  # idempotent re-prefill ok
+    logits0 = eng.tail(x[-1])
+    diff0 = (logits0.float() - ref_logits[0, -1].float()).abs().max()
+    assert diff0 < 2.0, float(diff0)
+
+    tok = ref_logits[0, -1].float().argmax()
+    pos = 12
+    agree = 0
+    for i in range(8):
+        ref = m(tok.view(1, 1), input_pos=pos, slot=0)[0, -1].float()
+        x = eng.decode_step_head(tok.to(torch.int32), slot=0)
+        got = eng.tail(x).float()
+        assert (got - ref).abs().max() < 2.0, i
+        if int(got.argmax()) == int(ref.argmax()):
+            agree += 1
+        tok = ref.argmax()
+        pos += 1
+    assert agree >= 6, agree
+    # torch path must refuse to write the fp8 pool
+    import pytest as _pytest
+    with _pytest.raises(RuntimeError):
+        stage.kv_pool.append(0, 0, torch.zeros(2, 1, 64, device=DEV,
+                                               dtype=torch.bfloat16),
+                             torch.zeros(2, 1, 64, device=DEV,
+                                         dtype=torch.bfloat16), pos)
+
+
+@torch.inference_mode()
+def test_engine_fp8_kv_long_context(monkeypatch):
+    """fp8 KV through the split-S + combine path (max_seq 8192)."""
+    import dataclasses
+
+    monkeypatch.setenv("MDI_KV_DTYPE", "fp8")
+    from mdi_llm_amd import GPT, ModelConfig
+    from mdi_llm_amd.models.stages import StarterStage
+    from mdi_llm_amd.ops.engine import DecodeEngine
+
+    cfg = dataclasses.replace(ModelConfig.from_name("nano-gpu"),
+                              block_size=8192)
+    torch.manual_seed(33)
+    m = GPT(cfg)
+    m.apply_init()
+    m = m.to(device=DEV, dtype=torch.bfloat16)
+    m.eval()
+    stage = StarterStage(cfg, cfg.n_layer).to(DEV, dtype=torch.bfloat16)
+    stage.load_state_dict(m.state_dict())
+    stage.max_seq_length = 8192
+    m.max_seq_length = 8192
+    stage.eval()
+    stage.set_kv_cache(1)
+    m.set_kv_cache(1)
+
+    torch.manual_seed(34)
+    S0 = 4100
+    prompt = torch.randint(0, 511, (S0,), device=DEV)
+    ref_logits = m(prompt.view(1, -1), input_pos=0, slot=0)
+
+    eng = DecodeEngine(stage, stage.kv_pool, use_graphs=False)
+    assert eng.kv8 and eng.n_chunks >= 256
+    eng.prefill_prompt(prompt, 0, 0)
+    eng.set_slot_pos(0, S0)
+
+    tok = ref_logits[0, -1].float().argmax()
+    pos = S0
+    agree = 0
+    for i in range(4):
+        ref = m(tok.view(1, 1), input_pos=pos, slot=0)[0, -1].float()
+        x = eng.decode_step_head(tok.to(torch.int32), slot=0)
+        got = eng.tail(x).float()
+        assert (got - ref).abs().max() < 2.0, (i, float((got - ref).abs().max()))
+        if int(got.argmax()) == int(ref.argmax()):
+            agree += 1
+        tok = ref.argmax()
+        pos += 1
+    assert agree >= 3, agree
