@@ -17,6 +17,12 @@ sys.path.insert(0, str(SCRIPT_DIR))
 
 
 def main(args):
+    import os as _os
+
+    if getattr(args, "weights", "bf16") == "fp8":
+        _os.environ["MDI_WEIGHT_DTYPE"] = "fp8"
+    if getattr(args, "kv", "bf16") == "fp8":
+        _os.environ["MDI_KV_DTYPE"] = "fp8"
     import torch
 
     from mdi_llm_amd.config import ModelConfig
@@ -94,6 +100,12 @@ def build_parser():
                    default=SCRIPT_DIR / "checkpoints" / "custom" / "NanoLlama")
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--dtype", type=str, default=None)
+    p.add_argument("--weights", choices=["bf16", "fp8"], default="bf16",
+                   help="decode weight dtype on the HIP engine (fp8 = "
+                        "e4m3 per-row-scaled)")
+    p.add_argument("--kv", choices=["bf16", "fp8"], default="bf16",
+                   help="KV-cache dtype on the HIP engine (fp8 halves "
+                        "cache memory per sample)")
     p.add_argument("--prompt", type=str, default="Who are you?")
     p.add_argument("--n-samples", type=int, default=1)
     p.add_argument("--n-tokens", type=int, default=300)

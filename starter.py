@@ -19,6 +19,12 @@ sys.path.insert(0, str(SCRIPT_DIR))
 
 
 def main(args):
+    import os as _os
+
+    if getattr(args, "weights", "bf16") == "fp8":
+        _os.environ["MDI_WEIGHT_DTYPE"] = "fp8"
+    if getattr(args, "kv", "bf16") == "fp8":
+        _os.environ["MDI_KV_DTYPE"] = "fp8"
     import torch
 
     from mdi_llm_amd.parallel.orchestrator import MDIRuntime
@@ -105,6 +111,12 @@ def build_parser():
     p.add_argument("--sequence-length", "--context-length", "--block-size",
                    type=int, default=None, dest="sequence_length")
     p.add_argument("--dtype", type=str, default=None)
+    p.add_argument("--weights", choices=["bf16", "fp8"], default="bf16",
+                   help="decode weight dtype on the HIP engine (fp8 = "
+                        "e4m3 per-row-scaled)")
+    p.add_argument("--kv", choices=["bf16", "fp8"], default="bf16",
+                   help="KV-cache dtype on the HIP engine (fp8 halves "
+                        "cache memory per sample)")
     p.add_argument("--time-run", type=Path, default=None)
     p.add_argument("--seed", type=int, default=10137)
     return p
