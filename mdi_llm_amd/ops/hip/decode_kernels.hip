@@ -102,6 +102,26 @@ DEVINL bf16x8_t fp8x8_to_bf16(const u8kv* p, float scl) {
   return r;
 }
 
+// PV accumulate straight from the fp8 pairs (skips the bf16 round trip
+// the MFMA-operand path needs)
+DEVINL void fp8x8_fma(float* o, float pw, const u8kv* p, float scl) {
+  const int lo = *reinterpret_cast<const int*>(p);
+  const int hi = *reinterpret_cast<const int*>(p + 4);
+  const v2fkv a = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
+  const v2fkv b = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
+  const v2fkv c = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
+  const v2fkv d = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
+  const float ps = pw * scl;
+  o[0] += ps * a.x;
+  o[1] += ps * a.y;
+  o[2] += ps * b.x;
+  o[3] += ps * b.y;
+  o[4] += ps * c.x;
+  o[5] += ps * c.y;
+  o[6] += ps * d.x;
+  o[7] += ps * d.y;
+}
+
 // atomicMax for a non-negative float held in LDS as its uint bits
 DEVINL void lds_fmax_u(unsigned* addr, float v) {
   atomicMax(addr, __float_as_uint(v));
@@ -1304,12 +1324,16 @@ __global__ void attn_decode_block_kernel(
       if constexpr (ODIM >= 8) {
 #pragma unroll
         for (int i = 0; i < ODIM; i += 8) {
+          if (!vcur) {
+            if constexpr (KV8) {
+              fp8x8_fma(&o_acc[i], p,
+                        reinterpret_cast<const u8kv*>(vrow_p) + i, vscl);
+              continue;
+            }
+          }
           bf16x8_t vv;
           if (vcur) {
             vv = *reinterpret_cast<const bf16x8_t*>(vrow_b + i);
-          } else if constexpr (KV8) {
-            vv = fp8x8_to_bf16(
-                reinterpret_cast<const u8kv*>(vrow_p) + i, vscl);
           } else {
             vv = *reinterpret_cast<const bf16x8_t*>(
                 reinterpret_cast<const bf16*>(vrow_p) + i);
@@ -1887,12 +1911,16 @@ __global__ void attn_decode_kernel(
       if constexpr (ODIM >= 8) {
 #pragma unroll
         for (int i = 0; i < ODIM; i += 8) {
+          if (!vcur) {
+            if constexpr (KV8) {
+              fp8x8_fma(&o_acc[i], p,
+                        reinterpret_cast<const u8kv*>(vrow_p) + i, vscl);
+              continue;
+            }
+          }
           bf16x8_t vv;
           if (vcur) {
             vv = *reinterpret_cast<const bf16x8_t*>(vrow_b + i);
-          } else if constexpr (KV8) {
-            vv = fp8x8_to_bf16(
-                reinterpret_cast<const u8kv*>(vrow_p) + i, vscl);
           } else {
             vv = *reinterpret_cast<const bf16x8_t*>(
                 reinterpret_cast<const bf16*>(vrow_p) + i);
