@@ -218,3 +218,51 @@ print("RSS_DELTA_KB", peak - base)
                     mmap=True)
     assert sd["transformer.wte.weight"].shape == (8192, 1024)
     assert "transformer.h.5.attn.attn.weight" in sd
+
+
+def test_moe_lit_hf_roundtrip(tmp_path):
+    """Mixtral-layout (block_sparse_moe experts) HF -> lit -> HF."""
+    cfg = ModelConfig.from_name("nano-test-moe")
+    torch.manual_seed(3)
+    E, I = cfg.n_embd, cfg.intermediate_size
+    hs = cfg.head_size
+    sd = {"model.embed_tokens.weight": torch.randn(cfg.vocab_size, E),
+          "model.norm.weight": torch.randn(E),
+          "lm_head.weight": torch.randn(cfg.vocab_size, E)}
+    for i in range(cfg.n_layer):
+        p = f"model.layers.{i}"
+        sd.update({
+            f"{p}.input_layernorm.weight": torch.randn(E),
+            f"{p}.self_attn.q_proj.weight":
+                torch.randn(cfg.n_head * hs, E),
+            f"{p}.self_attn.k_proj.weight":
+                torch.randn(cfg.n_query_groups * hs, E),
+            f"{p}.self_attn.v_proj.weight":
+                torch.randn(cfg.n_query_groups * hs, E),
+            f"{p}.self_attn.o_proj.weight": torch.randn(E, cfg.n_head * hs),
+            f"{p}.post_attention_layernorm.weight": torch.randn(E),
+            f"{p}.block_sparse_moe.gate.weight":
+                torch.randn(cfg.n_expert, E),
+        })
+        for e in range(cfg.n_expert):
+            q = f"{p}.block_sparse_moe.experts.{e}"
+            sd.update({
+                f"{q}.w1.weight": torch.randn(I, E),
+                f"{q}.w3.weight": torch.randn(I, E),
+                f"{q}.w2.weight": torch.randn(E, I),
+            })
+    hf_dir = tmp_path / "hf"
+    hf_dir.mkdir()
+    torch.save(sd, hf_dir / "pytorch_model.bin")
+    out = tmp_path / "lit"
+    convert_hf_checkpoint(hf_dir, out, model_name="nano-test-moe")
+    lit_sd = torch.load(out / "lit_model.pth", weights_only=True)
+    assert "transformer.h.0.mlp.experts.0.fc_1.weight" in lit_sd
+
+    back_path = tmp_path / "back.bin"
+    convert_lit_checkpoint(out, back_path, model_name="nano-test-moe")
+    back = torch.load(back_path, weights_only=True)
+    for k, v in sd.items():
+        assert k in back, k
+        assert torch.equal(back[k], v), k
+    assert not (set(back) - set(sd))
