@@ -2751,7 +2751,18 @@ static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
   // rows per block group: fit every block resident at once (2 blocks/CU
   // of 4 waves = the 8-wave/CU limit) AND the W tile in <=79 KB of LDS;
   // grid-stride covers any remainder
-  const int cap_blocks = 2 * 256 - n_kv_heads;
+  // MDI_ATTN_PROJ_PB caps the number of proj (staging) blocks: fewer
+  // co-resident stagers leave the attention blocks' CUs unloaded (the
+  // "thin the loader while latency-sensitive work runs" principle,
+  // block-granular) at the cost of a longer-but-overlapped stage
+  static int pb_cap = -1;
+  if (pb_cap < 0) {
+    const char* e = getenv("MDI_ATTN_PROJ_PB");
+    pb_cap = e ? atoi(e) : 0;
+    if (pb_cap <= 0) pb_cap = 2 * 256;
+  }
+  const int cap_blocks =
+      (pb_cap < 2 * 256 ? pb_cap : 2 * 256) - n_kv_heads;
   int BR = (M + cap_blocks - 1) / cap_blocks;
   const int br_lds = (79 * 1024 - 16) / (K * 2);
   if (br_lds < 1) return;  // K too large for the LDS tile (no instantiation)
