@@ -1419,7 +1419,7 @@ __global__ void attn_decode_block_kernel(
 // ---------------------------------------------------------------------------
 typedef __attribute__((address_space(1))) unsigned int gu32_t;
 
-template <int QPK, int HS>
+template <int QPK, int HS, int THROTTLE>
 __global__ void attn_proj_kernel(
     bf16* __restrict__ out,       // [M] = res + bias + W @ y
     const bf16* __restrict__ qkv, // [qkv_dim] interleaved, RAW
@@ -1459,9 +1459,16 @@ __global__ void attn_proj_kernel(
     const int PBe = (int)gridDim.x - n_kv_heads;
     bf16* wlds = reinterpret_cast<bf16*>(smem);            // [BR, K]
     int* vflag = reinterpret_cast<int*>(smem + (size_t)BR * K * 2);
+    volatile int* yready = vflag + 1;
     const bf16* ygl = reinterpret_cast<const bf16*>(gran); // y after fence
     bool polled = false;
     bool y_ok = true;
+    // temporal throttle (MDI_ATTN_PROJ_SLEEP s_sleep units between
+    // 64-B/lane staging batches while attention is still running): the
+    // PB sweep showed the attention stretch is GLOBAL memory pressure,
+    // so the staging rate — not placement — is the lever
+    if (tid == 0) *yready = 0;
+    __syncthreads();
     for (int row0 = pb * BR; row0 < M; row0 += PBe * BR) {
       if (row0 != pb * BR) __syncthreads();  // prev dot read wlds
       const int nr = min(BR, M - row0);
@@ -1473,6 +1480,15 @@ __global__ void attn_proj_kernel(
         const int chunks = nr * (K / 8);
         int i = tid;
         for (; i + 3 * 256 < chunks; i += 4 * 256) {
+          if (THROTTLE > 0 && !polled && !*yready) {
+            if (lane == 0) {
+              const unsigned f = __hip_atomic_load(
+                  &flags[wave & (n_kv_heads - 1)], __ATOMIC_RELAXED,
+                  __HIP_MEMORY_SCOPE_AGENT);
+              if (f == tag) *yready = 1;
+            }
+            __builtin_amdgcn_s_sleep(THROTTLE);
+          }
           bf16x8 w0, w1, w2, w3;
 #pragma unroll
           for (int u = 0; u < 4; ++u) {
@@ -2775,19 +2791,41 @@ static void attn_proj_dispatch2(void* out, const void* qkv, void* kpool,
                      (size_t)ATTN_WAVES * QPK * HS * 4 +
                      (size_t)ATTN_WAVES * QPK * 2 * 4;
   size_t smem = smem_proj > smem_attn ? smem_proj : smem_attn;
+  static int thr = -1;
+  if (thr < 0) {
+    const char* e = getenv("MDI_ATTN_PROJ_SLEEP");
+    thr = e ? atoi(e) : 0;
+    if (thr != 0 && thr != 4 && thr != 8 && thr != 16) thr = 8;
+  }
   static bool attr_set = false;
   if (!attr_set) {
-    hipFuncSetAttribute(
-        reinterpret_cast<const void*>(&attn_proj_kernel<QPK, HS>),
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&attn_proj_kernel<QPK, HS, 0>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&attn_proj_kernel<QPK, HS, 4>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&attn_proj_kernel<QPK, HS, 8>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    (void)hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&attn_proj_kernel<QPK, HS, 16>),
         hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
     attr_set = true;
   }
-  hipLaunchKernelGGL((attn_proj_kernel<QPK, HS>), dim3(n_kv_heads + PB),
-                     dim3(256), smem, stream, (bf16*)out, (const bf16*)qkv,
-                     (bf16*)kpool, (bf16*)vpool, cos_t, sin_t, rope_ne, pos,
-                     slot, layer, n_layers_pool, n_kv_heads, max_seq, scale,
-                     (const bf16*)W, (const bf16*)bias, (const bf16*)res,
-                     (unsigned int*)gran, M, BR);
+#define AP_LAUNCH(T)                                                        \
+  hipLaunchKernelGGL((attn_proj_kernel<QPK, HS, T>),                        \
+                     dim3(n_kv_heads + PB), dim3(256), smem, stream,        \
+                     (bf16*)out, (const bf16*)qkv, (bf16*)kpool,            \
+                     (bf16*)vpool, cos_t, sin_t, rope_ne, pos, slot,        \
+                     layer, n_layers_pool, n_kv_heads, max_seq, scale,      \
+                     (const bf16*)W, (const bf16*)bias, (const bf16*)res,   \
+                     (unsigned int*)gran, M, BR)
+  if (thr == 0) AP_LAUNCH(0);
+  else if (thr == 4) AP_LAUNCH(4);
+  else if (thr == 8) AP_LAUNCH(8);
+  else AP_LAUNCH(16);
+#undef AP_LAUNCH
 }
 
 template <int QPK>
