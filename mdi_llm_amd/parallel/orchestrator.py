@@ -15,7 +15,7 @@ import sys
 import time
 from datetime import timedelta
 from pathlib import Path
-from typing import List, Optional
+from typing import Optional
 
 import torch
 import torch.distributed as dist

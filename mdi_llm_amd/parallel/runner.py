@@ -9,12 +9,10 @@ hand-written kernels when available.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from ..config import ModelConfig
-from ..models.stages import SecondaryStage, StarterStage
+from ..models.stages import StarterStage
 
 __all__ = ["TorchRunner", "HipRunner", "make_runner"]
 

@@ -19,7 +19,6 @@ from typing import Dict, Iterable, Union
 import torch
 
 from ..config import ModelConfig
-from .checkpoint import save_checkpoint
 from .incremental import IncrementalSaver
 
 __all__ = ["convert_hf_checkpoint", "convert_lit_checkpoint"]

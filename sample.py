@@ -25,7 +25,6 @@ def main(args):
         _os.environ["MDI_KV_DTYPE"] = "fp8"
     import torch
 
-    from mdi_llm_amd.config import ModelConfig
     from mdi_llm_amd.models.stages import StarterStage
     from mdi_llm_amd.parallel.orchestrator import default_dtype
     from mdi_llm_amd.parallel.runner import make_runner
