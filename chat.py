@@ -69,6 +69,13 @@ def main(args):
         # prefill this turn at the current position
         if pos == 0:
             x = runner.prefill_head(toks, 0)
+        elif (hasattr(runner, "engine")
+              and runner.engine.supports_hip_prefill):
+            # continue the cache through the HIP prefill at the current
+            # position (also the only writer the fp8 KV cache allows)
+            x = runner.engine.prefill_prompt(toks, 0, pos)
+            runner.pos[0] = pos + toks.numel()
+            runner.engine.set_slot_pos(0, runner.pos[0])
         else:
             # continue the cache: feed tokens one batch at the current pos
             x = runner.stage.forward_head(toks.view(1, -1), slot=0,

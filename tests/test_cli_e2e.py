@@ -92,3 +92,18 @@ def test_bench_cpu_mode_wide(world):
     assert r.returncode == 0, r.stderr[-2000:]
     assert f'"n_gpus": {world}' in r.stdout
     assert '"value"' in r.stdout
+
+
+def test_chat_cli_two_turns(ckpt, tmp_path):
+    """chat.py streaming loop: two piped turns, clean EOF exit."""
+    import subprocess
+    import sys as _sys
+
+    r = subprocess.run(
+        [_sys.executable, "chat.py", "--ckpt", str(ckpt),
+         "--max-new-tokens", "8", "--device", "cpu"],
+        input="who are you?\nand what can you do?\n",
+        capture_output=True, text=True, timeout=300, cwd=ROOT,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert r.stdout.count(">>") >= 2

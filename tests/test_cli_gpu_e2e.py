@@ -110,3 +110,24 @@ def test_train_one_gpu(tmp_path):
         env=env, capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "loss" in r.stdout
+
+
+def test_chat_cli_gpu_multi_turn(tmp_path):
+    """chat.py on the HIP engine: multi-turn continuation goes through
+    the HIP prefill (also required for the fp8 KV cache)."""
+    sys.path.insert(0, str(ROOT))
+    from tests.helpers import make_toy_checkpoint
+
+    ckpt = tmp_path / "ckpt" / "nano-gpu"
+    make_toy_checkpoint(ckpt, name="nano-gpu")
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(ROOT)
+    for kv in ("bf16", "fp8"):
+        r = subprocess.run(
+            [sys.executable, str(ROOT / "chat.py"), "--ckpt", str(ckpt),
+             "--max-new-tokens", "8", "--device", "cuda:0", "--kv", kv],
+            input="who are you?\nand more?\n", capture_output=True,
+            text=True, timeout=300, env=env, cwd=str(tmp_path),
+        )
+        assert r.returncode == 0, (kv, r.stderr[-2000:])
+        assert "(hip)" in r.stdout, r.stdout[:300]
