@@ -15,6 +15,12 @@ sys.path.insert(0, str(SCRIPT_DIR))
 
 
 def main(args):
+    import os as _os
+
+    if args.weights == "fp8":
+        _os.environ["MDI_WEIGHT_DTYPE"] = "fp8"
+    if args.kv == "fp8":
+        _os.environ["MDI_KV_DTYPE"] = "fp8"
     import torch
 
     from mdi_llm_amd.models.stages import StarterStage
@@ -118,6 +124,10 @@ if __name__ == "__main__":
     p.add_argument("--temperature", type=float, default=0.8)
     p.add_argument("--top-k", type=int, default=200)
     p.add_argument("--sequence-length", type=int, default=None)
+    p.add_argument("--weights", choices=["bf16", "fp8"], default="bf16",
+                   help="decode weight dtype on the HIP engine")
+    p.add_argument("--kv", choices=["bf16", "fp8"], default="bf16",
+                   help="KV-cache dtype on the HIP engine")
     p.add_argument("--seed", type=int, default=10137)
     p.add_argument("-d", "--debug", action="store_true",
                    help="write debug logs to logs/logs_chat.log")
