@@ -21,11 +21,13 @@ def main(model="Meta-Llama-3-8B-Instruct", lens=(128, 1024, 4096, 8000)):
     with torch.no_grad():
         for p in stage.parameters():
             p.normal_(0.0, 0.02)
-    stage.max_seq_length = 8192
+    stage.max_seq_length = min(8192, cfg.block_size)
     stage.eval()
     runner = make_runner(stage, 1, torch.device("cuda:0"))
     print("backend", runner.backend)
     for T in lens:
+        if T >= cfg.block_size:
+            continue
         prompt = torch.randint(0, cfg.vocab_size - 1, (T,), device="cuda:0")
         for rep in range(3):
             runner.reset()
