@@ -2341,6 +2341,209 @@ __global__ void prefill_attn_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Prefill flash attention v2 (causal, GQA, all-MFMA): the default path.
+//
+// Two structural fixes over prefill_attn_kernel above (which remains as the
+// MDI_PREFILL_V1=1 fallback):
+//   1. The whole WORKGROUP shares one K/V LDS staging per 64-key chunk.  The
+//      4 waves are the GQA group (hpb = min(qpk,4) query heads of ONE kv
+//      head; leftover wave capacity covers extra q tiles), so K/V cross HBM
+//      once per group instead of once per query head.
+//   2. The P·V product is MFMA too (the v1 kernel spent ~2048 scalar VALU
+//      FMAs per wave per chunk on it — 74% of a T=4096 prefill).  P is
+//      round-tripped through a per-wave transposed LDS tile to match the
+//      A-fragment layout; V is staged transposed so B-fragments are single
+//      conflict-free ds_read_b128s (row strides padded to 16 B multiples
+//      that are coprime-ish with the 64 LDS banks).
+// Softmax state (m, l) lives in registers, replicated across the 4
+// sub-tiles of a wave and exchanged with __shfl — no LDS, no barriers
+// beyond the two per-chunk staging fences.
+// Reference behavior: /root/reference/src/sub/model.py:738-751 (SDPA with
+// causal bool mask at prefill), recomputed with online softmax.
+// ---------------------------------------------------------------------------
+#define PF_KCH 64  // keys staged per chunk
+
+template <int HS, int KV8>
+__global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
+    bf16* __restrict__ out,        // [T, n_head*HS]
+    const bf16* __restrict__ qkv,  // [T, qkv_dim], q already roped
+    const void* __restrict__ kpool_v, const void* __restrict__ vpool_v,
+    const float* __restrict__ kscale, const float* __restrict__ vscale,
+    int pos0, int slot, int layer, int n_layers_pool, int n_kv_heads,
+    int max_seq, int qpk, int T, float scale, int hpb, int qtpb, int n_hgrp,
+    int n_qtg) {
+  using kvt = std::conditional_t<KV8 != 0, u8kv, bf16>;
+  const kvt* kpool = reinterpret_cast<const kvt*>(kpool_v);
+  const kvt* vpool = reinterpret_cast<const kvt*>(vpool_v);
+  __shared__ __attribute__((aligned(16))) bf16 k_lds[PF_KCH][HS + 8];
+  __shared__ __attribute__((aligned(16))) bf16 v_t[HS][PF_KCH + 8];
+  __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
+  __shared__ __attribute__((aligned(16)))
+      bf16 p_t[ATTN_WAVES][16][PF_KCH + 8];
+
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int n_qtiles = (T + 15) / 16;
+  int bid = blockIdx.x;
+  const int g = bid / (n_hgrp * n_qtg);  // kv head
+  bid -= g * (n_hgrp * n_qtg);
+  const int hgrp = bid / n_qtg;
+  const int qtg = bid % n_qtg;
+  const int hj = hgrp * hpb + (wave % hpb);  // q row within the kv group
+  const int qtile = qtg * qtpb + (wave / hpb);
+  const bool active = (hj < qpk) && (qtile < n_qtiles);
+  const int h = g * qpk + hj;            // query head (output numbering)
+  const int q_base = qtile * 16;
+  const int qkv_dim = n_kv_heads * (qpk + 2) * HS;
+
+  const size_t cache_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq * HS;
+  const size_t scl_base =
+      (((size_t)slot * n_layers_pool + layer) * n_kv_heads + g) *
+      (size_t)max_seq;
+
+  // stage this wave's 16 q rows (zero-filled when inactive / past T)
+  for (int i = lane; i < 16 * (HS / 8); i += 64) {
+    const int r = i / (HS / 8);
+    const int d8 = (i % (HS / 8)) * 8;
+    int4 val = {0, 0, 0, 0};
+    const int t = q_base + r;
+    if (active && t < T)
+      val = *reinterpret_cast<const int4*>(
+          qkv + (size_t)t * qkv_dim + ((size_t)g * (qpk + 2) + hj) * HS + d8);
+    *reinterpret_cast<int4*>(&q_lds[wave][q_swz<HS>(r, d8)]) = val;
+  }
+
+  // causal bounds: block = staging bound (last q row any wave covers),
+  // wave = this wave's own tile
+  const int qt_last = min(qtg * qtpb + qtpb, n_qtiles) * 16 - 1;
+  const int k_last_blk = pos0 + min(qt_last, T - 1) + 1;
+  const int k_last_w = active ? (pos0 + min(q_base + 15, T - 1) + 1) : 0;
+
+  const int arow = lane & 15;
+  const int sub = lane >> 4;
+  const int koff = sub * 8;
+  const int q_abs = pos0 + q_base + arow;
+
+  float m_r = -1e30f, l_r = 0.f;
+  f32x4 o_pv[HS / 16];
+#pragma unroll
+  for (int s = 0; s < HS / 16; ++s) o_pv[s] = {0.f, 0.f, 0.f, 0.f};
+
+  const int n_ch = (k_last_blk + PF_KCH - 1) / PF_KCH;
+  for (int ch = 0; ch < n_ch; ++ch) {
+    const int key0 = ch * PF_KCH;
+    __syncthreads();  // previous chunk's readers done (q staging on ch 0)
+    // ---- cooperative K / V^T staging (fp8 dequantized once, here) ----
+    for (int i = threadIdx.x; i < PF_KCH * (HS / 8); i += 256) {
+      const int r = i / (HS / 8);
+      const int d8 = (i % (HS / 8)) * 8;
+      const int key = key0 + r;
+      bf16x8_t kk = {}, vv = {};
+      if (key < k_last_blk) {
+        const size_t off = cache_base + (size_t)key * HS + d8;
+        if constexpr (KV8) {
+          kk = fp8x8_to_bf16(kpool + off, kscale[scl_base + key]);
+          vv = fp8x8_to_bf16(vpool + off, vscale[scl_base + key]);
+        } else {
+          kk = *reinterpret_cast<const bf16x8_t*>(
+              reinterpret_cast<const bf16*>(kpool) + off);
+          vv = *reinterpret_cast<const bf16x8_t*>(
+              reinterpret_cast<const bf16*>(vpool) + off);
+        }
+      }
+      *reinterpret_cast<bf16x8_t*>(&k_lds[r][d8]) = kk;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<__bf16*>(&v_t[d8 + j][r]) = vv[j];
+    }
+    __syncthreads();
+    if (!active || key0 >= k_last_w) continue;  // barriers are at loop top
+
+    // ---- scores: 4 MFMA 16-key tiles against this wave's q tile ----
+    float sc[PF_KCH / 16][4];
+#pragma unroll
+    for (int t4 = 0; t4 < PF_KCH / 16; ++t4) {
+      f32x4 a4 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int c = 0; c < HS / 32; ++c) {
+        const bf16x8_t af = *reinterpret_cast<const bf16x8_t*>(
+            &k_lds[t4 * 16 + arow][c * 32 + koff]);
+        const bf16x8_t bfr = *reinterpret_cast<const bf16x8_t*>(
+            &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
+        a4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, a4, 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = key0 + t4 * 16 + sub * 4 + r;
+        sc[t4][r] =
+            (key <= q_abs && key < k_last_w) ? a4[r] * scale : -1e30f;
+      }
+    }
+    // ---- online softmax, state in registers (per qa = arow) ----
+    float tmax = -1e30f;
+#pragma unroll
+    for (int t4 = 0; t4 < PF_KCH / 16; ++t4)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) tmax = fmaxf(tmax, sc[t4][r]);
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 16, 64));
+    tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+    const float m_new = fmaxf(m_r, tmax);
+    const float alpha = __expf(m_r - m_new);
+    m_r = m_new;
+    float psum = 0.f;
+#pragma unroll
+    for (int t4 = 0; t4 < PF_KCH / 16; ++t4)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float p = __expf(sc[t4][r] - m_new);  // masked rows: exactly 0
+        psum += p;
+        p_t[wave][arow][t4 * 16 + sub * 4 + r] = f2b(p);
+      }
+    psum += __shfl_xor(psum, 16, 64);
+    psum += __shfl_xor(psum, 32, 64);
+    l_r = l_r * alpha + psum;
+
+    // ---- P·V via MFMA (A = P rows from p_t, B = V^T slices) ----
+    float al[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) al[r] = __shfl(alpha, sub * 4 + r, 64);
+#pragma unroll
+    for (int s = 0; s < HS / 16; ++s)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_pv[s][r] *= al[r];
+#pragma unroll
+    for (int k2 = 0; k2 < PF_KCH / 32; ++k2) {
+      const bf16x8_t ap = *reinterpret_cast<const bf16x8_t*>(
+          &p_t[wave][arow][k2 * 32 + koff]);
+#pragma unroll
+      for (int s = 0; s < HS / 16; ++s) {
+        const bf16x8_t vb = *reinterpret_cast<const bf16x8_t*>(
+            &v_t[s * 16 + arow][k2 * 32 + koff]);
+        o_pv[s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, vb, o_pv[s],
+                                                          0, 0, 0);
+      }
+    }
+  }
+
+  // ---- normalize + write (lane holds q rows sub*4+r, dim s*16+arow) ----
+  if (active) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int t_out = q_base + sub * 4 + r;
+      const float lq = __shfl(l_r, sub * 4 + r, 64);
+      if (t_out >= T) continue;
+      const float inv = 1.f / lq;
+      bf16* op = out + (size_t)t_out * (n_kv_heads * qpk * HS) +
+                 (size_t)h * HS + arow;
+#pragma unroll
+      for (int s = 0; s < HS / 16; ++s) op[s * 16] = f2b(o_pv[s][r] * inv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Residual add: out = a + b (bf16, fp32 math)
 // ---------------------------------------------------------------------------
 __global__ void add_kernel(bf16* __restrict__ out, const bf16* __restrict__ a,
@@ -3330,6 +3533,38 @@ int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
   const int n_qtiles = (T + 15) / 16;
   const int blocks = (n_head * n_qtiles + ATTN_WAVES - 1) / ATTN_WAVES;
   const int kv8 = kscale != nullptr;
+  static const bool v1 = [] {
+    const char* e = getenv("MDI_PREFILL_V1");
+    return e != nullptr && e[0] == '1';
+  }();
+  if (!v1) {
+    // v2: workgroup shares K/V staging across the GQA group; P.V by MFMA
+    const int hpb = qpk >= 4 ? 4 : qpk;  // q heads per block
+    const int qtpb = (hpb == 1) ? 4 : (hpb == 2 ? 2 : 1);  // q tiles / block
+    const int n_hgrp = (qpk + hpb - 1) / hpb;
+    const int n_qtg = (n_qtiles + qtpb - 1) / qtpb;
+    const int blocks2 = n_kv_heads * n_hgrp * n_qtg;
+#define PF2_CASE(H)                                                         \
+  if (head_size == H) {                                                     \
+    if (kv8)                                                                \
+      hipLaunchKernelGGL((prefill_attn_mfma_kernel<H, 1>), dim3(blocks2),   \
+                         dim3(256), 0, stream, (bf16*)out,                  \
+                         (const bf16*)qkv, kpool, vpool, kscale, vscale,    \
+                         pos0, slot, layer, n_layers_pool, n_kv_heads,      \
+                         max_seq, qpk, T, scale, hpb, qtpb, n_hgrp, n_qtg); \
+    else                                                                    \
+      hipLaunchKernelGGL((prefill_attn_mfma_kernel<H, 0>), dim3(blocks2),   \
+                         dim3(256), 0, stream, (bf16*)out,                  \
+                         (const bf16*)qkv, kpool, vpool, kscale, vscale,    \
+                         pos0, slot, layer, n_layers_pool, n_kv_heads,      \
+                         max_seq, qpk, T, scale, hpb, qtpb, n_hgrp, n_qtg); \
+    return 0;                                                               \
+  }
+    PF2_CASE(64)
+    PF2_CASE(128)
+    PF2_CASE(256)
+#undef PF2_CASE
+  }
 #define PF_CASE(H)                                                          \
   if (head_size == H) {                                                     \
     if (kv8)                                                                \
