@@ -2361,9 +2361,7 @@ __global__ void prefill_attn_kernel(
 // Reference behavior: /root/reference/src/sub/model.py:738-751 (SDPA with
 // causal bool mask at prefill), recomputed with online softmax.
 // ---------------------------------------------------------------------------
-#define PF_KCH 64  // keys staged per chunk
-
-template <int HS, int KV8>
+template <int HS, int KV8, int PF_KCH>  // PF_KCH: keys staged per chunk
 __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     bf16* __restrict__ out,        // [T, n_head*HS]
     const bf16* __restrict__ qkv,  // [T, qkv_dim], q already roped
@@ -3537,6 +3535,12 @@ int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
     const char* e = getenv("MDI_PREFILL_V1");
     return e != nullptr && e[0] == '1';
   }();
+  // chunk-size A/B knob: 64 (2 blocks/CU at HS=128) vs 32 (4 blocks/CU,
+  // twice the barriers) — 64 measured faster, see profiles
+  static const int kch = [] {
+    const char* e = getenv("MDI_PREFILL_KCH");
+    return (e != nullptr && atoi(e) == 32) ? 32 : 64;
+  }();
   if (!v1) {
     // v2: workgroup shares K/V staging across the GQA group; P.V by MFMA
     const int hpb = qpk >= 4 ? 4 : qpk;  // q heads per block
@@ -3544,26 +3548,26 @@ int launch_prefill_attn(void* out, const void* qkv, const void* kpool,
     const int n_hgrp = (qpk + hpb - 1) / hpb;
     const int n_qtg = (n_qtiles + qtpb - 1) / qtpb;
     const int blocks2 = n_kv_heads * n_hgrp * n_qtg;
+#define PF2_LAUNCH(H, K8, KC)                                               \
+  hipLaunchKernelGGL((prefill_attn_mfma_kernel<H, K8, KC>), dim3(blocks2),  \
+                     dim3(256), 0, stream, (bf16*)out, (const bf16*)qkv,    \
+                     kpool, vpool, kscale, vscale, pos0, slot, layer,       \
+                     n_layers_pool, n_kv_heads, max_seq, qpk, T, scale,     \
+                     hpb, qtpb, n_hgrp, n_qtg)
 #define PF2_CASE(H)                                                         \
   if (head_size == H) {                                                     \
-    if (kv8)                                                                \
-      hipLaunchKernelGGL((prefill_attn_mfma_kernel<H, 1>), dim3(blocks2),   \
-                         dim3(256), 0, stream, (bf16*)out,                  \
-                         (const bf16*)qkv, kpool, vpool, kscale, vscale,    \
-                         pos0, slot, layer, n_layers_pool, n_kv_heads,      \
-                         max_seq, qpk, T, scale, hpb, qtpb, n_hgrp, n_qtg); \
-    else                                                                    \
-      hipLaunchKernelGGL((prefill_attn_mfma_kernel<H, 0>), dim3(blocks2),   \
-                         dim3(256), 0, stream, (bf16*)out,                  \
-                         (const bf16*)qkv, kpool, vpool, kscale, vscale,    \
-                         pos0, slot, layer, n_layers_pool, n_kv_heads,      \
-                         max_seq, qpk, T, scale, hpb, qtpb, n_hgrp, n_qtg); \
+    if (kv8) {                                                              \
+      if (kch == 32) PF2_LAUNCH(H, 1, 32); else PF2_LAUNCH(H, 1, 64);       \
+    } else {                                                                \
+      if (kch == 32) PF2_LAUNCH(H, 0, 32); else PF2_LAUNCH(H, 0, 64);       \
+    }                                                                       \
     return 0;                                                               \
   }
     PF2_CASE(64)
     PF2_CASE(128)
     PF2_CASE(256)
 #undef PF2_CASE
+#undef PF2_LAUNCH
   }
 #define PF_CASE(H)                                                          \
   if (head_size == H) {                                                     \
