@@ -569,15 +569,22 @@ class DecodeEngine:
                              slot, li, scale,
                              kscale=self.kv_pool.kscale,
                              vscale=self.kv_pool.vscale)
-            a = x + F.linear(Y, w.proj_w, w.proj_b)
+            # residual rides the GEMM epilogue (addmm beta=1) when there is
+            # no bias; act(gate)*up is one fused launch instead of two
+            if w.proj_b is None:
+                a = torch.addmm(x, Y, w.proj_w.t())
+            else:
+                a = x + F.linear(Y, w.proj_w, w.proj_b)
             hn = xn
             ops.rmsnorm(hn, a, w.norm2_w, cfg.norm_eps)
             gelu_gate = cfg.mlp_class_name == "GemmaMLP"
             gate = F.linear(hn, w.fc1_w)
             up = F.linear(hn, w.fc2_w)
-            act = (F.gelu(gate, approximate="tanh") if gelu_gate
-                   else F.silu(gate)) * up
-            x = a + F.linear(act, w.mlp_proj_w, w.mlp_proj_b)
+            ops.swiglu_mul(up, gate, up, gelu_gate)
+            if w.mlp_proj_b is None:
+                x = torch.addmm(a, up, w.mlp_proj_w.t())
+            else:
+                x = a + F.linear(up, w.mlp_proj_w, w.mlp_proj_b)
         return x
 
     @torch.inference_mode()
