@@ -2389,7 +2389,10 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
   const int qtg = bid % n_qtg;
   const int hj = hgrp * hpb + (wave % hpb);  // q row within the kv group
   const int qtile = qtg * qtpb + (wave / hpb);
-  const bool active = (hj < qpk) && (qtile < n_qtiles);
+  // wave/hpb >= qtpb happens when hpb doesn't divide 4 (hpb=3): that wave
+  // would claim a q tile outside the block's staged causal bound — idle it
+  const bool active =
+      (hj < qpk) && (qtile < n_qtiles) && (wave / hpb < qtpb);
   const int h = g * qpk + hj;            // query head (output numbering)
   const int q_base = qtile * 16;
   const int qkv_dim = n_kv_heads * (qpk + 2) * HS;
