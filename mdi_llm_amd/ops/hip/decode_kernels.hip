@@ -2426,6 +2426,8 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
   const int sub = lane >> 4;
   const int koff = sub * 8;
   const int q_abs = pos0 + q_base + arow;
+  // p_t write/read column swizzle (same rationale as the v_t one)
+  const int pkx = ((arow >> 3) & (PF_KCH / 8 - 1)) << 3;
 
   float m_r = -1e30f, l_r = 0.f;
   f32x4 o_pv[HS / 16];
@@ -2455,9 +2457,17 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
         }
       }
       *reinterpret_cast<bf16x8_t*>(&k_lds[r][d8]) = kk;
+      // key-column XOR swizzle: v_t rows are 36 dwords apart, so rows 8
+      // apart hit one ds_write bank (mod 32) — the per-(row>>3) XOR on
+      // the key index spreads the 16-way write conflict to 2-way while
+      // keeping each b128 read's 8 keys contiguous (key0 stays a
+      // multiple of 8 under the XOR)
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *reinterpret_cast<__bf16*>(&v_t[d8 + j][r]) = vv[j];
+      for (int j = 0; j < 8; ++j) {
+        const int vrow = d8 + j;
+        const int kc = r ^ (((vrow >> 3) & (PF_KCH / 8 - 1)) << 3);
+        *reinterpret_cast<__bf16*>(&v_t[vrow][kc]) = vv[j];
+      }
     }
     __syncthreads();
     if (!active || key0 >= k_last_w) continue;  // barriers are at loop top
@@ -2500,7 +2510,7 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
       for (int r = 0; r < 4; ++r) {
         const float p = __expf(sc[t4][r] - m_new);  // masked rows: exactly 0
         psum += p;
-        p_t[wave][arow][t4 * 16 + sub * 4 + r] = f2b(p);
+        p_t[wave][arow][(t4 * 16 + sub * 4 + r) ^ pkx] = f2b(p);
       }
     psum += __shfl_xor(psum, 16, 64);
     psum += __shfl_xor(psum, 32, 64);
@@ -2517,11 +2527,14 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
 #pragma unroll
     for (int k2 = 0; k2 < PF_KCH / 32; ++k2) {
       const bf16x8_t ap = *reinterpret_cast<const bf16x8_t*>(
-          &p_t[wave][arow][k2 * 32 + koff]);
+          &p_t[wave][arow][(k2 * 32 + koff) ^ pkx]);
 #pragma unroll
       for (int s = 0; s < HS / 16; ++s) {
-        const bf16x8_t vb = *reinterpret_cast<const bf16x8_t*>(
-            &v_t[s * 16 + arow][k2 * 32 + koff]);
+        const int vrow = s * 16 + arow;
+        const int kc0 = (k2 * 32 + koff) ^
+                        (((vrow >> 3) & (PF_KCH / 8 - 1)) << 3);
+        const bf16x8_t vb =
+            *reinterpret_cast<const bf16x8_t*>(&v_t[vrow][kc0]);
         o_pv[s] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap, vb, o_pv[s],
                                                           0, 0, 0);
       }
