@@ -2361,6 +2361,16 @@ __global__ void prefill_attn_kernel(
 // Reference behavior: /root/reference/src/sub/model.py:738-751 (SDPA with
 // causal bool mask at prefill), recomputed with online softmax.
 // ---------------------------------------------------------------------------
+// q-tile swizzle for the v2 prefill kernel: the b128 read groups mix
+// arow 0-3/12-15 (sub 0) with arow 4-11 (sub 1), so the XOR key must
+// separate all 16 rows — (row&15)<<4 when the row stride allows (HS>=128)
+template <int HS>
+DEVINL int q_swz2(int row, int d) {
+  int byte = (row * HS + d) * 2;
+  byte ^= (row & (HS >= 128 ? 15 : 7)) << 4;
+  return byte >> 1;
+}
+
 template <int HS, int KV8, int PF_KCH>  // PF_KCH: keys staged per chunk
 __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     bf16* __restrict__ out,        // [T, n_head*HS]
@@ -2373,11 +2383,11 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
   using kvt = std::conditional_t<KV8 != 0, u8kv, bf16>;
   const kvt* kpool = reinterpret_cast<const kvt*>(kpool_v);
   const kvt* vpool = reinterpret_cast<const kvt*>(vpool_v);
-  __shared__ __attribute__((aligned(16))) bf16 k_lds[PF_KCH][HS + 8];
-  __shared__ __attribute__((aligned(16))) bf16 v_t[HS][PF_KCH + 8];
+  __shared__ __attribute__((aligned(16))) bf16 k_lds[PF_KCH][HS + 16];
+  __shared__ __attribute__((aligned(16))) bf16 v_t[HS][PF_KCH + 24];
   __shared__ __attribute__((aligned(16))) bf16 q_lds[ATTN_WAVES][16 * HS];
   __shared__ __attribute__((aligned(16)))
-      bf16 p_t[ATTN_WAVES][16][PF_KCH + 8];
+      bf16 p_t[ATTN_WAVES][16][PF_KCH + 24];
 
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -2413,7 +2423,7 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     if (active && t < T)
       val = *reinterpret_cast<const int4*>(
           qkv + (size_t)t * qkv_dim + ((size_t)g * (qpk + 2) + hj) * HS + d8);
-    *reinterpret_cast<int4*>(&q_lds[wave][q_swz<HS>(r, d8)]) = val;
+    *reinterpret_cast<int4*>(&q_lds[wave][q_swz2<HS>(r, d8)]) = val;
   }
 
   // causal bounds: block = staging bound (last q row any wave covers),
@@ -2482,7 +2492,7 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
         const bf16x8_t af = *reinterpret_cast<const bf16x8_t*>(
             &k_lds[t4 * 16 + arow][c * 32 + koff]);
         const bf16x8_t bfr = *reinterpret_cast<const bf16x8_t*>(
-            &q_lds[wave][q_swz<HS>(arow, c * 32 + koff)]);
+            &q_lds[wave][q_swz2<HS>(arow, c * 32 + koff)]);
         a4 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr, a4, 0, 0, 0);
       }
 #pragma unroll
