@@ -2482,6 +2482,9 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
     __syncthreads();
     if (!active || key0 >= k_last_w) continue;  // barriers are at loop top
 
+    // interior chunks (every key causally visible to every q row of the
+    // tile) skip the per-key mask compares — wave-uniform
+    const bool full = (key0 + PF_KCH - 1) <= (pos0 + q_base);
     // ---- scores: 4 MFMA 16-key tiles against this wave's q tile ----
     float sc[PF_KCH / 16][4];
 #pragma unroll
@@ -2499,7 +2502,9 @@ __global__ __launch_bounds__(256) void prefill_attn_mfma_kernel(
       for (int r = 0; r < 4; ++r) {
         const int key = key0 + t4 * 16 + sub * 4 + r;
         sc[t4][r] =
-            (key <= q_abs && key < k_last_w) ? a4[r] * scale : -1e30f;
+            full ? a4[r] * scale
+                 : ((key <= q_abs && key < k_last_w) ? a4[r] * scale
+                                                     : -1e30f);
       }
     }
     // ---- online softmax, state in registers (per qa = arow) ----
