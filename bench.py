@@ -136,6 +136,7 @@ def main():
     runner = make_runner(
         stage, n_samples, device, use_graphs=not args.no_graphs,
         force_torch=args.backend == "torch",
+        expected_s=args.prompt_len + args.warmup + args.steps + 2,
     )
     log(f"rank {rank}: runner backend={runner.backend}")
     if args.backend == "hip" and runner.backend != "hip":

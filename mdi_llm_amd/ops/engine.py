@@ -124,6 +124,7 @@ class DecodeEngine:
         kv_pool: KVCachePool,
         n_chunks: int = 16,
         use_graphs: bool = True,
+        expected_s: "int | None" = None,
     ) -> None:
         # adaptive split-S: 16 chunks up to 4k contexts (measured best at
         # short S; the one-launch block-local kernel is used there anyway),
@@ -131,8 +132,17 @@ class DecodeEngine:
         # 256 chunks = 2048 waves = 8/CU, the full wave occupancy (128
         # chunks measured only ~0.6 TB/s of KV streaming at S=8192: half
         # the waves, 4 serial tiles each)
+        # expected_s: the caller's upper bound on the sequence length this
+        # engine will actually decode at (prompt + tokens).  The one-launch
+        # block-local attention (grid = n_kv_heads blocks) wins at short S
+        # by the launch floor but serializes at S >~ 500 (measured: +2% at
+        # S~166, -14% at S~1900 vs split-S) — when the caller knows the
+        # generation runs past the crossover, force the split-S + combine
+        # path even at max_seq <= 4096
+        self.force_split = 1 if (expected_s is not None
+                                 and expected_s >= 512) else 0
         default_chunks = max(16, min(256, stage.max_seq_length // 32))
-        if stage.max_seq_length <= 4096:
+        if stage.max_seq_length <= 4096 and not self.force_split:
             default_chunks = 16
         if n_chunks == 16:
             n_chunks = default_chunks
@@ -386,6 +396,7 @@ class DecodeEngine:
                 self.kv_pool.v, self.cos, self.sin, self.pos, self.slot, li,
                 self.n_chunks, scale,
                 kscale=self.kv_pool.kscale, vscale=self.kv_pool.vscale,
+                force_split=self.force_split,
             )
             if cfg.parallel_residual:
                 # x = x + proj(y) + mlp(norm2(x) or norm1(x))

@@ -288,7 +288,8 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
                  torch::Tensor slot, int64_t layer, int64_t n_chunks,
                  double scale, int64_t n_batch,
                  c10::optional<torch::Tensor> kscale,
-                 c10::optional<torch::Tensor> vscale) {
+                 c10::optional<torch::Tensor> vscale,
+                 int64_t force_split) {
   check_bf16(out, "out");
   check_f32(part_o, "part_o");
   check_f32(part_ml, "part_ml");
@@ -336,7 +337,7 @@ void attn_decode(torch::Tensor out, torch::Tensor part_o,
       rope_ne ? sin_t.data_ptr<float>() : nullptr, rope_ne,
       pos.data_ptr<int>(), slot.data_ptr<int>(), (int)layer, n_layers_pool,
       n_kv, max_seq, hs, qpk, (int)n_chunks, (float)scale, nb,
-      cur_stream());
+      (int)force_split, cur_stream());
   TORCH_CHECK(rc == 0, "attn_decode: unsupported geometry qpk=", qpk,
               " head_size=", hs);
 }
@@ -612,7 +613,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("qkv"), py::arg("kpool"), py::arg("vpool"), py::arg("cos"),
         py::arg("sin"), py::arg("pos"), py::arg("slot"), py::arg("layer"),
         py::arg("n_chunks"), py::arg("scale"), py::arg("n_batch") = 0,
-        py::arg("kscale") = c10::nullopt, py::arg("vscale") = c10::nullopt);
+        py::arg("kscale") = c10::nullopt, py::arg("vscale") = c10::nullopt,
+        py::arg("force_split") = 0);
   m.def("add", &add, "bf16 residual add");
   m.def("rope_prefill_append", &rope_prefill_append,
         "prefill: rope q/k for T positions + append k/v to the pool",

@@ -75,7 +75,7 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
                        int head_size, int qpk, int n_chunks, float scale,
-                       int n_batch, hipStream_t stream);
+                       int n_batch, int force_split, hipStream_t stream);
 
 void launch_add(void* out, const void* a, const void* b, int n,
                 hipStream_t stream);

@@ -3386,7 +3386,8 @@ static void attn_dispatch2(void* out, float* part_o, float* part_ml,
                            int rope_ne, const int* pos, const int* slot,
                            int layer, int n_layers_pool, int n_kv_heads,
                            int max_seq, int n_chunks, float scale,
-                           int n_batch, int blocks, hipStream_t stream) {
+                           int n_batch, int blocks, int force_split,
+                           hipStream_t stream) {
   const int kv8 = kscale != nullptr;
   // short/medium contexts: one block per (batch, kv_head), no global
   // partials, no combine kernel — the whole attention step is ONE launch.
@@ -3398,7 +3399,7 @@ static void attn_dispatch2(void* out, float* part_o, float* part_ml,
     const char* e = getenv("MDI_ATTN_FORCE_SPLITS");
     force_splits = (e && e[0] == '1') ? 1 : 0;
   }
-  if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits) {
+  if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits && !force_split) {
     const int nb = (n_batch > 0 ? n_batch : 1) * n_kv_heads;
 #define ATTN_BLK(BATCHV, KV8V, NBATCH)                                      \
     hipLaunchKernelGGL((attn_decode_block_kernel<QPK, HS, BATCHV, KV8V>),   \
@@ -3441,7 +3442,7 @@ static int attn_dispatch1(int head_size, void* out, float* part_o,
                           const int* slot, int layer, int n_layers_pool,
                           int n_kv_heads, int max_seq, int n_chunks,
                           float scale, int n_batch, int blocks,
-                          hipStream_t stream) {
+                          int force_split, hipStream_t stream) {
 #define CASE_HS(H)                                                          \
   if (head_size == H) {                                                     \
     if constexpr (QPK * H >= 64) {                                          \
@@ -3449,7 +3450,7 @@ static int attn_dispatch1(int head_size, void* out, float* part_o,
                              kscale, vscale, cos_t, sin_t, rope_ne, pos,    \
                              slot, layer, n_layers_pool, n_kv_heads,        \
                              max_seq, n_chunks, scale, n_batch, blocks,     \
-                             stream);                                       \
+                             force_split, stream);                          \
       return 0;                                                             \
     }                                                                       \
   }
@@ -3469,7 +3470,7 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
                        const int* pos, const int* slot, int layer,
                        int n_layers_pool, int n_kv_heads, int max_seq,
                        int head_size, int qpk, int n_chunks, float scale,
-                       int n_batch, hipStream_t stream) {
+                       int n_batch, int force_split, hipStream_t stream) {
   const int n_wg = n_kv_heads * n_chunks * (n_batch > 0 ? n_batch : 1);
   const int blocks = (n_wg + ATTN_WAVES - 1) / ATTN_WAVES;
   int rc = -1;
@@ -3479,35 +3480,35 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
           head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
           vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, n_batch, blocks, stream);
+          n_chunks, scale, n_batch, blocks, force_split, stream);
       break;
     case 2:
       rc = attn_dispatch1<2>(
           head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
           vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, n_batch, blocks, stream);
+          n_chunks, scale, n_batch, blocks, force_split, stream);
       break;
     case 4:
       rc = attn_dispatch1<4>(
           head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
           vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, n_batch, blocks, stream);
+          n_chunks, scale, n_batch, blocks, force_split, stream);
       break;
     case 8:
       rc = attn_dispatch1<8>(
           head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
           vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, n_batch, blocks, stream);
+          n_chunks, scale, n_batch, blocks, force_split, stream);
       break;
     case 16:
       rc = attn_dispatch1<16>(
           head_size, out, part_o, part_ml, qkv, kpool, vpool, kscale,
           vscale, cos_t, sin_t,
           rope_ne, pos, slot, layer, n_layers_pool, n_kv_heads, max_seq,
-          n_chunks, scale, n_batch, blocks, stream);
+          n_chunks, scale, n_batch, blocks, force_split, stream);
       break;
     default:
       return -1;
@@ -3518,7 +3519,7 @@ int launch_attn_decode(void* out, float* part_o, float* part_ml,
     const char* e = getenv("MDI_ATTN_FORCE_SPLITS");
     force_splits2 = (e && e[0] == '1') ? 1 : 0;
   }
-  if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits2)
+  if (max_seq <= ATTN_BLOCK_MAX_SEQ && !force_splits2 && !force_split)
     return 0;  // block-local: no combine
   // the combine kernel is batch-agnostic: [B, n_head, chunks, hs] is just
   // B*n_head heads

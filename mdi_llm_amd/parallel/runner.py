@@ -74,12 +74,14 @@ class HipRunner(TorchRunner):
     backend = "hip"
 
     def __init__(self, stage, n_slots: int, n_chunks: int = 16,
-                 use_graphs: bool = True) -> None:
+                 use_graphs: bool = True,
+                 expected_s: "int | None" = None) -> None:
         super().__init__(stage, n_slots)
         from ..ops.engine import DecodeEngine
 
         self.engine = DecodeEngine(stage, stage.kv_pool, n_chunks=n_chunks,
-                                   use_graphs=use_graphs)
+                                   use_graphs=use_graphs,
+                                   expected_s=expected_s)
         if use_graphs:
             self.engine.capture_graphs()
             # graph warm-up scribbled on the cache pool; start clean
@@ -129,7 +131,8 @@ class HipRunner(TorchRunner):
 
 def make_runner(stage, n_slots: int, device: torch.device,
                 n_chunks: int = 16, use_graphs: bool = True,
-                force_torch: bool = False):
+                force_torch: bool = False,
+                expected_s: "int | None" = None):
     """Pick the HIP runner on GPU when the config is supported; fail loudly
     if a GPU is present but the extension is missing (no silent eager
     fallback for supported configs)."""
@@ -139,7 +142,7 @@ def make_runner(stage, n_slots: int, device: torch.device,
 
     if engine_supported(stage.config):
         return HipRunner(stage, n_slots, n_chunks=n_chunks,
-                         use_graphs=use_graphs)
+                         use_graphs=use_graphs, expected_s=expected_s)
     import sys
 
     print(
