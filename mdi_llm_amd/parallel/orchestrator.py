@@ -170,6 +170,9 @@ class MDIRuntime:
                 "master_addr": self.topology.master_addr,
                 "master_port": self.topology.master_port,
                 "env_serve": env_serve,
+                # upper bound on decoded S (prompt allowance + new tokens):
+                # lets the engine pick block-local vs split-S attention
+                "expected_s": min(seq_len, tokens_per_sample + 256),
             }
             # path hint works on shared/local filesystems; params are pushed
             # when the chunks were split just now (the secondary host may not
@@ -216,7 +219,10 @@ class MDIRuntime:
         self.stage.max_seq_length = seq_len
         self.stage.eval()
 
-        self.runner = make_runner(self.stage, n_samples, self.device)
+        self.runner = make_runner(
+            self.stage, n_samples, self.device,
+            expected_s=min(seq_len, tokens_per_sample + 256),
+        )
         if self.world > 1:
             self.comm = RingComm(config.n_embd, seq_len, self.device,
                                  n_samples, dtype=self.dtype)
@@ -305,7 +311,8 @@ class MDIRuntime:
         # scratch slot on device
         self.runner = make_runner(self.stage,
                                   n_samples + (1 if env_serve else 0),
-                                  self.device)
+                                  self.device,
+                                  expected_s=msg.get("expected_s"))
         self.comm = RingComm(config.n_embd, seq_len, self.device, n_samples,
                              dtype=self.dtype)
         self.runtime = PipelineRuntime(self.runner, msg["rank"],
